@@ -1,0 +1,208 @@
+"""Op dispatch: hand-written CDNA4 HIP kernels on GPU, reference PyTorch on CPU.
+
+Policy (deliberate): on a CUDA/HIP device these functions REQUIRE the in-tree
+extension ``polyrl_amd._hip`` — a missing extension raises instead of silently
+falling back to eager, so a GPU run can never pass on a non-native path.
+On CPU they use ops.ref (the same functions the numerics tests compare
+against).
+"""
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+from . import ref
+
+_EXT = None
+_EXT_ERR: Optional[str] = None
+try:
+    from polyrl_amd import _hip as _EXT  # type: ignore
+except Exception as e:  # pragma: no cover - import-environment dependent
+    _EXT_ERR = repr(e)
+
+
+def extension_loaded() -> bool:
+    return _EXT is not None
+
+
+def _require_ext():
+    if _EXT is None:
+        raise RuntimeError(
+            "polyrl_amd._hip extension is required on GPU but failed to load "
+            f"({_EXT_ERR}). Build it in-tree: "
+            "PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace")
+    return _EXT
+
+
+# --------------------------------------------------------------------- norms
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-6) -> torch.Tensor:
+    if x.is_cuda:
+        ext = _require_ext()
+        out = torch.empty_like(x)
+        ext.rmsnorm(out, x.contiguous(), weight.contiguous(), eps)
+        return out
+    return ref.rmsnorm(x, weight, eps)
+
+
+def fused_add_rmsnorm(x: torch.Tensor, residual: torch.Tensor,
+                      weight: torch.Tensor, eps: float = 1e-6
+                      ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """returns (normed, new_residual); on GPU, residual is updated IN PLACE
+    and returned as new_residual."""
+    if x.is_cuda:
+        ext = _require_ext()
+        out = torch.empty_like(x)
+        ext.fused_add_rmsnorm(out, residual, x.contiguous(), weight.contiguous(), eps)
+        return out, residual
+    return ref.fused_add_rmsnorm(x, residual, weight, eps)
+
+
+def silu_mul(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
+    if gate.is_cuda:
+        ext = _require_ext()
+        out = torch.empty_like(gate)
+        ext.silu_mul(out, gate.contiguous(), up.contiguous())
+        return out
+    return ref.silu_mul(gate, up)
+
+
+# ---------------------------------------------------------------------- RoPE
+
+
+class RopeTable:
+    """Host-precomputed cos/sin tables (guide App. B: no on-device trig)."""
+
+    def __init__(self, head_dim: int, max_positions: int, theta: float,
+                 device, dtype=torch.float32):
+        pos = torch.arange(max_positions, device=device)
+        cos, sin = ref.rope_cos_sin(pos, head_dim, theta, dtype=torch.float32)
+        self.cos = cos.contiguous()
+        self.sin = sin.contiguous()
+        self.head_dim = head_dim
+        self.theta = theta
+        self.max_positions = max_positions
+
+
+def apply_rope_inplace(q: torch.Tensor, k: torch.Tensor,
+                       positions: torch.Tensor, table: RopeTable
+                       ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """q: (N, Hq, D), k: (N, Hk, D), positions (N,) int32. In place on GPU."""
+    if q.is_cuda:
+        ext = _require_ext()
+        ext.rope_inplace(q, k, positions.int(), table.cos, table.sin)
+        return q, k
+    cos = table.cos[positions.long()]
+    sin = table.sin[positions.long()]
+    q2, k2 = ref.apply_rope(q, k, cos, sin)
+    q.copy_(q2)
+    k.copy_(k2)
+    return q, k
+
+
+# ------------------------------------------------------------------ logprobs
+
+
+def gather_logprobs(logits: torch.Tensor, labels: torch.Tensor,
+                    want_entropy: bool = False):
+    """(N, V), (N,) -> lp (N,) fp32 [, entropy (N,) fp32]"""
+    if logits.is_cuda:
+        ext = _require_ext()
+        N = logits.size(0)
+        lp = torch.empty(N, dtype=torch.float32, device=logits.device)
+        ent = torch.empty(N if want_entropy else 0, dtype=torch.float32,
+                          device=logits.device)
+        ext.gather_logprobs(lp, ent, logits.contiguous(), labels.long(),
+                            want_entropy)
+        return (lp, ent) if want_entropy else lp
+    if want_entropy:
+        return ref.gather_logprobs_entropy(logits, labels)
+    return ref.gather_logprobs(logits, labels)
+
+
+# ------------------------------------------------------------------ sampling
+
+
+def sample(logits: torch.Tensor, temperature: torch.Tensor,
+           top_k: torch.Tensor, top_p: torch.Tensor, seed: int,
+           generator: Optional[torch.Generator] = None
+           ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """(N, V) -> (tokens (N,) int64, logprobs (N,) fp32 under raw softmax)."""
+    if logits.is_cuda:
+        ext = _require_ext()
+        N = logits.size(0)
+        tokens = torch.empty(N, dtype=torch.int64, device=logits.device)
+        lps = torch.empty(N, dtype=torch.float32, device=logits.device)
+        ext.sample(tokens, lps, logits.contiguous(), temperature.float(),
+                   top_k.int(), top_p.float(), seed)
+        return tokens, lps
+    return ref.top_k_top_p_sample(logits, temperature, top_k, top_p, generator)
+
+
+# ------------------------------------------------------------------ KV cache
+
+
+def kv_cache_append(k_cache: torch.Tensor, v_cache: torch.Tensor,
+                    k: torch.Tensor, v: torch.Tensor,
+                    slot_mapping: torch.Tensor):
+    """Scatter new token K/V rows (N, Hk, D) into page slots."""
+    if k_cache.is_cuda:
+        ext = _require_ext()
+        ext.kv_cache_append(k_cache, v_cache, k.contiguous(), v.contiguous(),
+                            slot_mapping.int())
+        return
+    page_size = k_cache.shape[1]
+    flat_k = k_cache.view(-1, *k_cache.shape[2:])
+    flat_v = v_cache.view(-1, *v_cache.shape[2:])
+    valid = slot_mapping >= 0
+    flat_k[slot_mapping[valid].long()] = k[valid]
+    flat_v[slot_mapping[valid].long()] = v[valid]
+
+
+# ----------------------------------------------------------------- attention
+
+
+def paged_attention_decode(q: torch.Tensor, k_cache: torch.Tensor,
+                           v_cache: torch.Tensor, page_table: torch.Tensor,
+                           context_lens: torch.Tensor, scale: float
+                           ) -> torch.Tensor:
+    if q.is_cuda:
+        ext = _require_ext()
+        out = torch.empty_like(q)
+        ext.paged_attention_decode(out, q, k_cache, v_cache,
+                                   page_table.int(), context_lens.int(), scale)
+        return out
+    return ref.paged_attention_decode(q, k_cache, v_cache, page_table,
+                                      context_lens, scale)
+
+
+def _build_prefill_tiles(cu_seqlens_q: torch.Tensor, qtile: int = 64):
+    """tile tables (tile_seq, tile_q0) for the prefill kernel grid."""
+    cu = cu_seqlens_q.cpu()
+    tile_seq, tile_q0 = [], []
+    for s in range(cu.numel() - 1):
+        L = int(cu[s + 1] - cu[s])
+        for q0 in range(0, L, qtile):
+            tile_seq.append(s)
+            tile_q0.append(q0)
+    return (torch.tensor(tile_seq, dtype=torch.int32),
+            torch.tensor(tile_q0, dtype=torch.int32))
+
+
+def varlen_prefill_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                             cu_seqlens_q: torch.Tensor,
+                             cu_seqlens_k: torch.Tensor, scale: float,
+                             causal: bool = True) -> torch.Tensor:
+    if q.is_cuda:
+        ext = _require_ext()
+        out = torch.empty_like(q)
+        tile_seq, tile_q0 = _build_prefill_tiles(cu_seqlens_q)
+        ext.varlen_prefill_attention(
+            out, q, k.contiguous(), v.contiguous(), cu_seqlens_q.int(),
+            cu_seqlens_k.int(), tile_seq.to(q.device), tile_q0.to(q.device),
+            scale, causal)
+        return out
+    return ref.varlen_prefill_attention(q, k, v, cu_seqlens_q, cu_seqlens_k,
+                                        scale, causal)

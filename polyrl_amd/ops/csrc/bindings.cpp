@@ -1,0 +1,46 @@
+// pybind bindings for the polyrl_amd CDNA4 kernel suite.
+#include <torch/extension.h>
+
+void rmsnorm(torch::Tensor out, torch::Tensor x, torch::Tensor weight,
+             double eps);
+void fused_add_rmsnorm(torch::Tensor out, torch::Tensor residual,
+                       torch::Tensor x, torch::Tensor weight, double eps);
+void silu_mul(torch::Tensor out, torch::Tensor gate, torch::Tensor up);
+void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
+                  torch::Tensor cos_tab, torch::Tensor sin_tab);
+void gather_logprobs(torch::Tensor out_lp, torch::Tensor out_ent,
+                     torch::Tensor logits, torch::Tensor labels,
+                     bool want_entropy);
+void sample(torch::Tensor out_tokens, torch::Tensor out_logprobs,
+            torch::Tensor logits, torch::Tensor temperature,
+            torch::Tensor top_k, torch::Tensor top_p, int64_t seed);
+void kv_cache_append(torch::Tensor k_cache, torch::Tensor v_cache,
+                     torch::Tensor k, torch::Tensor v,
+                     torch::Tensor slot_mapping);
+void paged_attention_decode(torch::Tensor out, torch::Tensor q,
+                            torch::Tensor k_cache, torch::Tensor v_cache,
+                            torch::Tensor page_table,
+                            torch::Tensor context_lens, double scale);
+void varlen_prefill_attention(torch::Tensor out, torch::Tensor q,
+                              torch::Tensor k, torch::Tensor v,
+                              torch::Tensor cu_seqlens_q,
+                              torch::Tensor cu_seqlens_k,
+                              torch::Tensor tile_seq, torch::Tensor tile_q0,
+                              double scale, bool causal);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, CDNA4)");
+  m.def("fused_add_rmsnorm", &fused_add_rmsnorm,
+        "residual-add + RMSNorm, residual updated in place");
+  m.def("silu_mul", &silu_mul, "SiLU(gate) * up");
+  m.def("rope_inplace", &rope_inplace, "NEOX RoPE in place (table-driven)");
+  m.def("gather_logprobs", &gather_logprobs,
+        "log-softmax gather at labels (+optional entropy)");
+  m.def("sample", &sample,
+        "fused temperature/top-k/top-p sampling + logprob capture");
+  m.def("kv_cache_append", &kv_cache_append, "paged KV cache scatter-append");
+  m.def("paged_attention_decode", &paged_attention_decode,
+        "paged decode attention (GQA, bf16)");
+  m.def("varlen_prefill_attention", &varlen_prefill_attention,
+        "varlen causal prefill attention (MFMA, bf16)");
+}

@@ -1,0 +1,43 @@
+"""In-tree build of the polyrl_amd HIP extension for gfx950 (MI355X).
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+The built .so lands next to the package (polyrl_amd/_hip_C*.so) so it travels
+with a repo snapshot to GPU boxes.
+"""
+import os
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from setuptools import setup
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension
+
+CSRC = os.path.join("polyrl_amd", "ops", "csrc")
+
+sources = [
+    os.path.join(CSRC, f)
+    for f in (
+        "bindings.cpp",
+        "elementwise.hip",
+        "logprobs.hip",
+        "sampling.hip",
+        "kv_cache.hip",
+        "attention_decode.hip",
+        "attention_prefill.hip",
+    )
+]
+
+setup(
+    name="polyrl_amd_hip",
+    ext_modules=[
+        CUDAExtension(
+            name="polyrl_amd._hip",
+            sources=sources,
+            extra_compile_args={
+                "cxx": ["-O3", "-std=c++17"],
+                "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+            },
+        )
+    ],
+    cmdclass={"build_ext": BuildExtension.with_options(no_python_abi_suffix=False)},
+)

@@ -1,0 +1,245 @@
+"""GPU numerics: every HIP kernel vs the plain-PyTorch fp32 reference (ops.ref).
+
+All tests are @pytest.mark.gpu and run on a real MI355X via gpurun / the
+driver's round-end GPU tier.
+"""
+import math
+
+import pytest
+import torch
+
+import polyrl_amd.ops as ops
+from polyrl_amd.ops import ref
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda"
+
+
+def setup_module():
+    assert ops.extension_loaded(), "HIP extension must be loaded on GPU"
+
+
+def rel_err(a, b):
+    a, b = a.float(), b.float()
+    return ((a - b).abs().max() / (b.abs().max() + 1e-6)).item()
+
+
+# ---------------------------------------------------------------- norms/elem
+
+
+@pytest.mark.parametrize("shape", [(4, 512), (37, 4096), (128, 8192)])
+def test_rmsnorm(shape):
+    torch.manual_seed(0)
+    x = torch.randn(shape, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(shape[-1], dtype=torch.bfloat16, device=DEV)
+    out = ops.rmsnorm(x, w)
+    expect = ref.rmsnorm(x.cpu(), w.cpu())
+    assert rel_err(out.cpu(), expect) < 2e-2
+
+
+def test_fused_add_rmsnorm():
+    torch.manual_seed(1)
+    x = torch.randn(33, 2048, dtype=torch.bfloat16, device=DEV)
+    res = torch.randn_like(x)
+    w = torch.randn(2048, dtype=torch.bfloat16, device=DEV)
+    expect_norm, expect_res = ref.fused_add_rmsnorm(x.cpu(), res.cpu().clone(), w.cpu())
+    out, new_res = ops.fused_add_rmsnorm(x, res, w)
+    assert rel_err(new_res.cpu(), expect_res) < 2e-2
+    assert rel_err(out.cpu(), expect_norm) < 2e-2
+
+
+def test_silu_mul():
+    torch.manual_seed(2)
+    g = torch.randn(64, 5632, dtype=torch.bfloat16, device=DEV)
+    u = torch.randn_like(g)
+    out = ops.silu_mul(g, u)
+    expect = ref.silu_mul(g.cpu(), u.cpu())
+    assert rel_err(out.cpu(), expect) < 2e-2
+
+
+def test_rope():
+    torch.manual_seed(3)
+    N, Hq, Hk, D = 40, 8, 2, 128
+    q = torch.randn(N, Hq, D, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(N, Hk, D, dtype=torch.bfloat16, device=DEV)
+    pos = torch.randint(0, 1000, (N,), dtype=torch.int32, device=DEV)
+    table = ops.RopeTable(D, 2048, 10000.0, DEV)
+    cos, sin = ref.rope_cos_sin(pos.cpu(), D, 10000.0)
+    eq, ek = ref.apply_rope(q.cpu(), k.cpu(), cos, sin)
+    ops.apply_rope_inplace(q, k, pos, table)
+    assert rel_err(q.cpu(), eq) < 2e-2
+    assert rel_err(k.cpu(), ek) < 2e-2
+
+
+# ------------------------------------------------------------------ logprobs
+
+
+@pytest.mark.parametrize("V", [128, 32000, 151936])
+def test_gather_logprobs(V):
+    torch.manual_seed(4)
+    N = 33
+    logits = torch.randn(N, V, dtype=torch.bfloat16, device=DEV) * 4
+    labels = torch.randint(0, V, (N,), device=DEV)
+    lp, ent = ops.gather_logprobs(logits, labels, want_entropy=True)
+    elp, eent = ref.gather_logprobs_entropy(logits.cpu(), labels.cpu())
+    assert torch.allclose(lp.cpu(), elp, atol=1e-2, rtol=1e-3)
+    assert torch.allclose(ent.cpu(), eent, atol=1e-2, rtol=1e-3)
+
+
+def test_gather_logprobs_fp32():
+    torch.manual_seed(5)
+    logits = torch.randn(7, 5000, device=DEV) * 3
+    labels = torch.randint(0, 5000, (7,), device=DEV)
+    lp = ops.gather_logprobs(logits, labels)
+    elp = ref.gather_logprobs(logits.cpu(), labels.cpu())
+    assert torch.allclose(lp.cpu(), elp, atol=1e-4, rtol=1e-5)
+
+
+# ------------------------------------------------------------------ sampling
+
+
+def test_sample_greedy_exact():
+    torch.manual_seed(6)
+    N, V = 16, 32000
+    logits = torch.randn(N, V, dtype=torch.bfloat16, device=DEV) * 3
+    temp = torch.zeros(N, device=DEV)
+    tk = torch.full((N,), -1, dtype=torch.int32, device=DEV)
+    tp = torch.ones(N, device=DEV)
+    tokens, lps = ops.sample(logits, temp, tk, tp, seed=7)
+    expect = logits.float().argmax(-1)
+    assert torch.equal(tokens, expect)
+    elp = ref.gather_logprobs(logits.cpu(), expect.cpu())
+    assert torch.allclose(lps.cpu(), elp, atol=1e-2, rtol=1e-3)
+
+
+def test_sample_top_k_stays_in_set():
+    torch.manual_seed(7)
+    N, V, K = 64, 4096, 20
+    logits = (torch.randn(N, V, device=DEV) * 5).bfloat16()
+    temp = torch.ones(N, device=DEV)
+    tk = torch.full((N,), K, dtype=torch.int32, device=DEV)
+    tp = torch.ones(N, device=DEV)
+    topk_sets = torch.topk(logits.float(), K, dim=-1).indices
+    for seed in range(5):
+        tokens, _ = ops.sample(logits, temp, tk, tp, seed=seed)
+        for i in range(N):
+            # allow boundary ties: sampled logit >= k-th largest logit
+            kth = logits[i].float()[topk_sets[i][-1]]
+            assert logits[i, tokens[i]].float() >= kth - 1e-3
+
+
+def test_sample_top_p_mass():
+    torch.manual_seed(8)
+    N, V = 32, 1024
+    logits = (torch.randn(N, V, device=DEV) * 6).bfloat16()
+    temp = torch.ones(N, device=DEV)
+    tk = torch.full((N,), -1, dtype=torch.int32, device=DEV)
+    tp = torch.full((N,), 0.7, device=DEV)
+    probs = torch.softmax(logits.float(), -1)
+    sorted_p, sorted_idx = probs.sort(-1, descending=True)
+    cum = sorted_p.cumsum(-1)
+    for seed in range(5):
+        tokens, _ = ops.sample(logits, temp, tk, tp, seed=seed)
+        for i in range(N):
+            rank = (sorted_idx[i] == tokens[i]).nonzero().item()
+            # sampled token must be inside (or at the boundary of) the nucleus
+            before = cum[i, rank - 1].item() if rank > 0 else 0.0
+            assert before < 0.7 + 2e-2, f"row {i}: mass before token {before}"
+
+
+def test_sample_categorical_distribution():
+    """statistical check: sampled frequencies track softmax probs on small V"""
+    V = 8
+    logits = torch.tensor([[0., 1., 2., 0.5, -1., 3., 0., 1.5]],
+                          device=DEV).bfloat16()
+    logits = logits.repeat(4096, 1).contiguous()
+    temp = torch.ones(4096, device=DEV)
+    tk = torch.full((4096,), -1, dtype=torch.int32, device=DEV)
+    tp = torch.ones(4096, device=DEV)
+    counts = torch.zeros(V)
+    for seed in range(4):
+        tokens, _ = ops.sample(logits, temp, tk, tp, seed=seed * 7919 + 13)
+        counts += torch.bincount(tokens.cpu(), minlength=V).float()
+    freq = counts / counts.sum()
+    expect = torch.softmax(logits[0].float().cpu(), -1)
+    assert (freq - expect).abs().max().item() < 0.02
+
+
+# ------------------------------------------------------------------ KV cache
+
+
+def test_kv_cache_append():
+    torch.manual_seed(9)
+    num_pages, page_size, Hk, D = 16, 16, 4, 128
+    kc = torch.zeros(num_pages, page_size, Hk, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.zeros_like(kc)
+    N = 23
+    k = torch.randn(N, Hk, D, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn_like(k)
+    slots = torch.randperm(num_pages * page_size, device=DEV)[:N].int()
+    ops.kv_cache_append(kc, vc, k, v, slots)
+    flat = kc.view(-1, Hk, D)
+    for i in range(N):
+        assert torch.equal(flat[slots[i].long()], k[i])
+
+
+# ----------------------------------------------------------------- attention
+
+
+@pytest.mark.parametrize("Hq,Hk,D", [(8, 8, 128), (8, 2, 128), (32, 8, 128),
+                                     (12, 2, 128), (28, 4, 128), (12, 12, 64)])
+def test_paged_attention_decode(Hq, Hk, D):
+    torch.manual_seed(10)
+    B, page_size = 9, 16
+    lens = torch.tensor([1, 5, 16, 17, 60, 64, 100, 255, 300], dtype=torch.int32)
+    max_pages = int((-(-lens.max().item() // page_size)))
+    num_pages = int(sum(-(-int(l) // page_size) for l in lens)) + 4
+    kc = torch.randn(num_pages, page_size, Hk, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.randn_like(kc)
+    # random page table (distinct pages)
+    perm = torch.randperm(num_pages)
+    pt = torch.zeros(B, max_pages, dtype=torch.int32)
+    pi = 0
+    for b in range(B):
+        for j in range(-(-int(lens[b]) // page_size)):
+            pt[b, j] = perm[pi]
+            pi += 1
+    q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device=DEV)
+    scale = 1.0 / math.sqrt(D)
+    out = ops.paged_attention_decode(q, kc, vc, pt.to(DEV), lens.to(DEV), scale)
+    expect = ref.paged_attention_decode(q.cpu(), kc.cpu(), vc.cpu(), pt,
+                                        lens, scale)
+    assert rel_err(out.cpu(), expect) < 3e-2
+
+
+@pytest.mark.parametrize("Hq,Hk", [(8, 8), (8, 2), (32, 8)])
+def test_varlen_prefill_attention(Hq, Hk):
+    torch.manual_seed(11)
+    D = 128
+    seqlens = [1, 17, 64, 63, 200, 256]
+    cu = torch.tensor([0] + list(torch.tensor(seqlens).cumsum(0)), dtype=torch.int32)
+    total = int(cu[-1])
+    q = torch.randn(total, Hq, D, dtype=torch.bfloat16, device=DEV) / 4
+    k = torch.randn(total, Hk, D, dtype=torch.bfloat16, device=DEV) / 4
+    v = torch.randn(total, Hk, D, dtype=torch.bfloat16, device=DEV)
+    scale = 1.0 / math.sqrt(D)
+    out = ops.varlen_prefill_attention(q, k, v, cu.to(DEV), cu.to(DEV), scale)
+    expect = ref.varlen_prefill_attention(q.cpu(), k.cpu(), v.cpu(), cu, cu, scale)
+    assert rel_err(out.cpu(), expect) < 3e-2
+
+
+def test_varlen_prefill_chunked_alignment():
+    """seqlen_k > seqlen_q: query block aligned to the end of keys."""
+    torch.manual_seed(12)
+    Hq, Hk, D = 8, 2, 128
+    cu_q = torch.tensor([0, 32, 96], dtype=torch.int32)
+    cu_k = torch.tensor([0, 100, 228], dtype=torch.int32)
+    q = torch.randn(96, Hq, D, dtype=torch.bfloat16, device=DEV) / 4
+    k = torch.randn(228, Hk, D, dtype=torch.bfloat16, device=DEV) / 4
+    v = torch.randn(228, Hk, D, dtype=torch.bfloat16, device=DEV)
+    scale = 1.0 / math.sqrt(D)
+    out = ops.varlen_prefill_attention(q, k, v, cu_q.to(DEV), cu_k.to(DEV), scale)
+    expect = ref.varlen_prefill_attention(q.cpu(), k.cpu(), v.cpu(), cu_q, cu_k,
+                                          scale)
+    assert rel_err(out.cpu(), expect) < 3e-2
