@@ -75,7 +75,7 @@ def prepare_dynamic_batch(batch: TensorBatch, max_token_len: int
     Returns (micro_batches, index_lists); restore_dynamic_batch inverts it.
     """
     attention_mask = batch["attention_mask"]
-    seqlens = attention_mask.sum(dim=-1).tolist()
+    seqlens = [int(s) for s in attention_mask.sum(dim=-1).tolist()]
     max_seq = max(seqlens) if seqlens else 0
     assert max_seq <= max_token_len, \
         f"one sample has {max_seq} tokens > budget {max_token_len}"
