@@ -1,0 +1,256 @@
+"""Native dense decoder family (Llama-3 / Qwen2.5) for training and rollout.
+
+One weight set, two execution paths:
+  * training forward (this file): PyTorch-ROCm autograd path — SDPA attention,
+    used by the FSDP actor/critic workers for fwd/bwd.
+  * rollout forward (rollout/engine.py): no-autograd paged path on the
+    hand-written CDNA4 HIP kernels, reading the SAME parameters.
+
+Parameter names are HF-compatible (model.layers.N.self_attn.q_proj.weight ...)
+so the checkpoint layout interchanges (SURVEY.md §5.4 north star).
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from .registry import DecoderConfig
+
+
+def _rope_rotate_half(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor):
+    """NEOX-style RoPE: x (B, H, L, D), cos/sin (L, D/2)."""
+    d = x.shape[-1] // 2
+    x1, x2 = x[..., :d], x[..., d:]
+    c = cos.view(1, 1, *cos.shape)
+    s = sin.view(1, 1, *sin.shape)
+    return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], dim=-1)
+
+
+class RMSNorm(nn.Module):
+    def __init__(self, dim: int, eps: float):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(dim))
+        self.variance_epsilon = eps
+
+    def forward(self, x):
+        dt = x.dtype
+        xf = x.float()
+        var = xf.pow(2).mean(-1, keepdim=True)
+        return (xf * torch.rsqrt(var + self.variance_epsilon)).to(dt) * self.weight
+
+
+class RotaryCache(nn.Module):
+    """fp32 cos/sin cache, grown on demand."""
+
+    def __init__(self, head_dim: int, theta: float):
+        super().__init__()
+        self.head_dim = head_dim
+        self.theta = theta
+        self.register_buffer("cos_cached", torch.empty(0), persistent=False)
+        self.register_buffer("sin_cached", torch.empty(0), persistent=False)
+
+    def get(self, positions: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+        need = int(positions.max().item()) + 1 if positions.numel() else 1
+        if self.cos_cached.numel() == 0 or self.cos_cached.shape[0] < need or \
+                self.cos_cached.device != positions.device:
+            n = max(need, 512)
+            inv = 1.0 / (self.theta ** (
+                torch.arange(0, self.head_dim, 2, device=positions.device,
+                             dtype=torch.float32) / self.head_dim))
+            t = torch.arange(n, device=positions.device, dtype=torch.float32)
+            freqs = torch.outer(t, inv)
+            self.cos_cached = freqs.cos()
+            self.sin_cached = freqs.sin()
+        idx = positions.long()
+        return self.cos_cached[idx], self.sin_cached[idx]
+
+
+class Attention(nn.Module):
+    def __init__(self, cfg: DecoderConfig):
+        super().__init__()
+        self.num_heads = cfg.num_attention_heads
+        self.num_kv_heads = cfg.num_key_value_heads
+        self.head_dim = cfg.head_dim
+        h = cfg.hidden_size
+        bias = cfg.attention_bias
+        self.q_proj = nn.Linear(h, self.num_heads * self.head_dim, bias=bias)
+        self.k_proj = nn.Linear(h, self.num_kv_heads * self.head_dim, bias=bias)
+        self.v_proj = nn.Linear(h, self.num_kv_heads * self.head_dim, bias=bias)
+        self.o_proj = nn.Linear(self.num_heads * self.head_dim, h, bias=False)
+
+    def forward(self, x, cos, sin, attn_bias_mask: Optional[torch.Tensor]):
+        B, L, _ = x.shape
+        q = self.q_proj(x).view(B, L, self.num_heads, self.head_dim).transpose(1, 2)
+        k = self.k_proj(x).view(B, L, self.num_kv_heads, self.head_dim).transpose(1, 2)
+        v = self.v_proj(x).view(B, L, self.num_kv_heads, self.head_dim).transpose(1, 2)
+        cos = cos.to(q.dtype)
+        sin = sin.to(q.dtype)
+        q = _rope_rotate_half(q, cos, sin)
+        k = _rope_rotate_half(k, cos, sin)
+        if self.num_kv_heads != self.num_heads:
+            rep = self.num_heads // self.num_kv_heads
+            k = k.repeat_interleave(rep, dim=1)
+            v = v.repeat_interleave(rep, dim=1)
+        if attn_bias_mask is not None:
+            o = F.scaled_dot_product_attention(q, k, v, attn_mask=attn_bias_mask)
+        else:
+            o = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+        o = o.transpose(1, 2).reshape(B, L, -1)
+        return self.o_proj(o)
+
+
+class MLP(nn.Module):
+    def __init__(self, cfg: DecoderConfig):
+        super().__init__()
+        h, i = cfg.hidden_size, cfg.intermediate_size
+        self.gate_proj = nn.Linear(h, i, bias=cfg.mlp_bias)
+        self.up_proj = nn.Linear(h, i, bias=cfg.mlp_bias)
+        self.down_proj = nn.Linear(i, h, bias=cfg.mlp_bias)
+
+    def forward(self, x):
+        return self.down_proj(F.silu(self.gate_proj(x)) * self.up_proj(x))
+
+
+class DecoderLayer(nn.Module):
+    def __init__(self, cfg: DecoderConfig):
+        super().__init__()
+        self.self_attn = Attention(cfg)
+        self.mlp = MLP(cfg)
+        self.input_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
+        self.post_attention_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
+
+    def forward(self, x, cos, sin, attn_bias_mask):
+        x = x + self.self_attn(self.input_layernorm(x), cos, sin, attn_bias_mask)
+        x = x + self.mlp(self.post_attention_layernorm(x))
+        return x
+
+
+class DecoderModel(nn.Module):
+    def __init__(self, cfg: DecoderConfig):
+        super().__init__()
+        self.config = cfg
+        self.embed_tokens = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
+        self.layers = nn.ModuleList(
+            DecoderLayer(cfg) for _ in range(cfg.num_hidden_layers))
+        self.norm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
+        self.rotary = RotaryCache(cfg.head_dim, cfg.rope_theta)
+        self.gradient_checkpointing = False
+
+    def forward(self, input_ids, attention_mask=None, position_ids=None):
+        B, L = input_ids.shape
+        if position_ids is None:
+            position_ids = torch.arange(L, device=input_ids.device).expand(B, L)
+        # rope tables per position row (assume shared positions across batch
+        # when position_ids rows are identical — the common padded case needs
+        # per-row tables)
+        cos, sin = self.rotary.get(position_ids.reshape(-1))
+        cos = cos.view(B, L, -1)
+        sin = sin.view(B, L, -1)
+
+        attn_bias_mask = None
+        if attention_mask is not None and not bool(attention_mask.all()):
+            # build additive mask: causal + padding
+            causal = torch.ones(L, L, dtype=torch.bool,
+                                device=input_ids.device).tril()
+            pad = attention_mask.bool().view(B, 1, 1, L)
+            attn_bias_mask = (causal.view(1, 1, L, L) & pad)
+
+        x = self.embed_tokens(input_ids)
+        for layer in self.layers:
+            if self.gradient_checkpointing and self.training:
+                x = torch.utils.checkpoint.checkpoint(
+                    _layer_call, layer, x, cos, sin, attn_bias_mask,
+                    use_reentrant=False)
+            else:
+                x = _layer_call(layer, x, cos, sin, attn_bias_mask)
+        return self.norm(x)
+
+
+def _layer_call(layer, x, cos, sin, attn_bias_mask):
+    # per-row cos/sin: reshape to broadcast (B, 1, L, D/2)
+    B, L = x.shape[0], x.shape[1]
+    c = cos.view(B, 1, L, -1)
+    s = sin.view(B, 1, L, -1)
+    return _layer_forward(layer, x, c, s, attn_bias_mask)
+
+
+def _layer_forward(layer, x, c, s, attn_bias_mask):
+    h = layer.input_layernorm(x)
+    B, L, _ = h.shape
+    att = layer.self_attn
+    q = att.q_proj(h).view(B, L, att.num_heads, att.head_dim).transpose(1, 2)
+    k = att.k_proj(h).view(B, L, att.num_kv_heads, att.head_dim).transpose(1, 2)
+    v = att.v_proj(h).view(B, L, att.num_kv_heads, att.head_dim).transpose(1, 2)
+    cq = c.to(q.dtype)
+    sq = s.to(q.dtype)
+    d = q.shape[-1] // 2
+    q = torch.cat([q[..., :d] * cq - q[..., d:] * sq,
+                   q[..., d:] * cq + q[..., :d] * sq], dim=-1)
+    k = torch.cat([k[..., :d] * cq - k[..., d:] * sq,
+                   k[..., d:] * cq + k[..., :d] * sq], dim=-1)
+    if att.num_kv_heads != att.num_heads:
+        rep = att.num_heads // att.num_kv_heads
+        k = k.repeat_interleave(rep, dim=1)
+        v = v.repeat_interleave(rep, dim=1)
+    if attn_bias_mask is not None:
+        o = F.scaled_dot_product_attention(q, k, v, attn_mask=attn_bias_mask)
+    else:
+        o = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+    o = o.transpose(1, 2).reshape(B, L, -1)
+    x = x + att.o_proj(o)
+    x = x + layer.mlp(layer.post_attention_layernorm(x))
+    return x
+
+
+class CausalLM(nn.Module):
+    """Actor / reference policy."""
+
+    def __init__(self, cfg: DecoderConfig):
+        super().__init__()
+        self.config = cfg
+        self.model = DecoderModel(cfg)
+        self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias=False)
+        if cfg.tie_word_embeddings:
+            self.lm_head.weight = self.model.embed_tokens.weight
+        self.apply(self._init_weights)
+
+    def _init_weights(self, m):
+        if isinstance(m, nn.Linear):
+            nn.init.normal_(m.weight, std=0.02)
+            if m.bias is not None:
+                nn.init.zeros_(m.bias)
+        elif isinstance(m, nn.Embedding):
+            nn.init.normal_(m.weight, std=0.02)
+
+    def gradient_checkpointing_enable(self):
+        self.model.gradient_checkpointing = True
+
+    def forward(self, input_ids, attention_mask=None, position_ids=None):
+        hidden = self.model(input_ids, attention_mask, position_ids)
+        return self.lm_head(hidden)
+
+    @torch.no_grad()
+    def num_params(self) -> int:
+        return sum(p.numel() for p in self.parameters())
+
+
+class CausalLMWithValueHead(nn.Module):
+    """Critic: decoder trunk + scalar value head (token-level values)."""
+
+    def __init__(self, cfg: DecoderConfig):
+        super().__init__()
+        self.config = cfg
+        self.model = DecoderModel(cfg)
+        self.value_head = nn.Linear(cfg.hidden_size, 1, bias=False)
+        nn.init.normal_(self.value_head.weight, std=0.02 / math.sqrt(cfg.hidden_size))
+
+    def gradient_checkpointing_enable(self):
+        self.model.gradient_checkpointing = True
+
+    def forward(self, input_ids, attention_mask=None, position_ids=None):
+        hidden = self.model(input_ids, attention_mask, position_ids)
+        return self.value_head(hidden).squeeze(-1)  # (B, L)

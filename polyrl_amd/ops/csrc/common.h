@@ -55,6 +55,8 @@ DEV_INLINE float group16_reduce_sum(float x) {
 }
 
 // block reduce over NW waves through LDS; every thread returns the result.
+// The final value is re-broadcast through LDS: a shfl only reaches the wave
+// that computed it, not the other waves of the block.
 template <int NW>
 DEV_INLINE float block_reduce_sum(float x, float* lds_scratch) {
   int lane = threadIdx.x & (WAVE_SIZE - 1);
@@ -62,11 +64,14 @@ DEV_INLINE float block_reduce_sum(float x, float* lds_scratch) {
   x = wave_reduce_sum(x);
   if (lane == 0) lds_scratch[wid] = x;
   __syncthreads();
-  float r = (threadIdx.x < NW) ? lds_scratch[threadIdx.x] : 0.f;
+  if (threadIdx.x == 0) {
+    float r = lds_scratch[0];
 #pragma unroll
-  for (int off = NW / 2; off > 0; off >>= 1) r += __shfl_xor(r, off, WAVE_SIZE);
-  r = __shfl(r, 0, WAVE_SIZE);
-  return r;
+    for (int w = 1; w < NW; ++w) r += lds_scratch[w];
+    lds_scratch[0] = r;
+  }
+  __syncthreads();
+  return lds_scratch[0];
 }
 
 template <int NW>
@@ -76,12 +81,14 @@ DEV_INLINE float block_reduce_max(float x, float* lds_scratch) {
   x = wave_reduce_max(x);
   if (lane == 0) lds_scratch[wid] = x;
   __syncthreads();
-  float r = (threadIdx.x < NW) ? lds_scratch[threadIdx.x] : -INFINITY;
+  if (threadIdx.x == 0) {
+    float r = lds_scratch[0];
 #pragma unroll
-  for (int off = NW / 2; off > 0; off >>= 1)
-    r = fmaxf(r, __shfl_xor(r, off, WAVE_SIZE));
-  r = __shfl(r, 0, WAVE_SIZE);
-  return r;
+    for (int w = 1; w < NW; ++w) r = fmaxf(r, lds_scratch[w]);
+    lds_scratch[0] = r;
+  }
+  __syncthreads();
+  return lds_scratch[0];
 }
 
 #define HIP_CHECK_KERNEL()                                      \

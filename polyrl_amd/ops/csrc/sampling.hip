@@ -54,17 +54,24 @@ DEV_INLINE ArgMax block_argmax(float v, int idx, float* lds_v, int* lds_i) {
   }
   if (lane == 0) { lds_v[wid] = v; lds_i[wid] = idx; }
   __syncthreads();
-  float rv = (threadIdx.x < NW) ? lds_v[threadIdx.x] : -INFINITY;
-  int ri = (threadIdx.x < NW) ? lds_i[threadIdx.x] : 0;
+  // re-broadcast through LDS: a shfl would only reach wave 0
+  if (threadIdx.x == 0) {
+    float rv = lds_v[0];
+    int ri = lds_i[0];
 #pragma unroll
-  for (int off = NW / 2; off > 0; off >>= 1) {
-    float ov = __shfl_xor(rv, off, 64);
-    int oi = __shfl_xor(ri, off, 64);
-    if (ov > rv || (ov == rv && oi < ri)) { rv = ov; ri = oi; }
+    for (int w = 1; w < NW; ++w) {
+      if (lds_v[w] > rv || (lds_v[w] == rv && lds_i[w] < ri)) {
+        rv = lds_v[w];
+        ri = lds_i[w];
+      }
+    }
+    lds_v[0] = rv;
+    lds_i[0] = ri;
   }
+  __syncthreads();
   ArgMax out;
-  out.v = __shfl(rv, 0, 64);
-  out.i = __shfl(ri, 0, 64);
+  out.v = lds_v[0];
+  out.i = lds_i[0];
   return out;
 }
 

@@ -1,0 +1,478 @@
+"""In-process continuous-batching decoder for MI355X.
+
+Replaces the reference's external inference engine (SGLang
+behind HTTP).  Token-in / token-out.  The model forward here is a separate
+no-autograd path over the hand-written HIP kernels (ops/): fused-add-RMSNorm,
+table-RoPE, paged decode attention, MFMA varlen prefill attention, SiLU-mul,
+fused sampling with logprob capture; projections on hipBLASLt via torch.matmul
+against fused QKV / gate-up weight buffers.
+
+Engine weights live in their own contiguous per-layer buffers, decoupled from
+the trainer's FSDP parameters, so weight-version updates are plain copies
+(transfer/weight_transfer.py) gated against in-flight generation.
+"""
+from __future__ import annotations
+
+import math
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+import torch
+
+import polyrl_amd.ops as ops
+from ..models.registry import DecoderConfig
+from .kv_cache import PagedKVCache
+
+
+@dataclass
+class SamplingParams:
+    temperature: float = 1.0
+    top_k: int = -1
+    top_p: float = 1.0
+    max_new_tokens: int = 128
+    stop_token_ids: tuple = ()
+
+
+@dataclass
+class Request:
+    rid: str
+    input_ids: List[int]
+    sampling: SamplingParams
+    # runtime state
+    output_ids: List[int] = field(default_factory=list)
+    output_logprobs: List[float] = field(default_factory=list)
+    prefill_pos: int = 0          # tokens already prefilled
+    finished: bool = False
+    finish_reason: str = ""       # stop | length | abort
+    arrival_t: float = 0.0
+
+    @property
+    def seq_len(self) -> int:
+        return len(self.input_ids) + len(self.output_ids)
+
+
+@dataclass
+class RequestOutput:
+    rid: str
+    output_ids: List[int]
+    output_logprobs: List[float]
+    finish_reason: str
+
+
+class LayerWeights:
+    """Fused per-layer weight buffers (views kept for named updates)."""
+
+    def __init__(self, cfg: DecoderConfig, device, dtype):
+        h = cfg.hidden_size
+        hq = cfg.num_attention_heads * cfg.head_dim
+        hk = cfg.num_key_value_heads * cfg.head_dim
+        i = cfg.intermediate_size
+        self.wqkv = torch.empty(hq + 2 * hk, h, device=device, dtype=dtype)
+        self.bqkv = (torch.zeros(hq + 2 * hk, device=device, dtype=dtype)
+                     if cfg.attention_bias else None)
+        self.wo = torch.empty(h, hq, device=device, dtype=dtype)
+        self.w_gate_up = torch.empty(2 * i, h, device=device, dtype=dtype)
+        self.w_down = torch.empty(h, i, device=device, dtype=dtype)
+        self.input_ln = torch.empty(h, device=device, dtype=dtype)
+        self.post_ln = torch.empty(h, device=device, dtype=dtype)
+        self._hq, self._hk, self._i = hq, hk, i
+
+    def named_slices(self, prefix: str) -> Dict[str, torch.Tensor]:
+        hq, hk, i = self._hq, self._hk, self._i
+        out = {
+            f"{prefix}.self_attn.q_proj.weight": self.wqkv[:hq],
+            f"{prefix}.self_attn.k_proj.weight": self.wqkv[hq:hq + hk],
+            f"{prefix}.self_attn.v_proj.weight": self.wqkv[hq + hk:],
+            f"{prefix}.self_attn.o_proj.weight": self.wo,
+            f"{prefix}.mlp.gate_proj.weight": self.w_gate_up[:i],
+            f"{prefix}.mlp.up_proj.weight": self.w_gate_up[i:],
+            f"{prefix}.mlp.down_proj.weight": self.w_down,
+            f"{prefix}.input_layernorm.weight": self.input_ln,
+            f"{prefix}.post_attention_layernorm.weight": self.post_ln,
+        }
+        if self.bqkv is not None:
+            out[f"{prefix}.self_attn.q_proj.bias"] = self.bqkv[:hq]
+            out[f"{prefix}.self_attn.k_proj.bias"] = self.bqkv[hq:hq + hk]
+            out[f"{prefix}.self_attn.v_proj.bias"] = self.bqkv[hq + hk:]
+        return out
+
+
+class InferenceModel:
+    """No-autograd decoder forward over paged KV on the HIP kernel suite."""
+
+    def __init__(self, cfg: DecoderConfig, device="cuda",
+                 dtype=torch.bfloat16):
+        assert cfg.arch in ("llama", "qwen2"), \
+            "rollout engine serves the llama/qwen2 family"
+        self.cfg = cfg
+        self.device = device
+        self.dtype = dtype
+        self.layers = [LayerWeights(cfg, device, dtype)
+                       for _ in range(cfg.num_hidden_layers)]
+        h = cfg.hidden_size
+        self.embed = torch.empty(cfg.vocab_size, h, device=device, dtype=dtype)
+        self.final_norm = torch.empty(h, device=device, dtype=dtype)
+        self.lm_head = torch.empty(cfg.vocab_size, h, device=device, dtype=dtype)
+        self.rope = ops.RopeTable(cfg.head_dim, cfg.max_position_embeddings,
+                                  cfg.rope_theta, device)
+        # name -> engine buffer view, for weight updates
+        self._name_map: Dict[str, torch.Tensor] = {
+            "model.embed_tokens.weight": self.embed,
+            "model.norm.weight": self.final_norm,
+            "lm_head.weight": self.lm_head,
+        }
+        for li, lw in enumerate(self.layers):
+            self._name_map.update(lw.named_slices(f"model.layers.{li}"))
+
+    def weight_bytes(self) -> int:
+        return sum(v.numel() * v.element_size() for v in self._name_map.values())
+
+    @torch.no_grad()
+    def load_state_dict(self, sd: Dict[str, torch.Tensor], strict: bool = True):
+        seen = set()
+        tied = self.cfg.tie_word_embeddings
+        for name, buf in self._name_map.items():
+            src = sd.get(name)
+            if src is None and name == "lm_head.weight" and tied:
+                src = sd.get("model.embed_tokens.weight")
+            if src is None:
+                if strict:
+                    raise KeyError(f"missing weight {name}")
+                continue
+            buf.copy_(src.to(device=buf.device, dtype=buf.dtype,
+                             non_blocking=True))
+            seen.add(name)
+        if strict and len(seen) != len(self._name_map):
+            missing = set(self._name_map) - seen
+            raise KeyError(f"missing weights: {sorted(missing)[:5]} ...")
+
+    @torch.no_grad()
+    def update_named(self, name: str, tensor: torch.Tensor) -> bool:
+        buf = self._name_map.get(name)
+        if buf is None:
+            return False
+        buf.copy_(tensor.to(device=buf.device, dtype=buf.dtype,
+                            non_blocking=True))
+        return True
+
+    # ------------------------------------------------------------- forward
+    @torch.no_grad()
+    def forward_tokens(self, token_ids: torch.Tensor, positions: torch.Tensor,
+                       kv: PagedKVCache, slot_mapping: torch.Tensor,
+                       attn_fn) -> torch.Tensor:
+        """Shared trunk: embed -> L x (norm, qkv, rope, kv-append, attn(fn),
+        o-proj, norm, mlp) -> final norm.  Returns hidden (N, H)."""
+        cfg = self.cfg
+        Hq, Hk, D = cfg.num_attention_heads, cfg.num_key_value_heads, cfg.head_dim
+        x = self.embed[token_ids]                     # (N, H) gather
+        residual = x.clone()
+        hidden = None
+        for li, lw in enumerate(self.layers):
+            if li == 0:
+                hidden = ops.rmsnorm(residual, lw.input_ln, cfg.rms_norm_eps)
+            else:
+                hidden, residual = ops.fused_add_rmsnorm(
+                    hidden, residual, lw.input_ln, cfg.rms_norm_eps)
+            qkv = hidden @ lw.wqkv.t()
+            if lw.bqkv is not None:
+                qkv = qkv + lw.bqkv
+            q, k, v = qkv.split([Hq * D, Hk * D, Hk * D], dim=-1)
+            q = q.view(-1, Hq, D).contiguous()
+            k = k.view(-1, Hk, D).contiguous()
+            v = v.view(-1, Hk, D).contiguous()
+            ops.apply_rope_inplace(q, k, positions, self.rope)
+            ops.kv_cache_append(kv.k_cache[li], kv.v_cache[li], k, v,
+                                slot_mapping)
+            attn_out = attn_fn(li, q, k, v)           # (N, Hq, D)
+            hidden = attn_out.view(-1, Hq * D) @ lw.wo.t()
+            hidden, residual = ops.fused_add_rmsnorm(
+                hidden, residual, lw.post_ln, cfg.rms_norm_eps)
+            gate_up = hidden @ lw.w_gate_up.t()
+            gate, up = gate_up.split(cfg.intermediate_size, dim=-1)
+            hidden = ops.silu_mul(gate.contiguous(), up.contiguous()) @ lw.w_down.t()
+        # final residual add + norm
+        final = (hidden.float() + residual.float()).to(self.dtype)
+        return ops.rmsnorm(final, self.final_norm, cfg.rms_norm_eps)
+
+    @torch.no_grad()
+    def logits(self, hidden: torch.Tensor) -> torch.Tensor:
+        return hidden @ self.lm_head.t()
+
+
+class Engine:
+    """Continuous-batching engine: waiting queue + running batch + step()."""
+
+    def __init__(self, cfg: DecoderConfig, device="cuda",
+                 dtype=torch.bfloat16, page_size: int = 16,
+                 kv_bytes_budget: Optional[int] = None,
+                 max_running_requests: int = 256,
+                 max_num_batched_tokens: int = 8192,
+                 max_model_len: Optional[int] = None,
+                 seed: int = 0):
+        self.cfg = cfg
+        self.device = device
+        self.dtype = dtype
+        self.model = InferenceModel(cfg, device, dtype)
+        self.max_running = max_running_requests
+        self.max_batched_tokens = max_num_batched_tokens
+        self.max_model_len = max_model_len or cfg.max_position_embeddings
+        bt = PagedKVCache.bytes_per_token(cfg.num_hidden_layers,
+                                          cfg.num_key_value_heads, cfg.head_dim)
+        if kv_bytes_budget is None:
+            kv_bytes_budget = 1 << 30  # 1 GiB default (tests); callers size it
+        num_pages = max(int(kv_bytes_budget // (bt * page_size)), 8)
+        self.kv = PagedKVCache(cfg.num_hidden_layers, cfg.num_key_value_heads,
+                               cfg.head_dim, num_pages, page_size,
+                               dtype=dtype, device=device)
+        self.waiting: List[Request] = []
+        self.running: List[Request] = []
+        self._seq_counter = 0
+        self._seq_ids: Dict[str, int] = {}
+        self._step_counter = 0
+        self._seed = seed
+        self._abort_all = False
+        self._scale = 1.0 / math.sqrt(cfg.head_dim)
+        self._gen = torch.Generator().manual_seed(seed)  # CPU ref path RNG
+
+    # ------------------------------------------------------------ public API
+    def add_request(self, rid: str, input_ids: List[int],
+                    sampling: SamplingParams):
+        assert len(input_ids) > 0
+        req = Request(rid=rid, input_ids=list(input_ids), sampling=sampling,
+                      arrival_t=time.time())
+        self.waiting.append(req)
+        return rid
+
+    def abort_request(self, rid: Optional[str] = None, abort_all: bool = False):
+        """Mark requests aborted; they are emitted with partial output on the
+        next step (token-level continuation happens at the scheduler)."""
+        for r in self.running + self.waiting:
+            if abort_all or r.rid == rid:
+                r.finished = True
+                r.finish_reason = "abort"
+
+    def has_work(self) -> bool:
+        return bool(self.waiting or self.running)
+
+    def num_queued(self) -> int:
+        return len(self.waiting)
+
+    def num_running(self) -> int:
+        return len(self.running)
+
+    # ---------------------------------------------------------------- step
+    @torch.no_grad()
+    def step(self) -> List[RequestOutput]:
+        """One engine iteration: emit aborted/finished, admit+prefill (chunked),
+        decode one token for the running batch.  Returns finished outputs."""
+        finished: List[RequestOutput] = []
+
+        # sweep finished/aborted
+        still = []
+        for r in self.running:
+            if r.finished:
+                finished.append(self._emit(r))
+            else:
+                still.append(r)
+        self.running = still
+        self.waiting = [r for r in self.waiting if not r.finished or
+                        finished.append(self._emit(r))]  # emit aborted waiters
+
+        # ---- admission + chunked prefill -----------------------------------
+        token_budget = self.max_batched_tokens
+        prefill_reqs: List[Request] = []
+        prefill_lens: List[int] = []
+        # continue partially-prefilled first, then admit new
+        candidates = [r for r in self.running if r.prefill_pos < len(r.input_ids)]
+        admit_budget = token_budget - sum(
+            min(len(r.input_ids) - r.prefill_pos, token_budget) for r in candidates)
+        while self.waiting and len(self.running) < self.max_running \
+                and admit_budget > 0:
+            nxt = self.waiting[0]
+            need = min(len(nxt.input_ids) - nxt.prefill_pos, admit_budget)
+            if not self.kv.can_allocate(need):
+                break
+            admit_budget -= need
+            self.waiting.pop(0)
+            sid = self._seq_counter
+            self._seq_counter += 1
+            self._seq_ids[nxt.rid] = sid
+            self.running.append(nxt)
+            candidates.append(nxt)
+        for r in candidates:
+            if token_budget <= 0:
+                break
+            need = len(r.input_ids) - r.prefill_pos
+            take = min(need, token_budget)
+            if take <= 0:
+                continue
+            if not self.kv.allocate(self._seq_ids[r.rid], take):
+                break
+            prefill_reqs.append(r)
+            prefill_lens.append(take)
+            token_budget -= take
+        if prefill_reqs:
+            self._run_prefill(prefill_reqs, prefill_lens)
+
+        # ---- decode ---------------------------------------------------------
+        decode_reqs = [r for r in self.running
+                       if r.prefill_pos >= len(r.input_ids) and not r.finished]
+        if decode_reqs:
+            ok = []
+            for r in decode_reqs:
+                if self.kv.allocate(self._seq_ids[r.rid], 1):
+                    ok.append(r)
+                else:
+                    # out of KV pages: abort (scheduler continues elsewhere)
+                    r.finished = True
+                    r.finish_reason = "abort"
+                    finished.append(self._emit(r))
+            if ok:
+                self._run_decode(ok)
+
+        # sweep newly finished
+        still = []
+        for r in self.running:
+            if r.finished:
+                finished.append(self._emit(r))
+            else:
+                still.append(r)
+        self.running = still
+        self._step_counter += 1
+        return finished
+
+    def generate(self, prompts: List[List[int]], sampling: SamplingParams,
+                 rid_prefix: str = "req") -> List[RequestOutput]:
+        """Synchronous batch generate (test / co-located convenience)."""
+        for i, p in enumerate(prompts):
+            self.add_request(f"{rid_prefix}-{i}", p, sampling)
+        outs: Dict[str, RequestOutput] = {}
+        while self.has_work():
+            for o in self.step():
+                outs[o.rid] = o
+        return [outs[f"{rid_prefix}-{i}"] for i in range(len(prompts))]
+
+    # -------------------------------------------------------------- internals
+    def _emit(self, r: Request) -> RequestOutput:
+        sid = self._seq_ids.pop(r.rid, None)
+        if sid is not None:
+            self.kv.free_seq(sid)
+        if not r.finish_reason:
+            r.finish_reason = "length"
+        return RequestOutput(rid=r.rid, output_ids=list(r.output_ids),
+                             output_logprobs=list(r.output_logprobs),
+                             finish_reason=r.finish_reason)
+
+    def _sample_last(self, reqs: List[Request], hidden_rows: torch.Tensor):
+        """Sample the next token for each req from its last hidden row."""
+        logits = self.model.logits(hidden_rows)
+        n = len(reqs)
+        temp = torch.tensor([r.sampling.temperature for r in reqs],
+                            dtype=torch.float32, device=self.device)
+        tk = torch.tensor([r.sampling.top_k for r in reqs],
+                          dtype=torch.int32, device=self.device)
+        tp = torch.tensor([r.sampling.top_p for r in reqs],
+                          dtype=torch.float32, device=self.device)
+        seed = (self._seed * 0x9E3779B9 + self._step_counter) & 0x7FFFFFFFFFFF
+        tokens, lps = ops.sample(logits, temp, tk, tp, seed,
+                                 generator=self._gen)
+        tokens_l = tokens.tolist()
+        lps_l = lps.tolist()
+        for i, r in enumerate(reqs):
+            t = int(tokens_l[i])
+            r.output_ids.append(t)
+            r.output_logprobs.append(float(lps_l[i]))
+            if t in r.sampling.stop_token_ids:
+                r.finished = True
+                r.finish_reason = "stop"
+            elif len(r.output_ids) >= r.sampling.max_new_tokens:
+                r.finished = True
+                r.finish_reason = "length"
+            elif r.seq_len >= self.max_model_len:
+                r.finished = True
+                r.finish_reason = "length"
+
+    def _run_prefill(self, reqs: List[Request], lens: List[int]):
+        dev = self.device
+        tok_list, pos_list, slot_list = [], [], []
+        cu_q = [0]
+        cu_k = [0]
+        for r, take in zip(reqs, lens):
+            sid = self._seq_ids[r.rid]
+            start = r.prefill_pos
+            tok_list.extend(r.input_ids[start:start + take])
+            pos_list.extend(range(start, start + take))
+            slot_list.append(self.kv.slots_for(sid, start, take))
+            cu_q.append(cu_q[-1] + take)
+            cu_k.append(cu_k[-1] + start + take)
+        tokens = torch.tensor(tok_list, dtype=torch.long, device=dev)
+        positions = torch.tensor(pos_list, dtype=torch.int32, device=dev)
+        slots = torch.cat(slot_list).to(dev)
+        cu_q_t = torch.tensor(cu_q, dtype=torch.int32, device=dev)
+        cu_k_t = torch.tensor(cu_k, dtype=torch.int32, device=dev)
+
+        seq_ids = [self._seq_ids[r.rid] for r in reqs]
+        page_table = self.kv.page_table(seq_ids).to(dev)
+        ctx_lens = torch.tensor([s + t for s, t in
+                                 zip((r.prefill_pos for r in reqs), lens)],
+                                dtype=torch.int32, device=dev)
+
+        def attn_fn(li, q, k, v):
+            # chunked prefill: queries attend to the FULL cached history
+            # (cu_k spans include prior chunks) — gather K/V from the cache
+            # pages for this layer.
+            if all(r.prefill_pos == 0 for r in reqs):
+                return ops.varlen_prefill_attention(q, k, v, cu_q_t, cu_k_t,
+                                                    self._scale, causal=True)
+            kf, vf = self._gather_kv(li, seq_ids, ctx_lens)
+            return ops.varlen_prefill_attention(q, kf, vf, cu_q_t, cu_k_t,
+                                                self._scale, causal=True)
+
+        hidden = self.model.forward_tokens(tokens, positions, self.kv, slots,
+                                           attn_fn)
+        # requests whose prefill completes sample their first token
+        done_rows, done_reqs = [], []
+        for i, (r, take) in enumerate(zip(reqs, lens)):
+            r.prefill_pos += take
+            if r.prefill_pos >= len(r.input_ids):
+                done_rows.append(cu_q[i + 1] - 1)
+                done_reqs.append(r)
+        if done_reqs:
+            rows = hidden[torch.tensor(done_rows, dtype=torch.long, device=dev)]
+            self._sample_last(done_reqs, rows)
+
+    def _gather_kv(self, li: int, seq_ids: List[int], ctx_lens: torch.Tensor):
+        """Materialize contiguous K/V for chunked-prefill history (per layer)."""
+        ks, vs = [], []
+        flat_k = self.kv.k_cache[li].view(-1, self.cfg.num_key_value_heads,
+                                          self.cfg.head_dim)
+        flat_v = self.kv.v_cache[li].view(-1, self.cfg.num_key_value_heads,
+                                          self.cfg.head_dim)
+        for i, sid in enumerate(seq_ids):
+            L = int(ctx_lens[i])
+            slots = self.kv.slots_for(sid, 0, L).to(flat_k.device)
+            ks.append(flat_k[slots.long()])
+            vs.append(flat_v[slots.long()])
+        return torch.cat(ks), torch.cat(vs)
+
+    def _run_decode(self, reqs: List[Request]):
+        dev = self.device
+        tokens = torch.tensor([r.output_ids[-1] for r in reqs],
+                              dtype=torch.long, device=dev)
+        pos = [r.seq_len - 1 for r in reqs]
+        positions = torch.tensor(pos, dtype=torch.int32, device=dev)
+        seq_ids = [self._seq_ids[r.rid] for r in reqs]
+        slots = torch.cat([self.kv.slots_for(sid, p, 1)
+                           for sid, p in zip(seq_ids, pos)]).to(dev)
+        page_table = self.kv.page_table(seq_ids).to(dev)
+        ctx = torch.tensor([p + 1 for p in pos], dtype=torch.int32, device=dev)
+
+        def attn_fn(li, q, k, v):
+            return ops.paged_attention_decode(q, self.kv.k_cache[li],
+                                              self.kv.v_cache[li], page_table,
+                                              ctx, self._scale)
+
+        hidden = self.model.forward_tokens(tokens, positions, self.kv, slots,
+                                           attn_fn)
+        self._sample_last(reqs, hidden)

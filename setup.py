@@ -5,9 +5,16 @@
 The built .so lands next to the package (polyrl_amd/_hip_C*.so) so it travels
 with a repo snapshot to GPU boxes.
 """
+import glob
 import os
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+# hipify writes *_hip.hip copies of each source; a header-only edit does not
+# refresh them, which leaves stale objects in incremental builds. Remove the
+# copies so every build re-derives them from the true sources.
+for _f in glob.glob(os.path.join("polyrl_amd", "ops", "csrc", "*_hip.hip")):
+    os.remove(_f)
 
 from setuptools import setup
 from torch.utils.cpp_extension import BuildExtension, CUDAExtension
