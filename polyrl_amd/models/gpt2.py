@@ -67,7 +67,8 @@ class GPT2LMModel(nn.Module):
     def gradient_checkpointing_enable(self):
         pass
 
-    def forward(self, input_ids, attention_mask=None, position_ids=None):
+    def forward(self, input_ids, attention_mask=None, position_ids=None,
+                logits_slice=None):
         B, L = input_ids.shape
         if position_ids is None:
             position_ids = torch.arange(L, device=input_ids.device).expand(B, L)
@@ -79,7 +80,10 @@ class GPT2LMModel(nn.Module):
             attn_mask = causal.view(1, 1, L, L) & attention_mask.bool().view(B, 1, 1, L)
         for blk in self.h:
             x = blk(x, attn_mask)
-        return self.lm_head(self.ln_f(x))
+        x = self.ln_f(x)
+        if logits_slice is not None:
+            x = x[:, logits_slice]
+        return self.lm_head(x)
 
     def num_params(self):
         return sum(p.numel() for p in self.parameters())

@@ -21,15 +21,6 @@ import torch.nn.functional as F
 from .registry import DecoderConfig
 
 
-def _rope_rotate_half(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor):
-    """NEOX-style RoPE: x (B, H, L, D), cos/sin (L, D/2)."""
-    d = x.shape[-1] // 2
-    x1, x2 = x[..., :d], x[..., d:]
-    c = cos.view(1, 1, *cos.shape)
-    s = sin.view(1, 1, *sin.shape)
-    return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], dim=-1)
-
-
 class RMSNorm(nn.Module):
     def __init__(self, dim: int, eps: float):
         super().__init__()
@@ -83,14 +74,18 @@ class Attention(nn.Module):
         self.o_proj = nn.Linear(self.num_heads * self.head_dim, h, bias=False)
 
     def forward(self, x, cos, sin, attn_bias_mask: Optional[torch.Tensor]):
+        """cos/sin: (B, 1, L, D/2) fp32 (per-row positions supported)."""
         B, L, _ = x.shape
         q = self.q_proj(x).view(B, L, self.num_heads, self.head_dim).transpose(1, 2)
         k = self.k_proj(x).view(B, L, self.num_kv_heads, self.head_dim).transpose(1, 2)
         v = self.v_proj(x).view(B, L, self.num_kv_heads, self.head_dim).transpose(1, 2)
         cos = cos.to(q.dtype)
         sin = sin.to(q.dtype)
-        q = _rope_rotate_half(q, cos, sin)
-        k = _rope_rotate_half(k, cos, sin)
+        d = self.head_dim // 2
+        q = torch.cat([q[..., :d] * cos - q[..., d:] * sin,
+                       q[..., d:] * cos + q[..., :d] * sin], dim=-1)
+        k = torch.cat([k[..., :d] * cos - k[..., d:] * sin,
+                       k[..., d:] * cos + k[..., :d] * sin], dim=-1)
         if self.num_kv_heads != self.num_heads:
             rep = self.num_heads // self.num_kv_heads
             k = k.repeat_interleave(rep, dim=1)
@@ -144,12 +139,10 @@ class DecoderModel(nn.Module):
         B, L = input_ids.shape
         if position_ids is None:
             position_ids = torch.arange(L, device=input_ids.device).expand(B, L)
-        # rope tables per position row (assume shared positions across batch
-        # when position_ids rows are identical — the common padded case needs
-        # per-row tables)
+        # per-row rope tables shaped for broadcast: (B, 1, L, D/2)
         cos, sin = self.rotary.get(position_ids.reshape(-1))
-        cos = cos.view(B, L, -1)
-        sin = sin.view(B, L, -1)
+        cos = cos.view(B, 1, L, -1)
+        sin = sin.view(B, 1, L, -1)
 
         attn_bias_mask = None
         if attention_mask is not None and not bool(attention_mask.all()):
@@ -161,49 +154,14 @@ class DecoderModel(nn.Module):
 
         x = self.embed_tokens(input_ids)
         for layer in self.layers:
+            # NOTE: layers must be invoked via __call__ so FSDP2's
+            # unshard/reshard pre/post-forward hooks fire.
             if self.gradient_checkpointing and self.training:
                 x = torch.utils.checkpoint.checkpoint(
-                    _layer_call, layer, x, cos, sin, attn_bias_mask,
-                    use_reentrant=False)
+                    layer, x, cos, sin, attn_bias_mask, use_reentrant=False)
             else:
-                x = _layer_call(layer, x, cos, sin, attn_bias_mask)
+                x = layer(x, cos, sin, attn_bias_mask)
         return self.norm(x)
-
-
-def _layer_call(layer, x, cos, sin, attn_bias_mask):
-    # per-row cos/sin: reshape to broadcast (B, 1, L, D/2)
-    B, L = x.shape[0], x.shape[1]
-    c = cos.view(B, 1, L, -1)
-    s = sin.view(B, 1, L, -1)
-    return _layer_forward(layer, x, c, s, attn_bias_mask)
-
-
-def _layer_forward(layer, x, c, s, attn_bias_mask):
-    h = layer.input_layernorm(x)
-    B, L, _ = h.shape
-    att = layer.self_attn
-    q = att.q_proj(h).view(B, L, att.num_heads, att.head_dim).transpose(1, 2)
-    k = att.k_proj(h).view(B, L, att.num_kv_heads, att.head_dim).transpose(1, 2)
-    v = att.v_proj(h).view(B, L, att.num_kv_heads, att.head_dim).transpose(1, 2)
-    cq = c.to(q.dtype)
-    sq = s.to(q.dtype)
-    d = q.shape[-1] // 2
-    q = torch.cat([q[..., :d] * cq - q[..., d:] * sq,
-                   q[..., d:] * cq + q[..., :d] * sq], dim=-1)
-    k = torch.cat([k[..., :d] * cq - k[..., d:] * sq,
-                   k[..., d:] * cq + k[..., :d] * sq], dim=-1)
-    if att.num_kv_heads != att.num_heads:
-        rep = att.num_heads // att.num_kv_heads
-        k = k.repeat_interleave(rep, dim=1)
-        v = v.repeat_interleave(rep, dim=1)
-    if attn_bias_mask is not None:
-        o = F.scaled_dot_product_attention(q, k, v, attn_mask=attn_bias_mask)
-    else:
-        o = F.scaled_dot_product_attention(q, k, v, is_causal=True)
-    o = o.transpose(1, 2).reshape(B, L, -1)
-    x = x + att.o_proj(o)
-    x = x + layer.mlp(layer.post_attention_layernorm(x))
-    return x
 
 
 class CausalLM(nn.Module):
@@ -229,8 +187,14 @@ class CausalLM(nn.Module):
     def gradient_checkpointing_enable(self):
         self.model.gradient_checkpointing = True
 
-    def forward(self, input_ids, attention_mask=None, position_ids=None):
+    def forward(self, input_ids, attention_mask=None, position_ids=None,
+                logits_slice: Optional[slice] = None):
+        """logits_slice: sequence-dim slice applied to the hidden states
+        BEFORE lm_head — avoids materializing vocab logits over prompt
+        positions when only response logprobs are needed."""
         hidden = self.model(input_ids, attention_mask, position_ids)
+        if logits_slice is not None:
+            hidden = hidden[:, logits_slice]
         return self.lm_head(hidden)
 
     @torch.no_grad()

@@ -54,6 +54,10 @@ MODEL_CONFIGS = {
         num_hidden_layers=12, num_attention_heads=12, num_key_value_heads=12,
         max_position_embeddings=1024, tie_word_embeddings=True),
     # tiny debug models (CPU unit tests / fast GPU smoke)
+    "gpt2-debug": DecoderConfig(
+        arch="gpt2", vocab_size=256, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=4,
+        max_position_embeddings=512, tie_word_embeddings=True),
     "llama-tiny": DecoderConfig(
         arch="llama", vocab_size=1024, hidden_size=256, intermediate_size=688,
         num_hidden_layers=2, num_attention_heads=8, num_key_value_heads=2,

@@ -98,13 +98,40 @@ class LayerWeights:
         return out
 
 
+class Gpt2LayerWeights:
+    """GPT-2 layer buffers (CPU plumbing tier, BASELINE config #1)."""
+
+    def __init__(self, cfg: DecoderConfig, device, dtype):
+        h, i = cfg.hidden_size, cfg.intermediate_size
+        z = lambda *s: torch.empty(*s, device=device, dtype=dtype)
+        self.ln1_w, self.ln1_b = z(h), z(h)
+        self.ln2_w, self.ln2_b = z(h), z(h)
+        self.wqkv, self.bqkv = z(3 * h, h), z(3 * h)
+        self.wo, self.bo = z(h, h), z(h)
+        self.w_fc, self.b_fc = z(i, h), z(i)
+        self.w_proj, self.b_proj = z(h, i), z(h)
+
+    def named_slices(self, prefix: str):
+        return {
+            f"{prefix}.ln_1.weight": self.ln1_w, f"{prefix}.ln_1.bias": self.ln1_b,
+            f"{prefix}.ln_2.weight": self.ln2_w, f"{prefix}.ln_2.bias": self.ln2_b,
+            f"{prefix}.attn_qkv.weight": self.wqkv, f"{prefix}.attn_qkv.bias": self.bqkv,
+            f"{prefix}.attn_out.weight": self.wo, f"{prefix}.attn_out.bias": self.bo,
+            f"{prefix}.mlp_fc.weight": self.w_fc, f"{prefix}.mlp_fc.bias": self.b_fc,
+            f"{prefix}.mlp_proj.weight": self.w_proj, f"{prefix}.mlp_proj.bias": self.b_proj,
+        }
+
+
 class InferenceModel:
     """No-autograd decoder forward over paged KV on the HIP kernel suite."""
 
     def __init__(self, cfg: DecoderConfig, device="cuda",
                  dtype=torch.bfloat16):
+        if cfg.arch == "gpt2":
+            self._init_gpt2(cfg, device, dtype)
+            return
         assert cfg.arch in ("llama", "qwen2"), \
-            "rollout engine serves the llama/qwen2 family"
+            "rollout engine serves the llama/qwen2/gpt2 families"
         self.cfg = cfg
         self.device = device
         self.dtype = dtype
@@ -124,6 +151,29 @@ class InferenceModel:
         }
         for li, lw in enumerate(self.layers):
             self._name_map.update(lw.named_slices(f"model.layers.{li}"))
+
+    def _init_gpt2(self, cfg: DecoderConfig, device, dtype):
+        self.cfg = cfg
+        self.device = device
+        self.dtype = dtype
+        h = cfg.hidden_size
+        self.layers = [Gpt2LayerWeights(cfg, device, dtype)
+                       for _ in range(cfg.num_hidden_layers)]
+        self.embed = torch.empty(cfg.vocab_size, h, device=device, dtype=dtype)
+        self.wpe = torch.empty(cfg.max_position_embeddings, h, device=device,
+                               dtype=dtype)
+        self.lnf_w = torch.empty(h, device=device, dtype=dtype)
+        self.lnf_b = torch.empty(h, device=device, dtype=dtype)
+        self.lm_head = torch.empty(cfg.vocab_size, h, device=device, dtype=dtype)
+        self._name_map = {
+            "wte.weight": self.embed,
+            "wpe.weight": self.wpe,
+            "ln_f.weight": self.lnf_w,
+            "ln_f.bias": self.lnf_b,
+            "lm_head.weight": self.lm_head,
+        }
+        for li, lw in enumerate(self.layers):
+            self._name_map.update(lw.named_slices(f"h.{li}"))
 
     def weight_bytes(self) -> int:
         return sum(v.numel() * v.element_size() for v in self._name_map.values())
@@ -163,6 +213,9 @@ class InferenceModel:
                        attn_fn) -> torch.Tensor:
         """Shared trunk: embed -> L x (norm, qkv, rope, kv-append, attn(fn),
         o-proj, norm, mlp) -> final norm.  Returns hidden (N, H)."""
+        if self.cfg.arch == "gpt2":
+            return self._forward_tokens_gpt2(token_ids, positions, kv,
+                                             slot_mapping, attn_fn)
         cfg = self.cfg
         Hq, Hk, D = cfg.num_attention_heads, cfg.num_key_value_heads, cfg.head_dim
         x = self.embed[token_ids]                     # (N, H) gather
@@ -194,6 +247,32 @@ class InferenceModel:
         # final residual add + norm
         final = (hidden.float() + residual.float()).to(self.dtype)
         return ops.rmsnorm(final, self.final_norm, cfg.rms_norm_eps)
+
+    @torch.no_grad()
+    def _forward_tokens_gpt2(self, token_ids, positions, kv, slot_mapping,
+                             attn_fn):
+        import torch.nn.functional as F
+        cfg = self.cfg
+        H = cfg.num_attention_heads
+        D = cfg.head_dim
+        x = self.embed[token_ids] + self.wpe[positions.long()]
+        for li, lw in enumerate(self.layers):
+            h1 = F.layer_norm(x, (cfg.hidden_size,), lw.ln1_w, lw.ln1_b,
+                              cfg.layer_norm_eps)
+            qkv = h1 @ lw.wqkv.t() + lw.bqkv
+            q, k, v = qkv.split(H * D, dim=-1)
+            q = q.view(-1, H, D).contiguous()
+            k = k.view(-1, H, D).contiguous()
+            v = v.view(-1, H, D).contiguous()
+            ops.kv_cache_append(kv.k_cache[li], kv.v_cache[li], k, v,
+                                slot_mapping)
+            attn = attn_fn(li, q, k, v).view(-1, H * D)
+            x = x + attn @ lw.wo.t() + lw.bo
+            h2 = F.layer_norm(x, (cfg.hidden_size,), lw.ln2_w, lw.ln2_b,
+                              cfg.layer_norm_eps)
+            x = x + F.gelu(h2 @ lw.w_fc.t() + lw.b_fc) @ lw.w_proj.t() + lw.b_proj
+        return F.layer_norm(x, (cfg.hidden_size,), self.lnf_w, self.lnf_b,
+                            cfg.layer_norm_eps)
 
     @torch.no_grad()
     def logits(self, hidden: torch.Tensor) -> torch.Tensor:

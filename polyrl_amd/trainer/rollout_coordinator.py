@@ -1,0 +1,167 @@
+"""Rollout coordination: submit prompt groups, stream finished groups back.
+
+Streaming contract mirrors the reference: submission returns
+immediately; finished prompt GROUPS (all n samples of one prompt) stream back
+while the trainer updates on earlier ibatches; postprocess_samples builds the
+padded TensorBatch (sglang_rollout_remote.py:318-391 capability).
+
+Co-located mode: each trainer rank owns one in-process Engine on its GPU and
+generates its own shard (instance-level DP).  Split / elastic mode goes
+through the scheduler (polyrl_amd/scheduler) with the same group-stream
+interface.
+"""
+from __future__ import annotations
+
+from collections import defaultdict
+from typing import Dict, Iterator, List, Optional
+
+import numpy as np
+import torch
+
+from ..protocol import TensorBatch
+from ..rollout.engine import Engine, RequestOutput, SamplingParams
+
+
+def postprocess_groups(
+    prompt_ids: torch.Tensor,        # (G, Lp) left-padded local prompts
+    prompt_mask: torch.Tensor,       # (G, Lp)
+    group_uids: List[str],
+    outputs: List[List[RequestOutput]],  # per group: n sample outputs
+    response_length: int,
+    pad_token_id: int = 0,
+    device="cpu",
+) -> TensorBatch:
+    """Build the training batch for a set of finished prompt groups."""
+    G = len(outputs)
+    n = len(outputs[0])
+    Lp = prompt_ids.shape[1]
+    B = G * n
+    resp = torch.full((B, response_length), pad_token_id, dtype=torch.long)
+    resp_mask = torch.zeros(B, response_length, dtype=torch.long)
+    rollout_lp = torch.zeros(B, response_length, dtype=torch.float32)
+    for g in range(G):
+        for s, out in enumerate(outputs[g]):
+            i = g * n + s
+            L = min(len(out.output_ids), response_length)
+            resp[i, :L] = torch.tensor(out.output_ids[:L], dtype=torch.long)
+            resp_mask[i, :L] = 1
+            rollout_lp[i, :L] = torch.tensor(out.output_logprobs[:L])
+    prompts = prompt_ids.repeat_interleave(n, dim=0)
+    pmask = prompt_mask.repeat_interleave(n, dim=0)
+    input_ids = torch.cat([prompts, resp], dim=1)
+    attention_mask = torch.cat([pmask, resp_mask], dim=1)
+    position_ids = torch.clamp(torch.cumsum(attention_mask, dim=1) - 1, min=0)
+    uids = np.array([group_uids[g] for g in range(G) for _ in range(n)],
+                    dtype=object)
+    return TensorBatch.from_dict(
+        tensors={
+            "prompts": prompts,
+            "responses": resp,
+            "input_ids": input_ids,
+            "attention_mask": attention_mask,
+            "response_mask": resp_mask,
+            "position_ids": position_ids,
+            "rollout_log_probs": rollout_lp,
+        },
+        non_tensors={"uid": uids},
+    ).to(device)
+
+
+class LocalRolloutCoordinator:
+    """One in-process engine per trainer rank (co-located mode)."""
+
+    def __init__(self, engine: Engine, response_length: int,
+                 pad_token_id: int = 0, device="cpu"):
+        self.engine = engine
+        self.response_length = response_length
+        self.pad_token_id = pad_token_id
+        self.device = device
+        self._groups: Dict[str, dict] = {}
+        self._submit_counter = 0
+
+    # --------------------------------------------------------------- submit
+    def submit(self, prompts: TensorBatch, sampling: SamplingParams, n: int):
+        """Enqueue n samples per prompt.  prompts: tensors input_ids (G, Lp),
+        attention_mask (G, Lp), non_tensor uid (G,)."""
+        ids = prompts["input_ids"]
+        mask = prompts["attention_mask"]
+        uids = prompts["uid"]
+        G = ids.shape[0]
+        base = self._submit_counter
+        self._submit_counter += 1
+        for g in range(G):
+            gid = f"b{base}-g{g}"
+            raw = ids[g][mask[g].bool()].tolist()
+            self._groups[gid] = {
+                "uid": str(uids[g]),
+                "prompt_ids": ids[g].cpu(),
+                "prompt_mask": mask[g].cpu(),
+                "outputs": [None] * n,
+                "done": 0,
+                "n": n,
+            }
+            for s in range(n):
+                self.engine.add_request(f"{gid}-s{s}", raw, sampling)
+
+    # --------------------------------------------------------------- stream
+    def pending_groups(self) -> int:
+        return len(self._groups)
+
+    def has_work(self) -> bool:
+        return bool(self._groups)
+
+    def poll(self, max_steps: int = 1) -> List[dict]:
+        """Step the engine; return finished groups (dicts)."""
+        done = []
+        for _ in range(max_steps):
+            if not self.engine.has_work():
+                break
+            for out in self.engine.step():
+                gid, s = out.rid.rsplit("-s", 1)
+                grp = self._groups.get(gid)
+                if grp is None:
+                    continue
+                grp["outputs"][int(s)] = out
+                grp["done"] += 1
+                if grp["done"] == grp["n"]:
+                    done.append(self._groups.pop(gid))
+        return done
+
+    def stream_batches(self, stream_size: int) -> Iterator[TensorBatch]:
+        """Yield TensorBatches of EXACTLY stream_size samples (whole groups;
+        requires stream_size % n == 0) until all submitted groups are
+        consumed; the final batch carries any remainder.  Exact sizing keeps
+        every DP rank's ibatch count identical — a requirement for the SPMD
+        FSDP collectives (each ibatch triggers collective fwd/bwd)."""
+        ready: List[dict] = []
+        ready_samples = 0
+        while self._groups or ready:
+            if self._groups:
+                for grp in self.poll():
+                    ready.append(grp)
+                    ready_samples += grp["n"]
+            while ready_samples >= stream_size and ready:
+                take, taken = [], 0
+                while ready and taken < stream_size:
+                    g = ready.pop(0)
+                    take.append(g)
+                    taken += g["n"]
+                ready_samples -= taken
+                yield self._make_batch(take)
+            if not self._groups and ready:  # tail (partial final batch)
+                yield self._make_batch(ready)
+                ready = []
+                ready_samples = 0
+
+    def _make_batch(self, groups: List[dict]) -> TensorBatch:
+        prompt_ids = torch.stack([g["prompt_ids"] for g in groups])
+        prompt_mask = torch.stack([g["prompt_mask"] for g in groups])
+        uids = [g["uid"] for g in groups]
+        outputs = [g["outputs"] for g in groups]
+        return postprocess_groups(prompt_ids, prompt_mask, uids, outputs,
+                                  self.response_length, self.pad_token_id,
+                                  self.device)
+
+    # -------------------------------------------------------------- weights
+    def update_weights(self, state_dict):
+        self.engine.model.load_state_dict(state_dict, strict=False)

@@ -1,0 +1,345 @@
+"""StreamPPOTrainer — the streamed PPO/GRPO fit loop, SPMD one-process-per-GPU.
+
+Reference capability: StreamRayPPOTrainer.fit (stream_ray_trainer.py:282-689)
+re-designed without Ray: every rank runs the same loop; the rollout engine is
+in-process per rank; data-dependent control flow is deterministic by
+construction (equal per-rank sample counts, exact-size stream batches), so no
+cross-rank RPC is needed — only the FSDP/RCCL collectives inside the workers.
+
+Preserved semantics:
+  * streamed training: the actor updates on ibatch i while later samples are
+    still decoding (engine work interleaves between update calls),
+  * minibatch-boundary bookkeeping: optimizer steps fire when the cumulative
+    streamed sample count crosses each ppo_mini_batch boundary
+    (stream_ray_trainer.py:500-568), LR steps on the last slice,
+  * driver-side reward -> old/ref logprob -> values -> KL penalty ->
+    advantage (GAE or GRPO group-norm) per ibatch,
+  * weight publication to every rollout consumer each iteration (§3.3).
+"""
+from __future__ import annotations
+
+import os
+import time
+from typing import Dict, List, Optional
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+from ..config import PPOConfig
+from ..core import algos
+from ..core.metrics import (Tracking, compute_data_metrics,
+                            compute_throughput_metrics, compute_timing_metrics,
+                            marked_timer, reduce_metrics)
+from ..data import SyntheticPromptDataset, epoch_batches
+from ..models import create_model, get_model_config
+from ..protocol import TensorBatch
+from ..reward import load_reward_manager
+from ..rollout.engine import Engine, SamplingParams
+from ..transfer.weight_transfer import WeightPublisher
+from .checkpoint import CheckpointManager
+from .rollout_coordinator import LocalRolloutCoordinator
+from .workers import ActorWorker, CriticWorker
+
+
+def _dist_info():
+    if dist.is_available() and dist.is_initialized():
+        return dist.get_world_size(), dist.get_rank()
+    return 1, 0
+
+
+class StreamPPOTrainer:
+    def __init__(self, config: PPOConfig, device: Optional[str] = None,
+                 reward_fn=None, dataset=None):
+        self.config = config
+        self.world, self.rank = _dist_info()
+        if device is None:
+            device = config.trainer.device
+        if device == "cuda":
+            device = f"cuda:{torch.cuda.current_device()}"
+        self.device = device
+
+        self._validate_config()
+
+        arr = config.actor_rollout_ref
+        model_cfg = get_model_config(arr.model.path)
+        self.model_cfg = model_cfg
+        dtype = arr.model.dtype
+
+        torch.manual_seed(config.trainer.seed)
+        actor_model = create_model(model_cfg, kind="actor", dtype=dtype,
+                                   device=device)
+        if arr.model.enable_gradient_checkpointing:
+            actor_model.gradient_checkpointing_enable()
+        self.actor = ActorWorker(actor_model, arr.actor, device=device)
+
+        self.use_ref = arr.actor.use_kl_loss or config.algorithm.use_kl_in_reward
+        self.ref = None
+        if self.use_ref:
+            torch.manual_seed(config.trainer.seed)
+            ref_model = create_model(model_cfg, kind="actor", dtype=dtype,
+                                     device=device)
+            ref_model.load_state_dict(
+                {k: v for k, v in actor_model.state_dict().items()})
+            for p in ref_model.parameters():
+                p.requires_grad_(False)
+            self.ref = ActorWorker(ref_model, arr.actor, device=device,
+                                   is_ref=True)
+
+        self.use_critic = config.algorithm.adv_estimator == "gae"
+        self.critic = None
+        if self.use_critic:
+            critic_cfg = config.critic
+            critic_model_cfg = get_model_config(
+                critic_cfg.model.path if critic_cfg.model.path != "llama3-1b"
+                or arr.model.path == "llama3-1b"
+                else arr.model.path)
+            critic_model = create_model(critic_model_cfg, kind="critic",
+                                        dtype=dtype, device=device)
+            if critic_cfg.model.enable_gradient_checkpointing:
+                critic_model.gradient_checkpointing_enable()
+            self.critic = CriticWorker(critic_model, critic_cfg, device=device)
+
+        # ---------------- rollout engine (co-located, one per rank) ---------
+        ro = arr.rollout
+        eng_dtype = getattr(torch, ro.dtype)
+        kv_budget = self._kv_budget(ro)
+        self.engine = Engine(model_cfg, device=device, dtype=eng_dtype,
+                             page_size=ro.page_size,
+                             kv_bytes_budget=kv_budget,
+                             max_running_requests=ro.max_running_requests,
+                             max_num_batched_tokens=ro.max_num_batched_tokens,
+                             max_model_len=ro.prompt_length + ro.response_length,
+                             seed=config.trainer.seed * 1000 + self.rank)
+        self.coordinator = LocalRolloutCoordinator(
+            self.engine, ro.response_length, pad_token_id=0, device="cpu")
+        self.publisher = WeightPublisher(
+            self.actor.model, [self.engine.model],
+            tie_word_embeddings=model_cfg.tie_word_embeddings)
+
+        # ---------------- data + reward -------------------------------------
+        dcfg = config.data
+        if dataset is not None:
+            self.dataset = dataset
+        else:
+            self.dataset = SyntheticPromptDataset(
+                num_prompts=dcfg.synthetic_num_prompts,
+                vocab_size=model_cfg.vocab_size,
+                max_prompt_length=dcfg.max_prompt_length,
+                seed=dcfg.seed)
+        self.reward_fn = reward_fn or load_reward_manager("constant")
+
+        self.ckpt_actor = CheckpointManager(config.trainer.default_local_dir,
+                                            "actor")
+        self.ckpt_critic = CheckpointManager(config.trainer.default_local_dir,
+                                             "critic") if self.use_critic else None
+        self.tracking = Tracking(config.trainer.project_name,
+                                 config.trainer.experiment_name,
+                                 config.trainer.logger if self.rank == 0 else [],
+                                 config.trainer.default_local_dir)
+        self.global_step = 0
+        self._maybe_resume()
+
+    # ------------------------------------------------------------------ setup
+    def _kv_budget(self, ro) -> int:
+        if self.device.startswith("cuda"):
+            free, total = torch.cuda.mem_get_info()
+            return int(free * ro.gpu_memory_utilization * 0.5)
+        return 64 << 20
+
+    def _validate_config(self):
+        c = self.config
+        n = c.actor_rollout_ref.rollout.sampling.n
+        tb = c.data.train_batch_size
+        mini = c.actor_rollout_ref.actor.ppo_mini_batch_size
+        stream = c.actor_rollout_ref.rollout.min_stream_batch_size
+        world = self.world
+        assert tb % world == 0, f"train_batch_size {tb} % world {world}"
+        total = tb * n
+        assert total % mini == 0, \
+            f"total samples {total} must divide by ppo_mini_batch {mini}"
+        assert mini % world == 0
+        assert stream % n == 0, f"min_stream_batch {stream} % n {n}"
+        assert stream % world == 0
+        local_stream = stream // world
+        local_total = total // world
+        assert local_total % local_stream == 0, \
+            f"local total {local_total} % local stream {local_stream}"
+        mini_local = mini // world
+        assert mini_local % local_stream == 0 or local_stream % mini_local == 0 \
+            or mini_local % n == 0, "minibatch/stream sizes must compose"
+
+    # ------------------------------------------------------------------- fit
+    def fit(self, max_steps: Optional[int] = None):
+        c = self.config
+        total_steps = max_steps or c.trainer.total_training_steps or 1
+        ro = c.actor_rollout_ref.rollout
+        n = ro.sampling.n
+        local_bs = c.data.train_batch_size // self.world
+        local_stream = ro.min_stream_batch_size // self.world
+        mini_local = c.actor_rollout_ref.actor.ppo_mini_batch_size // self.world
+        local_total = local_bs * n
+
+        step_in_run = 0
+        for epoch in range(c.trainer.total_epochs):
+            for global_batch in epoch_batches(self.dataset,
+                                              c.data.train_batch_size,
+                                              shuffle=c.data.shuffle,
+                                              seed=c.data.seed + epoch):
+                if step_in_run >= total_steps:
+                    return
+                self.global_step += 1
+                step_in_run += 1
+                metrics: Dict[str, float] = {}
+                timing: Dict[str, float] = {}
+                with marked_timer("step", timing):
+                    batch_metrics = self._run_step(
+                        global_batch, n, local_bs, local_stream, mini_local,
+                        local_total, timing)
+                metrics.update(batch_metrics)
+                metrics.update(compute_timing_metrics(
+                    self._last_full_batch, timing))
+                metrics.update(compute_throughput_metrics(
+                    self._last_full_batch, timing,
+                    self.world))
+                metrics["training/global_step"] = self.global_step
+                if self.rank == 0:
+                    self.tracking.log(metrics, self.global_step)
+                if c.trainer.save_freq > 0 and \
+                        self.global_step % c.trainer.save_freq == 0:
+                    self.save_checkpoint()
+        return
+
+    # ------------------------------------------------------------- one step
+    def _run_step(self, global_batch: TensorBatch, n: int, local_bs: int,
+                  local_stream: int, mini_local: int, local_total: int,
+                  timing: Dict[str, float]) -> Dict[str, float]:
+        c = self.config
+        ro = c.actor_rollout_ref.rollout
+
+        # 1. publish current weights to the rollout plane (every iteration,
+        #    incl. bootstrap — §3.3)
+        with marked_timer("weight_sync", timing):
+            self.publisher.publish()
+
+        # 2. shard the global batch; submit local prompts
+        local = global_batch.slice(
+            slice(self.rank * local_bs, (self.rank + 1) * local_bs))
+        sampling = SamplingParams(
+            temperature=ro.sampling.temperature,
+            top_k=ro.sampling.top_k, top_p=ro.sampling.top_p,
+            max_new_tokens=ro.response_length)
+        with marked_timer("gen_submit", timing):
+            self.coordinator.submit(local, sampling, n)
+
+        # 3. stream loop
+        all_metrics: Dict[str, List[float]] = {}
+        ibatches: List[TensorBatch] = []
+        cum = 0                       # cumulative local samples trained
+        warmup = c.trainer.critic_warmup
+        for ibatch in self.coordinator.stream_batches(local_stream):
+            with marked_timer("prep", timing):
+                ibatch = self._prepare_ibatch(ibatch, timing)
+            ibatches.append(ibatch)
+            # 4. slice at minibatch boundaries (global bookkeeping via equal
+            #    per-rank counts)
+            with marked_timer("update", timing):
+                off = 0
+                bs = len(ibatch)
+                while off < bs:
+                    next_boundary = ((cum // mini_local) + 1) * mini_local
+                    take = min(bs - off, next_boundary - cum)
+                    sl = ibatch.slice(slice(off, off + take))
+                    cum += take
+                    off += take
+                    is_opt = (cum % mini_local == 0)
+                    is_lr = (cum >= local_total)
+                    scale = take / mini_local
+                    if self.use_critic:
+                        m = self.critic.update_critic_stream(
+                            sl, is_opt, is_lr, accum_scale=scale)
+                        for k, v in m.items():
+                            all_metrics.setdefault(k, []).extend(v)
+                    if warmup <= self.global_step:
+                        m = self.actor.update_policy_stream(
+                            sl, is_opt, is_lr, accum_scale=scale)
+                        for k, v in m.items():
+                            all_metrics.setdefault(k, []).extend(v)
+        full = TensorBatch.concat(ibatches) if ibatches else TensorBatch()
+        self._last_full_batch = full
+        metrics = reduce_metrics(all_metrics)
+        metrics.update(compute_data_metrics(full, self.use_critic))
+        return metrics
+
+    # ------------------------------------------------- per-ibatch preparation
+    def _prepare_ibatch(self, ibatch: TensorBatch,
+                        timing: Dict[str, float]) -> TensorBatch:
+        c = self.config
+        # reward
+        with marked_timer("reward", timing):
+            scores = self.reward_fn(ibatch)
+            ibatch["token_level_scores"] = scores
+        # old log probs (actor fwd, exact — reference :425-439)
+        with marked_timer("old_log_prob", timing):
+            old_lp, _ = self.actor.compute_log_prob(ibatch)
+            ibatch["old_log_probs"] = old_lp.cpu()
+            if "rollout_log_probs" in ibatch:
+                rm = ibatch["response_mask"].float()
+                diff = (old_lp.cpu() - ibatch["rollout_log_probs"]).abs()
+                ibatch.meta_info["rollout_logprob_diff"] = float(
+                    algos.masked_mean(diff, rm))
+        # ref log probs
+        if self.ref is not None:
+            with marked_timer("ref_log_prob", timing):
+                ref_lp, _ = self.ref.compute_log_prob(ibatch)
+                ibatch["ref_log_probs"] = ref_lp.cpu()
+        # values
+        if self.use_critic:
+            with marked_timer("values", timing):
+                ibatch["values"] = self.critic.compute_values(ibatch).cpu()
+        # KL penalty into reward + advantage (driver-side, :465-498)
+        with marked_timer("adv", timing):
+            scores = ibatch["token_level_scores"]
+            if c.algorithm.use_kl_in_reward and self.ref is not None:
+                rewards, kl = algos.apply_kl_penalty(
+                    scores, ibatch["old_log_probs"], ibatch["ref_log_probs"],
+                    ibatch["response_mask"].float(),
+                    c.algorithm.kl_ctrl.kl_coef, c.algorithm.kl_penalty)
+                ibatch["token_level_rewards"] = rewards
+            else:
+                ibatch["token_level_rewards"] = scores
+            adv, ret = algos.compute_advantage(
+                ibatch["token_level_rewards"],
+                ibatch["response_mask"].float(),
+                c.algorithm.adv_estimator,
+                values=ibatch.tensors.get("values"),
+                index=ibatch.non_tensors.get("uid"),
+                gamma=c.algorithm.gamma, lam=c.algorithm.lam,
+                norm_adv_by_std_in_grpo=c.algorithm.norm_adv_by_std_in_grpo)
+            ibatch["advantages"] = adv
+            ibatch["returns"] = ret
+        return ibatch
+
+    # ------------------------------------------------------------ checkpoint
+    def save_checkpoint(self):
+        extra = {"global_step": self.global_step}
+        self.ckpt_actor.save(self.global_step, self.actor.model,
+                             self.actor.optimizer, self.actor.lr_scheduler,
+                             extra)
+        if self.use_critic:
+            self.ckpt_critic.save(self.global_step, self.critic.model,
+                                  self.critic.optimizer,
+                                  self.critic.lr_scheduler, extra)
+
+    def _maybe_resume(self):
+        if self.config.trainer.resume_mode == "disable":
+            return
+        ex = self.ckpt_actor.load(self.actor.model, self.actor.optimizer,
+                                  self.actor.lr_scheduler)
+        if ex is not None:
+            self.global_step = int(ex.get("global_step", 0))
+            if self.use_critic:
+                self.ckpt_critic.load(self.critic.model, self.critic.optimizer,
+                                      self.critic.lr_scheduler)
+            if self.rank == 0:
+                print(f"[resume] restored global_step={self.global_step}")

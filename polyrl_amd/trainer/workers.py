@@ -1,0 +1,307 @@
+"""FSDP2 actor / critic / reference workers — SPMD, one process per GPU.
+
+Replaces the reference's Ray worker layer (StreamActorRolloutRefWorker /
+StreamCriticWorker, stream_fsdp_workers.py; StreamDataParallelPPOActor,
+stream_dp_actor.py:85-231) with a collective SPMD design: every rank runs the
+same method at the same time; FSDP2 (``fully_shard``) provides param
+all-gather / grad reduce-scatter on RCCL over xGMI.
+
+Streaming semantics preserved from the reference:
+  * gradient accumulation ACROSS update calls — optimizer steps only when the
+    cumulative streamed sample count crosses a ppo_mini_batch boundary
+    (is_opt_step; stream_dp_actor.py:226-230),
+  * dynamic micro-batching by token budget (ppo_max_token_len_per_gpu),
+  * loss scaled per-slice so a minibatch's gradient matches the non-streamed
+    equivalent.
+"""
+from __future__ import annotations
+
+import math
+import os
+from typing import Dict, List, Optional, Tuple
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from ..config import ActorConfig, CriticConfig, OptimConfig
+from ..core import algos
+from ..core.seqlen import fixed_micro_batches, prepare_dynamic_batch, restore_dynamic_batch
+from ..protocol import TensorBatch
+
+
+def _maybe_fully_shard(model: nn.Module, mixed_precision: bool = True):
+    """Apply FSDP2 per decoder layer + root.  Works for world_size 1..N on
+    nccl(RCCL) and gloo alike."""
+    if not (dist.is_available() and dist.is_initialized()):
+        return model
+    from torch.distributed.fsdp import MixedPrecisionPolicy, fully_shard
+    mp = None
+    if mixed_precision and next(model.parameters()).dtype == torch.bfloat16:
+        mp = MixedPrecisionPolicy(param_dtype=torch.bfloat16,
+                                  reduce_dtype=torch.float32)
+    kwargs = {"mp_policy": mp} if mp else {}
+    layers = None
+    if hasattr(model, "model") and hasattr(model.model, "layers"):
+        layers = model.model.layers
+    elif hasattr(model, "h"):
+        layers = model.h
+    elif hasattr(model, "trunk"):
+        layers = model.trunk.h
+    if layers is not None:
+        for layer in layers:
+            fully_shard(layer, **kwargs)
+    fully_shard(model, **kwargs)
+    return model
+
+
+def _build_optimizer(params, cfg: OptimConfig):
+    return torch.optim.AdamW(params, lr=cfg.lr, betas=tuple(cfg.betas),
+                             eps=cfg.eps, weight_decay=cfg.weight_decay)
+
+
+def _build_lr_scheduler(optimizer, cfg: OptimConfig):
+    warmup = cfg.lr_warmup_steps
+    if warmup <= 0 and cfg.lr_warmup_steps_ratio > 0 and cfg.total_training_steps > 0:
+        warmup = int(cfg.lr_warmup_steps_ratio * cfg.total_training_steps)
+
+    def lr_lambda(step):
+        if warmup > 0 and step < warmup:
+            return (step + 1) / warmup
+        if cfg.warmup_style == "cosine" and cfg.total_training_steps > 0:
+            prog = (step - warmup) / max(cfg.total_training_steps - warmup, 1)
+            return 0.5 * (1 + math.cos(math.pi * min(prog, 1.0)))
+        return 1.0
+
+    return torch.optim.lr_scheduler.LambdaLR(optimizer, lr_lambda)
+
+
+def _clip_grad_norm(model: nn.Module, max_norm: float) -> float:
+    """Grad-norm clip that understands DTensor (FSDP2) gradients."""
+    grads = [p.grad for p in model.parameters() if p.grad is not None]
+    if not grads:
+        return 0.0
+    total = torch.nn.utils.get_total_norm(grads)
+    if hasattr(total, "full_tensor"):
+        total = total.full_tensor()
+    torch.nn.utils.clip_grads_with_norm_(
+        (p for p in model.parameters() if p.grad is not None),
+        max_norm, total)
+    return float(total)
+
+
+def _sync_num_micro(k_local: int) -> int:
+    """FSDP2 unshard/reshard are collectives: every rank must run the SAME
+    number of micro-batch fwd (and bwd) passes.  All-reduce the max count;
+    ranks short of it run dummy passes (outputs discarded / zero-weighted)."""
+    if not (dist.is_available() and dist.is_initialized()) or \
+            dist.get_world_size() == 1:
+        return k_local
+    t = torch.tensor([k_local], dtype=torch.int64)
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    return int(t.item())
+
+
+def _model_inputs(batch: TensorBatch, device):
+    ids = batch["input_ids"].to(device)
+    am = batch["attention_mask"].to(device)
+    pos = batch["position_ids"].to(device)
+    resp = batch["responses"].to(device)
+    return ids, am, pos, resp
+
+
+class ActorWorker:
+    """Policy model under FSDP2: compute_log_prob + update_policy_stream."""
+
+    def __init__(self, model: nn.Module, cfg: ActorConfig, device="cpu",
+                 is_ref: bool = False):
+        self.cfg = cfg
+        self.device = device
+        self.is_ref = is_ref
+        self.model = _maybe_fully_shard(model)
+        if not is_ref:
+            self.optimizer = _build_optimizer(self.model.parameters(), cfg.optim)
+            self.lr_scheduler = _build_lr_scheduler(self.optimizer, cfg.optim)
+        self._accum_tokens = 0.0  # diagnostics
+
+    # ----------------------------------------------------------- log prob
+    @torch.no_grad()
+    def compute_log_prob(self, batch: TensorBatch, want_entropy: bool = False
+                         ) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
+        """old/ref log-probs of the sampled responses.  (B, resp_len) fp32."""
+        self.model.eval()
+        micro, parts = self._split(batch)
+        k = _sync_num_micro(len(micro))
+        lps, ents = [], []
+        for mb in micro:
+            lp, ent = self._forward_logprobs(mb, want_entropy)
+            lps.append(lp)
+            ents.append(ent)
+        for _ in range(k - len(micro)):  # dummy collective-alignment passes
+            self._forward_logprobs(micro[0].slice(slice(0, 1)), want_entropy)
+        lp = restore_dynamic_batch(torch.cat(lps), parts)
+        ent = restore_dynamic_batch(torch.cat(ents), parts) if want_entropy else None
+        return lp, ent
+
+    def _split(self, batch: TensorBatch):
+        if self.cfg.use_dynamic_bsz:
+            budget = self.cfg.ppo_max_token_len_per_gpu * \
+                self.cfg.ulysses_sequence_parallel_size
+            return prepare_dynamic_batch(batch, budget)
+        mbs = self.cfg.ppo_micro_batch_size_per_gpu or len(batch)
+        return fixed_micro_batches(batch, mbs)
+
+    def _forward_logprobs(self, mb: TensorBatch, want_entropy: bool,
+                          grad: bool = False):
+        ids, am, pos, resp = _model_inputs(mb, self.device)
+        Lr = resp.shape[1]
+        ctx = torch.enable_grad() if grad else torch.no_grad()
+        with ctx:
+            logits = self.model(ids, attention_mask=am, position_ids=pos,
+                                logits_slice=slice(-Lr - 1, -1))
+            logits = logits.float()
+            lp = algos.logprobs_from_logits(logits, resp)
+            ent = algos.entropy_from_logits(logits) if want_entropy else \
+                torch.zeros_like(lp)
+        return lp, ent
+
+    # ------------------------------------------------------------- update
+    def update_policy_stream(self, batch: TensorBatch, is_opt_step: bool,
+                             is_lr_step: bool, accum_scale: float
+                             ) -> Dict[str, List[float]]:
+        """fwd/bwd on one streamed slice; optimizer step at minibatch boundary.
+
+        accum_scale = slice_samples / minibatch_samples normalization factor so
+        the accumulated gradient equals the full-minibatch gradient.
+        """
+        assert not self.is_ref
+        self.model.train()
+        metrics: Dict[str, List[float]] = {}
+        micro, _ = self._split(batch)
+        k = _sync_num_micro(len(micro))
+        # dummy zero-weight passes keep FSDP fwd/bwd collectives aligned
+        dummies = [micro[0].slice(slice(0, 1)) for _ in range(k - len(micro))]
+        n_total = len(batch)
+        loss_fn = algos.get_policy_loss_fn(self.cfg.policy_loss_type)
+        for mi, mb in enumerate(micro + dummies):
+            is_dummy = mi >= len(micro)
+            ids, am, pos, resp = _model_inputs(mb, self.device)
+            Lr = resp.shape[1]
+            response_mask = mb["response_mask"].to(self.device)
+            old_log_prob = mb["old_log_probs"].to(self.device)
+            advantages = mb["advantages"].to(self.device)
+            logits = self.model(ids, attention_mask=am, position_ids=pos,
+                                logits_slice=slice(-Lr - 1, -1)).float()
+            log_prob = algos.logprobs_from_logits(logits, resp)
+            pg_loss, pg_clipfrac, ppo_kl, pg_clipfrac_lower = loss_fn(
+                old_log_prob=old_log_prob, log_prob=log_prob,
+                advantages=advantages, response_mask=response_mask,
+                clip_ratio=self.cfg.clip_ratio,
+                clip_ratio_low=self.cfg.clip_ratio_low,
+                clip_ratio_high=self.cfg.clip_ratio_high,
+                loss_agg_mode=self.cfg.loss_agg_mode)
+            loss = pg_loss
+            if self.cfg.entropy_coeff:
+                ent = algos.entropy_from_logits(logits)
+                loss = loss - self.cfg.entropy_coeff * algos.agg_loss(
+                    ent, response_mask, self.cfg.loss_agg_mode)
+            if self.cfg.use_kl_loss and "ref_log_probs" in mb.tensors:
+                ref_lp = mb["ref_log_probs"].to(self.device)
+                kld = algos.kl_penalty(log_prob, ref_lp, self.cfg.kl_loss_type)
+                kl_loss = algos.agg_loss(kld, response_mask, self.cfg.loss_agg_mode)
+                loss = loss + self.cfg.kl_loss_coef * kl_loss
+                if not is_dummy:
+                    metrics.setdefault("actor/kl_loss", []).append(kl_loss.item())
+            # micro-batch weight within slice x slice weight within minibatch
+            w = 0.0 if is_dummy else (len(mb) / n_total) * accum_scale
+            (loss * w).backward()
+            if not is_dummy:
+                metrics.setdefault("actor/pg_loss", []).append(pg_loss.item())
+                metrics.setdefault("actor/pg_clipfrac", []).append(pg_clipfrac.item())
+                metrics.setdefault("actor/ppo_kl", []).append(ppo_kl.item())
+        if is_opt_step:
+            gn = self._optimizer_step()
+            metrics.setdefault("actor/grad_norm", []).append(gn)
+        if is_lr_step:
+            self.lr_scheduler.step()
+            metrics.setdefault("actor/lr", []).append(
+                self.lr_scheduler.get_last_lr()[0])
+        return metrics
+
+    def _optimizer_step(self) -> float:
+        gn = _clip_grad_norm(self.model, self.cfg.optim.grad_clip)
+        if not math.isfinite(gn):
+            self.optimizer.zero_grad()
+            return gn
+        self.optimizer.step()
+        self.optimizer.zero_grad()
+        return gn
+
+
+class CriticWorker:
+    """Value model under FSDP2: compute_values + update_critic_stream."""
+
+    def __init__(self, model: nn.Module, cfg: CriticConfig, device="cpu"):
+        self.cfg = cfg
+        self.device = device
+        self.model = _maybe_fully_shard(model)
+        self.optimizer = _build_optimizer(self.model.parameters(), cfg.optim)
+        self.lr_scheduler = _build_lr_scheduler(self.optimizer, cfg.optim)
+
+    def _split(self, batch: TensorBatch):
+        if self.cfg.use_dynamic_bsz:
+            return prepare_dynamic_batch(batch, self.cfg.ppo_max_token_len_per_gpu)
+        mbs = self.cfg.ppo_micro_batch_size_per_gpu or len(batch)
+        return fixed_micro_batches(batch, mbs)
+
+    @torch.no_grad()
+    def compute_values(self, batch: TensorBatch) -> torch.Tensor:
+        self.model.eval()
+        micro, parts = self._split(batch)
+        k = _sync_num_micro(len(micro))
+        outs = []
+        for mi in range(k):
+            mb = micro[mi] if mi < len(micro) else micro[0].slice(slice(0, 1))
+            ids, am, pos, resp = _model_inputs(mb, self.device)
+            Lr = resp.shape[1]
+            values = self.model(ids, attention_mask=am, position_ids=pos)
+            if mi < len(micro):
+                outs.append(values[:, -Lr - 1:-1].float())
+        return restore_dynamic_batch(torch.cat(outs), parts)
+
+    def update_critic_stream(self, batch: TensorBatch, is_opt_step: bool,
+                             is_lr_step: bool, accum_scale: float
+                             ) -> Dict[str, List[float]]:
+        self.model.train()
+        metrics: Dict[str, List[float]] = {}
+        micro, _ = self._split(batch)
+        k = _sync_num_micro(len(micro))
+        dummies = [micro[0].slice(slice(0, 1)) for _ in range(k - len(micro))]
+        n_total = len(batch)
+        for mi, mb in enumerate(micro + dummies):
+            is_dummy = mi >= len(micro)
+            ids, am, pos, resp = _model_inputs(mb, self.device)
+            Lr = resp.shape[1]
+            response_mask = mb["response_mask"].to(self.device)
+            values = mb["values"].to(self.device)
+            returns = mb["returns"].to(self.device)
+            vpreds = self.model(ids, attention_mask=am, position_ids=pos)
+            vpreds = vpreds[:, -Lr - 1:-1].float()
+            vf_loss, vf_clipfrac = algos.compute_value_loss(
+                vpreds, returns, values, response_mask,
+                cliprange_value=self.cfg.cliprange_value,
+                loss_agg_mode=self.cfg.loss_agg_mode)
+            w = 0.0 if is_dummy else (len(mb) / n_total) * accum_scale
+            (vf_loss * w).backward()
+            if not is_dummy:
+                metrics.setdefault("critic/vf_loss", []).append(vf_loss.item())
+                metrics.setdefault("critic/vf_clipfrac", []).append(vf_clipfrac.item())
+        if is_opt_step:
+            gn = _clip_grad_norm(self.model, self.cfg.optim.grad_clip)
+            if math.isfinite(gn):
+                self.optimizer.step()
+            self.optimizer.zero_grad()
+            metrics.setdefault("critic/grad_norm", []).append(gn)
+        if is_lr_step:
+            self.lr_scheduler.step()
+        return metrics
