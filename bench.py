@@ -82,7 +82,9 @@ def main():
         global_batch * (args.steps + args.warmup + 1), 64)
     cfg.data.shuffle = False
     cfg.trainer.device = device
-    cfg.trainer.logger = []
+    # POLYRL_BENCH_LOG=1 prints per-step phase timings (gen/prep/update/...)
+    cfg.trainer.logger = (["console"]
+                          if os.environ.get("POLYRL_BENCH_LOG") else [])
     cfg.trainer.resume_mode = "disable"
     cfg.trainer.default_local_dir = "/tmp/polyrl_bench_ckpt"
 

@@ -1,0 +1,229 @@
+"""Scheduler contract tests (CPU tier): zero-queue RR dispatch, token-level
+continuation on instance failure, local time-box abort + remote tail,
+weight-version gating, adaptive balance — the behavioral contract of the
+reference Rust rollout-manager."""
+import asyncio
+
+import pytest
+
+from polyrl_amd.scheduler import (FakeInstance, GroupRequest, LoadBalanceState,
+                                  RolloutScheduler, SchedulerConfig)
+from polyrl_amd.scheduler.manager import StreamingBatchIterator
+from polyrl_amd.scheduler.types import MetricsUpdate, SamplingSpec
+
+
+def run(coro):
+    return asyncio.run(coro)
+
+
+def mk_req(gid=0, n=2, max_new=8, prompt=None):
+    return GroupRequest(gid=gid, input_ids=prompt or [5, 6, 7], n=n,
+                        sampling=SamplingSpec(max_new_tokens=max_new))
+
+
+def fake_expected(prompt, max_new):
+    """FakeInstance echo: token t = (prompt[-1] + 1 + t) % 50000."""
+    return [(prompt[-1] + 1 + t) % 50000 for t in range(max_new)]
+
+
+def test_basic_group_generation():
+    async def go():
+        sched = RolloutScheduler(SchedulerConfig(stats_interval_s=0.02))
+        inst = FakeInstance("i0")
+        await sched.register_instance(inst, skip_health_check=True)
+        res = await sched.process_group(mk_req(n=3, max_new=6))
+        await sched.close()
+        assert len(res.samples) == 3
+        for s in res.samples:
+            assert s.output_ids == fake_expected([5, 6, 7], 6)
+            assert s.finish_reason == "length"
+            assert len(s.output_logprobs) == 6
+            assert s.num_migrations == 0
+    run(go())
+
+
+def test_token_level_continuation_on_failure():
+    """Kill the serving instance mid-generation; the sample must continue
+    token-exactly on another instance (handlers.rs:330-418)."""
+    async def go():
+        sched = RolloutScheduler(SchedulerConfig(stats_interval_s=0.02))
+        bad = FakeInstance("bad", fail_after_tokens=3)
+        good = FakeInstance("good")
+        await sched.register_instance(bad, skip_health_check=True)
+        await sched.register_instance(good, skip_health_check=True)
+        # force the bad instance to serve first
+        sched._rr = len(sched._active) - 1
+        prompt = [10, 11]
+        res = await sched.process_group(mk_req(n=1, max_new=8, prompt=prompt))
+        await sched.close()
+        s = res.samples[0]
+        assert len(s.output_ids) == 8, s
+        # first 3 tokens from bad, then continuation: prompt' = prompt + 3 toks
+        first = fake_expected(prompt, 3)
+        cont = fake_expected(prompt + first, 5)
+        assert s.output_ids == first + cont
+        assert len(s.output_logprobs) == 8
+        assert s.num_migrations == 1
+        assert s.finish_reason == "length"
+        # bad instance evicted from the registry
+        assert "bad" not in [i.instance_id for i in sched.instances()]
+    run(go())
+
+
+def test_retry_cap_gives_error():
+    async def go():
+        cfg = SchedulerConfig(stats_interval_s=0.02, max_retries=2)
+        sched = RolloutScheduler(cfg)
+        # every instance dies immediately; with instant re-registration the
+        # retry cap must stop the loop
+        for k in range(6):
+            await sched.register_instance(
+                FakeInstance(f"f{k}", fail_after_tokens=0),
+                skip_health_check=True)
+        res = await sched.process_group(mk_req(n=1, max_new=4))
+        await sched.close()
+        assert res.samples[0].finish_reason in ("error", "abort")
+    run(go())
+
+
+def test_local_timebox_aborts_and_remote_continues():
+    """Submit with a tiny window: the slow local instance is aborted and
+    deactivated; the tail continues on the remote instance; the notifier is
+    the first streamed item (handlers.rs:442-564, :500-513)."""
+    async def go():
+        sched = RolloutScheduler(SchedulerConfig(stats_interval_s=0.02))
+        local = FakeInstance("local", is_local=True, token_time_s=0.05)
+        remote = FakeInstance("remote", is_local=False)
+        await sched.register_instance(local, skip_health_check=True)
+        await sched.register_instance(remote, skip_health_check=True)
+        # force dispatch to the local instance first
+        sched._states["remote"].assigned_batches = 10**9
+        items = []
+        async for it in sched.submit_batch([mk_req(n=1, max_new=20)],
+                                           max_local_gen_s=0.12):
+            items.append(it)
+        await sched.close()
+        assert items[0] == {"type": "notifier", "status": "success"}
+        res = items[1]
+        s = res.samples[0]
+        assert len(s.output_ids) == 20          # completed despite the abort
+        assert s.num_migrations >= 1            # local -> remote continuation
+        assert "remote" in res.instance_ids
+        # local removed from the active pool by the time-box
+        assert "local" not in sched._active
+        sched.reactivate_local()
+        assert "local" in sched._active
+    run(go())
+
+
+def test_timebox_noop_without_remote():
+    """With only local capacity the time-box must not strand the batch."""
+    async def go():
+        sched = RolloutScheduler(SchedulerConfig(stats_interval_s=0.02))
+        local = FakeInstance("l0", is_local=True, token_time_s=0.01)
+        await sched.register_instance(local, skip_health_check=True)
+        items = []
+        async for it in sched.submit_batch([mk_req(n=1, max_new=10)],
+                                           max_local_gen_s=0.03):
+            items.append(it)
+        await sched.close()
+        res = [i for i in items if not isinstance(i, dict)][0]
+        assert len(res.samples[0].output_ids) == 10
+        assert res.samples[0].finish_reason == "length"
+    run(go())
+
+
+def test_round_robin_spread():
+    async def go():
+        sched = RolloutScheduler(SchedulerConfig(
+            stats_interval_s=0.02, max_assigned_batches_per_stats_check=100))
+        insts = [FakeInstance(f"i{k}") for k in range(4)]
+        for i in insts:
+            await sched.register_instance(i, skip_health_check=True)
+        for g in range(8):
+            await sched.process_group(mk_req(gid=g, n=1, max_new=2))
+        await sched.close()
+        served = [len(i.served_gids) for i in insts]
+        assert sum(served) == 8
+        assert max(served) - min(served) <= 1, served  # even spread
+    run(go())
+
+
+def test_weight_version_gating():
+    """Version bump clears the pool and re-adds local; remote instances
+    rejoin only after a successful update (handlers.rs:566-795)."""
+    async def go():
+        sched = RolloutScheduler(SchedulerConfig(stats_interval_s=0.02))
+        local = FakeInstance("loc", is_local=True)
+        r_ok = FakeInstance("r_ok", is_local=False)
+        r_bad = FakeInstance("r_bad", is_local=False)
+        for i in (local, r_ok, r_bad):
+            await sched.register_instance(i, skip_health_check=True)
+        assert sched.num_active() == 3
+
+        await sched.update_weight_version(1)
+        assert sched._active == ["loc"]       # only local re-added
+
+        recv = sched.get_receive_instances()
+        assert {i.instance_id for i in recv} == {"r_ok", "r_bad"}
+        # CAS: a second call returns nothing while updating
+        assert sched.get_receive_instances() == []
+
+        r_bad.healthy = False                  # its update will fail
+        await sched.finish_weight_update("r_ok", 1, success=True)
+        await sched.finish_weight_update("r_bad", 1, success=True)
+        assert set(sched._active) == {"loc", "r_ok"}
+        assert r_ok.weight_version == 1
+        # stale instance never rejoins until a later successful update
+        assert "r_bad" not in sched._active
+        # monotonicity enforced
+        with pytest.raises(AssertionError):
+            await sched.update_weight_version(1)
+        await sched.close()
+    run(go())
+
+
+def test_streaming_iterator_sync_facade():
+    sched = RolloutScheduler(SchedulerConfig(stats_interval_s=0.02))
+
+    async def setup():
+        await sched.register_instance(FakeInstance("i0"),
+                                      skip_health_check=True)
+    asyncio.run(setup())
+    groups = [mk_req(gid=g, n=2, max_new=4) for g in range(3)]
+    it = StreamingBatchIterator(sched, groups, max_local_gen_s=0)
+    items = list(it)
+    assert items[0]["type"] == "notifier"
+    results = items[1:]
+    assert sorted(r.gid for r in results) == [0, 1, 2]
+    for r in results:
+        assert all(len(s.output_ids) == 4 for s in r.samples)
+
+
+def test_balance_adapts_window():
+    b = LoadBalanceState(initial_gen_s=100.0)
+    # trainer bubble large vs remote bubble -> grow local window
+    w1 = b.update(step_time_s=200.0, trainer_bubble_s=80.0,
+                  step_throughput=10.0, num_instances=2)
+    w2 = b.update(step_time_s=200.0, trainer_bubble_s=80.0,
+                  step_throughput=10.0, num_instances=2)
+    assert w2 > 100.0
+    # trainer bubble tiny vs remote bubble -> shrink, floored at 5 s
+    b2 = LoadBalanceState(initial_gen_s=30.0)
+    b2.update(1000.0, 0.0, 5.0, 3)
+    for _ in range(50):
+        w = b2.update(1000.0, 0.0, 5.0, 3)
+    assert w == pytest.approx(5.0)
+
+
+def test_scheduler_metrics_feedback():
+    async def go():
+        sched = RolloutScheduler(SchedulerConfig(stats_interval_s=0.02))
+        await sched.register_instance(FakeInstance("r0", is_local=False),
+                                      skip_health_check=True)
+        out = sched.update_metrics(MetricsUpdate(
+            step_time_s=100.0, trainer_bubble_time_s=10.0,
+            step_throughput=5.0))
+        assert "new_max_gen_s" in out and out["num_instances"] == 1
+        await sched.close()
+    run(go())
