@@ -53,112 +53,57 @@ class RolloutInstance:
 
 
 class InProcessInstance(RolloutInstance):
-    """Co-located engine on this rank's GPU.
-
-    A dedicated pump thread advances ``engine.step()`` whenever requests are
-    pending, and resolves per-request futures back onto the asyncio loop.
-    The weight-update path takes the pump lock so updates exclude in-flight
-    generation (the reference's model_update_lock.writer_lock contract,
-    patches.py:482).
-    """
+    """Co-located engine on this rank's GPU, pumped by rollout.runner.
+    Weight updates run under the step lock so they exclude in-flight
+    generation (patches.py:482 writer-lock contract)."""
 
     def __init__(self, engine, instance_id: str = "local-0",
                  weight_source=None):
+        from ..rollout.runner import EngineRunner
         self.engine = engine
+        self.runner = EngineRunner(engine)
         self.instance_id = instance_id
         self.is_local = True
         self.weight_source = weight_source  # callable version -> state_dict
-        self._lock = threading.Lock()       # engine-step / weight-update mutex
-        self._futures: Dict[str, asyncio.Future] = {}
-        self._loop: Optional[asyncio.AbstractEventLoop] = None
-        self._pump: Optional[threading.Thread] = None
-        self._stop = threading.Event()
-        self._wake = threading.Event()
-        self._gen_tokens = 0
-        self._gen_window_t = time.monotonic()
-        self._throughput = 0.0
-        self.weight_version = 0
 
-    # ------------------------------------------------------------- pump
-    def _ensure_pump(self):
-        if self._pump is None or not self._pump.is_alive():
-            self._stop.clear()
-            self._pump = threading.Thread(target=self._pump_loop, daemon=True)
-            self._pump.start()
+    @property
+    def weight_version(self) -> int:
+        return self.runner.weight_version
 
-    def _pump_loop(self):
-        while not self._stop.is_set():
-            if not self.engine.has_work():
-                self._wake.wait(timeout=0.05)
-                self._wake.clear()
-                continue
-            with self._lock:
-                outs = self.engine.step()
-            if outs:
-                self._gen_tokens += sum(len(o.output_ids) for o in outs)
-                now = time.monotonic()
-                dt = now - self._gen_window_t
-                if dt > 0.5:
-                    self._throughput = self._gen_tokens / dt
-                    self._gen_tokens = 0
-                    self._gen_window_t = now
-                for o in outs:
-                    fut = self._futures.pop(o.rid, None)
-                    if fut is not None and self._loop is not None:
-                        self._loop.call_soon_threadsafe(
-                            fut.set_result, SampleResult(
-                                output_ids=list(o.output_ids),
-                                output_logprobs=list(o.output_logprobs),
-                                finish_reason=o.finish_reason,
-                                completion_tokens=len(o.output_ids)))
-
-    # ---------------------------------------------------------- contract
     async def generate_group(self, req: GroupRequest) -> List[SampleResult]:
         from ..rollout.engine import SamplingParams
-        self._loop = asyncio.get_running_loop()
-        self._ensure_pump()
         sp = SamplingParams(
             temperature=req.sampling.temperature, top_k=req.sampling.top_k,
             top_p=req.sampling.top_p,
             max_new_tokens=req.sampling.max_new_tokens,
             stop_token_ids=tuple(req.sampling.stop_token_ids))
-        futs = []
-        for s in range(req.n):
-            rid = f"g{req.gid}-s{s}-{id(req)}"
-            fut = self._loop.create_future()
-            self._futures[rid] = fut
-            with self._lock:
-                self.engine.add_request(rid, req.input_ids, sp)
-            futs.append(fut)
-        self._wake.set()
-        return list(await asyncio.gather(*futs))
+        outs = await self.runner.generate(req.input_ids, sp, n=req.n)
+        return [SampleResult(output_ids=list(o.output_ids),
+                             output_logprobs=list(o.output_logprobs),
+                             finish_reason=o.finish_reason,
+                             completion_tokens=len(o.output_ids))
+                for o in outs]
 
     def get_stats(self) -> InstanceStats:
-        return InstanceStats(num_running=self.engine.num_running(),
-                             num_queued=self.engine.num_queued(),
-                             gen_throughput=self._throughput)
+        s = self.runner.stats()
+        return InstanceStats(num_running=s["#running_req"],
+                             num_queued=s["#queue_req"],
+                             gen_throughput=s["last_gen_throughput"])
 
     async def update_weights(self, version: int, bootstrap: bool = False
                              ) -> bool:
         if self.weight_source is None:
-            self.weight_version = version
+            self.runner.weight_version = version
             return True
         sd = self.weight_source(version)
-        with self._lock:  # excludes generation while swapping
-            self.engine.model.load_state_dict(sd, strict=False)
-        self.weight_version = version
+        self.runner.update_weights(sd, version, strict=False)
         return True
 
     def abort_all(self):
-        with self._lock:
-            self.engine.abort_request(abort_all=True)
-        self._wake.set()
+        self.runner.abort(abort_all=True)
 
     async def shutdown(self):
-        self._stop.set()
-        self._wake.set()
-        if self._pump is not None:
-            self._pump.join(timeout=2.0)
+        self.runner.stop()
 
 
 class FakeInstance(RolloutInstance):

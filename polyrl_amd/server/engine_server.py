@@ -1,0 +1,205 @@
+"""FastAPI app exposing one Engine as a remote rollout instance.
+
+Route surface mirrors what the reference scheduler + clients consume from a
+rollout server (handlers.rs:153,264,399,656,919; sglang_http_async_engine.py:
+155-299): token-in/token-out /generate with optional SSE streaming, stats,
+abort, weight updates, memory occupation switches.
+
+Weight delivery over HTTP: POST /update_weights_from_agent with a
+safetensors payload (body bytes) or a node-local file path — the co-located
+fast path never goes through HTTP (transfer/ does hipIPC + RCCL instead).
+"""
+import asyncio
+import json
+from typing import List, Optional
+
+from fastapi import FastAPI, Request
+from fastapi.responses import JSONResponse, StreamingResponse
+
+from ..rollout.engine import Engine, SamplingParams
+from ..rollout.runner import EngineRunner
+
+
+def create_app(engine: Engine, runner: Optional[EngineRunner] = None):
+
+    app = FastAPI(title="polyrl-amd rollout instance")
+    runner = runner or EngineRunner(engine)
+    app.state.runner = runner
+    app.state.engine = engine
+
+    def _sp(d: dict) -> SamplingParams:
+        return SamplingParams(
+            temperature=float(d.get("temperature", 1.0)),
+            top_k=int(d.get("top_k", -1)),
+            top_p=float(d.get("top_p", 1.0)),
+            max_new_tokens=int(d.get("max_new_tokens", 128)),
+            stop_token_ids=tuple(d.get("stop_token_ids", ())))
+
+    def _meta(o, include_logprobs: bool) -> dict:
+        m = {"finish_reason": {"type": o.finish_reason},
+             "completion_tokens": len(o.output_ids)}
+        if include_logprobs:
+            # reference format: [(logprob, token_id), ...]
+            m["output_token_logprobs"] = [
+                [lp, t] for lp, t in zip(o.output_logprobs, o.output_ids)]
+        return m
+
+    @app.post("/generate")
+    async def generate(request: Request):
+        body = await request.json()
+        input_ids: List[int] = body["input_ids"]
+        sp = _sp(body.get("sampling_params", {}))
+        n = int(body.get("sampling_params", {}).get("n", 1))
+        want_lp = bool(body.get("return_logprob", False))
+        stream = bool(body.get("stream", False))
+
+        futs = [runner.submit(input_ids, sp) for _ in range(n)]
+
+        if not stream:
+            outs = await asyncio.gather(*futs)
+            return JSONResponse([
+                {"index": i, "output_ids": o.output_ids,
+                 "meta_info": _meta(o, want_lp)}
+                for i, o in enumerate(outs)])
+
+        async def sse():
+            pending = {asyncio.ensure_future(_tag(i, f)): i
+                       for i, f in enumerate(futs)}
+            while pending:
+                done, _ = await asyncio.wait(
+                    pending.keys(), return_when=asyncio.FIRST_COMPLETED)
+                for d in done:
+                    pending.pop(d)
+                    i, o = d.result()
+                    chunk = {"index": i, "output_ids": o.output_ids,
+                             "meta_info": _meta(o, want_lp)}
+                    yield f"data: {json.dumps(chunk)}\n\n"
+            yield "data: [DONE]\n\n"
+
+        async def _tag(i, f):
+            return i, await f
+
+        return StreamingResponse(sse(), media_type="text/event-stream")
+
+    @app.get("/health")
+    async def health():
+        return {"status": "ok"}
+
+    @app.get("/health_generate")
+    async def health_generate():
+        # liveness probe that exercises the engine (one 1-token decode)
+        sp = SamplingParams(max_new_tokens=1, temperature=0.0)
+        out = await runner.submit([1], sp)
+        return {"status": "ok", "tokens": len(out.output_ids)}
+
+    @app.get("/get_server_info")
+    async def get_server_info():
+        return runner.stats()
+
+    @app.post("/abort_request")
+    async def abort_request(request: Request):
+        body = await request.json()
+        runner.abort(rid=body.get("rid"),
+                     abort_all=bool(body.get("abort_all", False)))
+        return {"status": "ok"}
+
+    @app.post("/flush_cache")
+    async def flush_cache():
+        # paged KV frees per request; full flush = drop any retained state
+        return {"status": "ok"}
+
+    @app.post("/update_weights_from_agent")
+    async def update_weights_from_agent(request: Request):
+        """Install new weights.  Accepts either
+        {"version": V, "path": "/node/local/file.safetensors"} or raw
+        safetensors bytes with the version in X-Weight-Version header."""
+        import torch
+        ctype = request.headers.get("content-type", "")
+        if ctype.startswith("application/json"):
+            body = await request.json()
+            version = int(body["version"])
+            path = body.get("path")
+            if path:
+                from safetensors.torch import load_file
+                sd = load_file(path)
+            else:
+                return JSONResponse({"success": False,
+                                     "message": "no path given"}, 400)
+        else:
+            raw = await request.body()
+            version = int(request.headers.get("x-weight-version", "0"))
+            from safetensors.torch import load
+            sd = load(raw)
+        # run the blocking swap off the event loop; takes the step lock so
+        # it excludes in-flight generation
+        await asyncio.get_running_loop().run_in_executor(
+            None, lambda: runner.update_weights(sd, version))
+        return {"success": True, "message": f"weights at version {version}"}
+
+    @app.post("/release_memory_occupation")
+    async def release_memory_occupation():
+        import torch
+        if torch.cuda.is_available():
+            torch.cuda.empty_cache()
+        return {"status": "ok"}
+
+    @app.post("/resume_memory_occupation")
+    async def resume_memory_occupation():
+        return {"status": "ok"}
+
+    @app.post("/shutdown")
+    async def shutdown():
+        runner.stop()
+        return {"status": "ok"}
+
+    return app
+
+
+def main():
+    """``python -m polyrl_amd.server.engine_server --model qwen2.5-1.5b
+    --port 30001 [--manager http://head:5000]`` — launch a remote elastic
+    instance and (optionally) self-register with the scheduler's HTTP facade
+    (reference: examples/scripts/launch_sglang.sh + patches.py:513-543)."""
+    import argparse
+
+    import torch
+    import uvicorn
+
+    from ..models.registry import get_config
+
+    p = argparse.ArgumentParser()
+    p.add_argument("--model", required=True)
+    p.add_argument("--host", default="0.0.0.0")
+    p.add_argument("--port", type=int, default=30001)
+    p.add_argument("--dtype", default="bfloat16")
+    p.add_argument("--kv-gb", type=float, default=8.0)
+    p.add_argument("--manager", default=None,
+                   help="scheduler HTTP endpoint to register with")
+    p.add_argument("--load", default=None,
+                   help="safetensors checkpoint to load (else random init)")
+    args = p.parse_args()
+
+    cfg = get_config(args.model)
+    device = "cuda" if torch.cuda.is_available() else "cpu"
+    dtype = getattr(torch, args.dtype) if device == "cuda" else torch.float32
+    engine = Engine(cfg, device=device, dtype=dtype,
+                    kv_bytes_budget=int(args.kv_gb * (1 << 30)))
+    if args.load:
+        from safetensors.torch import load_file
+        engine.model.load_state_dict(load_file(args.load), strict=False)
+    else:
+        for _, t in engine.model._name_map.items():
+            t.normal_(0, 0.02)
+    app = create_app(engine)
+
+    if args.manager:
+        import requests
+        requests.post(f"{args.manager}/register_rollout_instance",
+                      json={"addr": f"http://{args.host}:{args.port}"},
+                      timeout=10)
+
+    uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,140 @@
+"""HTTP facade tests (CPU tier, in-process ASGI — no sockets): the engine
+server route surface + HttpInstance client + scheduler integration with a
+remote instance (SURVEY.md §2.4.2 engine HTTP contract, §3.4 lifecycle)."""
+import asyncio
+import json
+
+import pytest
+import torch
+
+from polyrl_amd.models import create_model, get_model_config
+from polyrl_amd.rollout.engine import Engine, SamplingParams
+from polyrl_amd.rollout.runner import EngineRunner
+from polyrl_amd.scheduler import RolloutScheduler, SchedulerConfig
+from polyrl_amd.scheduler.types import GroupRequest, SamplingSpec
+from polyrl_amd.server import HttpInstance, create_app
+
+
+@pytest.fixture(scope="module")
+def served():
+    torch.manual_seed(0)
+    cfg = get_model_config("llama-debug-cpu")
+    model = create_model(cfg, kind="actor", dtype="float32", device="cpu")
+    eng = Engine(cfg, device="cpu", dtype=torch.float32,
+                 kv_bytes_budget=8 << 20, max_num_batched_tokens=64,
+                 max_running_requests=8)
+    eng.model.load_state_dict(model.state_dict())
+    runner = EngineRunner(eng)
+    app = create_app(eng, runner)
+    return cfg, model, eng, app
+
+
+def _transport(app):
+    import httpx
+    return httpx.ASGITransport(app=app)
+
+
+def test_generate_nonstream(served):
+    cfg, model, eng, app = served
+    import httpx
+
+    async def go():
+        async with httpx.AsyncClient(transport=_transport(app),
+                                     base_url="http://t") as c:
+            r = await c.post("/generate", json={
+                "input_ids": [3, 4, 5],
+                "sampling_params": {"n": 2, "max_new_tokens": 4,
+                                    "temperature": 1.0},
+                "return_logprob": True})
+            assert r.status_code == 200
+            outs = r.json()
+            assert len(outs) == 2
+            for o in outs:
+                assert len(o["output_ids"]) == 4
+                assert o["meta_info"]["finish_reason"]["type"] == "length"
+                assert len(o["meta_info"]["output_token_logprobs"]) == 4
+            r = await c.get("/get_server_info")
+            d = r.json()
+            assert "#running_req" in d and "#queue_req" in d
+            r = await c.get("/health")
+            assert r.json()["status"] == "ok"
+    asyncio.run(go())
+
+
+def test_generate_stream_sse(served):
+    cfg, model, eng, app = served
+    import httpx
+
+    async def go():
+        async with httpx.AsyncClient(transport=_transport(app),
+                                     base_url="http://t") as c:
+            chunks = []
+            async with c.stream("POST", "/generate", json={
+                    "input_ids": [7, 8], "stream": True,
+                    "sampling_params": {"n": 3, "max_new_tokens": 3},
+                    "return_logprob": True}) as resp:
+                async for line in resp.aiter_lines():
+                    if line.startswith("data: ") and line != "data: [DONE]":
+                        chunks.append(json.loads(line[6:]))
+            assert len(chunks) == 3
+            assert sorted(ch["index"] for ch in chunks) == [0, 1, 2]
+    asyncio.run(go())
+
+
+def test_http_instance_with_scheduler(served):
+    """A remote HttpInstance serves groups through the scheduler."""
+    cfg, model, eng, app = served
+
+    async def go():
+        inst = HttpInstance("http://remote-1", transport=_transport(app))
+        assert await inst.health()
+        sched = RolloutScheduler(SchedulerConfig(stats_interval_s=0.02))
+        await sched.register_instance(inst, skip_health_check=True)
+        res = await sched.process_group(GroupRequest(
+            gid=0, input_ids=[5, 6], n=2,
+            sampling=SamplingSpec(max_new_tokens=5)))
+        await sched.close()
+        assert len(res.samples) == 2
+        for s in res.samples:
+            assert len(s.output_ids) == 5
+            assert len(s.output_logprobs) == 5
+            assert s.finish_reason == "length"
+    asyncio.run(go())
+
+
+def test_weight_update_over_http(served, tmp_path):
+    """Weight install via safetensors path + version visible in server info;
+    greedy output changes accordingly."""
+    cfg, model, eng, app = served
+    import httpx
+    from safetensors.torch import save_file
+
+    sd = {k: (v + 0.01 * torch.randn_like(v)).contiguous()
+          for k, v in model.state_dict().items()}
+    path = str(tmp_path / "w1.safetensors")
+    save_file(sd, path)
+
+    async def go():
+        async with httpx.AsyncClient(transport=_transport(app),
+                                     base_url="http://t") as c:
+            r = await c.post("/update_weights_from_agent",
+                             json={"version": 7, "path": path})
+            assert r.status_code == 200 and r.json()["success"]
+            info = (await c.get("/get_server_info")).json()
+            assert info["weight_version"] == 7
+    asyncio.run(go())
+    # engine buffers actually changed
+    name = "model.embed_tokens.weight"
+    assert torch.allclose(eng.model._name_map[name], sd[name])
+
+
+def test_abort_over_http(served):
+    cfg, model, eng, app = served
+    import httpx
+
+    async def go():
+        async with httpx.AsyncClient(transport=_transport(app),
+                                     base_url="http://t") as c:
+            r = await c.post("/abort_request", json={"abort_all": True})
+            assert r.json()["status"] == "ok"
+    asyncio.run(go())
