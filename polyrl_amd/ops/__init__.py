@@ -61,10 +61,11 @@ def fused_add_rmsnorm(x: torch.Tensor, residual: torch.Tensor,
 
 
 def silu_mul(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
+    """gate/up may be strided row views into a fused gate_up projection."""
     if gate.is_cuda:
         ext = _require_ext()
-        out = torch.empty_like(gate)
-        ext.silu_mul(out, gate.contiguous(), up.contiguous())
+        out = torch.empty(gate.shape, dtype=gate.dtype, device=gate.device)
+        ext.silu_mul(out, gate, up)
         return out
     return ref.silu_mul(gate, up)
 
@@ -127,16 +128,24 @@ def gather_logprobs(logits: torch.Tensor, labels: torch.Tensor,
 
 def sample(logits: torch.Tensor, temperature: torch.Tensor,
            top_k: torch.Tensor, top_p: torch.Tensor, seed: int,
-           generator: Optional[torch.Generator] = None
+           generator: Optional[torch.Generator] = None,
+           no_filter: Optional[bool] = None
            ) -> Tuple[torch.Tensor, torch.Tensor]:
-    """(N, V) -> (tokens (N,) int64, logprobs (N,) fp32 under raw softmax)."""
+    """(N, V) -> (tokens (N,) int64, logprobs (N,) fp32 under raw softmax).
+
+    ``no_filter=True`` asserts every row has top_k<=0 and top_p>=1 (the
+    common RL rollout setting) and takes the fused single-pass kernel; pass
+    it from the caller to avoid a device sync here."""
     if logits.is_cuda:
         ext = _require_ext()
         N = logits.size(0)
         tokens = torch.empty(N, dtype=torch.int64, device=logits.device)
         lps = torch.empty(N, dtype=torch.float32, device=logits.device)
+        if no_filter is None:
+            no_filter = bool(((top_k <= 0) | (top_k >= logits.size(1))).all()
+                             and (top_p >= 1.0).all())
         ext.sample(tokens, lps, logits.contiguous(), temperature.float(),
-                   top_k.int(), top_p.float(), seed)
+                   top_k.int(), top_p.float(), seed, no_filter)
         return tokens, lps
     return ref.top_k_top_p_sample(logits, temperature, top_k, top_p, generator)
 
@@ -150,8 +159,7 @@ def kv_cache_append(k_cache: torch.Tensor, v_cache: torch.Tensor,
     """Scatter new token K/V rows (N, Hk, D) into page slots."""
     if k_cache.is_cuda:
         ext = _require_ext()
-        ext.kv_cache_append(k_cache, v_cache, k.contiguous(), v.contiguous(),
-                            slot_mapping.int())
+        ext.kv_cache_append(k_cache, v_cache, k, v, slot_mapping.int())
         return
     page_size = k_cache.shape[1]
     flat_k = k_cache.view(-1, *k_cache.shape[2:])
@@ -170,7 +178,7 @@ def paged_attention_decode(q: torch.Tensor, k_cache: torch.Tensor,
                            ) -> torch.Tensor:
     if q.is_cuda:
         ext = _require_ext()
-        out = torch.empty_like(q)
+        out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         ext.paged_attention_decode(out, q, k_cache, v_cache,
                                    page_table.int(), context_lens.int(), scale)
         return out
@@ -197,10 +205,10 @@ def varlen_prefill_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                              causal: bool = True) -> torch.Tensor:
     if q.is_cuda:
         ext = _require_ext()
-        out = torch.empty_like(q)
+        out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         tile_seq, tile_q0 = _build_prefill_tiles(cu_seqlens_q)
         ext.varlen_prefill_attention(
-            out, q, k.contiguous(), v.contiguous(), cu_seqlens_q.int(),
+            out, q, k, v, cu_seqlens_q.int(),
             cu_seqlens_k.int(), tile_seq.to(q.device), tile_q0.to(q.device),
             scale, causal)
         return out

@@ -28,7 +28,7 @@ __global__ __launch_bounds__(256) void paged_decode_attn_kernel(
     const bf16_t* __restrict__ v_cache,
     const int* __restrict__ page_table,    // (B, max_pages)
     const int* __restrict__ context_lens,  // (B,)
-    int Hq, int Hk, int page_size, int max_pages, float scale) {
+    int Hq, int Hk, int page_size, int max_pages, float scale, long ldq) {
   constexpr int GL = D / 8;        // lanes per key group
   constexpr int KPW = 64 / GL;     // keys per wave per iteration
   constexpr int NW = 4;            // waves per block
@@ -46,7 +46,7 @@ __global__ __launch_bounds__(256) void paged_decode_attn_kernel(
 #pragma unroll
   for (int h = 0; h < G; ++h) {
     const bf16x8 qv = *reinterpret_cast<const bf16x8*>(
-        q + (((long)b * Hq + hk * G + h) * D + d0));
+        q + ((long)b * ldq + (long)(hk * G + h) * D + d0));
 #pragma unroll
     for (int j = 0; j < 8; ++j) qr[h][j] = bf2f(qv.v[j]) * scale;
   }
@@ -173,7 +173,7 @@ static void dispatch_g(torch::Tensor& out, const torch::Tensor& q,
         (bf16_t*)out.data_ptr(), (const bf16_t*)q.data_ptr(),
         (const bf16_t*)k_cache.data_ptr(), (const bf16_t*)v_cache.data_ptr(),
         page_table.data_ptr<int>(), context_lens.data_ptr<int>(), Hq, Hk,
-        page_size, max_pages, scale);
+        page_size, max_pages, scale, q.stride(0));
   };
   switch (G) {
     case 1: args(paged_decode_attn_kernel<D, 1>); break;
@@ -193,7 +193,9 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
                             torch::Tensor k_cache, torch::Tensor v_cache,
                             torch::Tensor page_table,
                             torch::Tensor context_lens, double scale) {
-  TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.is_contiguous());
+  TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16);
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2),
+              "q heads/dims must be packed (token stride free)");
   TORCH_CHECK(page_table.dtype() == torch::kInt32 && page_table.is_contiguous());
   TORCH_CHECK(context_lens.dtype() == torch::kInt32);
   const int Hq = q.size(1), D = q.size(2);

@@ -46,7 +46,7 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
     const int* __restrict__ cu_k,
     const int* __restrict__ tile_seq,  // (ntiles,)
     const int* __restrict__ tile_q0,   // (ntiles,) local first q row
-    int Hq, int Hk, float scale, int causal) {
+    int Hq, int Hk, float scale, int causal, long ldq, long ldk, long ldv) {
   __shared__ bf16_t Ks[KVBLK * HEAD_DIM];          // swizzled
   __shared__ bf16_t Vs[KVBLK * HEAD_DIM];          // row-major
   __shared__ bf16_t Ps[NWAVE][QROWS_PER_WAVE * KVBLK];
@@ -75,7 +75,7 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
   {
     const int qrow_l = wq0 + l15;
     const long grow = (long)qbeg + ((qrow_l < Lq) ? qrow_l : 0);
-    const bf16_t* qp = q + (grow * Hq + hq) * HEAD_DIM + lhi * 8;
+    const bf16_t* qp = q + grow * ldq + (long)hq * HEAD_DIM + lhi * 8;
 #pragma unroll
     for (int ks = 0; ks < 4; ++ks)
       qfrag[ks] = *reinterpret_cast<const bf16x8_t*>(qp + ks * 32);
@@ -110,9 +110,9 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
         bf16x8 kv8, vv8;
         if (krow < Lk) {
           kv8 = *reinterpret_cast<const bf16x8*>(
-              k + ((long)(kbeg + krow) * Hk + hk) * HEAD_DIM + col);
+              k + (long)(kbeg + krow) * ldk + (long)hk * HEAD_DIM + col);
           vv8 = *reinterpret_cast<const bf16x8*>(
-              v + ((long)(kbeg + krow) * Hk + hk) * HEAD_DIM + col);
+              v + (long)(kbeg + krow) * ldv + (long)hk * HEAD_DIM + col);
         } else {
           for (int j = 0; j < 8; ++j) kv8.v[j] = f2bf(0.f), vv8.v[j] = f2bf(0.f);
         }
@@ -226,8 +226,13 @@ void varlen_prefill_attention(torch::Tensor out, torch::Tensor q,
                               torch::Tensor cu_seqlens_k,
                               torch::Tensor tile_seq, torch::Tensor tile_q0,
                               double scale, bool causal) {
-  TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.is_contiguous());
-  TORCH_CHECK(k.is_contiguous() && v.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16);
+  auto packed = [](const torch::Tensor& t) {
+    return t.stride(2) == 1 && t.stride(1) == t.size(2);
+  };
+  TORCH_CHECK(packed(q) && packed(k) && packed(v),
+              "q/k/v heads and dims must be packed (token stride free)");
+  TORCH_CHECK(out.is_contiguous());
   TORCH_CHECK(cu_seqlens_q.dtype() == torch::kInt32);
   TORCH_CHECK(tile_seq.dtype() == torch::kInt32 && tile_q0.dtype() == torch::kInt32);
   const int Hq = q.size(1), D = q.size(2), Hk = k.size(1);
@@ -240,6 +245,6 @@ void varlen_prefill_attention(torch::Tensor out, torch::Tensor q,
       (const bf16_t*)k.data_ptr(), (const bf16_t*)v.data_ptr(),
       cu_seqlens_q.data_ptr<int>(), cu_seqlens_k.data_ptr<int>(),
       tile_seq.data_ptr<int>(), tile_q0.data_ptr<int>(), Hq, Hk, (float)scale,
-      causal ? 1 : 0);
+      causal ? 1 : 0, q.stride(0), k.stride(0), v.stride(0));
   HIP_CHECK_KERNEL();
 }

@@ -86,14 +86,21 @@ void fused_add_rmsnorm(torch::Tensor out, torch::Tensor residual,
 }
 
 // ----------------------------------------------------------------- SiLU-mul
+// gate/up may be strided row views into a fused (N, 2I) gate_up projection
+// (row stride ldg/ldu elements, inner dim contiguous) — avoids the
+// .contiguous() copies in the engine forward.
 __global__ void silu_mul_kernel(bf16_t* __restrict__ out,
                                 const bf16_t* __restrict__ gate,
-                                const bf16_t* __restrict__ up, long nvec) {
+                                const bf16_t* __restrict__ up, long rows,
+                                int cvec /* cols/8 */, long ldg, long ldu) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
+  const long nvec = rows * cvec;
   for (; i < nvec; i += stride) {
-    bf16x8 g = reinterpret_cast<const bf16x8*>(gate)[i];
-    bf16x8 u = reinterpret_cast<const bf16x8*>(up)[i];
+    const long r = i / cvec;
+    const long c = i % cvec;
+    bf16x8 g = *reinterpret_cast<const bf16x8*>(gate + r * ldg + c * 8);
+    bf16x8 u = *reinterpret_cast<const bf16x8*>(up + r * ldu + c * 8);
     bf16x8 o;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
@@ -107,15 +114,20 @@ __global__ void silu_mul_kernel(bf16_t* __restrict__ out,
 
 void silu_mul(torch::Tensor out, torch::Tensor gate, torch::Tensor up) {
   TORCH_CHECK(gate.is_cuda() && gate.dtype() == torch::kBFloat16);
-  TORCH_CHECK(gate.is_contiguous() && up.is_contiguous() && out.is_contiguous());
-  const long n = gate.numel();
-  TORCH_CHECK(n % 8 == 0);
-  const long nvec = n / 8;
+  TORCH_CHECK(out.is_contiguous());
+  TORCH_CHECK(gate.dim() == 2 && up.dim() == 2);
+  TORCH_CHECK(gate.stride(1) == 1 && up.stride(1) == 1,
+              "inner dim must be contiguous");
+  const long rows = gate.size(0);
+  const long cols = gate.size(1);
+  TORCH_CHECK(cols % 8 == 0);
+  const long nvec = rows * (cols / 8);
   int blocks = (int)std::min<long>((nvec + 255) / 256, MAX_RESIDENT_BLOCKS);
   auto stream = at::hip::getCurrentHIPStream();
   silu_mul_kernel<<<dim3(blocks), dim3(256), 0, stream>>>(
       (bf16_t*)out.data_ptr(), (const bf16_t*)gate.data_ptr(),
-      (const bf16_t*)up.data_ptr(), nvec);
+      (const bf16_t*)up.data_ptr(), rows, (int)(cols / 8), gate.stride(0),
+      up.stride(0));
   HIP_CHECK_KERNEL();
 }
 
@@ -129,7 +141,7 @@ __global__ void rope_kernel(bf16_t* __restrict__ q, bf16_t* __restrict__ k,
                             const int* __restrict__ positions,
                             const float* __restrict__ cos_tab,
                             const float* __restrict__ sin_tab, int Hq, int Hk,
-                            int D) {
+                            int D, long ldq, long ldk /* token strides */) {
   const int tok = blockIdx.x;
   const int pos = positions[tok];
   const int half = D / 2;
@@ -141,8 +153,8 @@ __global__ void rope_kernel(bf16_t* __restrict__ q, bf16_t* __restrict__ k,
   for (int t = threadIdx.x; t < H * quads_per_head; t += blockDim.x) {
     const int h = t / quads_per_head;
     const int qd = (t % quads_per_head) * 4;
-    bf16_t* base = (h < Hq) ? q + ((long)tok * Hq + h) * D
-                            : k + ((long)tok * Hk + (h - Hq)) * D;
+    bf16_t* base = (h < Hq) ? q + (long)tok * ldq + (long)h * D
+                            : k + (long)tok * ldk + (long)(h - Hq) * D;
     bf16x4 x1 = *reinterpret_cast<bf16x4*>(base + qd);
     bf16x4 x2 = *reinterpret_cast<bf16x4*>(base + half + qd);
     float4 cv = *reinterpret_cast<const float4*>(c + qd);
@@ -163,8 +175,12 @@ __global__ void rope_kernel(bf16_t* __restrict__ q, bf16_t* __restrict__ k,
 
 void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
                   torch::Tensor cos_tab, torch::Tensor sin_tab) {
+  // q: (N, Hq, D), k: (N, Hk, D); token stride may exceed Hq*D (views into a
+  // fused qkv projection), heads/dims must be contiguous.
   TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16);
-  TORCH_CHECK(q.is_contiguous() && k.is_contiguous());
+  TORCH_CHECK(q.stride(2) == 1 && k.stride(2) == 1);
+  TORCH_CHECK(q.stride(1) == q.size(2) && k.stride(1) == k.size(2),
+              "head dim must be packed");
   TORCH_CHECK(positions.dtype() == torch::kInt32);
   const int N = q.size(0), Hq = q.size(1), D = q.size(2);
   const int Hk = k.size(1);
@@ -173,6 +189,6 @@ void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
   rope_kernel<<<dim3(N), dim3(256), 0, stream>>>(
       (bf16_t*)q.data_ptr(), (bf16_t*)k.data_ptr(),
       positions.data_ptr<int>(), cos_tab.data_ptr<float>(),
-      sin_tab.data_ptr<float>(), Hq, Hk, D);
+      sin_tab.data_ptr<float>(), Hq, Hk, D, q.stride(0), k.stride(0));
   HIP_CHECK_KERNEL();
 }

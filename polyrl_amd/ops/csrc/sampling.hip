@@ -130,6 +130,93 @@ DEV_INLINE uint32_t radix_select(const T* __restrict__ x, int V, int rounds,
   return prefix;
 }
 
+// Fast path: no top-k / top-p filter (the common RL rollout setting).
+// ONE streaming pass computes simultaneously, 8 elements per lane per step
+// (vectorized bf16x8/float4 loads, G13):
+//   * online max m + sumexp s (for the reported raw logprob),
+//   * raw argmax (greedy),
+//   * Gumbel-argmax z = x/T + G_i with the winner's logit tracked inline.
+// 1024 threads/block, one block per row: at decode batch >= 256 the whole
+// chip is busy; a second grid dim is unnecessary because decode sampling
+// runs alongside nothing.
+template <typename T, int VEC>
+__global__ __launch_bounds__(1024) void sample_fast_kernel(
+    int64_t* __restrict__ out_tokens, float* __restrict__ out_logprobs,
+    const T* __restrict__ logits, const float* __restrict__ temperature,
+    uint64_t seed, int V) {
+  __shared__ float red_f[16];
+  __shared__ int red_i[16];
+  const long row = blockIdx.x;
+  const T* x = logits + row * V;
+  const float temp = temperature[row];
+  const bool greedy = (temp == 0.f);
+  const float inv_temp = greedy ? 1.f : (1.0f / temp);
+  const uint64_t rowkey = seed + (uint64_t)row * 0x9e3779b97f4a7c15ull;
+
+  float m = -INFINITY, s = 0.f;       // online raw softmax state
+  float rmax = -INFINITY;             // raw argmax (greedy pick)
+  int rarg = 0;
+  float best = -INFINITY;             // gumbel winner
+  int best_i = 0;
+  float best_logit = 0.f;
+
+  const int nvec = V / VEC;
+  using vec_t = typename std::conditional<sizeof(T) == 2, bf16x8, float4>::type;
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    vec_t pack = reinterpret_cast<const vec_t*>(x)[i];
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float xi;
+      if constexpr (sizeof(T) == 2) xi = bf2f(((const bf16x8&)pack).v[j]);
+      else xi = ((const float*)&pack)[j];
+      const int idx = i * VEC + j;
+      if (xi > rmax) { rmax = xi; rarg = idx; }
+      if (xi > m) { s *= __expf(m - xi); m = xi; }
+      s += __expf(xi - m);
+      if (!greedy) {
+        float u = rng_uniform(rowkey, (uint64_t)idx);
+        float z = __fmaf_rn(xi, inv_temp, -__logf(-__logf(u)));
+        if (z > best) { best = z; best_i = idx; best_logit = xi; }
+      }
+    }
+  }
+  for (int idx = nvec * VEC + threadIdx.x; idx < V; idx += blockDim.x) {
+    float xi = (float)x[idx];
+    if (xi > rmax) { rmax = xi; rarg = idx; }
+    if (xi > m) { s *= __expf(m - xi); m = xi; }
+    s += __expf(xi - m);
+    if (!greedy) {
+      float u = rng_uniform(rowkey, (uint64_t)idx);
+      float z = __fmaf_rn(xi, inv_temp, -__logf(-__logf(u)));
+      if (z > best) { best = z; best_i = idx; best_logit = xi; }
+    }
+  }
+
+  // block reductions (16 waves)
+  ArgMax am = block_argmax<16>(rmax, rarg, red_f, red_i);
+  __syncthreads();
+  const float gm = am.v;
+  s *= (m == -INFINITY) ? 0.f : __expf(m - gm);
+  float gs = block_reduce_sum<16>(s, red_f);
+  __syncthreads();
+  const float lz_raw = gm + __logf(gs);
+
+  if (greedy) {
+    if (threadIdx.x == 0) {
+      out_tokens[row] = am.i;
+      out_logprobs[row] = (float)x[am.i] - lz_raw;
+    }
+    return;
+  }
+  // winner's raw logit rides along through a second argmax reduce keyed on z
+  ArgMax pick = block_argmax<16>(best, best_i, red_f, red_i);
+  if (threadIdx.x == 0) {
+    out_tokens[row] = pick.i;
+    out_logprobs[row] = (float)x[pick.i] - lz_raw;
+  }
+  (void)best_logit;
+}
+
 template <typename T>
 __global__ void sample_kernel(int64_t* __restrict__ out_tokens,
                               float* __restrict__ out_logprobs,
@@ -222,7 +309,8 @@ __global__ void sample_kernel(int64_t* __restrict__ out_tokens,
 
 void sample(torch::Tensor out_tokens, torch::Tensor out_logprobs,
             torch::Tensor logits, torch::Tensor temperature,
-            torch::Tensor top_k, torch::Tensor top_p, int64_t seed) {
+            torch::Tensor top_k, torch::Tensor top_p, int64_t seed,
+            bool no_filter) {
   TORCH_CHECK(logits.is_cuda() && logits.dim() == 2 && logits.is_contiguous());
   TORCH_CHECK(out_tokens.dtype() == torch::kInt64);
   TORCH_CHECK(out_logprobs.dtype() == torch::kFloat32);
@@ -232,6 +320,21 @@ void sample(torch::Tensor out_tokens, torch::Tensor out_logprobs,
   const long N = logits.size(0);
   const int V = logits.size(1);
   auto stream = at::hip::getCurrentHIPStream();
+  if (no_filter) {  // fused single-pass path (no top-k/top-p), G13-vectorized
+    if (logits.dtype() == torch::kBFloat16) {
+      sample_fast_kernel<bf16_t, 8><<<dim3(N), dim3(1024), 0, stream>>>(
+          out_tokens.data_ptr<int64_t>(), out_logprobs.data_ptr<float>(),
+          (const bf16_t*)logits.data_ptr(), temperature.data_ptr<float>(),
+          (uint64_t)seed, V);
+    } else {
+      sample_fast_kernel<float, 4><<<dim3(N), dim3(1024), 0, stream>>>(
+          out_tokens.data_ptr<int64_t>(), out_logprobs.data_ptr<float>(),
+          (const float*)logits.data_ptr(), temperature.data_ptr<float>(),
+          (uint64_t)seed, V);
+    }
+    HIP_CHECK_KERNEL();
+    return;
+  }
   if (logits.dtype() == torch::kBFloat16) {
     sample_kernel<bf16_t><<<dim3(N), dim3(256), 0, stream>>>(
         out_tokens.data_ptr<int64_t>(), out_logprobs.data_ptr<float>(),

@@ -230,10 +230,11 @@ class InferenceModel:
             qkv = hidden @ lw.wqkv.t()
             if lw.bqkv is not None:
                 qkv = qkv + lw.bqkv
-            q, k, v = qkv.split([Hq * D, Hk * D, Hk * D], dim=-1)
-            q = q.view(-1, Hq, D).contiguous()
-            k = k.view(-1, Hk, D).contiguous()
-            v = v.view(-1, Hk, D).contiguous()
+            # strided head views into the fused qkv row (no copies: the
+            # rope/kv-append/attention kernels take a free token stride)
+            q = qkv.narrow(1, 0, Hq * D).unflatten(1, (Hq, D))
+            k = qkv.narrow(1, Hq * D, Hk * D).unflatten(1, (Hk, D))
+            v = qkv.narrow(1, (Hq + Hk) * D, Hk * D).unflatten(1, (Hk, D))
             ops.apply_rope_inplace(q, k, positions, self.rope)
             ops.kv_cache_append(kv.k_cache[li], kv.v_cache[li], k, v,
                                 slot_mapping)
@@ -242,11 +243,13 @@ class InferenceModel:
             hidden, residual = ops.fused_add_rmsnorm(
                 hidden, residual, lw.post_ln, cfg.rms_norm_eps)
             gate_up = hidden @ lw.w_gate_up.t()
-            gate, up = gate_up.split(cfg.intermediate_size, dim=-1)
-            hidden = ops.silu_mul(gate.contiguous(), up.contiguous()) @ lw.w_down.t()
-        # final residual add + norm
-        final = (hidden.float() + residual.float()).to(self.dtype)
-        return ops.rmsnorm(final, self.final_norm, cfg.rms_norm_eps)
+            i_sz = cfg.intermediate_size
+            hidden = ops.silu_mul(gate_up.narrow(1, 0, i_sz),
+                                  gate_up.narrow(1, i_sz, i_sz)) @ lw.w_down.t()
+        # final residual add + norm (fused; residual buffer is dead after)
+        normed, _ = ops.fused_add_rmsnorm(hidden, residual, self.final_norm,
+                                          cfg.rms_norm_eps)
+        return normed
 
     @torch.no_grad()
     def _forward_tokens_gpt2(self, token_ids, positions, kv, slot_mapping,
@@ -454,8 +457,11 @@ class Engine:
         tp = torch.tensor([r.sampling.top_p for r in reqs],
                           dtype=torch.float32, device=self.device)
         seed = (self._seed * 0x9E3779B9 + self._step_counter) & 0x7FFFFFFFFFFF
+        no_filter = all(
+            (r.sampling.top_k <= 0 or r.sampling.top_k >= logits.shape[-1])
+            and r.sampling.top_p >= 1.0 for r in reqs)
         tokens, lps = ops.sample(logits, temp, tk, tp, seed,
-                                 generator=self._gen)
+                                 generator=self._gen, no_filter=no_filter)
         tokens_l = tokens.tolist()
         lps_l = lps.tolist()
         for i, r in enumerate(reqs):

@@ -248,3 +248,116 @@ def test_varlen_prefill_chunked_alignment():
     expect = ref.varlen_prefill_attention(q.cpu(), k.cpu(), v.cpu(), cu_q, cu_k,
                                           scale)
     assert rel_err(out.cpu(), expect) < 3e-2
+
+
+# ------------------------------------------------- strided (fused-qkv) views
+
+
+def test_silu_mul_strided_gate_up():
+    """gate/up as strided row views into a fused (N, 2I) projection."""
+    torch.manual_seed(31)
+    N, I = 33, 5632
+    gate_up = torch.randn(N, 2 * I, dtype=torch.bfloat16, device=DEV)
+    out = ops.silu_mul(gate_up.narrow(1, 0, I), gate_up.narrow(1, I, I))
+    expect = ref.silu_mul(gate_up[:, :I].cpu(), gate_up[:, I:].cpu())
+    assert rel_err(out.cpu(), expect) < 2e-2
+
+
+def test_rope_and_kv_append_strided_qkv():
+    """rope + kv-append on head views straight into a fused qkv row."""
+    torch.manual_seed(32)
+    N, Hq, Hk, D = 17, 8, 2, 128
+    table = ops.RopeTable(D, 256, 10000.0, DEV)
+    qkv = torch.randn(N, (Hq + 2 * Hk) * D, dtype=torch.bfloat16, device=DEV)
+    qkv_ref = qkv.clone()
+    pos = torch.randint(0, 256, (N,), dtype=torch.int32, device=DEV)
+
+    q = qkv.narrow(1, 0, Hq * D).unflatten(1, (Hq, D))
+    k = qkv.narrow(1, Hq * D, Hk * D).unflatten(1, (Hk, D))
+    v = qkv.narrow(1, (Hq + Hk) * D, Hk * D).unflatten(1, (Hk, D))
+    ops.apply_rope_inplace(q, k, pos, table)
+
+    qr = qkv_ref.narrow(1, 0, Hq * D).unflatten(1, (Hq, D)).contiguous()
+    kr = qkv_ref.narrow(1, Hq * D, Hk * D).unflatten(1, (Hk, D)).contiguous()
+    cos = table.cos[pos.long()]
+    sin = table.sin[pos.long()]
+    q2, k2 = ref.apply_rope(qr.cpu().float(), kr.cpu().float(),
+                            cos.cpu(), sin.cpu())
+    assert rel_err(q.cpu(), q2) < 2e-2
+    assert rel_err(k.cpu(), k2) < 2e-2
+
+    # kv append from the strided views
+    pages, psz = 8, 16
+    k_cache = torch.zeros(pages, psz, Hk, D, dtype=torch.bfloat16, device=DEV)
+    v_cache = torch.zeros_like(k_cache)
+    slots = torch.randperm(pages * psz, device=DEV)[:N].int()
+    ops.kv_cache_append(k_cache, v_cache, k, v, slots)
+    flat_k = k_cache.view(-1, Hk, D)
+    flat_v = v_cache.view(-1, Hk, D)
+    assert torch.equal(flat_k[slots.long()], k)
+    assert torch.equal(flat_v[slots.long()], v)
+
+
+def test_decode_attention_strided_q():
+    torch.manual_seed(33)
+    B, Hq, Hk, D = 5, 8, 2, 128
+    pages, psz = 16, 16
+    qkv = torch.randn(B, (Hq + 2 * Hk) * D, dtype=torch.bfloat16, device=DEV)
+    q = qkv.narrow(1, 0, Hq * D).unflatten(1, (Hq, D))
+    k_cache = torch.randn(pages, psz, Hk, D, dtype=torch.bfloat16, device=DEV)
+    v_cache = torch.randn_like(k_cache)
+    pt = torch.arange(pages, dtype=torch.int32, device=DEV).reshape(B, -1)
+    ctx = torch.tensor([7, 30, 48, 1, 16], dtype=torch.int32, device=DEV)
+    scale = 1.0 / math.sqrt(D)
+    out = ops.paged_attention_decode(q, k_cache, v_cache, pt, ctx, scale)
+    expect = ref.paged_attention_decode(
+        q.contiguous().cpu().float(), k_cache.cpu().float(),
+        v_cache.cpu().float(), pt.cpu(), ctx.cpu(), scale)
+    assert rel_err(out.cpu(), expect) < 2e-2
+
+
+def test_prefill_attention_strided_qkv():
+    torch.manual_seed(34)
+    Hq, Hk, D = 8, 2, 128
+    lens = [33, 64, 7]
+    total = sum(lens)
+    cu = torch.tensor([0] + list(torch.tensor(lens).cumsum(0)),
+                      dtype=torch.int32, device=DEV)
+    qkv = torch.randn(total, (Hq + 2 * Hk) * D, dtype=torch.bfloat16,
+                      device=DEV)
+    q = qkv.narrow(1, 0, Hq * D).unflatten(1, (Hq, D))
+    k = qkv.narrow(1, Hq * D, Hk * D).unflatten(1, (Hk, D))
+    v = qkv.narrow(1, (Hq + Hk) * D, Hk * D).unflatten(1, (Hk, D))
+    scale = 1.0 / math.sqrt(D)
+    out = ops.varlen_prefill_attention(q, k, v, cu, cu, scale, causal=True)
+    expect = ref.varlen_prefill_attention(
+        q.contiguous().cpu().float(), k.contiguous().cpu().float(),
+        v.contiguous().cpu().float(), cu.cpu(), cu.cpu(), scale, True)
+    assert rel_err(out.cpu(), expect) < 2e-2
+
+
+def test_sample_fast_path_statistics():
+    """Fused single-pass sampler: empirical distribution tracks softmax."""
+    torch.manual_seed(35)
+    N, V = 512, 4096
+    base = (torch.randn(V, device=DEV) * 3).bfloat16()
+    logits = base.expand(N, V).contiguous()
+    temp = torch.ones(N, device=DEV)
+    tk = torch.full((N,), -1, dtype=torch.int32, device=DEV)
+    tp = torch.ones(N, device=DEV)
+    counts = torch.zeros(V)
+    draws = 0
+    for seed in range(20):
+        tokens, lps = ops.sample(logits, temp, tk, tp, seed=seed * 977 + 1,
+                                 no_filter=True)
+        counts += torch.bincount(tokens.cpu(), minlength=V).float()
+        draws += N
+        # reported logprob is the raw softmax logprob of the sampled token
+        expect_lp = ref.gather_logprobs(logits.cpu(), tokens.cpu())
+        assert torch.allclose(lps.cpu(), expect_lp, atol=2e-2, rtol=1e-2)
+    probs = torch.softmax(base.float().cpu(), -1)
+    top = probs.topk(5).indices
+    emp = counts / draws
+    for t in top:
+        assert abs(emp[t] - probs[t]) < 0.05 + 0.3 * probs[t], \
+            (t, emp[t], probs[t])
