@@ -291,6 +291,7 @@ class Engine:
                  max_running_requests: int = 256,
                  max_num_batched_tokens: int = 8192,
                  max_model_len: Optional[int] = None,
+                 decode_chunk_size: int = 16,
                  seed: int = 0):
         self.cfg = cfg
         self.device = device
@@ -316,6 +317,7 @@ class Engine:
         self._abort_all = False
         self._scale = 1.0 / math.sqrt(cfg.head_dim)
         self._gen = torch.Generator().manual_seed(seed)  # CPU ref path RNG
+        self.decode_chunk_size = max(decode_chunk_size, 1)
 
     # ------------------------------------------------------------ public API
     def add_request(self, rid: str, input_ids: List[int],
@@ -541,23 +543,97 @@ class Engine:
             vs.append(flat_v[slots.long()])
         return torch.cat(ks), torch.cat(vs)
 
+    def _decode_chunk_len(self, reqs: List[Request]) -> int:
+        """How many decode iterations can run device-side without a host
+        sync: bounded by the chunk size, every request's remaining token
+        budget, and the model-len fence.  Stop-token requests decode 1 at a
+        time (token-exact stop)."""
+        if any(r.sampling.stop_token_ids for r in reqs):
+            return 1
+        rem = min(r.sampling.max_new_tokens - len(r.output_ids) for r in reqs)
+        rem = min(rem, min(self.max_model_len - r.seq_len for r in reqs))
+        return max(1, min(self.decode_chunk_size, rem))
+
     def _run_decode(self, reqs: List[Request]):
+        """C decode iterations with all state device-resident: one H2D
+        (initial tokens/positions/slots) and one D2H sync (the chunk's
+        sampled tokens + logprobs) per C tokens, instead of per token."""
         dev = self.device
+        C = self._decode_chunk_len(reqs)
+        B = len(reqs)
+        seq_ids = [self._seq_ids[r.rid] for r in reqs]
+        p0 = [r.seq_len - 1 for r in reqs]
+        if C > 1:
+            # slots for positions [p0, p0+C) were allocated by the caller
+            # for step 0 only; grow the allocation for the rest now
+            ok = []
+            for r, sid in zip(reqs, seq_ids):
+                if self.kv.allocate(sid, C - 1):
+                    ok.append(True)
+                else:
+                    ok.append(False)
+            if not all(ok):
+                # roll back to single-token decode for this round
+                C = 1
+
         tokens = torch.tensor([r.output_ids[-1] for r in reqs],
                               dtype=torch.long, device=dev)
-        pos = [r.seq_len - 1 for r in reqs]
-        positions = torch.tensor(pos, dtype=torch.int32, device=dev)
-        seq_ids = [self._seq_ids[r.rid] for r in reqs]
-        slots = torch.cat([self.kv.slots_for(sid, p, 1)
-                           for sid, p in zip(seq_ids, pos)]).to(dev)
+        pos0 = torch.tensor(p0, dtype=torch.int32, device=dev)
+        # (C, B) slot matrix: row s is the (contiguous) slot mapping of step s
+        slots_all = torch.stack(
+            [self.kv.slots_for(sid, p, C) for sid, p in zip(seq_ids, p0)],
+            dim=1).to(dev)
         page_table = self.kv.page_table(seq_ids).to(dev)
-        ctx = torch.tensor([p + 1 for p in pos], dtype=torch.int32, device=dev)
+        ctx = pos0 + 1
+
+        temp = torch.tensor([r.sampling.temperature for r in reqs],
+                            dtype=torch.float32, device=dev)
+        tk = torch.tensor([r.sampling.top_k for r in reqs],
+                          dtype=torch.int32, device=dev)
+        tp = torch.tensor([r.sampling.top_p for r in reqs],
+                          dtype=torch.float32, device=dev)
+        no_filter = all(
+            (r.sampling.top_k <= 0 or r.sampling.top_k >= self.cfg.vocab_size)
+            and r.sampling.top_p >= 1.0 for r in reqs)
+        out_tokens = torch.empty(C, B, dtype=torch.int64, device=dev)
+        out_lps = torch.empty(C, B, dtype=torch.float32, device=dev)
 
         def attn_fn(li, q, k, v):
             return ops.paged_attention_decode(q, self.kv.k_cache[li],
                                               self.kv.v_cache[li], page_table,
                                               ctx, self._scale)
 
-        hidden = self.model.forward_tokens(tokens, positions, self.kv, slots,
-                                           attn_fn)
-        self._sample_last(reqs, hidden)
+        for s in range(C):
+            positions = pos0 + s
+            hidden = self.model.forward_tokens(tokens, positions, self.kv,
+                                               slots_all[s], attn_fn)
+            logits = self.model.logits(hidden)
+            seed = (self._seed * 0x9E3779B9 + self._step_counter * 131 + s) \
+                & 0x7FFFFFFFFFFF
+            tokens, lps = ops.sample(logits, temp, tk, tp, seed,
+                                     generator=self._gen,
+                                     no_filter=no_filter)
+            out_tokens[s] = tokens
+            out_lps[s] = lps
+            if s + 1 < C:
+                ctx = ctx + 1
+
+        # single sync for the whole chunk
+        toks_h = out_tokens.t().cpu().tolist()
+        lps_h = out_lps.t().cpu().tolist()
+        for i, r in enumerate(reqs):
+            for s in range(C):
+                t = int(toks_h[i][s])
+                r.output_ids.append(t)
+                r.output_logprobs.append(float(lps_h[i][s]))
+                if t in r.sampling.stop_token_ids:
+                    r.finished = True
+                    r.finish_reason = "stop"
+                    break
+            if not r.finished:
+                if len(r.output_ids) >= r.sampling.max_new_tokens:
+                    r.finished = True
+                    r.finish_reason = "length"
+                elif r.seq_len >= self.max_model_len:
+                    r.finished = True
+                    r.finish_reason = "length"
