@@ -300,14 +300,14 @@ def test_rope_and_kv_append_strided_qkv():
 
 def test_decode_attention_strided_q():
     torch.manual_seed(33)
-    B, Hq, Hk, D = 5, 8, 2, 128
+    B, Hq, Hk, D = 4, 8, 2, 128
     pages, psz = 16, 16
     qkv = torch.randn(B, (Hq + 2 * Hk) * D, dtype=torch.bfloat16, device=DEV)
     q = qkv.narrow(1, 0, Hq * D).unflatten(1, (Hq, D))
     k_cache = torch.randn(pages, psz, Hk, D, dtype=torch.bfloat16, device=DEV)
     v_cache = torch.randn_like(k_cache)
     pt = torch.arange(pages, dtype=torch.int32, device=DEV).reshape(B, -1)
-    ctx = torch.tensor([7, 30, 48, 1, 16], dtype=torch.int32, device=DEV)
+    ctx = torch.tensor([7, 30, 48, 1], dtype=torch.int32, device=DEV)
     scale = 1.0 / math.sqrt(D)
     out = ops.paged_attention_decode(q, k_cache, v_cache, pt, ctx, scale)
     expect = ref.paged_attention_decode(
