@@ -31,7 +31,7 @@ def main():
     p.add_argument("--model", type=str, default="llama3-8b")
     p.add_argument("--batch-per-gpu", type=int, default=16,
                    help="prompts per GPU per step (weak scaling)")
-    p.add_argument("--n-samples", type=int, default=4,
+    p.add_argument("--n-samples", type=int, default=8,
                    help="rollout samples per prompt")
     p.add_argument("--prompt-len", type=int, default=256)
     p.add_argument("--response-len", type=int, default=256)

@@ -116,6 +116,7 @@ class RolloutConfig:
     max_num_batched_tokens: int = 8192
     max_running_requests: int = 256
     page_size: int = 16                  # KV tokens per page
+    decode_chunk_size: int = 16          # device-resident decode chunk
     sampling: SamplingConfig = field(default_factory=SamplingConfig)
     calculate_log_probs: bool = True
     min_stream_batch_size: int = 16

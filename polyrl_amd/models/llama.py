@@ -28,10 +28,12 @@ class RMSNorm(nn.Module):
         self.variance_epsilon = eps
 
     def forward(self, x):
+        # fused F.rms_norm (fp32 internal); weight applied in model dtype to
+        # match the HF/engine convention ((x_norm).to(dt) * w)
         dt = x.dtype
-        xf = x.float()
-        var = xf.pow(2).mean(-1, keepdim=True)
-        return (xf * torch.rsqrt(var + self.variance_epsilon)).to(dt) * self.weight
+        normed = F.rms_norm(x.float(), (x.shape[-1],), None,
+                            self.variance_epsilon)
+        return normed.to(dt) * self.weight
 
 
 class RotaryCache(nn.Module):
@@ -86,14 +88,13 @@ class Attention(nn.Module):
                        q[..., d:] * cos + q[..., :d] * sin], dim=-1)
         k = torch.cat([k[..., :d] * cos - k[..., d:] * sin,
                        k[..., d:] * cos + k[..., :d] * sin], dim=-1)
-        if self.num_kv_heads != self.num_heads:
-            rep = self.num_heads // self.num_kv_heads
-            k = k.repeat_interleave(rep, dim=1)
-            v = v.repeat_interleave(rep, dim=1)
+        gqa = self.num_kv_heads != self.num_heads
         if attn_bias_mask is not None:
-            o = F.scaled_dot_product_attention(q, k, v, attn_mask=attn_bias_mask)
+            o = F.scaled_dot_product_attention(q, k, v, attn_mask=attn_bias_mask,
+                                               enable_gqa=gqa)
         else:
-            o = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+            o = F.scaled_dot_product_attention(q, k, v, is_causal=True,
+                                               enable_gqa=gqa)
         o = o.transpose(1, 2).reshape(B, L, -1)
         return self.o_proj(o)
 

@@ -110,6 +110,7 @@ class StreamPPOTrainer:
                              max_running_requests=ro.max_running_requests,
                              max_num_batched_tokens=ro.max_num_batched_tokens,
                              max_model_len=ro.prompt_length + ro.response_length,
+                             decode_chunk_size=ro.decode_chunk_size,
                              seed=config.trainer.seed * 1000 + self.rank)
         self.coordinator = LocalRolloutCoordinator(
             self.engine, ro.response_length, pad_token_id=0, device="cpu")
