@@ -88,6 +88,7 @@ class CriticConfig:
     use_dynamic_bsz: bool = True
     cliprange_value: float = 0.5
     loss_agg_mode: str = "token-mean"
+    ulysses_sequence_parallel_size: int = 1
     model: ModelConfig = field(default_factory=ModelConfig)
     optim: OptimConfig = field(default_factory=lambda: OptimConfig(lr=1e-5))
     fsdp: FSDPConfig = field(default_factory=FSDPConfig)

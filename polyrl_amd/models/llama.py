@@ -75,12 +75,28 @@ class Attention(nn.Module):
         self.v_proj = nn.Linear(h, self.num_kv_heads * self.head_dim, bias=bias)
         self.o_proj = nn.Linear(self.num_heads * self.head_dim, h, bias=False)
 
-    def forward(self, x, cos, sin, attn_bias_mask: Optional[torch.Tensor]):
-        """cos/sin: (B, 1, L, D/2) fp32 (per-row positions supported)."""
-        B, L, _ = x.shape
-        q = self.q_proj(x).view(B, L, self.num_heads, self.head_dim).transpose(1, 2)
-        k = self.k_proj(x).view(B, L, self.num_kv_heads, self.head_dim).transpose(1, 2)
-        v = self.v_proj(x).view(B, L, self.num_kv_heads, self.head_dim).transpose(1, 2)
+    def forward(self, x, cos, sin, attn_bias_mask: Optional[torch.Tensor],
+                sp_group=None):
+        """cos/sin: (B, 1, L, D/2) fp32 over the FULL sequence.  With
+        ``sp_group`` (Ulysses SP) x is the rank's 1/sp sequence shard; the
+        all-to-all trades it for a head shard over the full sequence
+        (parallel/ulysses.py)."""
+        B, Ls, _ = x.shape
+        q = self.q_proj(x).view(B, Ls, self.num_heads, self.head_dim)
+        k = self.k_proj(x).view(B, Ls, self.num_kv_heads, self.head_dim)
+        v = self.v_proj(x).view(B, Ls, self.num_kv_heads, self.head_dim)
+        if sp_group is not None:
+            from ..parallel.ulysses import all_to_all_4d
+            import torch.distributed as _dist
+            sp = _dist.get_world_size(sp_group)
+            assert self.num_heads % sp == 0 and self.num_kv_heads % sp == 0, \
+                f"heads ({self.num_heads}/{self.num_kv_heads}) % sp {sp} != 0"
+            q = all_to_all_4d(q, 2, 1, sp_group)   # (B, L, Hq/sp, D)
+            k = all_to_all_4d(k, 2, 1, sp_group)
+            v = all_to_all_4d(v, 2, 1, sp_group)
+        q = q.transpose(1, 2)
+        k = k.transpose(1, 2)
+        v = v.transpose(1, 2)
         cos = cos.to(q.dtype)
         sin = sin.to(q.dtype)
         d = self.head_dim // 2
@@ -88,14 +104,18 @@ class Attention(nn.Module):
                        q[..., d:] * cos + q[..., :d] * sin], dim=-1)
         k = torch.cat([k[..., :d] * cos - k[..., d:] * sin,
                        k[..., d:] * cos + k[..., :d] * sin], dim=-1)
-        gqa = self.num_kv_heads != self.num_heads
+        gqa = k.shape[1] != q.shape[1]
         if attn_bias_mask is not None:
             o = F.scaled_dot_product_attention(q, k, v, attn_mask=attn_bias_mask,
                                                enable_gqa=gqa)
         else:
             o = F.scaled_dot_product_attention(q, k, v, is_causal=True,
                                                enable_gqa=gqa)
-        o = o.transpose(1, 2).reshape(B, L, -1)
+        o = o.transpose(1, 2)                      # (B, L[, Hq/sp], D)
+        if sp_group is not None:
+            from ..parallel.ulysses import all_to_all_4d
+            o = all_to_all_4d(o, 1, 2, sp_group)   # back to (B, Ls, Hq, D)
+        o = o.reshape(B, Ls, -1)
         return self.o_proj(o)
 
 
@@ -119,8 +139,9 @@ class DecoderLayer(nn.Module):
         self.input_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
         self.post_attention_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
 
-    def forward(self, x, cos, sin, attn_bias_mask):
-        x = x + self.self_attn(self.input_layernorm(x), cos, sin, attn_bias_mask)
+    def forward(self, x, cos, sin, attn_bias_mask, sp_group=None):
+        x = x + self.self_attn(self.input_layernorm(x), cos, sin,
+                               attn_bias_mask, sp_group)
         x = x + self.mlp(self.post_attention_layernorm(x))
         return x
 
@@ -135,8 +156,12 @@ class DecoderModel(nn.Module):
         self.norm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
         self.rotary = RotaryCache(cfg.head_dim, cfg.rope_theta)
         self.gradient_checkpointing = False
+        self.ulysses = None        # UlyssesContext set by the worker (SP>1)
 
     def forward(self, input_ids, attention_mask=None, position_ids=None):
+        """With Ulysses SP enabled, every SP rank passes the SAME full
+        (B, L) inputs (L divisible by sp); the returned hidden states are
+        this rank's (B, L/sp, H) sequence shard."""
         B, L = input_ids.shape
         if position_ids is None:
             position_ids = torch.arange(L, device=input_ids.device).expand(B, L)
@@ -153,15 +178,24 @@ class DecoderModel(nn.Module):
             pad = attention_mask.bool().view(B, 1, 1, L)
             attn_bias_mask = (causal.view(1, 1, L, L) & pad)
 
+        sp_group = None
+        if self.ulysses is not None and self.ulysses.enabled:
+            from ..parallel.ulysses import slice_for_rank
+            sp_group = self.ulysses.group
+            assert L % self.ulysses.size == 0, \
+                f"seq len {L} % sp {self.ulysses.size} != 0 (pad upstream)"
+            input_ids = slice_for_rank(input_ids, 1, sp_group)
+
         x = self.embed_tokens(input_ids)
         for layer in self.layers:
             # NOTE: layers must be invoked via __call__ so FSDP2's
             # unshard/reshard pre/post-forward hooks fire.
             if self.gradient_checkpointing and self.training:
                 x = torch.utils.checkpoint.checkpoint(
-                    layer, x, cos, sin, attn_bias_mask, use_reentrant=False)
+                    layer, x, cos, sin, attn_bias_mask, sp_group,
+                    use_reentrant=False)
             else:
-                x = layer(x, cos, sin, attn_bias_mask)
+                x = layer(x, cos, sin, attn_bias_mask, sp_group)
         return self.norm(x)
 
 
@@ -192,9 +226,11 @@ class CausalLM(nn.Module):
                 logits_slice: Optional[slice] = None):
         """logits_slice: sequence-dim slice applied to the hidden states
         BEFORE lm_head — avoids materializing vocab logits over prompt
-        positions when only response logprobs are needed."""
+        positions when only response logprobs are needed.  Ignored under
+        Ulysses SP (hidden is a sequence shard; the worker gathers)."""
         hidden = self.model(input_ids, attention_mask, position_ids)
-        if logits_slice is not None:
+        sp_on = self.model.ulysses is not None and self.model.ulysses.enabled
+        if logits_slice is not None and not sp_on:
             hidden = hidden[:, logits_slice]
         return self.lm_head(hidden)
 

@@ -110,6 +110,48 @@ def _model_inputs(batch: TensorBatch, device):
     return ids, am, pos, resp
 
 
+_SP_GROUPS: Dict[int, tuple] = {}   # sp_size -> (sp_group, dp_group)
+
+
+def _get_sp_groups(sp_size: int):
+    """Build (once) and cache the SP/DP process groups for this rank.
+    Safe to call from actor+critic+ref constructors: the group-creation
+    order is identical on every rank."""
+    if sp_size not in _SP_GROUPS:
+        from ..parallel.ulysses import build_sp_groups
+        _SP_GROUPS[sp_size] = build_sp_groups(sp_size)
+    return _SP_GROUPS[sp_size]
+
+
+def _setup_ulysses(model: nn.Module, sp_size: int):
+    """Install the Ulysses context on a llama-family trunk; returns the SP
+    group (or None)."""
+    if sp_size <= 1 or not (dist.is_available() and dist.is_initialized()):
+        return None
+    trunk = getattr(model, "model", None)
+    if trunk is None or not hasattr(trunk, "ulysses"):
+        raise ValueError("ulysses SP requires the llama-family trainer model")
+    from ..parallel.ulysses import UlyssesContext
+    sp_group, _ = _get_sp_groups(sp_size)
+    trunk.ulysses = UlyssesContext(sp_group)
+    return sp_group
+
+
+def _gather_rows(batch: TensorBatch, group):
+    """Pool batch rows across the SP group so every rank processes the SAME
+    rows (the reference slices sequences of a shared batch the same way —
+    verl FSDPUlyssesShardingManager preprocess capability).  Returns
+    (pooled_batch, slice_of_this_ranks_rows)."""
+    sp = dist.get_world_size(group)
+    r = dist.get_rank(group)
+    if sp == 1:
+        return batch, slice(0, len(batch))
+    boxes = [None] * sp
+    dist.all_gather_object(boxes, batch, group=group)
+    off = sum(len(b) for b in boxes[:r])
+    return TensorBatch.concat(boxes), slice(off, off + len(batch))
+
+
 class ActorWorker:
     """Policy model under FSDP2: compute_log_prob + update_policy_stream."""
 
@@ -119,6 +161,10 @@ class ActorWorker:
         self.device = device
         self.is_ref = is_ref
         self.model = _maybe_fully_shard(model)
+        self.sp_group = _setup_ulysses(model,
+                                       cfg.ulysses_sequence_parallel_size)
+        self.sp_size = cfg.ulysses_sequence_parallel_size \
+            if self.sp_group is not None else 1
         if not is_ref:
             self.optimizer = _build_optimizer(self.model.parameters(), cfg.optim)
             self.lr_scheduler = _build_lr_scheduler(self.optimizer, cfg.optim)
@@ -130,6 +176,9 @@ class ActorWorker:
                          ) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
         """old/ref log-probs of the sampled responses.  (B, resp_len) fp32."""
         self.model.eval()
+        my_rows = slice(0, len(batch))
+        if self.sp_group is not None:
+            batch, my_rows = _gather_rows(batch, self.sp_group)
         micro, parts = self._split(batch)
         k = _sync_num_micro(len(micro))
         lps, ents = [], []
@@ -139,8 +188,9 @@ class ActorWorker:
             ents.append(ent)
         for _ in range(k - len(micro)):  # dummy collective-alignment passes
             self._forward_logprobs(micro[0].slice(slice(0, 1)), want_entropy)
-        lp = restore_dynamic_batch(torch.cat(lps), parts)
-        ent = restore_dynamic_batch(torch.cat(ents), parts) if want_entropy else None
+        lp = restore_dynamic_batch(torch.cat(lps), parts)[my_rows]
+        ent = (restore_dynamic_batch(torch.cat(ents), parts)[my_rows]
+               if want_entropy else None)
         return lp, ent
 
     def _split(self, batch: TensorBatch):
@@ -157,12 +207,42 @@ class ActorWorker:
         Lr = resp.shape[1]
         ctx = torch.enable_grad() if grad else torch.no_grad()
         with ctx:
+            if self.sp_group is not None:
+                return self._forward_logprobs_sp(ids, am, pos, Lr,
+                                                 want_entropy)
             logits = self.model(ids, attention_mask=am, position_ids=pos,
                                 logits_slice=slice(-Lr - 1, -1))
             logits = logits.float()
             lp = algos.logprobs_from_logits(logits, resp)
             ent = algos.entropy_from_logits(logits) if want_entropy else \
                 torch.zeros_like(lp)
+        return lp, ent
+
+    def _forward_logprobs_sp(self, ids, am, pos, Lr, want_entropy):
+        """Ulysses path: full inputs on every SP rank; per-shard logits ->
+        per-shard logprobs -> autograd-aware sequence gather -> response
+        slice (verl's gather_outputs_and_unpad capability)."""
+        from ..parallel.ulysses import gather_seq, pad_to_multiple, slice_for_rank
+        sp = self.sp_size
+        B, L = ids.shape
+        ids_p = pad_to_multiple(ids, sp, 1)
+        am_p = pad_to_multiple(am, sp, 1)
+        pos_p = pad_to_multiple(pos, sp, 1)
+        logits = self.model(ids_p, attention_mask=am_p,
+                            position_ids=pos_p).float()   # (B, Lp/sp, V)
+        labels_p = torch.cat(
+            [ids_p[:, 1:], torch.zeros(B, 1, dtype=ids.dtype,
+                                       device=ids.device)], dim=1)
+        labels_shard = slice_for_rank(labels_p, 1, self.sp_group)
+        lp_shard = algos.logprobs_from_logits(logits, labels_shard)
+        lp = gather_seq(lp_shard, 1, self.sp_group)[:, :L]
+        lp = lp[:, L - Lr - 1:L - 1]
+        if want_entropy:
+            ent_shard = algos.entropy_from_logits(logits)
+            ent = gather_seq(ent_shard, 1, self.sp_group)[:, :L]
+            ent = ent[:, L - Lr - 1:L - 1]
+        else:
+            ent = torch.zeros_like(lp)
         return lp, ent
 
     # ------------------------------------------------------------- update
@@ -177,6 +257,8 @@ class ActorWorker:
         assert not self.is_ref
         self.model.train()
         metrics: Dict[str, List[float]] = {}
+        if self.sp_group is not None:
+            batch, _ = _gather_rows(batch, self.sp_group)
         micro, _ = self._split(batch)
         k = _sync_num_micro(len(micro))
         # dummy zero-weight passes keep FSDP fwd/bwd collectives aligned
@@ -185,14 +267,11 @@ class ActorWorker:
         loss_fn = algos.get_policy_loss_fn(self.cfg.policy_loss_type)
         for mi, mb in enumerate(micro + dummies):
             is_dummy = mi >= len(micro)
-            ids, am, pos, resp = _model_inputs(mb, self.device)
-            Lr = resp.shape[1]
             response_mask = mb["response_mask"].to(self.device)
             old_log_prob = mb["old_log_probs"].to(self.device)
             advantages = mb["advantages"].to(self.device)
-            logits = self.model(ids, attention_mask=am, position_ids=pos,
-                                logits_slice=slice(-Lr - 1, -1)).float()
-            log_prob = algos.logprobs_from_logits(logits, resp)
+            log_prob, ent = self._forward_logprobs(
+                mb, want_entropy=bool(self.cfg.entropy_coeff), grad=True)
             pg_loss, pg_clipfrac, ppo_kl, pg_clipfrac_lower = loss_fn(
                 old_log_prob=old_log_prob, log_prob=log_prob,
                 advantages=advantages, response_mask=response_mask,
@@ -202,7 +281,6 @@ class ActorWorker:
                 loss_agg_mode=self.cfg.loss_agg_mode)
             loss = pg_loss
             if self.cfg.entropy_coeff:
-                ent = algos.entropy_from_logits(logits)
                 loss = loss - self.cfg.entropy_coeff * algos.agg_loss(
                     ent, response_mask, self.cfg.loss_agg_mode)
             if self.cfg.use_kl_loss and "ref_log_probs" in mb.tensors:
@@ -212,8 +290,10 @@ class ActorWorker:
                 loss = loss + self.cfg.kl_loss_coef * kl_loss
                 if not is_dummy:
                     metrics.setdefault("actor/kl_loss", []).append(kl_loss.item())
-            # micro-batch weight within slice x slice weight within minibatch
-            w = 0.0 if is_dummy else (len(mb) / n_total) * accum_scale
+            # micro-batch weight within slice x slice weight within minibatch;
+            # x sp_size compensates FSDP's world-mean over SP-duplicated rows
+            w = 0.0 if is_dummy else \
+                (len(mb) / n_total) * accum_scale * self.sp_size
             (loss * w).backward()
             if not is_dummy:
                 metrics.setdefault("actor/pg_loss", []).append(pg_loss.item())
@@ -245,8 +325,28 @@ class CriticWorker:
         self.cfg = cfg
         self.device = device
         self.model = _maybe_fully_shard(model)
+        self.sp_group = _setup_ulysses(
+            model, getattr(cfg, "ulysses_sequence_parallel_size", 1))
+        self.sp_size = cfg.ulysses_sequence_parallel_size \
+            if self.sp_group is not None else 1
         self.optimizer = _build_optimizer(self.model.parameters(), cfg.optim)
         self.lr_scheduler = _build_lr_scheduler(self.optimizer, cfg.optim)
+
+    def _forward_values(self, mb: TensorBatch) -> torch.Tensor:
+        """(B, Lr) value predictions; SP-aware (shard -> gather -> slice)."""
+        ids, am, pos, resp = _model_inputs(mb, self.device)
+        Lr = resp.shape[1]
+        if self.sp_group is None:
+            values = self.model(ids, attention_mask=am, position_ids=pos)
+            return values[:, -Lr - 1:-1].float()
+        from ..parallel.ulysses import gather_seq, pad_to_multiple
+        sp = self.sp_size
+        L = ids.shape[1]
+        v_shard = self.model(pad_to_multiple(ids, sp, 1),
+                             attention_mask=pad_to_multiple(am, sp, 1),
+                             position_ids=pad_to_multiple(pos, sp, 1))
+        v = gather_seq(v_shard.float(), 1, self.sp_group)[:, :L]
+        return v[:, L - Lr - 1:L - 1]
 
     def _split(self, batch: TensorBatch):
         if self.cfg.use_dynamic_bsz:
@@ -257,41 +357,42 @@ class CriticWorker:
     @torch.no_grad()
     def compute_values(self, batch: TensorBatch) -> torch.Tensor:
         self.model.eval()
+        my_rows = slice(0, len(batch))
+        if self.sp_group is not None:
+            batch, my_rows = _gather_rows(batch, self.sp_group)
         micro, parts = self._split(batch)
         k = _sync_num_micro(len(micro))
         outs = []
         for mi in range(k):
             mb = micro[mi] if mi < len(micro) else micro[0].slice(slice(0, 1))
-            ids, am, pos, resp = _model_inputs(mb, self.device)
-            Lr = resp.shape[1]
-            values = self.model(ids, attention_mask=am, position_ids=pos)
+            values = self._forward_values(mb)
             if mi < len(micro):
-                outs.append(values[:, -Lr - 1:-1].float())
-        return restore_dynamic_batch(torch.cat(outs), parts)
+                outs.append(values)
+        return restore_dynamic_batch(torch.cat(outs), parts)[my_rows]
 
     def update_critic_stream(self, batch: TensorBatch, is_opt_step: bool,
                              is_lr_step: bool, accum_scale: float
                              ) -> Dict[str, List[float]]:
         self.model.train()
         metrics: Dict[str, List[float]] = {}
+        if self.sp_group is not None:
+            batch, _ = _gather_rows(batch, self.sp_group)
         micro, _ = self._split(batch)
         k = _sync_num_micro(len(micro))
         dummies = [micro[0].slice(slice(0, 1)) for _ in range(k - len(micro))]
         n_total = len(batch)
         for mi, mb in enumerate(micro + dummies):
             is_dummy = mi >= len(micro)
-            ids, am, pos, resp = _model_inputs(mb, self.device)
-            Lr = resp.shape[1]
             response_mask = mb["response_mask"].to(self.device)
             values = mb["values"].to(self.device)
             returns = mb["returns"].to(self.device)
-            vpreds = self.model(ids, attention_mask=am, position_ids=pos)
-            vpreds = vpreds[:, -Lr - 1:-1].float()
+            vpreds = self._forward_values(mb)
             vf_loss, vf_clipfrac = algos.compute_value_loss(
                 vpreds, returns, values, response_mask,
                 cliprange_value=self.cfg.cliprange_value,
                 loss_agg_mode=self.cfg.loss_agg_mode)
-            w = 0.0 if is_dummy else (len(mb) / n_total) * accum_scale
+            w = 0.0 if is_dummy else \
+                (len(mb) / n_total) * accum_scale * self.sp_size
             (vf_loss * w).backward()
             if not is_dummy:
                 metrics.setdefault("critic/vf_loss", []).append(vf_loss.item())
