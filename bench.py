@@ -61,7 +61,8 @@ def main():
     cfg = PPOConfig()
     cfg.actor_rollout_ref.model.path = args.model
     cfg.actor_rollout_ref.model.dtype = "bfloat16" if device == "cuda" else "float32"
-    cfg.actor_rollout_ref.model.enable_gradient_checkpointing = True
+    cfg.actor_rollout_ref.model.enable_gradient_checkpointing = False
+    cfg.critic.model.enable_gradient_checkpointing = False
     cfg.actor_rollout_ref.actor.ppo_mini_batch_size = total_samples // 2
     cfg.actor_rollout_ref.actor.ppo_max_token_len_per_gpu = 16384
     cfg.actor_rollout_ref.rollout.sampling.n = args.n_samples

@@ -144,8 +144,16 @@ class StreamPPOTrainer:
     # ------------------------------------------------------------------ setup
     def _kv_budget(self, ro) -> int:
         if self.device.startswith("cuda"):
+            from ..rollout.kv_cache import PagedKVCache
+            mc = get_model_config(self.config.actor_rollout_ref.model.path)
+            bt = PagedKVCache.bytes_per_token(
+                mc.num_hidden_layers, mc.num_key_value_heads, mc.head_dim)
+            # what max_running concurrent sequences at full length need
+            need = bt * ro.max_running_requests * \
+                (ro.prompt_length + ro.response_length)
             free, total = torch.cuda.mem_get_info()
-            return int(free * ro.gpu_memory_utilization * 0.5)
+            return int(min(free * ro.gpu_memory_utilization * 0.5,
+                           need * 1.125))
         return 64 << 20
 
     def _validate_config(self):
