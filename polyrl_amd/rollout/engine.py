@@ -61,19 +61,25 @@ class RequestOutput:
 
 
 class LayerWeights:
-    """Fused per-layer weight buffers (views kept for named updates)."""
+    """Fused per-layer weight buffers (views kept for named updates).
+    With tp > 1 the buffers hold this rank's TP shard (rows of qkv/gate/up,
+    cols of o/down — parallel/tp.py docstring)."""
 
-    def __init__(self, cfg: DecoderConfig, device, dtype):
+    def __init__(self, cfg: DecoderConfig, device, dtype, tp: int = 1):
         h = cfg.hidden_size
-        hq = cfg.num_attention_heads * cfg.head_dim
-        hk = cfg.num_key_value_heads * cfg.head_dim
-        i = cfg.intermediate_size
+        assert cfg.num_attention_heads % tp == 0 and \
+            cfg.num_key_value_heads % tp == 0, \
+            f"heads must divide tp={tp}"
+        hq = cfg.num_attention_heads * cfg.head_dim // tp
+        hk = cfg.num_key_value_heads * cfg.head_dim // tp
+        i = cfg.intermediate_size // tp
         self.wqkv = torch.empty(hq + 2 * hk, h, device=device, dtype=dtype)
         self.bqkv = (torch.zeros(hq + 2 * hk, device=device, dtype=dtype)
                      if cfg.attention_bias else None)
         self.wo = torch.empty(h, hq, device=device, dtype=dtype)
         self.w_gate_up = torch.empty(2 * i, h, device=device, dtype=dtype)
         self.w_down = torch.empty(h, i, device=device, dtype=dtype)
+        self.tp = tp
         self.input_ln = torch.empty(h, device=device, dtype=dtype)
         self.post_ln = torch.empty(h, device=device, dtype=dtype)
         self._hq, self._hk, self._i = hq, hk, i
@@ -123,24 +129,34 @@ class Gpt2LayerWeights:
 
 
 class InferenceModel:
-    """No-autograd decoder forward over paged KV on the HIP kernel suite."""
+    """No-autograd decoder forward over paged KV on the HIP kernel suite.
+
+    ``tp_ctx`` (parallel/tp.py) enables the tensor-parallel mode: buffers
+    hold this rank's shard, o/down projections all-reduce, lm_head is
+    vocab-parallel with an all-gather before sampling."""
 
     def __init__(self, cfg: DecoderConfig, device="cuda",
-                 dtype=torch.bfloat16):
+                 dtype=torch.bfloat16, tp_ctx=None):
+        from ..parallel.tp import TPContext
+        self.tp = tp_ctx if tp_ctx is not None else TPContext(None)
+        tp = self.tp.size
         if cfg.arch == "gpt2":
+            assert tp == 1, "gpt2 engine is the CPU plumbing tier (tp=1)"
             self._init_gpt2(cfg, device, dtype)
             return
         assert cfg.arch in ("llama", "qwen2"), \
             "rollout engine serves the llama/qwen2/gpt2 families"
+        assert cfg.vocab_size % tp == 0, "vocab must divide tp"
         self.cfg = cfg
         self.device = device
         self.dtype = dtype
-        self.layers = [LayerWeights(cfg, device, dtype)
+        self.layers = [LayerWeights(cfg, device, dtype, tp)
                        for _ in range(cfg.num_hidden_layers)]
         h = cfg.hidden_size
         self.embed = torch.empty(cfg.vocab_size, h, device=device, dtype=dtype)
         self.final_norm = torch.empty(h, device=device, dtype=dtype)
-        self.lm_head = torch.empty(cfg.vocab_size, h, device=device, dtype=dtype)
+        self.lm_head = torch.empty(cfg.vocab_size // tp, h, device=device,
+                                   dtype=dtype)
         self.rope = ops.RopeTable(cfg.head_dim, cfg.max_position_embeddings,
                                   cfg.rope_theta, device)
         # name -> engine buffer view, for weight updates
@@ -151,6 +167,18 @@ class InferenceModel:
         }
         for li, lw in enumerate(self.layers):
             self._name_map.update(lw.named_slices(f"model.layers.{li}"))
+
+    def _shard_incoming(self, name: str, src: torch.Tensor,
+                        buf: torch.Tensor) -> torch.Tensor:
+        """Slice a FULL tensor down to this rank's shard when the buffer is
+        TP-sharded (receiver-side resharding, patches.py:196-241 contract)."""
+        if self.tp.size == 1 or src.shape == buf.shape:
+            return src
+        from ..transfer.collective import tp_slice
+        if name == "lm_head.weight":
+            step = buf.shape[0]
+            return src[self.tp.rank * step:(self.tp.rank + 1) * step]
+        return tp_slice(name, src, self.tp.rank, self.tp.size, self.cfg.arch)
 
     def _init_gpt2(self, cfg: DecoderConfig, device, dtype):
         self.cfg = cfg
@@ -190,6 +218,7 @@ class InferenceModel:
                 if strict:
                     raise KeyError(f"missing weight {name}")
                 continue
+            src = self._shard_incoming(name, src, buf)
             buf.copy_(src.to(device=buf.device, dtype=buf.dtype,
                              non_blocking=True))
             seen.add(name)
@@ -202,6 +231,7 @@ class InferenceModel:
         buf = self._name_map.get(name)
         if buf is None:
             return False
+        tensor = self._shard_incoming(name, tensor, buf)
         buf.copy_(tensor.to(device=buf.device, dtype=buf.dtype,
                             non_blocking=True))
         return True
@@ -217,7 +247,10 @@ class InferenceModel:
             return self._forward_tokens_gpt2(token_ids, positions, kv,
                                              slot_mapping, attn_fn)
         cfg = self.cfg
-        Hq, Hk, D = cfg.num_attention_heads, cfg.num_key_value_heads, cfg.head_dim
+        tp = self.tp.size
+        Hq = cfg.num_attention_heads // tp
+        Hk = cfg.num_key_value_heads // tp
+        D = cfg.head_dim
         x = self.embed[token_ids]                     # (N, H) gather
         residual = x.clone()
         hidden = None
@@ -240,12 +273,14 @@ class InferenceModel:
                                 slot_mapping)
             attn_out = attn_fn(li, q, k, v)           # (N, Hq, D)
             hidden = attn_out.view(-1, Hq * D) @ lw.wo.t()
+            self.tp.all_reduce_(hidden)               # col-parallel o_proj
             hidden, residual = ops.fused_add_rmsnorm(
                 hidden, residual, lw.post_ln, cfg.rms_norm_eps)
             gate_up = hidden @ lw.w_gate_up.t()
-            i_sz = cfg.intermediate_size
+            i_sz = cfg.intermediate_size // tp
             hidden = ops.silu_mul(gate_up.narrow(1, 0, i_sz),
                                   gate_up.narrow(1, i_sz, i_sz)) @ lw.w_down.t()
+            self.tp.all_reduce_(hidden)               # col-parallel down_proj
         # final residual add + norm (fused; residual buffer is dead after)
         normed, _ = ops.fused_add_rmsnorm(hidden, residual, self.final_norm,
                                           cfg.rms_norm_eps)
@@ -279,7 +314,8 @@ class InferenceModel:
 
     @torch.no_grad()
     def logits(self, hidden: torch.Tensor) -> torch.Tensor:
-        return hidden @ self.lm_head.t()
+        local = hidden @ self.lm_head.t()
+        return self.tp.all_gather_cat(local, dim=-1)
 
 
 class Engine:
@@ -292,20 +328,25 @@ class Engine:
                  max_num_batched_tokens: int = 8192,
                  max_model_len: Optional[int] = None,
                  decode_chunk_size: int = 16,
+                 tp_ctx=None,
                  seed: int = 0):
         self.cfg = cfg
         self.device = device
         self.dtype = dtype
-        self.model = InferenceModel(cfg, device, dtype)
+        self.model = InferenceModel(cfg, device, dtype, tp_ctx=tp_ctx)
+        self.tp = self.model.tp
         self.max_running = max_running_requests
         self.max_batched_tokens = max_num_batched_tokens
         self.max_model_len = max_model_len or cfg.max_position_embeddings
+        # KV cache holds this rank's local KV heads (gpt2 has no GQA/TP)
+        hk_local = (cfg.num_attention_heads if cfg.arch == "gpt2"
+                    else cfg.num_key_value_heads // self.tp.size)
         bt = PagedKVCache.bytes_per_token(cfg.num_hidden_layers,
-                                          cfg.num_key_value_heads, cfg.head_dim)
+                                          hk_local, cfg.head_dim)
         if kv_bytes_budget is None:
             kv_bytes_budget = 1 << 30  # 1 GiB default (tests); callers size it
         num_pages = max(int(kv_bytes_budget // (bt * page_size)), 8)
-        self.kv = PagedKVCache(cfg.num_hidden_layers, cfg.num_key_value_heads,
+        self.kv = PagedKVCache(cfg.num_hidden_layers, hk_local,
                                cfg.head_dim, num_pages, page_size,
                                dtype=dtype, device=device)
         self.waiting: List[Request] = []
