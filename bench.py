@@ -64,7 +64,7 @@ def main():
     cfg.actor_rollout_ref.model.enable_gradient_checkpointing = False
     cfg.critic.model.enable_gradient_checkpointing = False
     cfg.actor_rollout_ref.actor.ppo_mini_batch_size = total_samples // 2
-    cfg.actor_rollout_ref.actor.ppo_max_token_len_per_gpu = 16384
+    cfg.actor_rollout_ref.actor.ppo_max_token_len_per_gpu = 8192
     cfg.actor_rollout_ref.rollout.sampling.n = args.n_samples
     cfg.actor_rollout_ref.rollout.sampling.temperature = 1.0
     cfg.actor_rollout_ref.rollout.prompt_length = args.prompt_len
@@ -76,7 +76,7 @@ def main():
     cfg.critic.model.path = args.model
     cfg.critic.model.dtype = cfg.actor_rollout_ref.model.dtype
     cfg.critic.ppo_mini_batch_size = total_samples // 2
-    cfg.critic.ppo_max_token_len_per_gpu = 16384
+    cfg.critic.ppo_max_token_len_per_gpu = 8192
     cfg.data.train_batch_size = global_batch
     cfg.data.max_prompt_length = args.prompt_len
     cfg.data.synthetic_num_prompts = max(
