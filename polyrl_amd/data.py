@@ -123,6 +123,15 @@ class ParquetRLHFDataset:
             extra["data_source"] = np.array(
                 [self.df.iloc[i][self.data_source_key] for i in indices],
                 dtype=object)
+        # ground truth for the rule-based scorers (reward_score/)
+        for col in ("ground_truth", "answer", "reward_model"):
+            if col in self.df.columns:
+                vals = [self.df.iloc[i][col] for i in indices]
+                if col == "reward_model":  # verl style: dict with ground_truth
+                    vals = [v.get("ground_truth") if isinstance(v, dict)
+                            else v for v in vals]
+                extra["ground_truth"] = np.array(vals, dtype=object)
+                break
         return TensorBatch.from_dict(
             tensors={
                 "input_ids": torch.stack([r["input_ids"] for r in rows]),

@@ -89,10 +89,44 @@ class FunctionReward:
         return _place_scores(batch, scores)
 
 
+class NaiveRewardManager:
+    """Detokenize each response and score it against its ground truth with
+    the data_source-dispatched rule scorer (reference: verl naive reward
+    manager + default_compute_score, SURVEY.md §2.1 'Reward loader/scores').
+
+    Needs ``tokenizer`` (anything with .decode(list[int]) -> str) and the
+    batch's non_tensors to carry 'data_source' and 'ground_truth'."""
+
+    def __init__(self, tokenizer=None, compute_score=None, **_):
+        from .reward_score import default_compute_score
+        self.tokenizer = tokenizer
+        self.compute_score = compute_score or default_compute_score
+
+    def __call__(self, batch: TensorBatch) -> torch.Tensor:
+        assert self.tokenizer is not None, "naive reward needs a tokenizer"
+        resp = batch["responses"]
+        mask = batch["response_mask"]
+        ds = batch.non_tensors.get("data_source")
+        gt = batch.non_tensors.get("ground_truth")
+        assert ds is not None and gt is not None, \
+            "naive reward needs data_source + ground_truth in non_tensors"
+        scores = torch.zeros(len(batch))
+        for i in range(len(batch)):
+            ids = resp[i][mask[i].bool()].tolist()
+            text = self.tokenizer.decode(ids)
+            try:
+                scores[i] = float(self.compute_score(str(ds[i]), text,
+                                                     str(gt[i])))
+            except (KeyError, NotImplementedError):
+                scores[i] = 0.0
+        return _place_scores(batch, scores)
+
+
 _REGISTRY: Dict[str, type] = {
     "constant": ConstantReward,
     "random": RandomReward,
     "length": LengthReward,
+    "naive": NaiveRewardManager,
 }
 
 
