@@ -75,6 +75,28 @@ class Attention(nn.Module):
         self.v_proj = nn.Linear(h, self.num_kv_heads * self.head_dim, bias=bias)
         self.o_proj = nn.Linear(self.num_heads * self.head_dim, h, bias=False)
 
+    def forward_packed(self, x, cos, sin, cu_seqlens):
+        """Packed varlen path (use_remove_padding): x (total, H); cos/sin
+        (total, D/2) per token; causal flash attention over the block-diag
+        layout via the hand-written MFMA fwd/bwd kernels (ops.flash_attn_
+        varlen — the reference's flash-attn capability)."""
+        import polyrl_amd.ops as pops
+        T = x.shape[0]
+        q = self.q_proj(x).view(T, self.num_heads, self.head_dim)
+        k = self.k_proj(x).view(T, self.num_kv_heads, self.head_dim)
+        v = self.v_proj(x).view(T, self.num_kv_heads, self.head_dim)
+        cos = cos.to(q.dtype).unsqueeze(1)         # (T, 1, D/2)
+        sin = sin.to(q.dtype).unsqueeze(1)
+        d = self.head_dim // 2
+        q = torch.cat([q[..., :d] * cos - q[..., d:] * sin,
+                       q[..., d:] * cos + q[..., :d] * sin], dim=-1)
+        k = torch.cat([k[..., :d] * cos - k[..., d:] * sin,
+                       k[..., d:] * cos + k[..., :d] * sin], dim=-1)
+        scale = 1.0 / math.sqrt(self.head_dim)
+        o = pops.flash_attn_varlen(q, k, v, cu_seqlens[0], scale,
+                                   causal=True, tiles=cu_seqlens[1])
+        return self.o_proj(o.to(x.dtype).reshape(T, -1))
+
     def forward(self, x, cos, sin, attn_bias_mask: Optional[torch.Tensor],
                 sp_group=None):
         """cos/sin: (B, 1, L, D/2) fp32 over the FULL sequence.  With
@@ -139,9 +161,14 @@ class DecoderLayer(nn.Module):
         self.input_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
         self.post_attention_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
 
-    def forward(self, x, cos, sin, attn_bias_mask, sp_group=None):
-        x = x + self.self_attn(self.input_layernorm(x), cos, sin,
-                               attn_bias_mask, sp_group)
+    def forward(self, x, cos, sin, attn_bias_mask, sp_group=None,
+                cu_seqlens=None):
+        if cu_seqlens is not None:
+            x = x + self.self_attn.forward_packed(
+                self.input_layernorm(x), cos, sin, cu_seqlens)
+        else:
+            x = x + self.self_attn(self.input_layernorm(x), cos, sin,
+                                   attn_bias_mask, sp_group)
         x = x + self.mlp(self.post_attention_layernorm(x))
         return x
 
@@ -157,12 +184,55 @@ class DecoderModel(nn.Module):
         self.rotary = RotaryCache(cfg.head_dim, cfg.rope_theta)
         self.gradient_checkpointing = False
         self.ulysses = None        # UlyssesContext set by the worker (SP>1)
+        self.use_remove_padding = False  # packed varlen path (worker-set)
+
+    def _forward_packed(self, input_ids, attention_mask, position_ids):
+        """Remove-padding path (reference: use_remove_padding + flash varlen,
+        SURVEY.md §5.7): pack the valid tokens of all rows, run every layer
+        on (total, H) with block-diagonal causal flash attention, scatter
+        back to (B, L, H).  12-25%% fewer tokens through every GEMM on
+        typical left-padded prompt batches, and no SDPA mask fallback."""
+        B, L = input_ids.shape
+        valid = attention_mask.bool()
+        seqlens = valid.sum(-1).int()
+        cu = torch.zeros(B + 1, dtype=torch.int32, device=input_ids.device)
+        torch.cumsum(seqlens, 0, out=cu[1:])
+        ids_p = input_ids[valid]                     # (total,)
+        pos_p = position_ids[valid]
+        cos, sin = self.rotary.get(pos_p)            # (total, D/2)
+        x = self.embed_tokens(ids_p)
+        # tile tables host-built ONCE per forward, shared by all layers
+        if input_ids.is_cuda:
+            import polyrl_amd.ops as pops
+            tiles = pops.build_varlen_tiles(cu.cpu(), input_ids.device)
+        else:
+            tiles = None
+        cu_pack = (cu, tiles)
+        for layer in self.layers:
+            if self.gradient_checkpointing and self.training:
+                x = torch.utils.checkpoint.checkpoint(
+                    layer, x, cos, sin, None, None, cu_pack,
+                    use_reentrant=False)
+            else:
+                x = layer(x, cos, sin, None, None, cu_pack)
+        x = self.norm(x)
+        out = torch.zeros(B, L, x.shape[-1], dtype=x.dtype, device=x.device)
+        out[valid] = x
+        return out
 
     def forward(self, input_ids, attention_mask=None, position_ids=None):
         """With Ulysses SP enabled, every SP rank passes the SAME full
         (B, L) inputs (L divisible by sp); the returned hidden states are
         this rank's (B, L/sp, H) sequence shard."""
         B, L = input_ids.shape
+        sp_on = self.ulysses is not None and self.ulysses.enabled
+        if self.use_remove_padding and not sp_on \
+                and attention_mask is not None:
+            if position_ids is None:
+                position_ids = torch.arange(
+                    L, device=input_ids.device).expand(B, L)
+            return self._forward_packed(input_ids, attention_mask,
+                                        position_ids)
         if position_ids is None:
             position_ids = torch.arange(L, device=input_ids.device).expand(B, L)
         # per-row rope tables shaped for broadcast: (B, 1, L, D/2)

@@ -202,15 +202,81 @@ def _build_prefill_tiles(cu_seqlens_q: torch.Tensor, qtile: int = 64):
 def varlen_prefill_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                              cu_seqlens_q: torch.Tensor,
                              cu_seqlens_k: torch.Tensor, scale: float,
-                             causal: bool = True) -> torch.Tensor:
+                             causal: bool = True, return_lse: bool = False):
     if q.is_cuda:
         ext = _require_ext()
         out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         tile_seq, tile_q0 = _build_prefill_tiles(cu_seqlens_q)
+        lse = (torch.empty(q.shape[0], q.shape[1], dtype=torch.float32,
+                           device=q.device) if return_lse else torch.Tensor())
         ext.varlen_prefill_attention(
             out, q, k, v, cu_seqlens_q.int(),
             cu_seqlens_k.int(), tile_seq.to(q.device), tile_q0.to(q.device),
-            scale, causal)
+            scale, causal, lse)
+        return (out, lse) if return_lse else out
+    out = ref.varlen_prefill_attention(q, k, v, cu_seqlens_q, cu_seqlens_k,
+                                       scale, causal)
+    if return_lse:
+        return out, ref.varlen_lse(q, k, cu_seqlens_q, cu_seqlens_k, scale,
+                                   causal)
+    return out
+
+
+class _FlashAttnVarlen(torch.autograd.Function):
+    """Training-path varlen causal attention: MFMA forward (with LSE) and
+    hand-written CDNA4 backward (attention_backward.hip) -- the flash-attn
+    capability of the reference training stack."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, cu_seqlens, scale, causal,
+                t64_seq, t64_q0, t32_seq, t32_k0):
+        ext = _require_ext()
+        qc, kc, vc = q.contiguous(), k.contiguous(), v.contiguous()
+        out = torch.empty(qc.shape, dtype=qc.dtype, device=qc.device)
+        lse = torch.empty(qc.shape[0], qc.shape[1], dtype=torch.float32,
+                          device=qc.device)
+        ext.varlen_prefill_attention(out, qc, kc, vc, cu_seqlens, cu_seqlens,
+                                     t64_seq, t64_q0, scale, causal, lse)
+        ctx.save_for_backward(qc, kc, vc, out, lse, cu_seqlens,
+                              t32_seq, t32_k0)
+        ctx.scale = scale
+        ctx.causal = causal
         return out
-    return ref.varlen_prefill_attention(q, k, v, cu_seqlens_q, cu_seqlens_k,
-                                        scale, causal)
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, out, lse, cu, t32_seq, t32_k0 = ctx.saved_tensors
+        ext = _require_ext()
+        dq = torch.zeros(q.shape, dtype=torch.float32, device=q.device)
+        dk = torch.zeros(k.shape, dtype=torch.float32, device=q.device)
+        dv = torch.zeros(v.shape, dtype=torch.float32, device=q.device)
+        ext.varlen_attention_backward(
+            dq, dk, dv, q, k, v, out, dout.contiguous(), lse,
+            cu, cu, t32_seq, t32_k0, ctx.scale, ctx.causal)
+        return (dq.to(q.dtype), dk.to(q.dtype), dv.to(q.dtype),
+                None, None, None, None, None, None, None)
+
+
+def build_varlen_tiles(cu_seqlens_cpu: torch.Tensor, device):
+    """Precompute the (64-row forward, 32-key backward) tile tables once per
+    packed batch; reused by every layer's flash_attn_varlen call."""
+    t64 = _build_prefill_tiles(cu_seqlens_cpu, qtile=64)
+    t32 = _build_prefill_tiles(cu_seqlens_cpu, qtile=32)
+    return (t64[0].to(device), t64[1].to(device),
+            t32[0].to(device), t32[1].to(device))
+
+
+def flash_attn_varlen(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                      cu_seqlens: torch.Tensor, scale: float,
+                      causal: bool = True, tiles=None) -> torch.Tensor:
+    """Differentiable varlen causal attention over packed (total, H, D).
+    GPU: custom MFMA fwd/bwd kernels (tiles from build_varlen_tiles, else
+    derived here with a device sync); CPU: the plain-torch reference (its
+    autograd supplies the backward)."""
+    if q.is_cuda:
+        if tiles is None:
+            tiles = build_varlen_tiles(cu_seqlens.cpu(), q.device)
+        return _FlashAttnVarlen.apply(q, k, v, cu_seqlens.int(), scale,
+                                      causal, *tiles)
+    return ref.varlen_prefill_attention(q, k, v, cu_seqlens, cu_seqlens,
+                                        scale, causal).to(q.dtype)

@@ -39,6 +39,7 @@ DEV_INLINE int kswz(int row, int byte_col) {
 
 __global__ __launch_bounds__(256) void prefill_attn_kernel(
     bf16_t* __restrict__ out,        // (total_q, Hq, 128)
+    float* __restrict__ lse,         // (total_q, Hq) log-sum-exp, or null
     const bf16_t* __restrict__ q,    // (total_q, Hq, 128)
     const bf16_t* __restrict__ k,    // (total_k, Hk, 128)
     const bf16_t* __restrict__ v,    // (total_k, Hk, 128)
@@ -207,7 +208,7 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
     __syncthreads();  // all waves done with Ks/Vs before restage
   }
 
-  // ---------------- epilogue: out = O / l --------------------------------
+  // ---------------- epilogue: out = O / l; lse = m + log l ----------------
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int qrow_l = wq0 + lhi * 4 + r;
@@ -217,6 +218,9 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
 #pragma unroll
     for (int nt = 0; nt < 8; ++nt)
       op[nt * 16 + l15] = f2bf(o_acc[nt][r] * inv);
+    if (lse != nullptr && l15 == 0)
+      lse[((long)qbeg + qrow_l) * Hq + hq] =
+          (l_run[r] > 0.f) ? (m_run[r] + __logf(l_run[r])) : -1e30f;
   }
 }
 
@@ -225,7 +229,12 @@ void varlen_prefill_attention(torch::Tensor out, torch::Tensor q,
                               torch::Tensor cu_seqlens_q,
                               torch::Tensor cu_seqlens_k,
                               torch::Tensor tile_seq, torch::Tensor tile_q0,
-                              double scale, bool causal) {
+                              double scale, bool causal,
+                              torch::Tensor lse /* undefined or (tq,Hq) f32 */) {
+  if (lse.defined()) {
+    TORCH_CHECK(lse.dtype() == torch::kFloat32 && lse.is_contiguous());
+    TORCH_CHECK(lse.size(0) == q.size(0) && lse.size(1) == q.size(1));
+  }
   TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16);
   auto packed = [](const torch::Tensor& t) {
     return t.stride(2) == 1 && t.stride(1) == t.size(2);
@@ -241,7 +250,9 @@ void varlen_prefill_attention(torch::Tensor out, torch::Tensor q,
   const int ntiles = tile_seq.size(0);
   auto stream = at::hip::getCurrentHIPStream();
   prefill_attn_kernel<<<dim3(ntiles, Hq), dim3(256), 0, stream>>>(
-      (bf16_t*)out.data_ptr(), (const bf16_t*)q.data_ptr(),
+      (bf16_t*)out.data_ptr(),
+      lse.defined() ? lse.data_ptr<float>() : nullptr,
+      (const bf16_t*)q.data_ptr(),
       (const bf16_t*)k.data_ptr(), (const bf16_t*)v.data_ptr(),
       cu_seqlens_q.data_ptr<int>(), cu_seqlens_k.data_ptr<int>(),
       tile_seq.data_ptr<int>(), tile_q0.data_ptr<int>(), Hq, Hk, (float)scale,

@@ -27,7 +27,14 @@ void varlen_prefill_attention(torch::Tensor out, torch::Tensor q,
                               torch::Tensor cu_seqlens_q,
                               torch::Tensor cu_seqlens_k,
                               torch::Tensor tile_seq, torch::Tensor tile_q0,
-                              double scale, bool causal);
+                              double scale, bool causal, torch::Tensor lse);
+void varlen_attention_backward(
+    torch::Tensor dq, torch::Tensor dk, torch::Tensor dv,
+    torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    torch::Tensor out, torch::Tensor dout, torch::Tensor lse,
+    torch::Tensor cu_seqlens_q, torch::Tensor cu_seqlens_k,
+    torch::Tensor tile_seq, torch::Tensor tile_k0,
+    double scale, bool causal);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, CDNA4)");
@@ -43,5 +50,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("paged_attention_decode", &paged_attention_decode,
         "paged decode attention (GQA, bf16)");
   m.def("varlen_prefill_attention", &varlen_prefill_attention,
-        "varlen causal prefill attention (MFMA, bf16)");
+        "varlen causal prefill attention (MFMA, bf16; optional LSE out)");
+  m.def("varlen_attention_backward", &varlen_attention_backward,
+        "varlen causal flash-attention backward (MFMA, bf16)");
 }

@@ -177,3 +177,27 @@ def varlen_prefill_attention(
             p = torch.softmax(s, dim=-1)
             out[q0:q1, h] = p @ vi[:, hk]
     return out.to(q.dtype)
+
+
+def varlen_lse(q, k, cu_seqlens_q, cu_seqlens_k, scale, causal=True):
+    """Log-sum-exp of the (masked) attention scores per (token, q-head) —
+    reference for the prefill kernel's LSE output."""
+    Hq = q.shape[1]
+    Hk = k.shape[1]
+    g = Hq // Hk
+    out = torch.empty(q.shape[0], Hq, dtype=torch.float32)
+    B = cu_seqlens_q.shape[0] - 1
+    for b in range(B):
+        q0, q1 = int(cu_seqlens_q[b]), int(cu_seqlens_q[b + 1])
+        k0, k1 = int(cu_seqlens_k[b]), int(cu_seqlens_k[b + 1])
+        Lq, Lk = q1 - q0, k1 - k0
+        qi = q[q0:q1].float()
+        ki = k[k0:k1].float()
+        for h in range(Hq):
+            s = (qi[:, h] @ ki[:, h // g].T) * scale
+            if causal:
+                qpos = torch.arange(Lq).unsqueeze(1) + (Lk - Lq)
+                kpos = torch.arange(Lk).unsqueeze(0)
+                s = s.masked_fill(kpos > qpos, float("-inf"))
+            out[q0:q1, h] = torch.logsumexp(s, dim=-1)
+    return out

@@ -66,11 +66,21 @@ class StreamPPOTrainer:
         self.model_cfg = model_cfg
         dtype = arr.model.dtype
 
+        def _set_remove_padding(m, mc, enabled):
+            # packed varlen path: GPU kernels need head_dim 128 (llama/qwen
+            # families); the CPU tier runs the torch reference at any dim
+            trunk = getattr(m, "model", None)
+            if trunk is None or not hasattr(trunk, "use_remove_padding"):
+                return
+            if enabled and (device == "cpu" or mc.head_dim == 128):
+                trunk.use_remove_padding = True
+
         torch.manual_seed(config.trainer.seed)
         actor_model = create_model(model_cfg, kind="actor", dtype=dtype,
                                    device=device)
         if arr.model.enable_gradient_checkpointing:
             actor_model.gradient_checkpointing_enable()
+        _set_remove_padding(actor_model, model_cfg, arr.model.use_remove_padding)
         self.actor = ActorWorker(actor_model, arr.actor, device=device)
 
         self.use_ref = arr.actor.use_kl_loss or config.algorithm.use_kl_in_reward
@@ -83,6 +93,8 @@ class StreamPPOTrainer:
                 {k: v for k, v in actor_model.state_dict().items()})
             for p in ref_model.parameters():
                 p.requires_grad_(False)
+            _set_remove_padding(ref_model, model_cfg,
+                                arr.model.use_remove_padding)
             self.ref = ActorWorker(ref_model, arr.actor, device=device,
                                    is_ref=True)
 
@@ -98,6 +110,8 @@ class StreamPPOTrainer:
                                         dtype=dtype, device=device)
             if critic_cfg.model.enable_gradient_checkpointing:
                 critic_model.gradient_checkpointing_enable()
+            _set_remove_padding(critic_model, critic_model_cfg,
+                                critic_cfg.model.use_remove_padding)
             self.critic = CriticWorker(critic_model, critic_cfg, device=device)
 
         # ---------------- rollout engine (co-located, one per rank) ---------

@@ -31,6 +31,7 @@ sources = [
         "kv_cache.hip",
         "attention_decode.hip",
         "attention_prefill.hip",
+        "attention_backward.hip",
     )
 ]
 
