@@ -17,12 +17,20 @@ import torch
 
 @contextmanager
 def marked_timer(name: str, timing_raw: Dict[str, float]):
-    """Accumulating wall-clock span: timing_raw[name] += elapsed."""
+    """Accumulating wall-clock span: timing_raw[name] += elapsed.
+    Also emits a roctx range so rocprofv3 marker traces group kernels by
+    trainer phase (SURVEY.md §5.1 capability)."""
+    import torch
+    on_gpu = torch.cuda.is_available()
+    if on_gpu:
+        torch.cuda.nvtx.range_push(f"polyrl/{name}")
     start = time.perf_counter()
     try:
         yield
     finally:
         timing_raw[name] = timing_raw.get(name, 0.0) + (time.perf_counter() - start)
+        if on_gpu:
+            torch.cuda.nvtx.range_pop()
 
 
 def reduce_metrics(metrics: Dict[str, List[float]]) -> Dict[str, float]:
