@@ -231,7 +231,8 @@ void varlen_prefill_attention(torch::Tensor out, torch::Tensor q,
                               torch::Tensor tile_seq, torch::Tensor tile_q0,
                               double scale, bool causal,
                               torch::Tensor lse /* undefined or (tq,Hq) f32 */) {
-  if (lse.defined()) {
+  const bool want_lse = lse.defined() && lse.numel() > 0;
+  if (want_lse) {
     TORCH_CHECK(lse.dtype() == torch::kFloat32 && lse.is_contiguous());
     TORCH_CHECK(lse.size(0) == q.size(0) && lse.size(1) == q.size(1));
   }
@@ -251,7 +252,7 @@ void varlen_prefill_attention(torch::Tensor out, torch::Tensor q,
   auto stream = at::hip::getCurrentHIPStream();
   prefill_attn_kernel<<<dim3(ntiles, Hq), dim3(256), 0, stream>>>(
       (bf16_t*)out.data_ptr(),
-      lse.defined() ? lse.data_ptr<float>() : nullptr,
+      want_lse ? lse.data_ptr<float>() : nullptr,
       (const bf16_t*)q.data_ptr(),
       (const bf16_t*)k.data_ptr(), (const bf16_t*)v.data_ptr(),
       cu_seqlens_q.data_ptr<int>(), cu_seqlens_k.data_ptr<int>(),
