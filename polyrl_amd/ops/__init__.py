@@ -129,23 +129,32 @@ def gather_logprobs(logits: torch.Tensor, labels: torch.Tensor,
 def sample(logits: torch.Tensor, temperature: torch.Tensor,
            top_k: torch.Tensor, top_p: torch.Tensor, seed: int,
            generator: Optional[torch.Generator] = None,
-           no_filter: Optional[bool] = None
+           no_filter: Optional[bool] = None,
+           seed_dev: Optional[torch.Tensor] = None,
+           out: Optional[Tuple[torch.Tensor, torch.Tensor]] = None
            ) -> Tuple[torch.Tensor, torch.Tensor]:
     """(N, V) -> (tokens (N,) int64, logprobs (N,) fp32 under raw softmax).
 
     ``no_filter=True`` asserts every row has top_k<=0 and top_p>=1 (the
     common RL rollout setting) and takes the fused single-pass kernel; pass
-    it from the caller to avoid a device sync here."""
+    it from the caller to avoid a device sync here.  ``seed_dev`` ((1,)
+    int64 device tensor) overrides ``seed`` so the call is hipGraph-
+    capturable with fresh randomness per replay; ``out`` supplies static
+    output buffers for the same reason."""
     if logits.is_cuda:
         ext = _require_ext()
         N = logits.size(0)
-        tokens = torch.empty(N, dtype=torch.int64, device=logits.device)
-        lps = torch.empty(N, dtype=torch.float32, device=logits.device)
+        if out is not None:
+            tokens, lps = out
+        else:
+            tokens = torch.empty(N, dtype=torch.int64, device=logits.device)
+            lps = torch.empty(N, dtype=torch.float32, device=logits.device)
         if no_filter is None:
             no_filter = bool(((top_k <= 0) | (top_k >= logits.size(1))).all()
                              and (top_p >= 1.0).all())
         ext.sample(tokens, lps, logits.contiguous(), temperature.float(),
-                   top_k.int(), top_p.float(), seed, no_filter)
+                   top_k.int(), top_p.float(), seed, no_filter,
+                   seed_dev if seed_dev is not None else torch.Tensor())
         return tokens, lps
     return ref.top_k_top_p_sample(logits, temperature, top_k, top_p, generator)
 

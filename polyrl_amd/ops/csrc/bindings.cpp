@@ -14,7 +14,7 @@ void gather_logprobs(torch::Tensor out_lp, torch::Tensor out_ent,
 void sample(torch::Tensor out_tokens, torch::Tensor out_logprobs,
             torch::Tensor logits, torch::Tensor temperature,
             torch::Tensor top_k, torch::Tensor top_p, int64_t seed,
-            bool no_filter);
+            bool no_filter, torch::Tensor seed_dev);
 void kv_cache_append(torch::Tensor k_cache, torch::Tensor v_cache,
                      torch::Tensor k, torch::Tensor v,
                      torch::Tensor slot_mapping);

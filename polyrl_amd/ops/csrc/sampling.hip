@@ -143,7 +143,8 @@ template <typename T, int VEC>
 __global__ __launch_bounds__(1024) void sample_fast_kernel(
     int64_t* __restrict__ out_tokens, float* __restrict__ out_logprobs,
     const T* __restrict__ logits, const float* __restrict__ temperature,
-    uint64_t seed, int V) {
+    uint64_t seed, const long long* __restrict__ seed_ptr, int V) {
+  if (seed_ptr != nullptr) seed = (uint64_t)*seed_ptr;
   __shared__ float red_f[16];
   __shared__ int red_i[16];
   const long row = blockIdx.x;
@@ -310,7 +311,7 @@ __global__ void sample_kernel(int64_t* __restrict__ out_tokens,
 void sample(torch::Tensor out_tokens, torch::Tensor out_logprobs,
             torch::Tensor logits, torch::Tensor temperature,
             torch::Tensor top_k, torch::Tensor top_p, int64_t seed,
-            bool no_filter) {
+            bool no_filter, torch::Tensor seed_dev /* optional (1,) i64 */) {
   TORCH_CHECK(logits.is_cuda() && logits.dim() == 2 && logits.is_contiguous());
   TORCH_CHECK(out_tokens.dtype() == torch::kInt64);
   TORCH_CHECK(out_logprobs.dtype() == torch::kFloat32);

@@ -359,6 +359,13 @@ class Engine:
         self._scale = 1.0 / math.sqrt(cfg.head_dim)
         self._gen = torch.Generator().manual_seed(seed)  # CPU ref path RNG
         self.decode_chunk_size = max(decode_chunk_size, 1)
+        # hipGraph-captured decode iteration (one replay per token):
+        # removes ~300 kernel-launch round-trips per decode step
+        self.enable_hip_graphs = (device != "cpu") and \
+            bool(int(__import__("os").environ.get("POLYRL_HIP_GRAPHS", "1")))
+        self._graphs: Dict[int, dict] = {}      # batch size -> capture state
+        self._graph_pool = None
+        self._graph_max_pages = (self.max_model_len + page_size - 1) // page_size
 
     # ------------------------------------------------------------ public API
     def add_request(self, rid: str, input_ids: List[int],
@@ -639,28 +646,134 @@ class Engine:
         out_tokens = torch.empty(C, B, dtype=torch.int64, device=dev)
         out_lps = torch.empty(C, B, dtype=torch.float32, device=dev)
 
-        def attn_fn(li, q, k, v):
-            return ops.paged_attention_decode(q, self.kv.k_cache[li],
-                                              self.kv.v_cache[li], page_table,
-                                              ctx, self._scale)
+        seed0 = (self._seed * 0x9E3779B9 + self._step_counter * 131) \
+            & 0x7FFFFFFFFFFF
+        if self.enable_hip_graphs and no_filter and dev != "cpu" \
+                and self.cfg.arch != "gpt2":
+            try:
+                self._decode_graphed(B, C, tokens, pos0, slots_all,
+                                     page_table, temp, tk, tp, seed0,
+                                     out_tokens, out_lps)
+                C_done = True
+            except Exception as e:
+                import traceback
+                print(f"[engine] hipGraph decode capture failed, falling "
+                      f"back to eager: {e!r}", flush=True)
+                traceback.print_exc()
+                self.enable_hip_graphs = False   # fall back permanently
+                C_done = False
+        else:
+            C_done = False
+        if not C_done:
+            def attn_fn(li, q, k, v):
+                return ops.paged_attention_decode(
+                    q, self.kv.k_cache[li], self.kv.v_cache[li], page_table,
+                    ctx, self._scale)
 
-        for s in range(C):
-            positions = pos0 + s
-            hidden = self.model.forward_tokens(tokens, positions, self.kv,
-                                               slots_all[s], attn_fn)
-            logits = self.model.logits(hidden)
-            seed = (self._seed * 0x9E3779B9 + self._step_counter * 131 + s) \
-                & 0x7FFFFFFFFFFF
-            tokens, lps = ops.sample(logits, temp, tk, tp, seed,
-                                     generator=self._gen,
-                                     no_filter=no_filter)
-            out_tokens[s] = tokens
-            out_lps[s] = lps
-            if s + 1 < C:
-                ctx = ctx + 1
+            for s in range(C):
+                positions = pos0 + s
+                hidden = self.model.forward_tokens(tokens, positions, self.kv,
+                                                   slots_all[s], attn_fn)
+                logits = self.model.logits(hidden)
+                tokens, lps = ops.sample(logits, temp, tk, tp, seed0 + s,
+                                         generator=self._gen,
+                                         no_filter=no_filter)
+                out_tokens[s] = tokens
+                out_lps[s] = lps
+                if s + 1 < C:
+                    ctx = ctx + 1
 
         # single sync for the whole chunk
         toks_h = out_tokens.t().cpu().tolist()
+        self._finish_decode_chunk(reqs, C, toks_h, out_lps)
+
+    def _decode_graphed(self, B, C, tokens, pos0, slots_all, page_table,
+                        temp, tk, tp, seed0, out_tokens, out_lps):
+        """hipGraph decode: capture one decode iteration per batch size and
+        replay it per token.  All state (tokens/positions/ctx/slots/page
+        table/seed) lives in static device buffers; positions, ctx and the
+        sampling seed advance INSIDE the graph, the slot column is a tiny
+        D2D copy per replay."""
+        dev = self.device
+        st = self._graphs.get(B)
+        if st is None:
+            st = {
+                "tokens": torch.zeros(B, dtype=torch.int64, device=dev),
+                "pos": torch.zeros(B, dtype=torch.int32, device=dev),
+                "ctx": torch.zeros(B, dtype=torch.int32, device=dev),
+                "slots": torch.zeros(B, dtype=torch.int32, device=dev),
+                "ptab": torch.zeros(B, self._graph_max_pages,
+                                    dtype=torch.int32, device=dev),
+                "temp": torch.ones(B, dtype=torch.float32, device=dev),
+                "tk": torch.full((B,), -1, dtype=torch.int32, device=dev),
+                "tp": torch.ones(B, dtype=torch.float32, device=dev),
+                "seed": torch.zeros(1, dtype=torch.int64, device=dev),
+                "out_t": torch.zeros(B, dtype=torch.int64, device=dev),
+                "out_l": torch.zeros(B, dtype=torch.float32, device=dev),
+            }
+
+            def body():
+                def attn_fn(li, q, k, v):
+                    return ops.paged_attention_decode(
+                        q, self.kv.k_cache[li], self.kv.v_cache[li],
+                        st["ptab"], st["ctx"], self._scale)
+                hidden = self.model.forward_tokens(
+                    st["tokens"], st["pos"], self.kv, st["slots"], attn_fn)
+                logits = self.model.logits(hidden)
+                ops.sample(logits, st["temp"], st["tk"], st["tp"], 0,
+                           no_filter=True, seed_dev=st["seed"],
+                           out=(st["out_t"], st["out_l"]))
+                st["tokens"].copy_(st["out_t"])
+                st["pos"] += 1
+                st["ctx"] += 1
+                st["seed"] += 1
+
+            st["body"] = body
+            # warmup on a side stream (kv writes land in slots that the
+            # first real replay overwrites with identical data)
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            st["tokens"].copy_(tokens)
+            st["pos"].copy_(pos0)
+            st["ctx"].copy_(pos0 + 1)
+            st["slots"].copy_(slots_all[0])
+            mp = min(page_table.shape[1], self._graph_max_pages)
+            st["ptab"].zero_()
+            st["ptab"][:, :mp].copy_(page_table[:, :mp])
+            with torch.cuda.stream(side):
+                for _ in range(2):
+                    body()
+            torch.cuda.current_stream().wait_stream(side)
+            g = torch.cuda.CUDAGraph()
+            if self._graph_pool is None:
+                with torch.cuda.graph(g):
+                    body()
+                self._graph_pool = g.pool()
+            else:
+                with torch.cuda.graph(g, pool=self._graph_pool):
+                    body()
+            st["graph"] = g
+            self._graphs[B] = st
+
+        # load chunk state into the static buffers
+        st["tokens"].copy_(tokens)
+        st["pos"].copy_(pos0)
+        st["ctx"].copy_(pos0 + 1)
+        mp = min(page_table.shape[1], self._graph_max_pages)
+        st["ptab"].zero_()
+        st["ptab"][:, :mp].copy_(page_table[:, :mp])
+        st["temp"].copy_(temp)
+        st["tk"].copy_(tk)
+        st["tp"].copy_(tp)
+        st["seed"].fill_(seed0)
+        g = st["graph"]
+        for s in range(C):
+            st["slots"].copy_(slots_all[s])
+            g.replay()
+            out_tokens[s].copy_(st["out_t"])
+            out_lps[s].copy_(st["out_l"])
+
+    def _finish_decode_chunk(self, reqs, C, toks_h, out_lps):
         lps_h = out_lps.t().cpu().tolist()
         for i, r in enumerate(reqs):
             for s in range(C):
