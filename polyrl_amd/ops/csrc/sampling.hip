@@ -322,16 +322,19 @@ void sample(torch::Tensor out_tokens, torch::Tensor out_logprobs,
   const int V = logits.size(1);
   auto stream = at::hip::getCurrentHIPStream();
   if (no_filter) {  // fused single-pass path (no top-k/top-p), G13-vectorized
+    const long long* sp = seed_dev.defined() && seed_dev.numel() > 0
+                              ? seed_dev.data_ptr<long long>()
+                              : nullptr;
     if (logits.dtype() == torch::kBFloat16) {
       sample_fast_kernel<bf16_t, 8><<<dim3(N), dim3(1024), 0, stream>>>(
           out_tokens.data_ptr<int64_t>(), out_logprobs.data_ptr<float>(),
           (const bf16_t*)logits.data_ptr(), temperature.data_ptr<float>(),
-          (uint64_t)seed, V);
+          (uint64_t)seed, sp, V);
     } else {
       sample_fast_kernel<float, 4><<<dim3(N), dim3(1024), 0, stream>>>(
           out_tokens.data_ptr<int64_t>(), out_logprobs.data_ptr<float>(),
           (const float*)logits.data_ptr(), temperature.data_ptr<float>(),
-          (uint64_t)seed, V);
+          (uint64_t)seed, sp, V);
     }
     HIP_CHECK_KERNEL();
     return;
