@@ -323,7 +323,7 @@ void sample(torch::Tensor out_tokens, torch::Tensor out_logprobs,
   auto stream = at::hip::getCurrentHIPStream();
   if (no_filter) {  // fused single-pass path (no top-k/top-p), G13-vectorized
     const long long* sp = seed_dev.defined() && seed_dev.numel() > 0
-                              ? seed_dev.data_ptr<long long>()
+                              ? (const long long*)seed_dev.data_ptr<int64_t>()
                               : nullptr;
     if (logits.dtype() == torch::kBFloat16) {
       sample_fast_kernel<bf16_t, 8><<<dim3(N), dim3(1024), 0, stream>>>(
