@@ -191,6 +191,20 @@ class StreamPPOTrainer:
         mini_local = mini // world
         assert mini_local % local_stream == 0 or local_stream % mini_local == 0 \
             or mini_local % n == 0, "minibatch/stream sizes must compose"
+        # parallelism fences (reference: config/rollout.py:193-202 —
+        # EP = TP x DP constraint, PP declared but NotImplemented for >1)
+        ro = c.actor_rollout_ref.rollout
+        if ro.pipeline_model_parallel_size > 1:
+            raise NotImplementedError(
+                "pipeline_model_parallel_size > 1 is config-declared but not "
+                "implemented (matches the reference fence)")
+        if ro.expert_parallel_size > 1:
+            assert ro.expert_parallel_size == \
+                ro.tensor_model_parallel_size * ro.data_parallel_size, \
+                "expert_parallel_size must equal TP x DP"
+        sp = c.actor_rollout_ref.actor.ulysses_sequence_parallel_size
+        if sp > 1:
+            assert world % sp == 0, f"world {world} % sp {sp} != 0"
 
     # ------------------------------------------------------------------- fit
     def fit(self, max_steps: Optional[int] = None):
@@ -231,7 +245,49 @@ class StreamPPOTrainer:
                 if c.trainer.save_freq > 0 and \
                         self.global_step % c.trainer.save_freq == 0:
                     self.save_checkpoint()
+                if c.trainer.test_freq > 0 and \
+                        self.global_step % c.trainer.test_freq == 0:
+                    val = self.validate()
+                    if self.rank == 0 and val:
+                        self.tracking.log(val, self.global_step)
         return
+
+    @torch.no_grad()
+    def validate(self, num_prompts: Optional[int] = None) -> Dict[str, float]:
+        """Greedy validation rollouts + reward scoring (the reference's
+        _validate capability, stream_ray_trainer.py:305-313; validation
+        falls back to local colocated generation like
+        sglang_rollout_remote.py:184-196)."""
+        c = self.config
+        ro = c.actor_rollout_ref.rollout
+        nval = num_prompts or max(c.data.train_batch_size // self.world, 1)
+        idx = list(range(min(nval, len(self.dataset))))
+        batch = self.dataset.batch(idx)
+        sampling = SamplingParams(temperature=0.0,
+                                  max_new_tokens=ro.response_length)
+        self.publisher.publish()
+        self.coordinator.submit(batch, sampling, 1)
+        groups = []
+        for b in self.coordinator.stream_batches(len(idx)):
+            groups.append(b)
+        if not groups:
+            return {}
+        full = TensorBatch.concat(groups)
+        scores = self.reward_fn(full)
+        seq_scores = scores.sum(-1)
+        lens = full["response_mask"].sum(-1).float()
+        out = {
+            "val/score/mean": float(seq_scores.mean()),
+            "val/score/max": float(seq_scores.max()),
+            "val/response_length/mean": float(lens.mean()),
+            "val/n": float(len(full)),
+        }
+        if dist.is_available() and dist.is_initialized():
+            t = torch.tensor([out["val/score/mean"], out["val/n"]])
+            dist.all_reduce(t)
+            out["val/score/mean"] = float(t[0] / self.world)
+            out["val/n"] = float(t[1])
+        return out
 
     # ------------------------------------------------------------- one step
     def _run_step(self, global_batch: TensorBatch, n: int, local_bs: int,

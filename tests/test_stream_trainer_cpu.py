@@ -125,3 +125,14 @@ def test_minibatch_boundary_math(tmp_path):
     total = cfg.data.train_batch_size * cfg.actor_rollout_ref.rollout.sampling.n
     expect = total // cfg.actor_rollout_ref.actor.ppo_mini_batch_size
     assert steps["n"] == expect, f"{steps['n']} opt steps, expected {expect}"
+
+
+def test_validate_greedy_rollouts(tmp_path):
+    cfg = tiny_config(tmp_path)
+    cfg.trainer.test_freq = 1
+    trainer = StreamPPOTrainer(cfg, reward_fn=load_reward_manager("random"))
+    trainer.fit(max_steps=1)          # runs validate() at step 1 via test_freq
+    val = trainer.validate(num_prompts=4)
+    assert "val/score/mean" in val
+    assert val["val/n"] == 4.0
+    assert 0.0 <= val["val/score/mean"] <= 1.0
