@@ -118,6 +118,11 @@ class RolloutConfig:
     max_running_requests: int = 256
     page_size: int = 16                  # KV tokens per page
     decode_chunk_size: int = 16          # device-resident decode chunk
+    # disaggregated split (BASELINE config #4): the LAST num_rollout_ranks
+    # ranks of the world serve rollout; 0 = co-located
+    num_rollout_ranks: int = 0
+    rollout_port_base: int = 30000
+    max_local_gen_s: float = 0.0         # scheduler time-box (0 = off)
     sampling: SamplingConfig = field(default_factory=SamplingConfig)
     calculate_log_probs: bool = True
     min_stream_batch_size: int = 16

@@ -118,6 +118,12 @@ def create_app(engine: Engine, runner: Optional[EngineRunner] = None):
         if ctype.startswith("application/json"):
             body = await request.json()
             version = int(body["version"])
+            if body.get("ack_only"):
+                # weights were delivered out of band (collective broadcast
+                # over xGMI, transfer/collective.py): version acknowledgment
+                runner.weight_version = version
+                return {"success": True,
+                        "message": f"ack version {version} (out-of-band)"}
             path = body.get("path")
             if path:
                 from safetensors.torch import load_file

@@ -110,9 +110,12 @@ class HttpInstance(RolloutInstance):
                     "/update_weights_from_agent",
                     json={"version": version, "path": path})
             else:
+                # no file transport: weights arrive out of band (collective
+                # broadcast); acknowledge the version so the scheduler
+                # reactivates this instance
                 r = await self._client.post(
                     "/update_weights_from_agent",
-                    json={"version": version})
+                    json={"version": version, "ack_only": True})
             ok = r.status_code == 200 and r.json().get("success", False)
         except Exception:
             ok = False

@@ -1,0 +1,291 @@
+"""Disaggregated (split) mode: trainer ranks + dedicated rollout ranks on
+one RCCL world (BASELINE config #4: e.g. 4 trainer GPUs + 4 rollout GPUs).
+
+Mirrors the reference's disaggregation with MI355X-
+native transports:
+  * generation: trainer rank 0 drives the in-process RolloutScheduler over
+    HttpInstances of the rollout ranks' engine servers (NDJSON-equivalent
+    group streaming, token-level continuation, weight-version gating);
+    finished groups are sliced into equal per-trainer-rank ibatches and
+    broadcast over the trainer group (the reference's RANK_ZERO
+    get_stream_batches pump, stream_fsdp_workers.py:497-507).
+  * weights: bucketed RCCL broadcast over the WORLD group
+    (transfer/collective.py) — trainer ranks all-gather their FSDP shards,
+    rank 0's buckets stream over xGMI to every rollout rank, which installs
+    them under the engine step lock.  Control plane = one
+    broadcast_object_list opcode per step ("publish" | "exit").
+"""
+from __future__ import annotations
+
+import asyncio
+import threading
+import time
+from typing import Dict, Iterator, List, Optional
+
+import torch
+import torch.distributed as dist
+
+from ..protocol import TensorBatch
+from ..scheduler import RolloutScheduler, SchedulerConfig
+from ..scheduler.manager import StreamingBatchIterator
+from ..scheduler.types import GroupRequest, SamplingSpec
+from .rollout_coordinator import postprocess_groups
+
+
+def split_roles(world: int, num_rollout: int):
+    assert 0 < num_rollout < world, \
+        f"num_rollout_ranks {num_rollout} must be in (0, world {world})"
+    t = world - num_rollout
+    return list(range(t)), list(range(t, world))
+
+
+def rollout_port(rank: int, base: int = 30000) -> int:
+    return base + rank
+
+
+# ------------------------------------------------------------ rollout rank
+
+
+def rollout_serve_loop(cfg, model_cfg, rank: int, device: str, dtype,
+                       port_base: int = 30000):
+    """Main loop of a dedicated rollout rank: engine + HTTP server thread;
+    the main thread participates in the control/weight collectives."""
+    import uvicorn
+
+    from ..rollout.engine import Engine
+    from ..rollout.runner import EngineRunner
+    from ..server import create_app
+    from ..transfer.collective import CollectiveWeightPlane
+
+    ro = cfg.actor_rollout_ref.rollout
+    kv_budget = 64 << 20 if device == "cpu" else None
+    if device != "cpu":
+        free, _ = torch.cuda.mem_get_info()
+        kv_budget = int(free * ro.gpu_memory_utilization * 0.6)
+    engine = Engine(model_cfg, device=device, dtype=dtype,
+                    page_size=ro.page_size, kv_bytes_budget=kv_budget,
+                    max_running_requests=ro.max_running_requests,
+                    max_num_batched_tokens=ro.max_num_batched_tokens,
+                    max_model_len=ro.prompt_length + ro.response_length,
+                    decode_chunk_size=ro.decode_chunk_size,
+                    seed=cfg.trainer.seed)
+    runner = EngineRunner(engine)
+    app = create_app(engine, runner)
+    port = rollout_port(rank, port_base)
+    server = uvicorn.Server(uvicorn.Config(app, host="127.0.0.1", port=port,
+                                           log_level="error"))
+    th = threading.Thread(target=server.run, daemon=True)
+    th.start()
+    print(f"[rollout rank {rank}] engine serving on 127.0.0.1:{port}",
+          flush=True)
+
+    plane = CollectiveWeightPlane(src=0, device=device)
+
+    def apply_weight(name, tensor):
+        engine.model.update_named(name, tensor)
+
+    while True:
+        box = [None]
+        dist.broadcast_object_list(box, src=0)
+        op = box[0]
+        if op == "publish":
+            with runner.lock:  # excludes in-flight generation during swap
+                version = plane.receive(apply_weight)
+            runner.weight_version = version
+        elif op == "exit":
+            break
+        else:
+            raise RuntimeError(f"unknown ctrl opcode {op!r}")
+    server.should_exit = True
+    runner.stop()
+
+
+# ------------------------------------------------------------ trainer side
+
+
+class DisaggPublisher:
+    """Weight publication for the split mode: ctrl opcode + bucketed world
+    broadcast; scheduler's weight version is bumped first so stale remote
+    instances are version-gated (handlers.rs:566-600 contract)."""
+
+    def __init__(self, model, device: str, scheduler: Optional[RolloutScheduler],
+                 loop: Optional[asyncio.AbstractEventLoop], rank: int):
+        from ..transfer.collective import CollectiveWeightPlane
+        self.model = model
+        self.plane = CollectiveWeightPlane(src=0, device=device)
+        self.scheduler = scheduler
+        self.loop = loop
+        self.rank = rank
+        self.version = 0
+        self.last_publish_s = 0.0
+
+    @torch.no_grad()
+    def publish(self) -> int:
+        t0 = time.perf_counter()
+        self.version += 1
+        if self.rank == 0:
+            dist.broadcast_object_list(["publish"], src=0)
+            if self.scheduler is not None:
+                fut = asyncio.run_coroutine_threadsafe(
+                    self.scheduler.update_weight_version(self.version),
+                    self.loop)
+                fut.result(timeout=60)
+        else:
+            box = [None]
+            dist.broadcast_object_list(box, src=0)
+        sd = dict(self.model.state_dict())
+        self.plane.publish(sd, version=self.version)
+        if self.rank == 0 and self.scheduler is not None:
+            # remote instances got the bytes via the broadcast; mark them
+            # current so they rejoin the active pool
+            async def _activate():
+                for inst in self.scheduler.get_receive_instances():
+                    await self.scheduler.finish_weight_update(
+                        inst.instance_id, self.version, success=True)
+            asyncio.run_coroutine_threadsafe(_activate(), self.loop) \
+                .result(timeout=60)
+        self.last_publish_s = time.perf_counter() - t0
+        return self.version
+
+    def shutdown(self):
+        """Release the rollout ranks.  The ctrl broadcast is a WORLD
+        collective — every trainer rank must take part."""
+        if self.rank == 0:
+            dist.broadcast_object_list(["exit"], src=0)
+        else:
+            box = [None]
+            dist.broadcast_object_list(box, src=0)
+
+
+class DisaggCoordinator:
+    """Trainer-side coordinator for split mode (LocalRolloutCoordinator
+    surface: submit / stream_batches).  Rank 0 talks to the rollout pool via
+    the scheduler; every trainer rank receives equal ibatch slices."""
+
+    def __init__(self, response_length: int, trainer_group,
+                 rollout_urls: List[str], rank: int, n_trainer: int,
+                 pad_token_id: int = 0, device="cpu",
+                 max_local_gen_s: float = 0.0):
+        self.response_length = response_length
+        self.group = trainer_group
+        self.rank = rank
+        self.n_trainer = n_trainer
+        self.pad = pad_token_id
+        self.device = device
+        self.max_local_gen_s = max_local_gen_s
+        self._iter: Optional[StreamingBatchIterator] = None
+        self._meta: Dict[int, dict] = {}
+        self._gid = 0
+        self.loop: Optional[asyncio.AbstractEventLoop] = None
+        self.scheduler: Optional[RolloutScheduler] = None
+        if rank == 0:
+            self.loop = asyncio.new_event_loop()
+            t = threading.Thread(target=self.loop.run_forever, daemon=True)
+            t.start()
+            self.scheduler = RolloutScheduler(SchedulerConfig(
+                stats_interval_s=0.2))
+
+            async def _register():
+                from ..server import HttpInstance
+                for url in rollout_urls:
+                    inst = HttpInstance(url)
+                    # remote engines start at version 0 == scheduler's
+                    await self.scheduler.register_instance(inst)
+            asyncio.run_coroutine_threadsafe(_register(), self.loop) \
+                .result(timeout=600)
+
+    # --------------------------------------------------------------- submit
+    def submit(self, prompts: TensorBatch, sampling, n: int):
+        """Called with the GLOBAL batch on rank 0 (other ranks no-op)."""
+        if self.rank != 0:
+            return
+        ids = prompts["input_ids"]
+        mask = prompts["attention_mask"]
+        uids = prompts["uid"]
+        reqs = []
+        for g in range(ids.shape[0]):
+            gid = self._gid
+            self._gid += 1
+            self._meta[gid] = {
+                "uid": str(uids[g]),
+                "prompt_ids": ids[g].cpu(),
+                "prompt_mask": mask[g].cpu(),
+            }
+            raw = ids[g][mask[g].bool()].tolist()
+            reqs.append(GroupRequest(
+                gid=gid, input_ids=raw, n=n,
+                sampling=SamplingSpec(
+                    temperature=sampling.temperature,
+                    top_k=sampling.top_k, top_p=sampling.top_p,
+                    max_new_tokens=sampling.max_new_tokens,
+                    stop_token_ids=tuple(sampling.stop_token_ids))))
+        self._iter = StreamingBatchIterator(
+            self.scheduler, reqs,
+            max_local_gen_s=self.max_local_gen_s, loop=self.loop)
+
+    # --------------------------------------------------------------- stream
+    def stream_batches(self, local_stream: int) -> Iterator[TensorBatch]:
+        """Yield equal per-rank batches of local_stream samples until the
+        submitted batch is consumed.  Rank 0 pulls from the scheduler stream
+        and fans slices out over the trainer group."""
+        while True:
+            if self.rank == 0:
+                shard_box = self._next_shards(local_stream)
+            else:
+                shard_box = [None]
+            dist.broadcast_object_list(shard_box, src=0, group=self.group)
+            shards = shard_box[0]
+            if shards is None:
+                return
+            yield shards[self.rank if self.group is None
+                         else dist.get_rank(self.group)]
+
+    def _next_shards(self, local_stream: int):
+        """Rank 0: gather n_trainer*local_stream finished samples, build one
+        TensorBatch per trainer rank.  Returns [None] when drained."""
+        need = self.n_trainer * local_stream
+        got, groups = 0, []
+        for item in self._iter:
+            if isinstance(item, dict):     # submit notifier
+                continue
+            res = item
+            meta = self._meta.pop(res.gid)
+            groups.append((meta, res))
+            got += len(res.samples)
+            if got >= need:
+                break
+        if not groups:
+            return [None]
+        # partition into EQUAL sample shards — required so every trainer
+        # rank sees the same cumulative counts (the minibatch-boundary
+        # is_opt_step flags must agree across ranks or FSDP deadlocks);
+        # config validation guarantees full rounds divide evenly
+        assert got % self.n_trainer == 0, \
+            f"stream round of {got} samples !% {self.n_trainer} trainer ranks"
+        per = got // self.n_trainer
+        out = []
+        gi = 0
+        for _ in range(self.n_trainer):
+            take, taken = [], 0
+            while gi < len(groups) and taken < per:
+                take.append(groups[gi])
+                taken += len(groups[gi][1].samples)
+                gi += 1
+            assert taken == per, "group sizes must tile the shard evenly"
+            out.append(self._make_batch(take))
+        return [out]
+
+    def _make_batch(self, pairs) -> TensorBatch:
+        prompt_ids = torch.stack([m["prompt_ids"] for m, _ in pairs])
+        prompt_mask = torch.stack([m["prompt_mask"] for m, _ in pairs])
+        uids = [m["uid"] for m, _ in pairs]
+
+        class _O:                      # adapt SampleResult -> RequestOutput
+            def __init__(self, s):
+                self.output_ids = s.output_ids
+                self.output_logprobs = s.output_logprobs
+                self.finish_reason = s.finish_reason
+        outputs = [[_O(s) for s in res.samples] for _, res in pairs]
+        return postprocess_groups(prompt_ids, prompt_mask, uids, outputs,
+                                  self.response_length, self.pad,
+                                  self.device)

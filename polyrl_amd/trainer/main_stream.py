@@ -55,12 +55,39 @@ def main(argv=None):
     if rank == 0:
         print(f"[main_stream] world={world} device={device_type} "
               f"model={cfg.actor_rollout_ref.model.path} "
-              f"adv={cfg.algorithm.adv_estimator}")
+              f"adv={cfg.algorithm.adv_estimator}", flush=True)
     reward_fn = load_reward_manager(reward_name)
-    trainer = StreamPPOTrainer(cfg, reward_fn=reward_fn)
-    trainer.fit(max_steps=max_steps)
-    if max_steps is None or cfg.trainer.save_freq == -2:
-        pass
+
+    ro = cfg.actor_rollout_ref.rollout
+    if ro.num_rollout_ranks > 0:
+        # disaggregated split (BASELINE config #4): trailing ranks serve
+        # rollout, the rest train.  The trainer subgroup must be created on
+        # EVERY rank (collective), before the roles branch.
+        from ..models import get_model_config
+        from .disagg import rollout_serve_loop, split_roles
+        assert world > 1, "disaggregated mode needs a multi-rank world"
+        trainer_ranks, rollout_ranks = split_roles(world,
+                                                   ro.num_rollout_ranks)
+        tg = dist.new_group(trainer_ranks)
+        model_cfg = get_model_config(cfg.actor_rollout_ref.model.path)
+        print(f"[main_stream] rank {rank}: role="
+              f"{'rollout' if rank in rollout_ranks else 'trainer'}",
+              flush=True)
+        if rank in rollout_ranks:
+            dtype = getattr(torch, ro.dtype) if device_type == "cuda" \
+                else torch.float32
+            rollout_serve_loop(cfg, model_cfg, rank, device_type, dtype,
+                               port_base=ro.rollout_port_base)
+        else:
+            trainer = StreamPPOTrainer(cfg, reward_fn=reward_fn,
+                                       process_group=tg)
+            print(f"[main_stream] rank {rank}: trainer ready", flush=True)
+            trainer.fit(max_steps=max_steps)
+            print(f"[main_stream] rank {rank}: fit done", flush=True)
+            trainer.publisher.shutdown()   # releases the rollout ranks
+    else:
+        trainer = StreamPPOTrainer(cfg, reward_fn=reward_fn)
+        trainer.fit(max_steps=max_steps)
     if dist.is_available() and dist.is_initialized():
         dist.barrier()
         dist.destroy_process_group()

@@ -50,9 +50,14 @@ def _dist_info():
 
 class StreamPPOTrainer:
     def __init__(self, config: PPOConfig, device: Optional[str] = None,
-                 reward_fn=None, dataset=None):
+                 reward_fn=None, dataset=None, process_group=None):
         self.config = config
-        self.world, self.rank = _dist_info()
+        self.pg = process_group          # trainer subgroup (disagg) or None
+        if process_group is not None:
+            self.world = dist.get_world_size(process_group)
+            self.rank = dist.get_rank(process_group)
+        else:
+            self.world, self.rank = _dist_info()
         if device is None:
             device = config.trainer.device
         if device == "cuda":
@@ -81,7 +86,8 @@ class StreamPPOTrainer:
         if arr.model.enable_gradient_checkpointing:
             actor_model.gradient_checkpointing_enable()
         _set_remove_padding(actor_model, model_cfg, arr.model.use_remove_padding)
-        self.actor = ActorWorker(actor_model, arr.actor, device=device)
+        self.actor = ActorWorker(actor_model, arr.actor, device=device,
+                                 pg=self.pg)
 
         self.use_ref = arr.actor.use_kl_loss or config.algorithm.use_kl_in_reward
         self.ref = None
@@ -96,7 +102,7 @@ class StreamPPOTrainer:
             _set_remove_padding(ref_model, model_cfg,
                                 arr.model.use_remove_padding)
             self.ref = ActorWorker(ref_model, arr.actor, device=device,
-                                   is_ref=True)
+                                   is_ref=True, pg=self.pg)
 
         self.use_critic = config.algorithm.adv_estimator == "gae"
         self.critic = None
@@ -112,25 +118,46 @@ class StreamPPOTrainer:
                 critic_model.gradient_checkpointing_enable()
             _set_remove_padding(critic_model, critic_model_cfg,
                                 critic_cfg.model.use_remove_padding)
-            self.critic = CriticWorker(critic_model, critic_cfg, device=device)
+            self.critic = CriticWorker(critic_model, critic_cfg,
+                                       device=device, pg=self.pg)
 
-        # ---------------- rollout engine (co-located, one per rank) ---------
+        # ---------------- rollout plane -------------------------------------
         ro = arr.rollout
-        eng_dtype = getattr(torch, ro.dtype)
-        kv_budget = self._kv_budget(ro)
-        self.engine = Engine(model_cfg, device=device, dtype=eng_dtype,
-                             page_size=ro.page_size,
-                             kv_bytes_budget=kv_budget,
-                             max_running_requests=ro.max_running_requests,
-                             max_num_batched_tokens=ro.max_num_batched_tokens,
-                             max_model_len=ro.prompt_length + ro.response_length,
-                             decode_chunk_size=ro.decode_chunk_size,
-                             seed=config.trainer.seed * 1000 + self.rank)
-        self.coordinator = LocalRolloutCoordinator(
-            self.engine, ro.response_length, pad_token_id=0, device="cpu")
-        self.publisher = WeightPublisher(
-            self.actor.model, [self.engine.model],
-            tie_word_embeddings=model_cfg.tie_word_embeddings)
+        self.disagg = ro.num_rollout_ranks > 0
+        if self.disagg:
+            # split mode (BASELINE config #4): no local engine; the rollout
+            # pool lives on the trailing ranks, reached via the scheduler
+            from .disagg import (DisaggCoordinator, DisaggPublisher,
+                                 rollout_port, split_roles)
+            gworld = dist.get_world_size()
+            _, rollout_ranks = split_roles(gworld, ro.num_rollout_ranks)
+            urls = [f"http://127.0.0.1:{rollout_port(r, ro.rollout_port_base)}"
+                    for r in rollout_ranks]
+            self.engine = None
+            self.coordinator = DisaggCoordinator(
+                ro.response_length, self.pg, urls, rank=self.rank,
+                n_trainer=self.world, pad_token_id=0, device="cpu",
+                max_local_gen_s=ro.max_local_gen_s)
+            self.publisher = DisaggPublisher(
+                self.actor.model, device,
+                self.coordinator.scheduler, self.coordinator.loop,
+                rank=dist.get_rank())
+        else:
+            eng_dtype = getattr(torch, ro.dtype)
+            kv_budget = self._kv_budget(ro)
+            self.engine = Engine(model_cfg, device=device, dtype=eng_dtype,
+                                 page_size=ro.page_size,
+                                 kv_bytes_budget=kv_budget,
+                                 max_running_requests=ro.max_running_requests,
+                                 max_num_batched_tokens=ro.max_num_batched_tokens,
+                                 max_model_len=ro.prompt_length + ro.response_length,
+                                 decode_chunk_size=ro.decode_chunk_size,
+                                 seed=config.trainer.seed * 1000 + self.rank)
+            self.coordinator = LocalRolloutCoordinator(
+                self.engine, ro.response_length, pad_token_id=0, device="cpu")
+            self.publisher = WeightPublisher(
+                self.actor.model, [self.engine.model],
+                tie_word_embeddings=model_cfg.tie_word_embeddings)
 
         # ---------------- data + reward -------------------------------------
         dcfg = config.data
@@ -261,14 +288,17 @@ class StreamPPOTrainer:
         c = self.config
         ro = c.actor_rollout_ref.rollout
         nval = num_prompts or max(c.data.train_batch_size // self.world, 1)
+        if self.disagg:
+            nval = max(nval - nval % self.world, self.world)
         idx = list(range(min(nval, len(self.dataset))))
         batch = self.dataset.batch(idx)
         sampling = SamplingParams(temperature=0.0,
                                   max_new_tokens=ro.response_length)
         self.publisher.publish()
         self.coordinator.submit(batch, sampling, 1)
+        stream = len(idx) // self.world if self.disagg else len(idx)
         groups = []
-        for b in self.coordinator.stream_batches(len(idx)):
+        for b in self.coordinator.stream_batches(stream):
             groups.append(b)
         if not groups:
             return {}
@@ -284,7 +314,7 @@ class StreamPPOTrainer:
         }
         if dist.is_available() and dist.is_initialized():
             t = torch.tensor([out["val/score/mean"], out["val/n"]])
-            dist.all_reduce(t)
+            dist.all_reduce(t, group=self.pg)
             out["val/score/mean"] = float(t[0] / self.world)
             out["val/n"] = float(t[1])
         return out
@@ -301,15 +331,19 @@ class StreamPPOTrainer:
         with marked_timer("weight_sync", timing):
             self.publisher.publish()
 
-        # 2. shard the global batch; submit local prompts
-        local = global_batch.slice(
-            slice(self.rank * local_bs, (self.rank + 1) * local_bs))
+        # 2. submit prompts: co-located = this rank's shard to its engine;
+        #    disagg = the whole batch through the scheduler (rank 0)
         sampling = SamplingParams(
             temperature=ro.sampling.temperature,
             top_k=ro.sampling.top_k, top_p=ro.sampling.top_p,
             max_new_tokens=ro.response_length)
+        if self.disagg:
+            submit_batch = global_batch
+        else:
+            submit_batch = global_batch.slice(
+                slice(self.rank * local_bs, (self.rank + 1) * local_bs))
         with marked_timer("gen_submit", timing):
-            self.coordinator.submit(local, sampling, n)
+            self.coordinator.submit(submit_batch, sampling, n)
 
         # 3. stream loop
         all_metrics: Dict[str, List[float]] = {}

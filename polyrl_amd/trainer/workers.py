@@ -30,9 +30,11 @@ from ..core.seqlen import fixed_micro_batches, prepare_dynamic_batch, restore_dy
 from ..protocol import TensorBatch
 
 
-def _maybe_fully_shard(model: nn.Module, mixed_precision: bool = True):
+def _maybe_fully_shard(model: nn.Module, mixed_precision: bool = True,
+                       pg=None):
     """Apply FSDP2 per decoder layer + root.  Works for world_size 1..N on
-    nccl(RCCL) and gloo alike."""
+    nccl(RCCL) and gloo alike.  ``pg`` restricts sharding to a subgroup
+    (disaggregated split: FSDP over the trainer ranks only)."""
     if not (dist.is_available() and dist.is_initialized()):
         return model
     from torch.distributed.fsdp import MixedPrecisionPolicy, fully_shard
@@ -41,6 +43,10 @@ def _maybe_fully_shard(model: nn.Module, mixed_precision: bool = True):
         mp = MixedPrecisionPolicy(param_dtype=torch.bfloat16,
                                   reduce_dtype=torch.float32)
     kwargs = {"mp_policy": mp} if mp else {}
+    if pg is not None:
+        from torch.distributed.device_mesh import DeviceMesh
+        dev_type = "cuda" if torch.cuda.is_available() else "cpu"
+        kwargs["mesh"] = DeviceMesh.from_group(pg, dev_type)
     layers = None
     if hasattr(model, "model") and hasattr(model.model, "layers"):
         layers = model.model.layers
@@ -90,15 +96,15 @@ def _clip_grad_norm(model: nn.Module, max_norm: float) -> float:
     return float(total)
 
 
-def _sync_num_micro(k_local: int) -> int:
+def _sync_num_micro(k_local: int, pg=None) -> int:
     """FSDP2 unshard/reshard are collectives: every rank must run the SAME
     number of micro-batch fwd (and bwd) passes.  All-reduce the max count;
     ranks short of it run dummy passes (outputs discarded / zero-weighted)."""
     if not (dist.is_available() and dist.is_initialized()) or \
-            dist.get_world_size() == 1:
+            dist.get_world_size(pg) == 1:
         return k_local
     t = torch.tensor([k_local], dtype=torch.int64)
-    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    dist.all_reduce(t, op=dist.ReduceOp.MAX, group=pg)
     return int(t.item())
 
 
@@ -156,11 +162,15 @@ class ActorWorker:
     """Policy model under FSDP2: compute_log_prob + update_policy_stream."""
 
     def __init__(self, model: nn.Module, cfg: ActorConfig, device="cpu",
-                 is_ref: bool = False):
+                 is_ref: bool = False, pg=None):
         self.cfg = cfg
         self.device = device
         self.is_ref = is_ref
-        self.model = _maybe_fully_shard(model)
+        self.pg = pg
+        self.model = _maybe_fully_shard(model, pg=pg)
+        if pg is not None and cfg.ulysses_sequence_parallel_size > 1:
+            raise NotImplementedError("Ulysses SP inside a disaggregated "
+                                      "trainer subgroup is not wired yet")
         self.sp_group = _setup_ulysses(model,
                                        cfg.ulysses_sequence_parallel_size)
         self.sp_size = cfg.ulysses_sequence_parallel_size \
@@ -180,7 +190,7 @@ class ActorWorker:
         if self.sp_group is not None:
             batch, my_rows = _gather_rows(batch, self.sp_group)
         micro, parts = self._split(batch)
-        k = _sync_num_micro(len(micro))
+        k = _sync_num_micro(len(micro), self.pg)
         lps, ents = [], []
         for mb in micro:
             lp, ent = self._forward_logprobs(mb, want_entropy)
@@ -260,7 +270,7 @@ class ActorWorker:
         if self.sp_group is not None:
             batch, _ = _gather_rows(batch, self.sp_group)
         micro, _ = self._split(batch)
-        k = _sync_num_micro(len(micro))
+        k = _sync_num_micro(len(micro), self.pg)
         # dummy zero-weight passes keep FSDP fwd/bwd collectives aligned
         dummies = [micro[0].slice(slice(0, 1)) for _ in range(k - len(micro))]
         n_total = len(batch)
@@ -321,10 +331,16 @@ class ActorWorker:
 class CriticWorker:
     """Value model under FSDP2: compute_values + update_critic_stream."""
 
-    def __init__(self, model: nn.Module, cfg: CriticConfig, device="cpu"):
+    def __init__(self, model: nn.Module, cfg: CriticConfig, device="cpu",
+                 pg=None):
         self.cfg = cfg
         self.device = device
-        self.model = _maybe_fully_shard(model)
+        self.pg = pg
+        self.model = _maybe_fully_shard(model, pg=pg)
+        if pg is not None and \
+                getattr(cfg, "ulysses_sequence_parallel_size", 1) > 1:
+            raise NotImplementedError("Ulysses SP inside a disaggregated "
+                                      "trainer subgroup is not wired yet")
         self.sp_group = _setup_ulysses(
             model, getattr(cfg, "ulysses_sequence_parallel_size", 1))
         self.sp_size = cfg.ulysses_sequence_parallel_size \
@@ -361,7 +377,7 @@ class CriticWorker:
         if self.sp_group is not None:
             batch, my_rows = _gather_rows(batch, self.sp_group)
         micro, parts = self._split(batch)
-        k = _sync_num_micro(len(micro))
+        k = _sync_num_micro(len(micro), self.pg)
         outs = []
         for mi in range(k):
             mb = micro[mi] if mi < len(micro) else micro[0].slice(slice(0, 1))
@@ -378,7 +394,7 @@ class CriticWorker:
         if self.sp_group is not None:
             batch, _ = _gather_rows(batch, self.sp_group)
         micro, _ = self._split(batch)
-        k = _sync_num_micro(len(micro))
+        k = _sync_num_micro(len(micro), self.pg)
         dummies = [micro[0].slice(slice(0, 1)) for _ in range(k - len(micro))]
         n_total = len(batch)
         for mi, mb in enumerate(micro + dummies):
