@@ -19,6 +19,16 @@ import json
 import os
 import time
 
+# hipBLASLt algo selections tuned offline on MI355X (tunableop/*.csv —
+# PyTorch TunableOp); must be configured before torch initializes
+_TUNDIR = os.path.join(os.path.dirname(os.path.abspath(__file__)), "tunableop")
+if os.path.isdir(_TUNDIR) and not os.environ.get("POLYRL_NO_TUNABLEOP"):
+    os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME",
+                          os.path.join(_TUNDIR, "tunableop_llama8b.csv"))
+os.environ.setdefault("PYTORCH_ALLOC_CONF", "expandable_segments:True")
+
 import torch
 import torch.distributed as dist
 
