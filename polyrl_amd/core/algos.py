@@ -134,6 +134,65 @@ def compute_grpo_outcome_advantage(
     return adv, adv
 
 
+def compute_rloo_outcome_advantage(
+    token_level_rewards: torch.Tensor,
+    response_mask: torch.Tensor,
+    index,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """RLOO: leave-one-out baseline within each prompt group
+    (verl core_algos rloo capability)."""
+    scores = token_level_rewards.sum(dim=-1)
+    id2score = defaultdict(list)
+    bsz = scores.shape[0]
+    with torch.no_grad():
+        for i in range(bsz):
+            id2score[index[i]].append((i, scores[i]))
+        adv = scores.clone()
+        for idx, pairs in id2score.items():
+            k = len(pairs)
+            if k == 1:
+                adv[pairs[0][0]] = 0.0
+                continue
+            total = torch.stack([s for _, s in pairs]).sum()
+            for i, s in pairs:
+                adv[i] = s - (total - s) / (k - 1)
+        adv = adv.unsqueeze(-1) * response_mask
+    return adv, adv
+
+
+def compute_reinforce_plus_plus_advantage(
+    token_level_rewards: torch.Tensor,
+    response_mask: torch.Tensor,
+    gamma: float = 1.0,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """REINFORCE++: discounted reward-to-go, globally whitened
+    (verl core_algos reinforce_plus_plus capability)."""
+    with torch.no_grad():
+        returns = torch.zeros_like(token_level_rewards)
+        run = torch.zeros(token_level_rewards.shape[0],
+                          device=token_level_rewards.device)
+        for t in reversed(range(token_level_rewards.shape[1])):
+            run = token_level_rewards[:, t] + gamma * run
+            returns[:, t] = run
+        returns = returns * response_mask
+        adv = masked_whiten(returns, response_mask) * response_mask
+    return adv, returns
+
+
+def compute_remax_outcome_advantage(
+    token_level_rewards: torch.Tensor,
+    response_mask: torch.Tensor,
+    reward_baselines: torch.Tensor,      # (bsz,) greedy-rollout baseline
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """ReMax: outcome reward minus the greedy-decode baseline reward
+    (verl core_algos remax capability; the trainer must supply baselines
+    from a greedy rollout of the same prompts)."""
+    with torch.no_grad():
+        scores = token_level_rewards.sum(dim=-1)
+        adv = (scores - reward_baselines).unsqueeze(-1) * response_mask
+    return adv, adv
+
+
 # --------------------------------------------------------------- KL penalties
 
 
@@ -356,4 +415,12 @@ def compute_advantage(batch_rewards: torch.Tensor,
         return compute_grpo_outcome_advantage(
             batch_rewards, response_mask, index,
             norm_adv_by_std_in_grpo=norm_adv_by_std_in_grpo)
+    if adv_estimator == "rloo":
+        assert index is not None
+        return compute_rloo_outcome_advantage(batch_rewards, response_mask,
+                                              index)
+    if adv_estimator == "reinforce_plus_plus":
+        return compute_reinforce_plus_plus_advantage(batch_rewards,
+                                                     response_mask,
+                                                     gamma=gamma)
     raise NotImplementedError(f"adv_estimator {adv_estimator!r}")

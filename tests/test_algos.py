@@ -152,3 +152,58 @@ def test_apply_kl_penalty():
     rewards, kl = algos.apply_kl_penalty(scores, lp, ref_lp, mask, kl_coef=0.1)
     assert torch.allclose(rewards, scores)
     assert kl.item() == 0.0
+
+
+def test_rloo_advantage():
+    import numpy as np
+    import torch
+    from polyrl_amd.core import algos
+    rewards = torch.zeros(4, 3)
+    rewards[:, -1] = torch.tensor([1.0, 3.0, 2.0, 6.0])
+    mask = torch.ones(4, 3)
+    idx = np.array(["a", "a", "b", "b"], dtype=object)
+    adv, ret = algos.compute_rloo_outcome_advantage(rewards, mask, idx)
+    # group a: baselines are each other's score
+    assert torch.allclose(adv[0], torch.full((3,), 1.0 - 3.0))
+    assert torch.allclose(adv[1], torch.full((3,), 3.0 - 1.0))
+    assert torch.allclose(adv[2], torch.full((3,), 2.0 - 6.0))
+    assert torch.allclose(adv[3], torch.full((3,), 6.0 - 2.0))
+
+
+def test_reinforce_plus_plus_advantage():
+    import torch
+    from polyrl_amd.core import algos
+    rewards = torch.zeros(2, 4)
+    rewards[0, 3] = 1.0
+    rewards[1, 3] = -1.0
+    mask = torch.ones(2, 4)
+    adv, ret = algos.compute_reinforce_plus_plus_advantage(rewards, mask,
+                                                           gamma=1.0)
+    # reward-to-go is constant over tokens; whitening centers the two rows
+    assert torch.allclose(ret[0], torch.ones(4))
+    assert torch.allclose(ret[1], -torch.ones(4))
+    assert (adv[0] > 0).all() and (adv[1] < 0).all()
+
+
+def test_remax_advantage():
+    import torch
+    from polyrl_amd.core import algos
+    rewards = torch.zeros(2, 3)
+    rewards[:, -1] = torch.tensor([2.0, 0.5])
+    mask = torch.ones(2, 3)
+    base = torch.tensor([1.0, 1.0])
+    adv, _ = algos.compute_remax_outcome_advantage(rewards, mask, base)
+    assert torch.allclose(adv[0], torch.full((3,), 1.0))
+    assert torch.allclose(adv[1], torch.full((3,), -0.5))
+
+
+def test_compute_advantage_dispatch_new_estimators():
+    import numpy as np
+    import torch
+    from polyrl_amd.core import algos
+    rewards = torch.rand(4, 5)
+    mask = torch.ones(4, 5)
+    idx = np.array(["a", "a", "b", "b"], dtype=object)
+    for est in ("rloo", "reinforce_plus_plus"):
+        adv, ret = algos.compute_advantage(rewards, mask, est, index=idx)
+        assert adv.shape == rewards.shape
