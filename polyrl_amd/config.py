@@ -25,6 +25,7 @@ class ModelConfig:
     enable_gradient_checkpointing: bool = True
     use_remove_padding: bool = True
     lora_rank: int = 0
+    lora_alpha: float = 16.0
     trust_remote_code: bool = False
 
 

@@ -85,6 +85,10 @@ class StreamPPOTrainer:
                                    device=device)
         if arr.model.enable_gradient_checkpointing:
             actor_model.gradient_checkpointing_enable()
+        if arr.model.lora_rank > 0:
+            from ..models.lora import apply_lora
+            apply_lora(actor_model, arr.model.lora_rank,
+                       alpha=arr.model.lora_alpha)
         _set_remove_padding(actor_model, model_cfg, arr.model.use_remove_padding)
         self.actor = ActorWorker(actor_model, arr.actor, device=device,
                                  pg=self.pg)

@@ -45,7 +45,11 @@ class WeightPublisher:
         state dict into a host buffer)."""
         t0 = time.perf_counter()
         self.version += 1
-        sd = self.model.state_dict()
+        from ..models.lora import LoRALinear, merged_state_dict
+        if any(isinstance(m, LoRALinear) for m in self.model.modules()):
+            sd = merged_state_dict(self.model)   # LoRA deltas folded in
+        else:
+            sd = self.model.state_dict()
         for name, param in sd.items():
             if hasattr(param, "full_tensor"):      # DTensor -> all-gather
                 full = param.full_tensor()

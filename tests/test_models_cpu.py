@@ -111,3 +111,55 @@ def test_remove_padding_backward_cpu():
     out.float().pow(2).mean().backward()
     g = m.model.layers[0].self_attn.q_proj.weight.grad
     assert g is not None and torch.isfinite(g).all()
+
+
+def test_lora_adapters_and_merge():
+    """LoRA: only adapters train; merged publication matches manual math and
+    round-trips into the rollout engine under HF names."""
+    import torch
+    from polyrl_amd.models import create_model, get_model_config
+    from polyrl_amd.models.lora import LoRALinear, apply_lora, merged_state_dict
+    from polyrl_amd.rollout.engine import Engine
+
+    cfg = get_model_config("llama-debug-cpu")
+    torch.manual_seed(20)
+    m = create_model(cfg, kind="actor", dtype="float32", device="cpu")
+    wrapped = apply_lora(m, r=4, alpha=8.0)
+    assert any(w.endswith("q_proj") for w in wrapped)
+    trainable = [n for n, p in m.named_parameters() if p.requires_grad]
+    assert all("lora_" in n or "value_head" in n for n in trainable)
+
+    # nudge an adapter; merged weight = base + scale * B A
+    lin = m.model.layers[0].self_attn.q_proj
+    assert isinstance(lin, LoRALinear)
+    with torch.no_grad():
+        lin.lora_B.normal_(0, 0.1)
+    merged = merged_state_dict(m)
+    name = "model.layers.0.self_attn.q_proj.weight"
+    expect = lin.base.weight + lin.lora_B @ lin.lora_A * lin.scaling
+    assert torch.allclose(merged[name], expect, atol=1e-5)
+    assert not any(".lora_" in k or ".base." in k for k in merged)
+
+    # engine ingests the merged dict under plain HF names
+    eng = Engine(cfg, device="cpu", dtype=torch.float32,
+                 kv_bytes_budget=8 << 20)
+    eng.model.load_state_dict(merged, strict=True)
+    assert torch.allclose(
+        eng.model.layers[0].wqkv[:lin.base.out_features], expect)
+
+
+def test_lora_backward_only_adapters():
+    import torch
+    from polyrl_amd.models import create_model, get_model_config
+    from polyrl_amd.models.lora import apply_lora
+    cfg = get_model_config("llama-debug-cpu")
+    m = create_model(cfg, kind="actor", dtype="float32", device="cpu")
+    apply_lora(m, r=2)
+    ids = torch.randint(0, cfg.vocab_size, (2, 8))
+    out = m(ids)
+    out.float().pow(2).mean().backward()
+    for n, p in m.named_parameters():
+        if "lora_" in n:
+            assert p.grad is not None and torch.isfinite(p.grad).all(), n
+        elif "value_head" not in n:
+            assert p.grad is None, n
