@@ -20,9 +20,11 @@ import os
 import time
 
 # hipBLASLt algo selections tuned offline on MI355X (tunableop/*.csv —
-# PyTorch TunableOp); must be configured before torch initializes
+# PyTorch TunableOp).  OPT-IN via POLYRL_TUNABLEOP=1: a table tuned on a
+# different box measured SLOWER on a fresh one (old_log_prob 0.94->3.2 s),
+# so selections do not transfer reliably across boxes/clock states.
 _TUNDIR = os.path.join(os.path.dirname(os.path.abspath(__file__)), "tunableop")
-if os.path.isdir(_TUNDIR) and not os.environ.get("POLYRL_NO_TUNABLEOP"):
+if os.path.isdir(_TUNDIR) and os.environ.get("POLYRL_TUNABLEOP") == "1":
     os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
     os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
     os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME",
