@@ -249,11 +249,15 @@ class StreamPPOTrainer:
         local_total = local_bs * n
 
         step_in_run = 0
-        for epoch in range(c.trainer.total_epochs):
+        skip = self.global_step   # resume: replay the deterministic batch
+        for epoch in range(c.trainer.total_epochs):   # order up to the ckpt
             for global_batch in epoch_batches(self.dataset,
                                               c.data.train_batch_size,
                                               shuffle=c.data.shuffle,
                                               seed=c.data.seed + epoch):
+                if skip > 0:
+                    skip -= 1
+                    continue
                 if step_in_run >= total_steps:
                     return
                 self.global_step += 1
