@@ -139,7 +139,8 @@ def main():
 
     if rank == 0:
         print(json.dumps({
-            "metric": "end-to-end PPO samples/sec (rollout+update), Llama-3-8B",
+            "metric": "end-to-end PPO samples/sec (rollout+update), "
+                      + args.model,
             "value": samples_per_s,
             "unit": "samples/s",
             "n_gpus": world,
