@@ -29,6 +29,15 @@ if os.path.isdir(_TUNDIR) and os.environ.get("POLYRL_TUNABLEOP") == "1":
     os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
     os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME",
                           os.path.join(_TUNDIR, "tunableop_llama8b.csv"))
+elif os.environ.get("POLYRL_TUNABLEOP_TUNE") == "1":
+    # tune hipBLASLt algo selection ON THIS BOX during init/warmup (tight
+    # per-shape budget); selections stay in-process for the timed region
+    os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME",
+                          "/tmp/polyrl_tunableop.csv")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_MAX_TUNING_DURATION_MS", "10")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_MAX_WARMUP_DURATION_MS", "5")
 os.environ.setdefault("PYTORCH_ALLOC_CONF", "expandable_segments:True")
 
 import torch
