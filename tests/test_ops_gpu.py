@@ -439,3 +439,78 @@ def test_packed_model_forward_gpu():
     valid = am.bool()
     err = rel_err(packed[valid], dense[valid])
     assert err < 5e-2, err
+
+
+# ----------------------------------------------------- engine-level on GPU
+
+
+def test_engine_greedy_matches_trainer_model_gpu():
+    """Greedy engine decode (HIP kernel path + hipGraphs) must match the
+    trainer model's argmax chain on GPU."""
+    from polyrl_amd.models import create_model
+    from polyrl_amd.models.registry import DecoderConfig
+    from polyrl_amd.rollout.engine import Engine, SamplingParams
+    cfg = DecoderConfig(arch="llama", vocab_size=512, hidden_size=256,
+                        intermediate_size=512, num_hidden_layers=2,
+                        num_attention_heads=2, num_key_value_heads=1,
+                        head_dim=128, max_position_embeddings=128,
+                        rope_theta=10000.0, rms_norm_eps=1e-6)
+    torch.manual_seed(50)
+    model = create_model(cfg, kind="actor", dtype="bfloat16", device=DEV)
+    eng = Engine(cfg, device=DEV, dtype=torch.bfloat16,
+                 kv_bytes_budget=32 << 20, decode_chunk_size=4)
+    eng.model.load_state_dict(model.state_dict())
+    torch.manual_seed(51)
+    prompt = torch.randint(0, cfg.vocab_size, (9,)).tolist()
+    outs = eng.generate([prompt], SamplingParams(temperature=0.0,
+                                                 max_new_tokens=8), "g")
+    got = outs[0].output_ids
+    ids = list(prompt)
+    expect = []
+    with torch.no_grad():
+        for _ in range(8):
+            x = torch.tensor([ids], device=DEV)
+            logits = model(x).float()
+            t = int(logits[0, -1].argmax())
+            expect.append(t)
+            ids.append(t)
+    # bf16 kernel-vs-eager forward differences can flip near-tie argmaxes in
+    # a 2-layer random model; require the prefixes to agree
+    agree = 0
+    for a, b in zip(got, expect):
+        if a != b:
+            break
+        agree += 1
+    assert agree >= 4, (got, expect)
+
+
+def test_engine_sampling_replay_determinism_gpu():
+    """Same seed + same submission order => identical sampled outputs
+    (counter-based RNG, deterministic kernels; hipGraph replays included).
+    Note: outputs are keyed by batch-row like other engines, so different
+    chunk sizes/batch compositions legitimately draw differently."""
+    from polyrl_amd.models import create_model
+    from polyrl_amd.models.registry import DecoderConfig
+    from polyrl_amd.rollout.engine import Engine, SamplingParams
+    cfg = DecoderConfig(arch="llama", vocab_size=512, hidden_size=256,
+                        intermediate_size=512, num_hidden_layers=2,
+                        num_attention_heads=2, num_key_value_heads=1,
+                        head_dim=128, max_position_embeddings=256,
+                        rope_theta=10000.0, rms_norm_eps=1e-6)
+    torch.manual_seed(52)
+    model = create_model(cfg, kind="actor", dtype="bfloat16", device=DEV)
+    prompts = [torch.randint(0, cfg.vocab_size, (n,)).tolist()
+               for n in (5, 11)]
+    results = []
+    for _ in range(2):
+        eng = Engine(cfg, device=DEV, dtype=torch.bfloat16,
+                     kv_bytes_budget=32 << 20, decode_chunk_size=16,
+                     seed=7)
+        eng.model.load_state_dict(model.state_dict())
+        outs = eng.generate(prompts,
+                            SamplingParams(temperature=1.0,
+                                           max_new_tokens=12), "c")
+        results.append([o.output_ids for o in outs])
+    assert results[0] == results[1], results
+    lens = [len(o) for o in results[0]]
+    assert lens == [12, 12], lens
