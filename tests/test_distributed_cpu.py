@@ -75,3 +75,30 @@ def test_world2_ppo_gae(tmp_path):
         "max_steps=1",
     ])
     assert r.returncode == 0, f"stdout:\n{r.stdout[-3000:]}\nstderr:\n{r.stderr[-3000:]}"
+
+
+@pytest.mark.timeout(600)
+def test_bench_contract_world2():
+    """The driver's N>1 launch contract: torch.distributed.run bench.py
+    --gpus 2 prints one whole-job JSON line from rank 0."""
+    import json
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", "--nproc-per-node=2",
+        "--master-addr", "127.0.0.1", "--master-port", "29672",
+        os.path.join(repo, "bench.py"),
+        "--gpus", "2", "--steps", "1", "--warmup", "1",
+        "--model", "llama-debug-cpu", "--batch-per-gpu", "4",
+        "--n-samples", "2", "--prompt-len", "16", "--response-len", "8",
+    ]
+    env = dict(os.environ)
+    env["PYTHONPATH"] = repo
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=540,
+                      env=env)
+    assert r.returncode == 0, f"{r.stdout[-2000:]}\n{r.stderr[-2000:]}"
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    d = json.loads(line)
+    assert d["n_gpus"] == 2
+    assert d["config"]["parallelism"] == "dp2"
+    assert d["value"] > 0
