@@ -87,14 +87,61 @@ def compute_timing_metrics(batch, timing_raw: Dict[str, float]) -> Dict[str, flo
     return m
 
 
+class FlopsCounter:
+    """Dense-decoder FLOP estimates (the reference's verl FlopsCounter
+    capability): attainable-vs-achieved MFU from model geometry."""
+
+    # MI355X dense bf16 MFMA peak (AMD spec without 2:1 sparsity)
+    MI355X_BF16_DENSE_PEAK = 2.5e15
+
+    def __init__(self, model_cfg):
+        c = model_cfg
+        h = c.hidden_size
+        self.per_token_params_flops = 2 * (
+            c.num_hidden_layers * (
+                h * (c.num_attention_heads + 2 * c.num_key_value_heads)
+                * c.head_dim                      # qkv
+                + c.num_attention_heads * c.head_dim * h   # o
+                + 3 * h * c.intermediate_size)    # gate/up/down
+            + h * c.vocab_size)                   # lm head
+        self.cfg = c
+
+    def forward_flops(self, total_tokens: int, avg_seqlen: float) -> float:
+        attn = (2 * 2 * self.cfg.num_hidden_layers
+                * self.cfg.num_attention_heads * self.cfg.head_dim
+                * total_tokens * avg_seqlen)      # QK^T + PV (causal ~ /2)
+        return self.per_token_params_flops * total_tokens + attn / 2
+
+    def train_step_flops(self, total_tokens: int, avg_seqlen: float,
+                         grad_ckpt: bool = False) -> float:
+        mult = 4.0 if grad_ckpt else 3.0          # fwd + (recompute) + bwd
+        return mult * self.forward_flops(total_tokens, avg_seqlen)
+
+    def mfu(self, flops: float, seconds: float, n_gpus: int = 1,
+            peak: float = MI355X_BF16_DENSE_PEAK) -> float:
+        if seconds <= 0:
+            return 0.0
+        return flops / seconds / (peak * max(n_gpus, 1))
+
+
 def compute_throughput_metrics(batch, timing_raw: Dict[str, float],
-                               n_gpus: int) -> Dict[str, float]:
+                               n_gpus: int,
+                               model_cfg=None,
+                               use_critic: bool = False) -> Dict[str, float]:
     m = {}
     if "attention_mask" in batch.tensors and timing_raw.get("step", 0) > 0:
         total_tokens = batch.tensors["attention_mask"].sum().item()
         m["perf/total_tokens"] = total_tokens
         m["perf/throughput_tokens_per_s_all_gpus"] = total_tokens / timing_raw["step"]
         m["perf/throughput_tokens_per_s_per_gpu"] = total_tokens / timing_raw["step"] / max(n_gpus, 1)
+        if model_cfg is not None and timing_raw.get("update", 0) > 0:
+            fc = FlopsCounter(model_cfg)
+            L = batch.tensors["attention_mask"].shape[1]
+            models = 2 if use_critic else 1
+            flops = models * fc.train_step_flops(total_tokens, L)
+            m["perf/update_tflops_per_gpu"] = \
+                flops / timing_raw["update"] / max(n_gpus, 1) / 1e12
+            m["perf/update_mfu"] = fc.mfu(flops, timing_raw["update"], n_gpus)
     return m
 
 

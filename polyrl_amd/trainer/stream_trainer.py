@@ -273,7 +273,8 @@ class StreamPPOTrainer:
                     self._last_full_batch, timing))
                 metrics.update(compute_throughput_metrics(
                     self._last_full_batch, timing,
-                    self.world))
+                    self.world, model_cfg=self.model_cfg,
+                    use_critic=self.use_critic))
                 metrics["training/global_step"] = self.global_step
                 if self.rank == 0:
                     self.tracking.log(metrics, self.global_step)

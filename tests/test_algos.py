@@ -207,3 +207,14 @@ def test_compute_advantage_dispatch_new_estimators():
     for est in ("rloo", "reinforce_plus_plus"):
         adv, ret = algos.compute_advantage(rewards, mask, est, index=idx)
         assert adv.shape == rewards.shape
+
+
+def test_flops_counter_mfu():
+    from polyrl_amd.core.metrics import FlopsCounter
+    from polyrl_amd.models import get_model_config
+    fc = FlopsCounter(get_model_config("llama3-8b"))
+    # ~8B params -> per-token param flops ~ 2 * 8e9 (embedding-free)
+    assert 1.2e10 < fc.per_token_params_flops < 1.8e10
+    f = fc.train_step_flops(65536, 512)
+    mfu = fc.mfu(f, seconds=7.0, n_gpus=1)
+    assert 0.0 < mfu < 1.0
