@@ -186,13 +186,3 @@ class Tracking:
         if self._jsonl is not None:
             self._jsonl.close()
 
-
-class FlopsCounter:
-    """Model-FLOPs/step estimator for dense decoder LMs (6*N*T fwd+bwd rule)."""
-
-    def __init__(self, num_params: int):
-        self.num_params = num_params
-
-    def estimate_flops(self, tokens: int, backward: bool = True) -> float:
-        mult = 6 if backward else 2
-        return mult * self.num_params * tokens
