@@ -360,8 +360,9 @@ class Engine:
         self._gen = torch.Generator().manual_seed(seed)  # CPU ref path RNG
         self.decode_chunk_size = max(decode_chunk_size, 1)
         # hipGraph-captured decode iteration (one replay per token):
-        # removes ~300 kernel-launch round-trips per decode step
-        self.enable_hip_graphs = (device != "cpu") and \
+        # removes ~300 kernel-launch round-trips per decode step.  Disabled
+        # under TP (capturing RCCL collectives in a graph is unvalidated).
+        self.enable_hip_graphs = (device != "cpu") and self.tp.size == 1 and \
             bool(int(__import__("os").environ.get("POLYRL_HIP_GRAPHS", "1")))
         self._graphs: Dict[int, dict] = {}      # batch size -> capture state
         self._graph_pool = None
