@@ -142,6 +142,70 @@ def create_app(engine: Engine, runner: Optional[EngineRunner] = None):
             None, lambda: runner.update_weights(sd, version))
         return {"success": True, "message": f"weights at version {version}"}
 
+    @app.post("/weights_handshake")
+    async def weights_handshake(request: Request):
+        """Arm a TCP bulk receive (the reference's receiver-agent bootstrap,
+        receiver_agent.py:184-240): allocate/reuse a registered CPU buffer
+        sized to the incoming state dict, listen on N stream ports, return
+        them to the sender."""
+        import torch
+
+        from ..transfer.tcp_engine import TcpWeightReceiver
+        body = await request.json()
+        metas = body["metas"]            # [(name, shape, dtype_str), ...]
+        num_streams = int(body.get("num_streams", 4))
+        total = 0
+        for _, shape, dts in metas:
+            n = 1
+            for s in shape:
+                n *= s
+            total += n * torch.tensor([], dtype=getattr(torch, dts)) \
+                .element_size()
+        st = app.state
+        if getattr(st, "wt_buffer", None) is None or \
+                st.wt_buffer.numel() < total:
+            st.wt_buffer = torch.empty(total, dtype=torch.uint8)
+        if getattr(st, "wt_rx", None) is not None:
+            st.wt_rx.close()
+        st.wt_rx = TcpWeightReceiver(st.wt_buffer[:total],
+                                     host=body.get("bind_host", "0.0.0.0"),
+                                     num_streams=num_streams)
+        st.wt_rx.expect(total)
+        st.wt_metas = metas
+        return {"ports": st.wt_rx.ports, "total_bytes": total}
+
+    @app.post("/update_weights_from_tcp")
+    async def update_weights_from_tcp(request: Request):
+        """Install weights from the armed TCP receive: wait for all spans,
+        reconstruct named views (patches.py:205-215 capability), swap under
+        the step lock."""
+        import torch
+        body = await request.json()
+        version = int(body["version"])
+        st = app.state
+        assert getattr(st, "wt_rx", None) is not None, "handshake first"
+        ok = await asyncio.get_running_loop().run_in_executor(
+            None, lambda: st.wt_rx.wait(timeout=float(
+                body.get("timeout_s", 600.0))))
+        if not ok:
+            return JSONResponse({"success": False,
+                                 "message": "tcp receive timed out"}, 500)
+        sd = {}
+        off = 0
+        for name, shape, dts in st.wt_metas:
+            dt = getattr(torch, dts)
+            n = 1
+            for s in shape:
+                n *= s
+            nbytes = n * torch.tensor([], dtype=dt).element_size()
+            sd[name] = st.wt_buffer[off:off + nbytes].view(dt).view(shape)
+            off += nbytes
+        await asyncio.get_running_loop().run_in_executor(
+            None, lambda: runner.update_weights(sd, version))
+        st.wt_rx.close()
+        st.wt_rx = None
+        return {"success": True, "message": f"weights at version {version}"}
+
     @app.post("/release_memory_occupation")
     async def release_memory_occupation():
         import torch

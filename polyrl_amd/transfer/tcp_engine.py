@@ -187,3 +187,39 @@ class TcpWeightSender:
                 return st
             time.sleep(0.005)
         return "timeout"
+
+
+def push_state_dict_tcp(state_dict, client, host: str,
+                        num_streams: int = 4, version: int = 1,
+                        timeout_s: float = 600.0) -> bool:
+    """Sender-agent data path to ONE remote instance (sender_agent.py:390-427
+    capability): handshake -> N-stream TCP push of the flattened state dict
+    -> install call.  ``client`` is an httpx.Client/AsyncClient-compatible
+    SYNC client bound to the instance's base URL; ``host`` is the address
+    its TCP ports are reachable at."""
+    import torch
+    metas = []
+    flats = []
+    for name in sorted(state_dict):
+        t = state_dict[name]
+        t = t.full_tensor() if hasattr(t, "full_tensor") else t
+        t = t.detach().cpu().contiguous()
+        metas.append((name, list(t.shape), str(t.dtype).replace("torch.", "")))
+        flats.append(t.reshape(-1).view(torch.uint8).reshape(-1)
+                     if t.dtype == torch.uint8
+                     else t.reshape(-1).contiguous().view(torch.uint8))
+    buf = torch.cat([f.reshape(-1) for f in flats])
+    r = client.post("/weights_handshake",
+                    json={"metas": metas, "num_streams": num_streams},
+                    timeout=60.0)
+    if r.status_code != 200:
+        return False
+    ports = r.json()["ports"]
+    tx = TcpWeightSender(num_streams=num_streams)
+    bid = tx.submit(buf, host, ports)
+    if tx.wait(bid, timeout=timeout_s) != "done":
+        return False
+    r = client.post("/update_weights_from_tcp",
+                    json={"version": version, "timeout_s": timeout_s},
+                    timeout=timeout_s)
+    return r.status_code == 200 and r.json().get("success", False)

@@ -138,3 +138,43 @@ def test_abort_over_http(served):
             r = await c.post("/abort_request", json={"abort_all": True})
             assert r.json()["status"] == "ok"
     asyncio.run(go())
+
+
+def test_tcp_weight_push_into_served_engine(served):
+    """Full cross-node weight path: handshake -> N-stream TCP push ->
+    install under the step lock -> greedy output reflects the new weights
+    (sender/receiver agent data path)."""
+    import threading
+
+    import uvicorn
+
+    from polyrl_amd.transfer.tcp_engine import push_state_dict_tcp
+    cfg, model, eng, app = served
+
+    # run the facade on a real socket (TCP ports must be reachable)
+    server = uvicorn.Server(uvicorn.Config(app, host="127.0.0.1", port=31970,
+                                           log_level="error"))
+    t = threading.Thread(target=server.run, daemon=True)
+    t.start()
+    import time
+
+    import httpx
+    for _ in range(100):
+        try:
+            if httpx.get("http://127.0.0.1:31970/health",
+                         timeout=1.0).status_code == 200:
+                break
+        except Exception:
+            time.sleep(0.1)
+
+    sd = {k: v + 0.01 * torch.randn_like(v)
+          for k, v in model.state_dict().items()}
+    with httpx.Client(base_url="http://127.0.0.1:31970") as c:
+        ok = push_state_dict_tcp(sd, c, "127.0.0.1", num_streams=3,
+                                 version=9)
+        assert ok
+        info = c.get("/get_server_info").json()
+        assert info["weight_version"] == 9
+    name = "model.embed_tokens.weight"
+    assert torch.allclose(eng.model._name_map[name], sd[name])
+    server.should_exit = True
