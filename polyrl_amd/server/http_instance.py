@@ -18,17 +18,23 @@ from ..scheduler.types import GroupRequest, InstanceStats, SampleResult
 class HttpInstance(RolloutInstance):
     def __init__(self, base_url: str, instance_id: Optional[str] = None,
                  transport=None, timeout: float = 600.0,
-                 weight_file_fn=None):
+                 weight_file_fn=None, weight_state_fn=None,
+                 is_local: bool = False):
         import httpx
         self.base_url = base_url.rstrip("/")
         self.instance_id = instance_id or self.base_url
-        self.is_local = False
+        self.is_local = is_local
         self._client = httpx.AsyncClient(transport=transport,
                                          base_url=self.base_url,
                                          timeout=timeout)
         self._stats = InstanceStats()
-        # callable version -> safetensors file path (for the HTTP weight path)
+        # weight delivery (checked in order):
+        #  * weight_state_fn: version -> state dict, pushed over the
+        #    N-stream TCP plane (sender-agent path, tcp_engine.py)
+        #  * weight_file_fn: version -> node-local safetensors path
+        #  * neither: version ack only (bytes arrive out of band)
         self.weight_file_fn = weight_file_fn
+        self.weight_state_fn = weight_state_fn
         self.weight_version = 0
 
     async def generate_group(self, req: GroupRequest) -> List[SampleResult]:
@@ -104,6 +110,12 @@ class HttpInstance(RolloutInstance):
     async def update_weights(self, version: int, bootstrap: bool = False
                              ) -> bool:
         try:
+            if self.weight_state_fn is not None:
+                ok = await asyncio.get_running_loop().run_in_executor(
+                    None, self._push_tcp, version)
+                if ok:
+                    self.weight_version = version
+                return ok
             if self.weight_file_fn is not None:
                 path = self.weight_file_fn(version)
                 r = await self._client.post(
@@ -122,6 +134,18 @@ class HttpInstance(RolloutInstance):
         if ok:
             self.weight_version = version
         return ok
+
+    def _push_tcp(self, version: int) -> bool:
+        """Blocking sender-agent push over the TCP plane (run off-loop)."""
+        import urllib.parse
+
+        import httpx
+
+        from ..transfer.tcp_engine import push_state_dict_tcp
+        sd = self.weight_state_fn(version)
+        host = urllib.parse.urlparse(self.base_url).hostname or "127.0.0.1"
+        with httpx.Client(base_url=self.base_url) as c:
+            return push_state_dict_tcp(sd, c, host, version=version)
 
     def abort_all(self):
         # fire-and-forget from sync context
