@@ -124,6 +124,9 @@ class RolloutConfig:
     num_rollout_ranks: int = 0
     rollout_port_base: int = 30000
     max_local_gen_s: float = 0.0         # scheduler time-box (0 = off)
+    # elastic mode: manager HTTP facade port on rank 0 (0 = off) — remote
+    # instances join a running job here (reference manager port 5000)
+    rollout_manager_port: int = 0
     sampling: SamplingConfig = field(default_factory=SamplingConfig)
     calculate_log_probs: bool = True
     min_stream_batch_size: int = 16

@@ -16,9 +16,11 @@ from fastapi import FastAPI, Request
 
 
 def create_manager_app(scheduler, loop: Optional[asyncio.AbstractEventLoop]
-                       = None) -> "FastAPI":
+                       = None, remote_weight_state_fn=None) -> "FastAPI":
     """App over a RolloutScheduler.  If the scheduler's asyncio loop is a
-    different thread's loop, pass it so mutations run there."""
+    different thread's loop, pass it so mutations run there.  Newly joining
+    remote instances get ``remote_weight_state_fn`` as their TCP-plane
+    weight source (version -> state dict)."""
     app = FastAPI(title="polyrl-amd rollout manager")
     app.state.scheduler = scheduler
 
@@ -39,7 +41,7 @@ def create_manager_app(scheduler, loop: Optional[asyncio.AbstractEventLoop]
         body = await request.json()
         addr = body["addr"]
         from ..server import HttpInstance
-        inst = HttpInstance(addr)
+        inst = HttpInstance(addr, weight_state_fn=remote_weight_state_fn)
         await _run(scheduler.register_instance(inst))
         return {"status": "registered", "instance_id": inst.instance_id,
                 "weight_version": scheduler.latest_weight_version}
@@ -112,13 +114,14 @@ def create_manager_app(scheduler, loop: Optional[asyncio.AbstractEventLoop]
 
 
 def serve_manager(scheduler, host: str = "0.0.0.0", port: int = 5000,
-                  loop: Optional[asyncio.AbstractEventLoop] = None):
+                  loop: Optional[asyncio.AbstractEventLoop] = None,
+                  remote_weight_state_fn=None):
     """Run the facade in a daemon thread (the reference spawns the Rust
     manager on the head node, launcher.py:14-51); returns the server."""
     import threading
 
     import uvicorn
-    app = create_manager_app(scheduler, loop)
+    app = create_manager_app(scheduler, loop, remote_weight_state_fn)
     server = uvicorn.Server(uvicorn.Config(app, host=host, port=port,
                                            log_level="error"))
     t = threading.Thread(target=server.run, daemon=True)

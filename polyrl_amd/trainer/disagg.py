@@ -163,9 +163,16 @@ class DisaggCoordinator:
     the scheduler; every trainer rank receives equal ibatch slices."""
 
     def __init__(self, response_length: int, trainer_group,
-                 rollout_urls: List[str], rank: int, n_trainer: int,
+                 rollout_urls: List, rank: int, n_trainer: int,
                  pad_token_id: int = 0, device="cpu",
-                 max_local_gen_s: float = 0.0):
+                 max_local_gen_s: float = 0.0,
+                 manager_port: int = 0,
+                 remote_weight_state_fn=None):
+        """rollout_urls entries are either plain url strings (remote) or
+        (url, is_local) tuples.  ``manager_port`` > 0 serves the manager
+        HTTP facade (scheduler/http_api.py) on rank 0 so elastic remote
+        instances can join at runtime; new joins get
+        ``remote_weight_state_fn`` as their TCP weight source."""
         self.response_length = response_length
         self.group = trainer_group
         self.rank = rank
@@ -178,6 +185,7 @@ class DisaggCoordinator:
         self._gid = 0
         self.loop: Optional[asyncio.AbstractEventLoop] = None
         self.scheduler: Optional[RolloutScheduler] = None
+        self.manager_server = None
         if rank == 0:
             self.loop = asyncio.new_event_loop()
             t = threading.Thread(target=self.loop.run_forever, daemon=True)
@@ -187,12 +195,21 @@ class DisaggCoordinator:
 
             async def _register():
                 from ..server import HttpInstance
-                for url in rollout_urls:
-                    inst = HttpInstance(url)
-                    # remote engines start at version 0 == scheduler's
+                for spec in rollout_urls:
+                    url, is_local = spec if isinstance(spec, tuple) \
+                        else (spec, False)
+                    inst = HttpInstance(
+                        url, is_local=is_local,
+                        weight_state_fn=None if is_local
+                        else remote_weight_state_fn)
+                    # engines start at version 0 == scheduler's
                     await self.scheduler.register_instance(inst)
             asyncio.run_coroutine_threadsafe(_register(), self.loop) \
                 .result(timeout=600)
+            if manager_port > 0:
+                from ..scheduler.http_api import serve_manager
+                self.manager_server = serve_manager(
+                    self.scheduler, port=manager_port, loop=self.loop)
 
     # --------------------------------------------------------------- submit
     def submit(self, prompts: TensorBatch, sampling, n: int):
@@ -289,3 +306,59 @@ class DisaggCoordinator:
         return postprocess_groups(prompt_ids, prompt_mask, uids, outputs,
                                   self.response_length, self.pad,
                                   self.device)
+
+
+class ElasticPublisher:
+    """Publisher for the elastic co-located mode: every rank copies full
+    params straight into ITS engine (the xGMI-free fast path), rank 0 bumps
+    the scheduler version (local instances rejoin immediately — their bytes
+    are already current), and stale REMOTE instances receive an async TCP
+    push of a host-cached copy (the reference's async sender-agent
+    behavior, sender_agent.py:324-647)."""
+
+    def __init__(self, model, engine_model, coordinator, tie: bool,
+                 trainer_group=None):
+        from ..transfer.weight_transfer import WeightPublisher
+        self.inner = WeightPublisher(model, [engine_model],
+                                     tie_word_embeddings=tie)
+        self.coordinator = coordinator
+        self.group = trainer_group
+        self.cpu_cache: Dict[str, torch.Tensor] = {}
+        self.last_publish_s = 0.0
+
+    @property
+    def version(self) -> int:
+        return self.inner.version
+
+    @torch.no_grad()
+    def publish(self) -> int:
+        t0 = time.time()
+        sched = self.coordinator.scheduler
+        # all trainer ranks must agree on whether to host-cache (the gather
+        # loop is collective); only rank 0 knows the remote roster
+        need = [0]
+        if sched is not None:
+            need = [int(any(not i.is_local for i in sched.instances()))]
+        if dist.is_available() and dist.is_initialized():
+            dist.broadcast_object_list(need, src=0, group=self.group)
+        cache = self.cpu_cache if (sched is not None and need[0]) else None
+        if cache is not None:
+            cache.clear()
+        v = self.inner.publish(cpu_cache=cache)
+        if sched is not None:
+            loop = self.coordinator.loop
+
+            async def _bump_and_push():
+                await sched.update_weight_version(v)
+                for inst in sched.get_receive_instances():
+                    # TCP push + activation, async w.r.t. training
+                    await sched.finish_weight_update(inst.instance_id, v,
+                                                     success=True)
+            fut = asyncio.run_coroutine_threadsafe(_bump_and_push(), loop)
+            if not need[0]:
+                fut.result(timeout=60)   # no remotes: cheap, keep ordering
+        self.last_publish_s = time.time() - t0
+        return v
+
+    def shutdown(self):
+        pass

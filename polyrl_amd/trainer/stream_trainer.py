@@ -128,6 +128,9 @@ class StreamPPOTrainer:
         # ---------------- rollout plane -------------------------------------
         ro = arr.rollout
         self.disagg = ro.num_rollout_ranks > 0
+        self.elastic = ro.name == "elastic" and not self.disagg
+        # scheduler-coordinated generation (rank-0 pump + shard broadcast)
+        self.sched_coordinated = self.disagg or self.elastic
         if self.disagg:
             # split mode (BASELINE config #4): no local engine; the rollout
             # pool lives on the trailing ranks, reached via the scheduler
@@ -147,7 +150,8 @@ class StreamPPOTrainer:
                 self.coordinator.scheduler, self.coordinator.loop,
                 rank=dist.get_rank())
         else:
-            eng_dtype = getattr(torch, ro.dtype)
+            eng_dtype = getattr(torch, ro.dtype) if device != "cpu" \
+                else torch.float32
             kv_budget = self._kv_budget(ro)
             self.engine = Engine(model_cfg, device=device, dtype=eng_dtype,
                                  page_size=ro.page_size,
@@ -157,11 +161,15 @@ class StreamPPOTrainer:
                                  max_model_len=ro.prompt_length + ro.response_length,
                                  decode_chunk_size=ro.decode_chunk_size,
                                  seed=config.trainer.seed * 1000 + self.rank)
-            self.coordinator = LocalRolloutCoordinator(
-                self.engine, ro.response_length, pad_token_id=0, device="cpu")
-            self.publisher = WeightPublisher(
-                self.actor.model, [self.engine.model],
-                tie_word_embeddings=model_cfg.tie_word_embeddings)
+            if self.elastic:
+                self._setup_elastic(ro, model_cfg, device)
+            else:
+                self.coordinator = LocalRolloutCoordinator(
+                    self.engine, ro.response_length, pad_token_id=0,
+                    device="cpu")
+                self.publisher = WeightPublisher(
+                    self.actor.model, [self.engine.model],
+                    tie_word_embeddings=model_cfg.tie_word_embeddings)
 
         # ---------------- data + reward -------------------------------------
         dcfg = config.data
@@ -187,6 +195,42 @@ class StreamPPOTrainer:
         self._maybe_resume()
 
     # ------------------------------------------------------------------ setup
+    def _setup_elastic(self, ro, model_cfg, device):
+        """Elastic co-located mode (the reference's primary shape, §3.4):
+        every rank serves its engine over HTTP; rank 0's scheduler drives
+        the whole pool as LOCAL instances (time-boxed when remotes exist);
+        elastic remote instances join at runtime through the manager facade
+        and receive weights over the TCP plane."""
+        import threading
+
+        import uvicorn
+
+        from ..rollout.runner import EngineRunner
+        from ..server import create_app
+        from .disagg import DisaggCoordinator, ElasticPublisher, rollout_port
+
+        grank = dist.get_rank() if dist.is_initialized() else 0
+        gworld = dist.get_world_size() if dist.is_initialized() else 1
+        self._engine_runner = EngineRunner(self.engine)
+        app = create_app(self.engine, self._engine_runner)
+        port = rollout_port(grank, ro.rollout_port_base)
+        self._engine_server = uvicorn.Server(uvicorn.Config(
+            app, host="127.0.0.1", port=port, log_level="error"))
+        threading.Thread(target=self._engine_server.run, daemon=True).start()
+        if dist.is_initialized():
+            dist.barrier(group=self.pg)
+        specs = [(f"http://127.0.0.1:{rollout_port(r, ro.rollout_port_base)}",
+                  True) for r in range(gworld)]
+        self.coordinator = DisaggCoordinator(
+            ro.response_length, self.pg, specs, rank=self.rank,
+            n_trainer=self.world, pad_token_id=0, device="cpu",
+            max_local_gen_s=ro.max_local_gen_s,
+            manager_port=ro.rollout_manager_port if self.rank == 0 else 0,
+            remote_weight_state_fn=lambda v: self.publisher.cpu_cache)
+        self.publisher = ElasticPublisher(
+            self.actor.model, self.engine.model, self.coordinator,
+            tie=model_cfg.tie_word_embeddings, trainer_group=self.pg)
+
     def _kv_budget(self, ro) -> int:
         if self.device.startswith("cuda"):
             from ..rollout.kv_cache import PagedKVCache
@@ -297,7 +341,7 @@ class StreamPPOTrainer:
         c = self.config
         ro = c.actor_rollout_ref.rollout
         nval = num_prompts or max(c.data.train_batch_size // self.world, 1)
-        if self.disagg:
+        if self.sched_coordinated:
             nval = max(nval - nval % self.world, self.world)
         idx = list(range(min(nval, len(self.dataset))))
         batch = self.dataset.batch(idx)
@@ -305,7 +349,8 @@ class StreamPPOTrainer:
                                   max_new_tokens=ro.response_length)
         self.publisher.publish()
         self.coordinator.submit(batch, sampling, 1)
-        stream = len(idx) // self.world if self.disagg else len(idx)
+        stream = (len(idx) // self.world if self.sched_coordinated
+                  else len(idx))
         groups = []
         for b in self.coordinator.stream_batches(stream):
             groups.append(b)
@@ -346,7 +391,7 @@ class StreamPPOTrainer:
             temperature=ro.sampling.temperature,
             top_k=ro.sampling.top_k, top_p=ro.sampling.top_p,
             max_new_tokens=ro.response_length)
-        if self.disagg:
+        if self.sched_coordinated:
             submit_batch = global_batch
         else:
             submit_batch = global_batch.slice(

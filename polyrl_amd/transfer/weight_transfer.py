@@ -38,7 +38,8 @@ class WeightPublisher:
         self.consumers.append(c)
 
     @torch.no_grad()
-    def publish(self) -> int:
+    def publish(self, cpu_cache: Optional[Dict[str, torch.Tensor]] = None
+                ) -> int:
         """All-gather each FSDP shard to a full tensor and copy into every
         consumer.  Per-parameter streaming keeps peak memory at one full
         param (cf. fsdp_interface.py:186-207 which materializes the whole
@@ -57,6 +58,9 @@ class WeightPublisher:
                 full = param
             for c in self.consumers:
                 c.update_named(name, full)
+            if cpu_cache is not None:
+                # keep a host copy for async TCP pushes to elastic remotes
+                cpu_cache[name] = full.detach().to("cpu", copy=True)
         if self.tie:
             for c in self.consumers:
                 emb = sd.get("model.embed_tokens.weight")
