@@ -10,7 +10,7 @@ fwd/bwd, not these.
 from __future__ import annotations
 
 from collections import defaultdict
-from typing import Dict, Optional, Tuple
+from typing import Optional, Tuple
 
 import torch
 

@@ -6,7 +6,7 @@ the verl-style sharded checkpoint layout.
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional
 
 

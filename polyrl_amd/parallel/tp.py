@@ -15,8 +15,6 @@ broadcast.
 """
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 import torch.distributed as dist
 

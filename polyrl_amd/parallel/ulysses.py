@@ -14,8 +14,6 @@ backward of a sequence gather is a slice.
 """
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 import torch.distributed as dist
 

@@ -13,7 +13,7 @@ from __future__ import annotations
 import asyncio
 import threading
 import time
-from typing import Callable, Dict, List, Optional
+from typing import Dict, List, Optional
 
 from .engine import Engine, RequestOutput, SamplingParams
 

@@ -14,9 +14,7 @@ contract, so co-located engines take no network hop.  Implementations:
 from __future__ import annotations
 
 import asyncio
-import threading
-import time
-from typing import Dict, List, Optional
+from typing import List
 
 from .types import GroupRequest, InstanceStats, SampleResult
 

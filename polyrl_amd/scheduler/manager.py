@@ -26,11 +26,11 @@ from __future__ import annotations
 import asyncio
 import threading
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import AsyncIterator, Dict, List, Optional
 
 from .balance import LoadBalanceState
-from .instances import FakeInstance, RolloutInstance
+from .instances import RolloutInstance
 from .types import (GroupRequest, GroupResult, MetricsUpdate, SampleResult,
                     continuation_request, merge_sample)
 

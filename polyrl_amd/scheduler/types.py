@@ -8,7 +8,7 @@ utils.rs:256-291 adjust_sampling_params_for_used_tokens).
 from __future__ import annotations
 
 from dataclasses import dataclass, field, replace
-from typing import Dict, List, Optional
+from typing import List
 
 
 @dataclass

@@ -12,8 +12,7 @@ interface.
 """
 from __future__ import annotations
 
-from collections import defaultdict
-from typing import Dict, Iterator, List, Optional
+from typing import Dict, Iterator, List
 
 import numpy as np
 import torch

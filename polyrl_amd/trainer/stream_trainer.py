@@ -18,11 +18,8 @@ Preserved semantics:
 """
 from __future__ import annotations
 
-import os
-import time
 from typing import Dict, List, Optional
 
-import numpy as np
 import torch
 import torch.distributed as dist
 

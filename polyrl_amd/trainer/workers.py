@@ -17,7 +17,6 @@ Streaming semantics preserved from the reference:
 from __future__ import annotations
 
 import math
-import os
 from typing import Dict, List, Optional, Tuple
 
 import torch

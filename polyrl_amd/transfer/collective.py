@@ -16,7 +16,7 @@ weight-transfer round-trip, trainer buffer -> instance tensors bitwise).
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import Callable, Dict, Iterable, List, Optional, Tuple
+from typing import Callable, Dict, List, Optional, Tuple
 
 import torch
 import torch.distributed as dist

@@ -11,7 +11,7 @@ import contextlib
 import functools
 import logging
 import os
-from typing import Iterable, Optional
+from typing import Optional
 
 import torch
 
