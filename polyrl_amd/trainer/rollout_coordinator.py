@@ -67,14 +67,22 @@ def postprocess_groups(
 
 
 class LocalRolloutCoordinator:
-    """One in-process engine per trainer rank (co-located mode)."""
+    """One in-process engine per trainer rank (co-located mode).
+
+    ``shard=(rank, world)`` enables the TP-rollout mode (BASELINE config
+    #5): every rank submits the SAME global batch to its shard of the ONE
+    tensor-parallel engine (SPMD lockstep — the engine's collectives align
+    because scheduling is deterministic and identical), and stream_batches
+    yields this rank's 1/world slice of each finished round for training."""
 
     def __init__(self, engine: Engine, response_length: int,
-                 pad_token_id: int = 0, device="cpu"):
+                 pad_token_id: int = 0, device="cpu",
+                 shard=None):
         self.engine = engine
         self.response_length = response_length
         self.pad_token_id = pad_token_id
         self.device = device
+        self.shard = shard
         self._groups: Dict[str, dict] = {}
         self._submit_counter = 0
 
@@ -131,7 +139,12 @@ class LocalRolloutCoordinator:
         requires stream_size % n == 0) until all submitted groups are
         consumed; the final batch carries any remainder.  Exact sizing keeps
         every DP rank's ibatch count identical — a requirement for the SPMD
-        FSDP collectives (each ibatch triggers collective fwd/bwd)."""
+        FSDP collectives (each ibatch triggers collective fwd/bwd).
+
+        With ``shard=(r, w)``: collect w*stream_size samples per round and
+        yield rank r's group-slice (deterministic, identical on all ranks —
+        no communication)."""
+        round_size = stream_size * (self.shard[1] if self.shard else 1)
         ready: List[dict] = []
         ready_samples = 0
         while self._groups or ready:
@@ -139,18 +152,37 @@ class LocalRolloutCoordinator:
                 for grp in self.poll():
                     ready.append(grp)
                     ready_samples += grp["n"]
-            while ready_samples >= stream_size and ready:
+            while ready_samples >= round_size and ready:
                 take, taken = [], 0
-                while ready and taken < stream_size:
+                while ready and taken < round_size:
                     g = ready.pop(0)
                     take.append(g)
                     taken += g["n"]
                 ready_samples -= taken
-                yield self._make_batch(take)
+                yield self._emit(take)
             if not self._groups and ready:  # tail (partial final batch)
-                yield self._make_batch(ready)
+                yield self._emit(ready)
                 ready = []
                 ready_samples = 0
+
+    def _emit(self, groups: List[dict]) -> TensorBatch:
+        if not self.shard:
+            return self._make_batch(groups)
+        r, w = self.shard
+        total = sum(g["n"] for g in groups)
+        assert total % w == 0, f"round of {total} samples !% {w} shards"
+        per = total // w
+        start = 0
+        for k in range(w):
+            take, taken = [], 0
+            while start < len(groups) and taken < per:
+                take.append(groups[start])
+                taken += groups[start]["n"]
+                start += 1
+            assert taken == per, "groups must tile shards evenly"
+            if k == r:
+                return self._make_batch(take)
+        raise AssertionError("shard index out of range")
 
     def _make_batch(self, groups: List[dict]) -> TensorBatch:
         prompt_ids = torch.stack([g["prompt_ids"] for g in groups])

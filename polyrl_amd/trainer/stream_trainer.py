@@ -126,6 +126,10 @@ class StreamPPOTrainer:
         ro = arr.rollout
         self.disagg = ro.num_rollout_ranks > 0
         self.elastic = ro.name == "elastic" and not self.disagg
+        # one tensor-parallel engine across the trainer ranks (config #5:
+        # FSDP trainer + TP rollout, on-device FSDP->TP reshard)
+        self.tp_rollout = (not self.disagg and not self.elastic
+                           and ro.tensor_model_parallel_size > 1)
         # scheduler-coordinated generation (rank-0 pump + shard broadcast)
         self.sched_coordinated = self.disagg or self.elastic
         if self.disagg:
@@ -150,6 +154,17 @@ class StreamPPOTrainer:
             eng_dtype = getattr(torch, ro.dtype) if device != "cpu" \
                 else torch.float32
             kv_budget = self._kv_budget(ro)
+            tp_ctx = None
+            eng_seed = config.trainer.seed * 1000 + self.rank
+            if ro.tensor_model_parallel_size > 1 and not self.elastic:
+                from ..parallel.tp import TPContext
+                tp = ro.tensor_model_parallel_size
+                assert dist.is_initialized() and \
+                    dist.get_world_size() == tp, \
+                    f"TP rollout currently requires tp == world ({tp})"
+                tp_ctx = TPContext(dist.group.WORLD)
+                eng_seed = config.trainer.seed * 1000   # rank-uniform: the
+                # TP engine's sampling must agree bitwise across ranks
             self.engine = Engine(model_cfg, device=device, dtype=eng_dtype,
                                  page_size=ro.page_size,
                                  kv_bytes_budget=kv_budget,
@@ -157,13 +172,17 @@ class StreamPPOTrainer:
                                  max_num_batched_tokens=ro.max_num_batched_tokens,
                                  max_model_len=ro.prompt_length + ro.response_length,
                                  decode_chunk_size=ro.decode_chunk_size,
-                                 seed=config.trainer.seed * 1000 + self.rank)
+                                 tp_ctx=tp_ctx,
+                                 seed=eng_seed)
             if self.elastic:
                 self._setup_elastic(ro, model_cfg, device)
             else:
+                shard = None
+                if self.tp_rollout:
+                    shard = (self.rank, self.world)
                 self.coordinator = LocalRolloutCoordinator(
                     self.engine, ro.response_length, pad_token_id=0,
-                    device="cpu")
+                    device="cpu", shard=shard)
                 self.publisher = WeightPublisher(
                     self.actor.model, [self.engine.model],
                     tie_word_embeddings=model_cfg.tie_word_embeddings)
@@ -338,7 +357,7 @@ class StreamPPOTrainer:
         c = self.config
         ro = c.actor_rollout_ref.rollout
         nval = num_prompts or max(c.data.train_batch_size // self.world, 1)
-        if self.sched_coordinated:
+        if self.sched_coordinated or self.tp_rollout:
             nval = max(nval - nval % self.world, self.world)
         idx = list(range(min(nval, len(self.dataset))))
         batch = self.dataset.batch(idx)
@@ -346,7 +365,8 @@ class StreamPPOTrainer:
                                   max_new_tokens=ro.response_length)
         self.publisher.publish()
         self.coordinator.submit(batch, sampling, 1)
-        stream = (len(idx) // self.world if self.sched_coordinated
+        stream = (len(idx) // self.world
+                  if (self.sched_coordinated or self.tp_rollout)
                   else len(idx))
         groups = []
         for b in self.coordinator.stream_batches(stream):
@@ -388,8 +408,8 @@ class StreamPPOTrainer:
             temperature=ro.sampling.temperature,
             top_k=ro.sampling.top_k, top_p=ro.sampling.top_p,
             max_new_tokens=ro.response_length)
-        if self.sched_coordinated:
-            submit_batch = global_batch
+        if self.sched_coordinated or self.tp_rollout:
+            submit_batch = global_batch     # identical on every rank
         else:
             submit_batch = global_batch.slice(
                 slice(self.rank * local_bs, (self.rank + 1) * local_bs))

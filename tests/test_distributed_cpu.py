@@ -102,3 +102,31 @@ def test_bench_contract_world2():
     assert d["n_gpus"] == 2
     assert d["config"]["parallelism"] == "dp2"
     assert d["value"] > 0
+
+
+@pytest.mark.timeout(600)
+def test_tp2_rollout_fsdp_trainer_world2(tmp_path):
+    """BASELINE config #5 shape on the CPU tier: FSDP trainer over 2 ranks
+    + ONE TP=2 rollout engine spanning them, with on-device FSDP->TP
+    resharding at every weight publication (full_tensor gather -> per-rank
+    TP slice into the sharded engine buffers)."""
+    r = _run_torchrun([
+        "actor_rollout_ref.model.path=llama-debug-cpu",
+        "actor_rollout_ref.model.dtype=float32",
+        "actor_rollout_ref.model.enable_gradient_checkpointing=false",
+        "actor_rollout_ref.actor.ppo_mini_batch_size=8",
+        "actor_rollout_ref.actor.ppo_max_token_len_per_gpu=512",
+        "actor_rollout_ref.rollout.sampling.n=2",
+        "actor_rollout_ref.rollout.response_length=8",
+        "actor_rollout_ref.rollout.min_stream_batch_size=4",
+        "actor_rollout_ref.rollout.tensor_model_parallel_size=2",
+        "data.train_batch_size=8",
+        "data.max_prompt_length=16",
+        "data.synthetic_num_prompts=32",
+        f"trainer.default_local_dir={tmp_path}/ckpt",
+        "trainer.resume_mode=disable",
+        "reward=random",
+        "max_steps=2",
+    ])
+    assert r.returncode == 0, \
+        f"stdout:\n{r.stdout[-3000:]}\nstderr:\n{r.stderr[-3000:]}"
