@@ -227,3 +227,62 @@ def test_scheduler_metrics_feedback():
         assert "new_max_gen_s" in out and out["num_instances"] == 1
         await sched.close()
     run(go())
+
+
+def test_http_stream_failure_continuation():
+    """HTTP-layer fault injection: an instance whose SSE stream dies
+    mid-generation surfaces partials; the scheduler evicts it and the
+    sample continues token-exactly on a healthy instance
+    (handlers.rs:152-328 partial capture + :363-414 continuation)."""
+    import httpx
+
+    class CutTransport(httpx.AsyncBaseTransport):
+        """ASGI transport that breaks every /generate response stream."""
+
+        def __init__(self, inner):
+            self.inner = inner
+
+        async def handle_async_request(self, request):
+            resp = await self.inner.handle_async_request(request)
+            if request.url.path == "/generate":
+                class CutStream(httpx.AsyncByteStream):
+                    async def __aiter__(self):
+                        raise httpx.ReadError("injected stream cut")
+                        yield b""  # pragma: no cover (makes it a generator)
+                return httpx.Response(resp.status_code,
+                                      headers=resp.headers,
+                                      stream=CutStream(),
+                                      request=request)
+            return resp
+
+    async def go():
+        import torch
+
+        from polyrl_amd.models import create_model, get_model_config
+        from polyrl_amd.rollout.engine import Engine
+        from polyrl_amd.server import HttpInstance, create_app
+        cfg = get_model_config("llama-debug-cpu")
+        torch.manual_seed(3)
+        model = create_model(cfg, kind="actor", dtype="float32", device="cpu")
+        eng = Engine(cfg, device="cpu", dtype=torch.float32,
+                     kv_bytes_budget=8 << 20)
+        eng.model.load_state_dict(model.state_dict())
+        app = create_app(eng)
+        flaky = HttpInstance(
+            "http://flaky", transport=CutTransport(httpx.ASGITransport(app=app)))
+        good = FakeInstance("good", is_local=False)
+        sched = RolloutScheduler(SchedulerConfig(stats_interval_s=0.02))
+        await sched.register_instance(flaky, skip_health_check=True)
+        await sched.register_instance(good, skip_health_check=True)
+        sched._states["good"].assigned_batches = 10**9  # flaky serves first
+        res = await sched.process_group(mk_req(n=1, max_new=6,
+                                               prompt=[4, 5]))
+        await sched.close()
+        s = res.samples[0]
+        assert len(s.output_ids) == 6          # completed despite the cut
+        assert s.finish_reason == "length"
+        assert s.num_migrations == 1
+        assert "good" in res.instance_ids      # continuation landed there
+        assert "http://flaky" not in [i.instance_id
+                                      for i in sched.instances()]  # evicted
+    asyncio.run(go())
