@@ -46,6 +46,9 @@ class Request:
     finished: bool = False
     finish_reason: str = ""       # stop | length | abort
     arrival_t: float = 0.0
+    # prefix sharing: a group parent prefills once for n samples; children
+    # are forked at first-token sampling (radix-cache capability)
+    group_children: Optional[List[str]] = None
 
     @property
     def seq_len(self) -> int:
@@ -359,6 +362,7 @@ class Engine:
         self._scale = 1.0 / math.sqrt(cfg.head_dim)
         self._gen = torch.Generator().manual_seed(seed)  # CPU ref path RNG
         self.decode_chunk_size = max(decode_chunk_size, 1)
+        self.enable_prefix_sharing = True
         # hipGraph-captured decode iteration (one replay per token):
         # removes ~300 kernel-launch round-trips per decode step.  Disabled
         # under TP (capturing RCCL collectives in a graph is unvalidated).
@@ -376,6 +380,25 @@ class Engine:
                       arrival_t=time.time())
         self.waiting.append(req)
         return rid
+
+    def add_request_group(self, rid_prefix: str, input_ids: List[int],
+                          sampling: SamplingParams, n: int):
+        """n samples of one prompt with a SHARED prompt prefill + shared
+        full KV pages (SURVEY.md §2.4.3 radix/prefix-cache capability): one
+        request prefills; at its first sampling step the engine draws n
+        first tokens and forks n-1 children whose page tables reference the
+        parent's full prompt pages (refcounted; a partial trailing page is
+        copied per child so decode writes never touch shared pages).
+        Child rids are '<prefix>-s0' .. '<prefix>-s<n-1>'."""
+        assert n >= 1
+        if n == 1 or not self.enable_prefix_sharing:
+            for s_ in range(n):
+                self.add_request(f"{rid_prefix}-s{s_}", input_ids, sampling)
+            return
+        parent = Request(rid=f"{rid_prefix}-s0", input_ids=list(input_ids),
+                         sampling=sampling, arrival_t=time.time())
+        parent.group_children = [f"{rid_prefix}-s{s_}" for s_ in range(1, n)]
+        self.waiting.append(parent)
 
     def abort_request(self, rid: Optional[str] = None, abort_all: bool = False):
         """Mark requests aborted; they are emitted with partial output on the
@@ -405,12 +428,12 @@ class Engine:
         still = []
         for r in self.running:
             if r.finished:
-                finished.append(self._emit(r))
+                finished.extend(self._emit_all(r))
             else:
                 still.append(r)
         self.running = still
         self.waiting = [r for r in self.waiting if not r.finished or
-                        finished.append(self._emit(r))]  # emit aborted waiters
+                        finished.extend(self._emit_all(r))]  # aborted waiters
 
         # ---- admission + chunked prefill -----------------------------------
         token_budget = self.max_batched_tokens
@@ -460,7 +483,7 @@ class Engine:
                     # out of KV pages: abort (scheduler continues elsewhere)
                     r.finished = True
                     r.finish_reason = "abort"
-                    finished.append(self._emit(r))
+                    finished.extend(self._emit_all(r))
             if ok:
                 self._run_decode(ok)
 
@@ -468,7 +491,7 @@ class Engine:
         still = []
         for r in self.running:
             if r.finished:
-                finished.append(self._emit(r))
+                finished.extend(self._emit_all(r))
             else:
                 still.append(r)
         self.running = still
@@ -496,6 +519,18 @@ class Engine:
         return RequestOutput(rid=r.rid, output_ids=list(r.output_ids),
                              output_logprobs=list(r.output_logprobs),
                              finish_reason=r.finish_reason)
+
+    def _emit_all(self, r: Request) -> List[RequestOutput]:
+        """Emit r; a group parent finishing BEFORE its fork (abort / OOM)
+        also emits one aborted output per unforked child rid."""
+        outs = [self._emit(r)]
+        if r.group_children:
+            for crid in r.group_children:
+                outs.append(RequestOutput(rid=crid, output_ids=[],
+                                          output_logprobs=[],
+                                          finish_reason=outs[0].finish_reason))
+            r.group_children = None
+        return outs
 
     def _sample_last(self, reqs: List[Request], hidden_rows: torch.Tensor):
         """Sample the next token for each req from its last hidden row."""
@@ -567,16 +602,66 @@ class Engine:
 
         hidden = self.model.forward_tokens(tokens, positions, self.kv, slots,
                                            attn_fn)
-        # requests whose prefill completes sample their first token
+        # requests whose prefill completes sample their first token;
+        # group parents fork their children first (shared prompt pages)
         done_rows, done_reqs = [], []
+        fork_list = []
         for i, (r, take) in enumerate(zip(reqs, lens)):
             r.prefill_pos += take
             if r.prefill_pos >= len(r.input_ids):
-                done_rows.append(cu_q[i + 1] - 1)
-                done_reqs.append(r)
+                if r.group_children:
+                    fork_list.append((r, cu_q[i + 1] - 1))
+                else:
+                    done_rows.append(cu_q[i + 1] - 1)
+                    done_reqs.append(r)
         if done_reqs:
             rows = hidden[torch.tensor(done_rows, dtype=torch.long, device=dev)]
             self._sample_last(done_reqs, rows)
+        for r, row in fork_list:
+            self._fork_group(r, hidden[row])
+
+    def _fork_group(self, parent: Request, last_hidden: torch.Tensor):
+        """Fork n-1 children off a freshly prefilled group parent: shared
+        full prompt pages (refcounted), per-child copy of the partial
+        trailing page, then ONE logits row sampled n times (per-row RNG
+        keys give independent draws)."""
+        child_rids = parent.group_children
+        n = len(child_rids) + 1
+        Lp = len(parent.input_ids)
+        children = [Request(rid=crid, input_ids=parent.input_ids,
+                            sampling=parent.sampling, prefill_pos=Lp,
+                            arrival_t=parent.arrival_t)
+                    for crid in child_rids]
+        child_sids = []
+        for c in children:
+            sid = self._seq_counter
+            self._seq_counter += 1
+            self._seq_ids[c.rid] = sid
+            child_sids.append(sid)
+        copies = self.kv.fork_seq(self._seq_ids[parent.rid], child_sids, Lp)
+        if copies is None:
+            # out of pages for the partial-page copies: children re-prefill
+            # independently (correct, just unshared)
+            for c, sid in zip(children, child_sids):
+                self._seq_ids.pop(c.rid, None)
+                c.prefill_pos = 0
+                self.waiting.append(c)
+            parent.group_children = None
+            self._sample_last([parent], last_hidden.unsqueeze(0))
+            return
+        if copies:
+            src = torch.tensor([a for a, _ in copies], dtype=torch.long,
+                               device=self.device)
+            dst = torch.tensor([b for _, b in copies], dtype=torch.long,
+                               device=self.device)
+            rem = Lp % self.kv.page_size
+            for li in range(self.cfg.num_hidden_layers):
+                self.kv.k_cache[li][dst, :rem] = self.kv.k_cache[li][src, :rem]
+                self.kv.v_cache[li][dst, :rem] = self.kv.v_cache[li][src, :rem]
+        parent.group_children = None
+        rows = last_hidden.unsqueeze(0).expand(n, -1).contiguous()
+        self.running.extend(children)
+        self._sample_last([parent] + children, rows)
 
     def _gather_kv(self, li: int, seq_ids: List[int], ctx_lens: torch.Tensor):
         """Materialize contiguous K/V for chunked-prefill history (per layer)."""

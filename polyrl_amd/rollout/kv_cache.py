@@ -8,7 +8,7 @@ split between weights / KV / activations — SURVEY.md §5.9).
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Dict, List, Optional, Tuple
 
 import torch
 
@@ -36,6 +36,9 @@ class PagedKVCache:
         self._free: List[int] = list(range(num_pages - 1, -1, -1))
         self._seq_pages: Dict[int, List[int]] = {}
         self._seq_len: Dict[int, int] = {}
+        # page refcounts for prefix sharing (radix-cache capability,
+        # SURVEY.md §2.4.3): n samples of one prompt share its full pages
+        self._ref: Dict[int, int] = {}
 
     # ------------------------------------------------------------- accounting
     @classmethod
@@ -64,14 +67,50 @@ class PagedKVCache:
         if grow > len(self._free):
             return False
         for _ in range(grow):
-            pages.append(self._free.pop())
+            p = self._free.pop()
+            self._ref[p] = 1
+            pages.append(p)
         self._seq_len[seq_id] = cur_len + num_tokens
         return True
 
     def free_seq(self, seq_id: int):
         for p in self._seq_pages.pop(seq_id, []):
-            self._free.append(p)
+            self._ref[p] = self._ref.get(p, 1) - 1
+            if self._ref[p] <= 0:
+                self._ref.pop(p, None)
+                self._free.append(p)
         self._seq_len.pop(seq_id, None)
+
+    def fork_seq(self, parent_id: int, child_ids: List[int],
+                 prefix_len: int) -> Optional[List[Tuple[int, int]]]:
+        """Prefix sharing: every child references the parent's FULL pages
+        of the first ``prefix_len`` tokens; a partial trailing page is given
+        to each child as a fresh page (the caller copies its KV content —
+        decode writes must never touch shared pages).  The parent keeps its
+        own pages (it becomes sample 0).
+
+        Returns [(src_page, dst_page), ...] partial-page copies the caller
+        must perform (empty when prefix_len %% page_size == 0), or None if
+        pages ran out (nothing allocated)."""
+        parent_pages = self._seq_pages[parent_id]
+        full = prefix_len // self.page_size
+        rem = prefix_len % self.page_size
+        need_fresh = len(child_ids) * (1 if rem else 0)
+        if need_fresh > len(self._free):
+            return None
+        copies: List[Tuple[int, int]] = []
+        for cid in child_ids:
+            pages = list(parent_pages[:full])
+            for p in pages:
+                self._ref[p] = self._ref.get(p, 1) + 1
+            if rem:
+                fresh = self._free.pop()
+                self._ref[fresh] = 1
+                copies.append((parent_pages[full], fresh))
+                pages.append(fresh)
+            self._seq_pages[cid] = pages
+            self._seq_len[cid] = prefix_len
+        return copies
 
     def seq_len(self, seq_id: int) -> int:
         return self._seq_len.get(seq_id, 0)

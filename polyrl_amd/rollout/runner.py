@@ -93,10 +93,30 @@ class EngineRunner:
         self._wake.set()
         return fut
 
+    def submit_group(self, input_ids: List[int], sp: SamplingParams,
+                     n: int) -> List["asyncio.Future[RequestOutput]"]:
+        """n samples of one prompt with a shared prompt prefill (prefix
+        sharing); returns one future per sample."""
+        if n == 1 or not getattr(self.engine, "enable_prefix_sharing", False):
+            return [self.submit(input_ids, sp) for _ in range(n)]
+        loop = asyncio.get_running_loop()
+        self._rid_counter += 1
+        prefix = f"g{self._rid_counter}"
+        futs = []
+        for s_ in range(n):
+            fut = loop.create_future()
+            self._futures[f"{prefix}-s{s_}"] = (loop, fut)
+            futs.append(fut)
+        with self.lock:
+            self.engine.add_request_group(prefix, input_ids, sp, n)
+        self.start()
+        self._wake.set()
+        return futs
+
     async def generate(self, input_ids: List[int], sp: SamplingParams,
                        n: int = 1) -> List[RequestOutput]:
-        futs = [self.submit(input_ids, sp) for _ in range(n)]
-        return list(await asyncio.gather(*futs))
+        return list(await asyncio.gather(*self.submit_group(input_ids, sp,
+                                                            n)))
 
     # ---------------------------------------------------------------- admin
     def abort(self, rid: Optional[str] = None, abort_all: bool = False):

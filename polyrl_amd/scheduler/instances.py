@@ -75,7 +75,7 @@ class InProcessInstance(RolloutInstance):
             top_p=req.sampling.top_p,
             max_new_tokens=req.sampling.max_new_tokens,
             stop_token_ids=tuple(req.sampling.stop_token_ids))
-        outs = await self.runner.generate(req.input_ids, sp, n=req.n)
+        outs = await self.runner.generate(req.input_ids, sp, n=req.n)  # group-shared prefill
         return [SampleResult(output_ids=list(o.output_ids),
                              output_logprobs=list(o.output_logprobs),
                              finish_reason=o.finish_reason,

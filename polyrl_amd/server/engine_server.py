@@ -53,7 +53,7 @@ def create_app(engine: Engine, runner: Optional[EngineRunner] = None):
         want_lp = bool(body.get("return_logprob", False))
         stream = bool(body.get("stream", False))
 
-        futs = [runner.submit(input_ids, sp) for _ in range(n)]
+        futs = runner.submit_group(input_ids, sp, n)
 
         if not stream:
             outs = await asyncio.gather(*futs)

@@ -107,8 +107,8 @@ class LocalRolloutCoordinator:
                 "done": 0,
                 "n": n,
             }
-            for s in range(n):
-                self.engine.add_request(f"{gid}-s{s}", raw, sampling)
+            # shared prompt prefill + shared full KV pages for the group
+            self.engine.add_request_group(gid, raw, sampling, n)
 
     # --------------------------------------------------------------- stream
     def pending_groups(self) -> int:

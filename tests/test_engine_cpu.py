@@ -150,3 +150,65 @@ def test_many_concurrent_requests(setup):
     assert len(outs) == 20
     assert all(len(o.output_ids) == 3 for o in outs)
     assert eng.kv.free_pages == eng.kv.num_pages
+
+
+def test_group_prefill_sharing_matches_separate(setup):
+    """Group API (shared prompt prefill + shared KV pages) must produce
+    the same GREEDY outputs as n separate requests, use fewer pages while
+    running, and return every page at the end."""
+    cfg, model, eng = setup
+    torch.manual_seed(9)
+    prompt = torch.randint(0, cfg.vocab_size, (13,)).tolist()  # 13 % 16 != 0
+    sp = SamplingParams(temperature=0.0, max_new_tokens=5)
+
+    # separate requests (no sharing)
+    eng.enable_prefix_sharing = False
+    outs_sep = eng.generate([prompt] * 3, sp, "sep")
+    eng.enable_prefix_sharing = True
+
+    # group API
+    eng.add_request_group("grp", prompt, sp, 3)
+    outs = {}
+    min_free = eng.kv.num_pages
+    while eng.has_work():
+        for o in eng.step():
+            outs[o.rid] = o
+        min_free = min(min_free, eng.kv.free_pages)
+    assert len(outs) == 3
+    for s in range(3):
+        assert outs[f"grp-s{s}"].output_ids == outs_sep[s].output_ids, s
+    assert eng.kv.free_pages == eng.kv.num_pages   # refcounts all released
+    # shared full prompt pages were not duplicated per child
+    sep_pages_at_peak = 3 * ((13 + 5 + 15) // 16 + 1)
+    assert min_free >= eng.kv.num_pages - sep_pages_at_peak
+
+
+def test_group_sampled_outputs_differ(setup):
+    """Stochastic group samples draw independently (per-row RNG keys)."""
+    cfg, model, eng = setup
+    torch.manual_seed(10)
+    prompt = torch.randint(0, cfg.vocab_size, (8,)).tolist()
+    eng.add_request_group("st", prompt, SamplingParams(temperature=1.0,
+                                                       max_new_tokens=6), 4)
+    outs = {}
+    while eng.has_work():
+        for o in eng.step():
+            outs[o.rid] = o
+    seqs = [tuple(outs[f"st-s{s}"].output_ids) for s in range(4)]
+    assert len(outs) == 4
+    assert len(set(seqs)) > 1, seqs    # not all identical
+
+
+def test_group_abort_before_fork_emits_all(setup):
+    """A group parent aborted before its fork must emit one aborted output
+    per child rid (the coordinator expects n results per group)."""
+    cfg, model, eng = setup
+    prompt = [3, 4, 5]
+    eng.add_request_group("ab", prompt, SamplingParams(max_new_tokens=50), 3)
+    eng.abort_request(abort_all=True)
+    outs = {}
+    while eng.has_work():
+        for o in eng.step():
+            outs[o.rid] = o
+    assert set(outs) == {"ab-s0", "ab-s1", "ab-s2"}
+    assert all(o.finish_reason == "abort" for o in outs.values())
