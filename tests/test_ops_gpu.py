@@ -514,3 +514,36 @@ def test_engine_sampling_replay_determinism_gpu():
     assert results[0] == results[1], results
     lens = [len(o) for o in results[0]]
     assert lens == [12, 12], lens
+
+
+def test_group_prefix_sharing_gpu():
+    """Group API on GPU: shared-prefill greedy == separate requests (fork
+    copies + shared page tables through the HIP decode kernels)."""
+    from polyrl_amd.models import create_model
+    from polyrl_amd.models.registry import DecoderConfig
+    from polyrl_amd.rollout.engine import Engine, SamplingParams
+    cfg = DecoderConfig(arch="llama", vocab_size=512, hidden_size=256,
+                        intermediate_size=512, num_hidden_layers=2,
+                        num_attention_heads=2, num_key_value_heads=1,
+                        head_dim=128, max_position_embeddings=128,
+                        rope_theta=10000.0, rms_norm_eps=1e-6)
+    torch.manual_seed(60)
+    model = create_model(cfg, kind="actor", dtype="bfloat16", device=DEV)
+    eng = Engine(cfg, device=DEV, dtype=torch.bfloat16,
+                 kv_bytes_budget=32 << 20, decode_chunk_size=4, page_size=16)
+    eng.model.load_state_dict(model.state_dict())
+    prompt = torch.randint(0, cfg.vocab_size, (13,)).tolist()  # partial page
+    sp = SamplingParams(temperature=0.0, max_new_tokens=6)
+
+    eng.enable_prefix_sharing = False
+    sep = eng.generate([prompt] * 3, sp, "sep")
+    eng.enable_prefix_sharing = True
+    eng.add_request_group("grp", prompt, sp, 3)
+    outs = {}
+    while eng.has_work():
+        for o in eng.step():
+            outs[o.rid] = o
+    for s in range(3):
+        assert outs[f"grp-s{s}"].output_ids == sep[s].output_ids, \
+            (s, outs[f"grp-s{s}"].output_ids, sep[s].output_ids)
+    assert eng.kv.free_pages == eng.kv.num_pages
