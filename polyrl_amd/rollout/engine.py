@@ -411,6 +411,29 @@ class Engine:
     def has_work(self) -> bool:
         return bool(self.waiting or self.running)
 
+    def release_memory(self):
+        """Free the KV cache tensors (the reference's
+        /release_memory_occupation capability, sglang_http_async_engine.py:
+        257-284 — used when trainer and rollout time-share a tight GPU).
+        Requires an idle engine; resume_memory() reallocates."""
+        assert not self.has_work(), "release_memory needs an idle engine"
+        self._kv_shape = (self.kv.num_layers, self.kv.num_kv_heads,
+                          self.kv.head_dim, self.kv.num_pages,
+                          self.kv.page_size)
+        self.kv.k_cache = []
+        self.kv.v_cache = []
+        self._kv_released = True
+        if str(self.device).startswith("cuda"):
+            torch.cuda.empty_cache()
+
+    def resume_memory(self):
+        if not getattr(self, "_kv_released", False):
+            return
+        L, Hk, D, P, ps = self._kv_shape
+        self.kv = PagedKVCache(L, Hk, D, P, ps, dtype=self.dtype,
+                               device=self.device)
+        self._kv_released = False
+
     def num_queued(self) -> int:
         return len(self.waiting)
 

@@ -130,8 +130,18 @@ class EngineRunner:
                 "last_gen_throughput": self.throughput,
                 "weight_version": self.weight_version}
 
-    def update_weights(self, state_dict, version: int, strict: bool = False):
-        """Swap engine weights under the step lock (excludes generation)."""
+    def update_weights(self, state_dict, version: int, strict: bool = False,
+                       abort_in_flight: bool = True):
+        """Swap engine weights under the step lock (excludes generation).
+
+        ``abort_in_flight`` mirrors the reference's KV flush after an update
+        (patches.py:374-380): requests mid-generation are aborted rather
+        than resumed on different weights; the scheduler continues them
+        token-exactly on the new version.  Co-located publication happens
+        between steps (no in-flight), so this is a no-op there."""
         with self.lock:
             self.engine.model.load_state_dict(state_dict, strict=strict)
+            if abort_in_flight and self.engine.has_work():
+                self.engine.abort_request(abort_all=True)
             self.weight_version = version
+        self._wake.set()

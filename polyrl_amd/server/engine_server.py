@@ -208,13 +208,16 @@ def create_app(engine: Engine, runner: Optional[EngineRunner] = None):
 
     @app.post("/release_memory_occupation")
     async def release_memory_occupation():
-        import torch
-        if torch.cuda.is_available():
-            torch.cuda.empty_cache()
-        return {"status": "ok"}
+        with runner.lock:
+            if not engine.has_work():
+                engine.release_memory()
+                return {"status": "ok", "released": True}
+        return {"status": "busy", "released": False}
 
     @app.post("/resume_memory_occupation")
     async def resume_memory_occupation():
+        with runner.lock:
+            engine.resume_memory()
         return {"status": "ok"}
 
     @app.post("/shutdown")

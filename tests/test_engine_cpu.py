@@ -212,3 +212,16 @@ def test_group_abort_before_fork_emits_all(setup):
             outs[o.rid] = o
     assert set(outs) == {"ab-s0", "ab-s1", "ab-s2"}
     assert all(o.finish_reason == "abort" for o in outs.values())
+
+
+def test_release_resume_memory(setup):
+    cfg, model, eng = setup
+    torch.manual_seed(30)
+    prompt = torch.randint(0, cfg.vocab_size, (6,)).tolist()
+    eng.generate([prompt], SamplingParams(max_new_tokens=3), "pre")
+    eng.release_memory()
+    assert eng.kv.k_cache == []
+    eng.resume_memory()
+    outs = eng.generate([prompt], SamplingParams(temperature=0.0,
+                                                 max_new_tokens=3), "post")
+    assert len(outs[0].output_ids) == 3
