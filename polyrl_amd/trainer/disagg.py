@@ -209,7 +209,8 @@ class DisaggCoordinator:
             if manager_port > 0:
                 from ..scheduler.http_api import serve_manager
                 self.manager_server = serve_manager(
-                    self.scheduler, port=manager_port, loop=self.loop)
+                    self.scheduler, port=manager_port, loop=self.loop,
+                    remote_weight_state_fn=remote_weight_state_fn)
 
     # --------------------------------------------------------------- submit
     def submit(self, prompts: TensorBatch, sampling, n: int):

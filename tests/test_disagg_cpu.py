@@ -75,3 +75,100 @@ def test_elastic_colocated_world2(tmp_path):
         capture_output=True, text=True, timeout=540, env=env)
     assert r.returncode == 0, \
         f"stdout:\n{r.stdout[-4000:]}\nstderr:\n{r.stderr[-4000:]}"
+
+
+@pytest.mark.timeout(900)
+def test_elastic_runtime_join(tmp_path):
+    """A remote instance joins a RUNNING elastic training job through the
+    manager facade (reference §3.4 lifecycle: register -> health gate ->
+    version gate -> TCP weight push -> active pool) and the run completes."""
+    import threading
+    import time as _time
+
+    import httpx
+    import torch
+    import uvicorn
+
+    from polyrl_amd.models import create_model, get_model_config
+    from polyrl_amd.rollout.engine import Engine
+    from polyrl_amd.server import create_app
+
+    env = dict(os.environ)
+    env["PYTHONPATH"] = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node=2", "--master-addr", "127.0.0.1",
+         "--master-port", "29676",
+         "-m", "polyrl_amd.trainer.main_stream",
+         "actor_rollout_ref.model.path=llama-debug-cpu",
+         "actor_rollout_ref.model.dtype=float32",
+         "actor_rollout_ref.model.enable_gradient_checkpointing=false",
+         "actor_rollout_ref.actor.ppo_mini_batch_size=8",
+         "actor_rollout_ref.actor.ppo_max_token_len_per_gpu=512",
+         "actor_rollout_ref.rollout.name=elastic",
+         "actor_rollout_ref.rollout.sampling.n=2",
+         "actor_rollout_ref.rollout.response_length=16",
+         "actor_rollout_ref.rollout.min_stream_batch_size=4",
+         "actor_rollout_ref.rollout.rollout_port_base=31880",
+         "actor_rollout_ref.rollout.rollout_manager_port=31879",
+         "data.train_batch_size=8",
+         "data.max_prompt_length=16",
+         "data.synthetic_num_prompts=64",
+         f"trainer.default_local_dir={tmp_path}/ckpt",
+         "trainer.resume_mode=disable",
+         "reward=random",
+         "max_steps=6",
+         ], env=env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+        text=True)
+
+    joined = {"ok": False, "active": False, "version": -1}
+    try:
+        # our "spot node": a served engine on this test process
+        cfg = get_model_config("llama-debug-cpu")
+        torch.manual_seed(1)
+        eng = Engine(cfg, device="cpu", dtype=torch.float32,
+                     kv_bytes_budget=16 << 20)
+        for t in eng.model._name_map.values():
+            t.normal_(0, 0.02)
+        embed_before = eng.model.embed.clone()
+        app = create_app(eng)
+        server = uvicorn.Server(uvicorn.Config(app, host="127.0.0.1",
+                                               port=31885,
+                                               log_level="error"))
+        threading.Thread(target=server.run, daemon=True).start()
+
+        mgr = "http://127.0.0.1:31879"
+        deadline = _time.monotonic() + 120
+        registered = False
+        while _time.monotonic() < deadline and proc.poll() is None:
+            try:
+                if not registered:
+                    r = httpx.post(f"{mgr}/register_rollout_instance",
+                                   json={"addr": "http://127.0.0.1:31885"},
+                                   timeout=30.0)
+                    registered = r.status_code == 200
+                    continue
+                d = httpx.get(f"{mgr}/get_instances_status",
+                              timeout=5.0).json()
+                me = [i for i in d["instances"]
+                      if i["id"].endswith(":31885")]
+                if me:
+                    joined["ok"] = True
+                    if me[0]["active"]:
+                        joined["active"] = True
+                        joined["version"] = me[0]["weight_version"]
+                        break
+            except Exception:
+                pass
+            _time.sleep(0.3)
+        out, _ = proc.communicate(timeout=600)
+        server.should_exit = True
+        assert proc.returncode == 0, out[-4000:]
+        assert joined["ok"], "instance never appeared in the manager roster"
+        assert joined["active"], "joined instance never became active"
+        assert joined["version"] >= 1   # received a weight version via TCP
+        # TCP push actually replaced our random init with trainer weights
+        assert not torch.equal(eng.model.embed, embed_before)
+    finally:
+        if proc.poll() is None:
+            proc.kill()
