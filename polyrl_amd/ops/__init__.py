@@ -259,9 +259,11 @@ class _FlashAttnVarlen(torch.autograd.Function):
         dq = torch.zeros(q.shape, dtype=torch.float32, device=q.device)
         dk = torch.zeros(k.shape, dtype=torch.float32, device=q.device)
         dv = torch.zeros(v.shape, dtype=torch.float32, device=q.device)
+        import os as _os
+        use_v2 = _os.environ.get("POLYRL_ATTN_BWD_V2", "0") == "1"
         ext.varlen_attention_backward(
             dq, dk, dv, q, k, v, out, dout.contiguous(), lse,
-            cu, cu, t32_seq, t32_k0, ctx.scale, ctx.causal)
+            cu, cu, t32_seq, t32_k0, ctx.scale, ctx.causal, use_v2)
         return (dq.to(q.dtype), dk.to(q.dtype), dv.to(q.dtype),
                 None, None, None, None, None, None, None)
 

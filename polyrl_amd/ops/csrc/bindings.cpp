@@ -34,7 +34,7 @@ void varlen_attention_backward(
     torch::Tensor out, torch::Tensor dout, torch::Tensor lse,
     torch::Tensor cu_seqlens_q, torch::Tensor cu_seqlens_k,
     torch::Tensor tile_seq, torch::Tensor tile_k0,
-    double scale, bool causal);
+    double scale, bool causal, bool use_v2);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, CDNA4)");
