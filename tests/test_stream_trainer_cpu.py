@@ -136,3 +136,19 @@ def test_validate_greedy_rollouts(tmp_path):
     assert "val/score/mean" in val
     assert val["val/n"] == 4.0
     assert 0.0 <= val["val/score/mean"] <= 1.0
+
+
+def test_gpt2_constant_reward_config1(tmp_path):
+    """BASELINE config #1: GPT-2 family, GRPO, constant reward, world 1 on
+    CPU — the plumbing tier the reference defines for no-GPU runs."""
+    cfg = tiny_config(tmp_path, model="gpt2-debug")
+    trainer = StreamPPOTrainer(cfg, reward_fn=load_reward_manager("constant"))
+    before = snapshot(trainer.actor.model)
+    trainer.fit(max_steps=2)
+    after = snapshot(trainer.actor.model)
+    # constant reward => zero advantage => pg_loss 0; params may still move
+    # only via optimizer side effects, so just assert the loop is stable
+    for k in before:
+        assert torch.isfinite(after[k]).all()
+    val = trainer.validate(num_prompts=4)
+    assert val["val/score/mean"] == 1.0       # constant reward
