@@ -70,3 +70,50 @@ def test_tp2_engine_matches_tp1(tmp_path):
         capture_output=True, text=True, timeout=540, env=env)
     assert r.returncode == 0, f"stdout:\n{r.stdout[-3000:]}\nstderr:\n{r.stderr[-3000:]}"
     assert "TP OK" in r.stdout
+
+
+@pytest.mark.timeout(600)
+def test_tp2_group_prefix_sharing(tmp_path):
+    """TP engine + prefix-sharing group API: greedy equality with separate
+    requests and cross-rank agreement (the two features compose)."""
+    script = tmp_path / "tp_group_worker.py"
+    script.write_text(r"""
+import torch, torch.distributed as dist
+dist.init_process_group("gloo")
+from polyrl_amd.models import create_model, get_model_config
+from polyrl_amd.parallel.tp import TPContext
+from polyrl_amd.rollout.engine import Engine, SamplingParams
+cfg = get_model_config("llama-debug-cpu")
+torch.manual_seed(0)
+model = create_model(cfg, kind="actor", dtype="float32", device="cpu")
+tp = TPContext(dist.group.WORLD)
+eng = Engine(cfg, device="cpu", dtype=torch.float32, kv_bytes_budget=8 << 20,
+             tp_ctx=tp)
+eng.model.load_state_dict(model.state_dict())
+prompt = [5, 9, 2, 7, 11]
+sp = SamplingParams(temperature=0.0, max_new_tokens=5)
+eng.enable_prefix_sharing = False
+sep = eng.generate([prompt] * 2, sp, "sep")
+eng.enable_prefix_sharing = True
+eng.add_request_group("grp", prompt, sp, 2)
+outs = {}
+while eng.has_work():
+    for o in eng.step():
+        outs[o.rid] = o
+for s in range(2):
+    assert outs[f"grp-s{s}"].output_ids == sep[s].output_ids
+ids = [outs[f"grp-s{s}"].output_ids for s in range(2)]
+box = [None, None]
+dist.all_gather_object(box, ids)
+assert box[0] == box[1], box
+print(f"rank {dist.get_rank()}: TP+group OK")
+""")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node=2", "--master-addr", "127.0.0.1",
+         "--master-port", "29678", str(script)],
+        capture_output=True, text=True, timeout=540, env=env)
+    assert r.returncode == 0, f"{r.stdout[-3000:]}\n{r.stderr[-3000:]}"
+    assert "TP+group OK" in r.stdout
