@@ -84,6 +84,10 @@ def main():
     cfg.actor_rollout_ref.model.dtype = "bfloat16" if device == "cuda" else "float32"
     cfg.actor_rollout_ref.model.enable_gradient_checkpointing = False
     cfg.critic.model.enable_gradient_checkpointing = False
+    # keep gathered params resident across the step's micro passes (one
+    # all-gather per step; +full-param memory, fine at 288 GB)
+    cfg.actor_rollout_ref.actor.fsdp.reshard_after_forward = False
+    cfg.critic.fsdp.reshard_after_forward = False
     cfg.actor_rollout_ref.actor.ppo_mini_batch_size = total_samples // 2
     cfg.actor_rollout_ref.actor.ppo_max_token_len_per_gpu = 8192
     cfg.actor_rollout_ref.rollout.sampling.n = args.n_samples

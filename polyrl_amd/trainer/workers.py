@@ -30,10 +30,14 @@ from ..protocol import TensorBatch
 
 
 def _maybe_fully_shard(model: nn.Module, mixed_precision: bool = True,
-                       pg=None):
+                       pg=None, reshard_after_forward: bool = True):
     """Apply FSDP2 per decoder layer + root.  Works for world_size 1..N on
     nccl(RCCL) and gloo alike.  ``pg`` restricts sharding to a subgroup
-    (disaggregated split: FSDP over the trainer ranks only)."""
+    (disaggregated split: FSDP over the trainer ranks only).
+    ``reshard_after_forward=False`` keeps gathered params resident across
+    the step's many micro passes — ONE all-gather per step instead of one
+    per micro fwd/bwd, paid with +full-param memory (288 GB affords it for
+    the 8B benchmark config)."""
     if not (dist.is_available() and dist.is_initialized()):
         return model
     from torch.distributed.fsdp import MixedPrecisionPolicy, fully_shard
@@ -42,6 +46,7 @@ def _maybe_fully_shard(model: nn.Module, mixed_precision: bool = True,
         mp = MixedPrecisionPolicy(param_dtype=torch.bfloat16,
                                   reduce_dtype=torch.float32)
     kwargs = {"mp_policy": mp} if mp else {}
+    kwargs["reshard_after_forward"] = reshard_after_forward
     if pg is not None:
         from torch.distributed.device_mesh import DeviceMesh
         dev_type = "cuda" if torch.cuda.is_available() else "cpu"
@@ -166,7 +171,9 @@ class ActorWorker:
         self.device = device
         self.is_ref = is_ref
         self.pg = pg
-        self.model = _maybe_fully_shard(model, pg=pg)
+        self.model = _maybe_fully_shard(
+            model, pg=pg,
+            reshard_after_forward=cfg.fsdp.reshard_after_forward)
         if pg is not None and cfg.ulysses_sequence_parallel_size > 1:
             raise NotImplementedError("Ulysses SP inside a disaggregated "
                                       "trainer subgroup is not wired yet")
@@ -335,7 +342,9 @@ class CriticWorker:
         self.cfg = cfg
         self.device = device
         self.pg = pg
-        self.model = _maybe_fully_shard(model, pg=pg)
+        self.model = _maybe_fully_shard(
+            model, pg=pg,
+            reshard_after_forward=cfg.fsdp.reshard_after_forward)
         if pg is not None and \
                 getattr(cfg, "ulysses_sequence_parallel_size", 1) > 1:
             raise NotImplementedError("Ulysses SP inside a disaggregated "
