@@ -325,7 +325,14 @@ class ElasticPublisher:
         self.coordinator = coordinator
         self.group = trainer_group
         self.cpu_cache: Dict[str, torch.Tensor] = {}
+        self._cache_lock = threading.Lock()
         self.last_publish_s = 0.0
+
+    def snapshot_cache(self) -> Dict[str, torch.Tensor]:
+        """Consistent snapshot for async TCP pushes (the next publish may
+        refill the cache while a push is in flight)."""
+        with self._cache_lock:
+            return dict(self.cpu_cache)
 
     @property
     def version(self) -> int:
@@ -342,10 +349,14 @@ class ElasticPublisher:
             need = [int(any(not i.is_local for i in sched.instances()))]
         if dist.is_available() and dist.is_initialized():
             dist.broadcast_object_list(need, src=0, group=self.group)
-        cache = self.cpu_cache if (sched is not None and need[0]) else None
-        if cache is not None:
-            cache.clear()
-        v = self.inner.publish(cpu_cache=cache)
+        use_cache = sched is not None and need[0]
+        if use_cache:
+            staging: Dict[str, torch.Tensor] = {}
+            v = self.inner.publish(cpu_cache=staging)
+            with self._cache_lock:
+                self.cpu_cache = staging
+        else:
+            v = self.inner.publish()
         if sched is not None:
             loop = self.coordinator.loop
 
