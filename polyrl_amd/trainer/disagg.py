@@ -241,6 +241,21 @@ class DisaggCoordinator:
             self.scheduler, reqs,
             max_local_gen_s=self.max_local_gen_s, loop=self.loop)
 
+    def update_metrics(self, step_time_s: float, trainer_bubble_s: float,
+                       step_throughput: float) -> float:
+        """Trainer feedback -> adaptive local-gen window (the reference's
+        /update_metrics loop, stream_ray_trainer.py:691-704).  Rank 0 only;
+        returns (and adopts) the new window."""
+        if self.scheduler is None:
+            return self.max_local_gen_s
+        from ..scheduler.types import MetricsUpdate
+        out = self.scheduler.update_metrics(MetricsUpdate(
+            step_time_s=step_time_s, trainer_bubble_time_s=trainer_bubble_s,
+            step_throughput=step_throughput))
+        if self.max_local_gen_s > 0:          # adaptive only when time-boxed
+            self.max_local_gen_s = out["new_max_gen_s"]
+        return self.max_local_gen_s
+
     # --------------------------------------------------------------- stream
     def stream_batches(self, local_stream: int) -> Iterator[TensorBatch]:
         """Yield equal per-rank batches of local_stream samples until the

@@ -453,6 +453,19 @@ class StreamPPOTrainer:
         self._last_full_batch = full
         metrics = reduce_metrics(all_metrics)
         metrics.update(compute_data_metrics(full, self.use_critic))
+        # feedback to the scheduler's adaptive local-gen time-box
+        # (stream_ray_trainer.py:691-704 capability)
+        if self.sched_coordinated and self.rank == 0 and \
+                hasattr(self.coordinator, "update_metrics"):
+            step_t = timing.get("step", 0.0)
+            busy = sum(timing.get(k, 0.0) for k in
+                       ("update", "prep", "weight_sync", "reward"))
+            bubble = max(step_t - busy, 0.0)
+            thr = len(full) / step_t if step_t > 0 else 0.0
+            new_window = self.coordinator.update_metrics(step_t, bubble, thr)
+            metrics["training/max_local_gen_s"] = new_window
+            metrics["training/num_rollout_instances"] = float(
+                len(self.coordinator.scheduler.instances()))
         return metrics
 
     # ------------------------------------------------- per-ibatch preparation
