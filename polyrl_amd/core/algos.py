@@ -404,7 +404,8 @@ def compute_advantage(batch_rewards: torch.Tensor,
                       index=None,
                       gamma: float = 1.0,
                       lam: float = 1.0,
-                      norm_adv_by_std_in_grpo: bool = True
+                      norm_adv_by_std_in_grpo: bool = True,
+                      reward_baselines: Optional[torch.Tensor] = None,
                       ) -> Tuple[torch.Tensor, torch.Tensor]:
     if adv_estimator == "gae":
         assert values is not None
@@ -423,4 +424,9 @@ def compute_advantage(batch_rewards: torch.Tensor,
         return compute_reinforce_plus_plus_advantage(batch_rewards,
                                                      response_mask,
                                                      gamma=gamma)
+    if adv_estimator == "remax":
+        assert reward_baselines is not None, \
+            "remax needs greedy-rollout baselines (trainer supplies them)"
+        return compute_remax_outcome_advantage(batch_rewards, response_mask,
+                                               reward_baselines)
     raise NotImplementedError(f"adv_estimator {adv_estimator!r}")

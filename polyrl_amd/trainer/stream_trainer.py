@@ -413,6 +413,12 @@ class StreamPPOTrainer:
         else:
             submit_batch = global_batch.slice(
                 slice(self.rank * local_bs, (self.rank + 1) * local_bs))
+        # ReMax: an extra greedy rollout of the same prompts supplies the
+        # per-prompt reward baseline (verl remax recipe; estimator in
+        # core/algos.py).  Runs on the freshly published weights.
+        if c.algorithm.adv_estimator == "remax":
+            with marked_timer("gen_baseline", timing):
+                self._collect_remax_baselines(submit_batch, local_bs)
         with marked_timer("gen_submit", timing):
             self.coordinator.submit(submit_batch, sampling, n)
 
@@ -468,6 +474,29 @@ class StreamPPOTrainer:
                 len(self.coordinator.scheduler.instances()))
         return metrics
 
+    def _collect_remax_baselines(self, submit_batch: TensorBatch,
+                                 local_bs: int) -> None:
+        """Greedy (temperature=0, n=1) rollout of this step's prompts;
+        sequence reward per prompt becomes the ReMax baseline, keyed by uid
+        and all-gathered so every rank can score any sample."""
+        ro = self.config.actor_rollout_ref.rollout
+        greedy = SamplingParams(temperature=0.0,
+                                max_new_tokens=ro.response_length)
+        self.coordinator.submit(submit_batch, greedy, 1)
+        parts = [b for b in self.coordinator.stream_batches(local_bs)]
+        batch = TensorBatch.concat(parts)
+        seq_scores = self.reward_fn(batch).sum(dim=-1)
+        local = {str(u): float(v)
+                 for u, v in zip(batch["uid"], seq_scores.tolist())}
+        if dist.is_initialized() and dist.get_world_size(self.pg) > 1:
+            box = [None] * dist.get_world_size(self.pg)
+            dist.all_gather_object(box, local, group=self.pg)
+            merged: Dict[str, float] = {}
+            for d in box:
+                merged.update(d)
+            local = merged
+        self._remax_baselines = local
+
     # ------------------------------------------------- per-ibatch preparation
     def _prepare_ibatch(self, ibatch: TensorBatch,
                         timing: Dict[str, float]) -> TensorBatch:
@@ -505,6 +534,11 @@ class StreamPPOTrainer:
                 ibatch["token_level_rewards"] = rewards
             else:
                 ibatch["token_level_rewards"] = scores
+            baselines = None
+            if c.algorithm.adv_estimator == "remax":
+                baselines = torch.tensor(
+                    [self._remax_baselines[str(u)] for u in ibatch["uid"]],
+                    dtype=torch.float32)
             adv, ret = algos.compute_advantage(
                 ibatch["token_level_rewards"],
                 ibatch["response_mask"].float(),
@@ -512,7 +546,8 @@ class StreamPPOTrainer:
                 values=ibatch.tensors.get("values"),
                 index=ibatch.non_tensors.get("uid"),
                 gamma=c.algorithm.gamma, lam=c.algorithm.lam,
-                norm_adv_by_std_in_grpo=c.algorithm.norm_adv_by_std_in_grpo)
+                norm_adv_by_std_in_grpo=c.algorithm.norm_adv_by_std_in_grpo,
+                reward_baselines=baselines)
             ibatch["advantages"] = adv
             ibatch["returns"] = ret
         return ibatch

@@ -152,3 +152,25 @@ def test_gpt2_constant_reward_config1(tmp_path):
         assert torch.isfinite(after[k]).all()
     val = trainer.validate(num_prompts=4)
     assert val["val/score/mean"] == 1.0       # constant reward
+
+
+def test_remax_greedy_baseline_path(tmp_path):
+    """ReMax: every step runs an extra greedy rollout whose sequence reward
+    becomes the per-prompt baseline; advantage = score - baseline on every
+    response token (core/algos.py::compute_remax_outcome_advantage)."""
+    cfg = tiny_config(tmp_path, adv="remax")
+    trainer = StreamPPOTrainer(cfg, reward_fn=load_reward_manager("random"))
+    before = snapshot(trainer.actor.model)
+    trainer.fit(max_steps=2)
+    after = snapshot(trainer.actor.model)
+    assert any(not torch.equal(before[k], after[k]) for k in before)
+    # every trained sample's uid had a baseline from the greedy pass
+    full = trainer._last_full_batch
+    assert all(str(u) in trainer._remax_baselines for u in full["uid"])
+    # advantage really is (seq score - baseline) broadcast over the mask
+    scores = full["token_level_rewards"].sum(-1)
+    base = torch.tensor([trainer._remax_baselines[str(u)]
+                         for u in full["uid"]])
+    mask = full["response_mask"].float()
+    expect = (scores - base).unsqueeze(-1) * mask
+    assert torch.allclose(full["advantages"], expect, atol=1e-6)
