@@ -14,6 +14,8 @@ backward of a sequence gather is a slice.
 """
 from __future__ import annotations
 
+from typing import List, Optional
+
 import torch
 import torch.distributed as dist
 
@@ -126,22 +128,29 @@ class UlyssesContext:
         return dist.get_rank(self.group) if self.group is not None else 0
 
 
-def build_sp_groups(sp_size: int):
-    """Partition the world into contiguous SP groups of ``sp_size`` ranks;
-    returns (sp_group, dp_group) for this rank.  dp groups connect the
-    same-sp-position ranks across SP groups (the FSDP data dimension)."""
-    world = dist.get_world_size()
-    assert world % sp_size == 0, f"world {world} % sp {sp_size} != 0"
+def build_sp_groups(sp_size: int, ranks: Optional[List[int]] = None):
+    """Partition ``ranks`` (default: the whole world) into contiguous SP
+    groups of ``sp_size``; returns (sp_group, dp_group) for this rank —
+    (None, None) if this rank is not in ``ranks``.  dp groups connect the
+    same-sp-position ranks across SP groups (the FSDP data dimension).
+
+    ``dist.new_group`` is a WORLD collective: EVERY rank of the default
+    group must call this with the same arguments, even ranks outside
+    ``ranks`` (e.g. disaggregated rollout ranks)."""
+    if ranks is None:
+        ranks = list(range(dist.get_world_size()))
+    assert len(ranks) % sp_size == 0, \
+        f"len(ranks) {len(ranks)} % sp {sp_size} != 0"
     rank = dist.get_rank()
     sp_group = dp_group = None
-    for g0 in range(0, world, sp_size):
-        ranks = list(range(g0, g0 + sp_size))
-        g = dist.new_group(ranks)
-        if rank in ranks:
+    for g0 in range(0, len(ranks), sp_size):
+        rs = ranks[g0:g0 + sp_size]
+        g = dist.new_group(rs)
+        if rank in rs:
             sp_group = g
     for pos in range(sp_size):
-        ranks = list(range(pos, world, sp_size))
-        g = dist.new_group(ranks)
-        if rank in ranks:
+        rs = ranks[pos::sp_size]
+        g = dist.new_group(rs)
+        if rank in rs:
             dp_group = g
     return sp_group, dp_group

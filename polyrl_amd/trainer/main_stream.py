@@ -69,6 +69,15 @@ def main(argv=None):
         trainer_ranks, rollout_ranks = split_roles(world,
                                                    ro.num_rollout_ranks)
         tg = dist.new_group(trainer_ranks)
+        sp = cfg.actor_rollout_ref.actor.ulysses_sequence_parallel_size
+        if sp > 1:
+            # SP groups over the TRAINER subgroup — created on every rank
+            # (world collective), used only by trainer ranks
+            from ..parallel.ulysses import build_sp_groups
+            from .workers import register_sp_groups
+            spg, dpg = build_sp_groups(sp, ranks=trainer_ranks)
+            if rank in trainer_ranks:
+                register_sp_groups(sp, spg, dpg)
         model_cfg = get_model_config(cfg.actor_rollout_ref.model.path)
         print(f"[main_stream] rank {rank}: role="
               f"{'rollout' if rank in rollout_ranks else 'trainer'}",

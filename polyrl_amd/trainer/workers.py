@@ -123,6 +123,14 @@ def _model_inputs(batch: TensorBatch, device):
 _SP_GROUPS: Dict[int, tuple] = {}   # sp_size -> (sp_group, dp_group)
 
 
+def register_sp_groups(sp_size: int, sp_group, dp_group) -> None:
+    """Pre-register SP/DP groups built over a rank SUBSET (disaggregated
+    trainer subgroup).  dist.new_group is a world collective, so the groups
+    must be created on every rank BEFORE the trainer/rollout role branch;
+    the trainer ranks then hand them to the workers through this hook."""
+    _SP_GROUPS[sp_size] = (sp_group, dp_group)
+
+
 def _get_sp_groups(sp_size: int):
     """Build (once) and cache the SP/DP process groups for this rank.
     Safe to call from actor+critic+ref constructors: the group-creation
@@ -175,8 +183,13 @@ class ActorWorker:
             model, pg=pg,
             reshard_after_forward=cfg.fsdp.reshard_after_forward)
         if pg is not None and cfg.ulysses_sequence_parallel_size > 1:
-            raise NotImplementedError("Ulysses SP inside a disaggregated "
-                                      "trainer subgroup is not wired yet")
+            # disagg: SP groups over the trainer subgroup must have been
+            # pre-built (a world collective) before the role branch —
+            # main_stream.py does this via register_sp_groups()
+            assert cfg.ulysses_sequence_parallel_size in _SP_GROUPS, \
+                ("Ulysses SP inside a disaggregated trainer subgroup needs "
+                 "pre-built groups: call workers.register_sp_groups() on the "
+                 "trainer ranks (main_stream.py wires this)")
         self.sp_group = _setup_ulysses(model,
                                        cfg.ulysses_sequence_parallel_size)
         self.sp_size = cfg.ulysses_sequence_parallel_size \
@@ -347,8 +360,10 @@ class CriticWorker:
             reshard_after_forward=cfg.fsdp.reshard_after_forward)
         if pg is not None and \
                 getattr(cfg, "ulysses_sequence_parallel_size", 1) > 1:
-            raise NotImplementedError("Ulysses SP inside a disaggregated "
-                                      "trainer subgroup is not wired yet")
+            assert cfg.ulysses_sequence_parallel_size in _SP_GROUPS, \
+                ("Ulysses SP inside a disaggregated trainer subgroup needs "
+                 "pre-built groups: call workers.register_sp_groups() on the "
+                 "trainer ranks (main_stream.py wires this)")
         self.sp_group = _setup_ulysses(
             model, getattr(cfg, "ulysses_sequence_parallel_size", 1))
         self.sp_size = cfg.ulysses_sequence_parallel_size \

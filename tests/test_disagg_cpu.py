@@ -172,3 +172,41 @@ def test_elastic_runtime_join(tmp_path):
     finally:
         if proc.poll() is None:
             proc.kill()
+
+
+@pytest.mark.timeout(600)
+def test_disagg_with_ulysses_sp2(tmp_path):
+    """Disagg split + Ulysses SP inside the trainer subgroup (world=3:
+    2 trainer ranks forming one SP=2 group + 1 rollout rank).  The SP
+    groups are a world collective built before the role branch
+    (main_stream.py / workers.register_sp_groups)."""
+    env = dict(os.environ)
+    env["PYTHONPATH"] = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node=3", "--master-addr", "127.0.0.1",
+         "--master-port", "29691",
+         "-m", "polyrl_amd.trainer.main_stream",
+         "actor_rollout_ref.model.path=llama-debug-cpu",
+         "actor_rollout_ref.model.dtype=float32",
+         "actor_rollout_ref.model.enable_gradient_checkpointing=false",
+         "actor_rollout_ref.actor.ppo_mini_batch_size=8",
+         "actor_rollout_ref.actor.ppo_max_token_len_per_gpu=512",
+         "actor_rollout_ref.actor.ulysses_sequence_parallel_size=2",
+         "actor_rollout_ref.rollout.sampling.n=2",
+         "actor_rollout_ref.rollout.response_length=8",
+         "actor_rollout_ref.rollout.min_stream_batch_size=4",
+         "actor_rollout_ref.rollout.num_rollout_ranks=1",
+         "actor_rollout_ref.rollout.rollout_port_base=31830",
+         "data.train_batch_size=8",
+         "data.max_prompt_length=16",
+         "data.synthetic_num_prompts=32",
+         f"trainer.default_local_dir={tmp_path}/ckpt",
+         "trainer.resume_mode=disable",
+         "reward=random",
+         "max_steps=2",
+         ],
+        capture_output=True, text=True, timeout=540, env=env)
+    assert r.returncode == 0, \
+        f"stdout:\n{r.stdout[-4000:]}\nstderr:\n{r.stderr[-4000:]}"
+    assert "fit done" in r.stdout
