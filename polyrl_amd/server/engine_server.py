@@ -248,6 +248,9 @@ def main():
     p.add_argument("--kv-gb", type=float, default=8.0)
     p.add_argument("--manager", default=None,
                    help="scheduler HTTP endpoint to register with")
+    p.add_argument("--advertise-addr", default=None,
+                   help="address the manager should reach us at (default: "
+                        "the interface that routes to the manager)")
     p.add_argument("--load", default=None,
                    help="safetensors checkpoint to load (else random init)")
     args = p.parse_args()
@@ -267,8 +270,23 @@ def main():
 
     if args.manager:
         import requests
+        adv = args.advertise_addr
+        if adv is None and args.host not in ("0.0.0.0", "::", ""):
+            adv = args.host
+        if adv is None:
+            # the interface that routes to the manager (multi-node safe)
+            import socket
+            from urllib.parse import urlparse
+            u = urlparse(args.manager)
+            try:
+                sk = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
+                sk.connect((u.hostname, u.port or 80))
+                adv = sk.getsockname()[0]
+                sk.close()
+            except OSError:
+                adv = "127.0.0.1"
         requests.post(f"{args.manager}/register_rollout_instance",
-                      json={"addr": f"http://{args.host}:{args.port}"},
+                      json={"addr": f"http://{adv}:{args.port}"},
                       timeout=10)
 
     uvicorn.run(app, host=args.host, port=args.port, log_level="warning")

@@ -18,6 +18,7 @@ native transports:
 from __future__ import annotations
 
 import asyncio
+import os
 import threading
 import time
 from typing import Dict, Iterator, List, Optional
@@ -41,6 +42,25 @@ def split_roles(world: int, num_rollout: int):
 
 def rollout_port(rank: int, base: int = 30000) -> int:
     return base + rank
+
+
+def advertise_addr() -> str:
+    """The address other nodes reach this rank's HTTP engine server at.
+    Multi-node: the interface that routes to MASTER_ADDR (or an explicit
+    POLYRL_ADVERTISE_ADDR); single node this resolves to 127.0.0.1."""
+    a = os.environ.get("POLYRL_ADVERTISE_ADDR")
+    if a:
+        return a
+    master = os.environ.get("MASTER_ADDR", "127.0.0.1")
+    try:
+        import socket
+        sk = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
+        sk.connect((master, int(os.environ.get("MASTER_PORT", "29500"))))
+        ip = sk.getsockname()[0]
+        sk.close()
+        return ip
+    except OSError:
+        return "127.0.0.1"
 
 
 # ------------------------------------------------------------ rollout rank
@@ -72,12 +92,14 @@ def rollout_serve_loop(cfg, model_cfg, rank: int, device: str, dtype,
     runner = EngineRunner(engine)
     app = create_app(engine, runner)
     port = rollout_port(rank, port_base)
-    server = uvicorn.Server(uvicorn.Config(app, host="127.0.0.1", port=port,
+    # bind all interfaces so trainer ranks on OTHER nodes can reach us;
+    # the advertised address travels through main_stream's address exchange
+    server = uvicorn.Server(uvicorn.Config(app, host="0.0.0.0", port=port,
                                            log_level="error"))
     th = threading.Thread(target=server.run, daemon=True)
     th.start()
-    print(f"[rollout rank {rank}] engine serving on 127.0.0.1:{port}",
-          flush=True)
+    print(f"[rollout rank {rank}] engine serving on "
+          f"{advertise_addr()}:{port}", flush=True)
 
     plane = CollectiveWeightPlane(src=0, device=device)
 

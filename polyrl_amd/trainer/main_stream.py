@@ -64,11 +64,18 @@ def main(argv=None):
         # rollout, the rest train.  The trainer subgroup must be created on
         # EVERY rank (collective), before the roles branch.
         from ..models import get_model_config
-        from .disagg import rollout_serve_loop, split_roles
+        from .disagg import advertise_addr, rollout_serve_loop, split_roles
         assert world > 1, "disaggregated mode needs a multi-rank world"
         trainer_ranks, rollout_ranks = split_roles(world,
                                                    ro.num_rollout_ranks)
         tg = dist.new_group(trainer_ranks)
+        # address map: each rollout rank advertises where its HTTP engine
+        # server is reachable (multi-node: its xGMI-node IP; single node:
+        # 127.0.0.1).  Exchanged over the world group before the branch.
+        box = [None] * world
+        dist.all_gather_object(
+            box, advertise_addr() if rank in rollout_ranks else None)
+        rollout_addrs = {r: box[r] for r in rollout_ranks}
         sp = cfg.actor_rollout_ref.actor.ulysses_sequence_parallel_size
         if sp > 1:
             # SP groups over the TRAINER subgroup — created on every rank
@@ -89,7 +96,8 @@ def main(argv=None):
                                port_base=ro.rollout_port_base)
         else:
             trainer = StreamPPOTrainer(cfg, reward_fn=reward_fn,
-                                       process_group=tg)
+                                       process_group=tg,
+                                       rollout_addrs=rollout_addrs)
             print(f"[main_stream] rank {rank}: trainer ready", flush=True)
             trainer.fit(max_steps=max_steps)
             print(f"[main_stream] rank {rank}: fit done", flush=True)

@@ -47,7 +47,8 @@ def _dist_info():
 
 class StreamPPOTrainer:
     def __init__(self, config: PPOConfig, device: Optional[str] = None,
-                 reward_fn=None, dataset=None, process_group=None):
+                 reward_fn=None, dataset=None, process_group=None,
+                 rollout_addrs: Optional[Dict[int, str]] = None):
         self.config = config
         self.pg = process_group          # trainer subgroup (disagg) or None
         if process_group is not None:
@@ -139,7 +140,9 @@ class StreamPPOTrainer:
                                  rollout_port, split_roles)
             gworld = dist.get_world_size()
             _, rollout_ranks = split_roles(gworld, ro.num_rollout_ranks)
-            urls = [f"http://127.0.0.1:{rollout_port(r, ro.rollout_port_base)}"
+            addrs = rollout_addrs or {}
+            urls = [f"http://{addrs.get(r) or '127.0.0.1'}:"
+                    f"{rollout_port(r, ro.rollout_port_base)}"
                     for r in rollout_ranks]
             self.engine = None
             self.coordinator = DisaggCoordinator(
