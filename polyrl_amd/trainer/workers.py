@@ -30,14 +30,17 @@ from ..protocol import TensorBatch
 
 
 def _maybe_fully_shard(model: nn.Module, mixed_precision: bool = True,
-                       pg=None, reshard_after_forward: bool = True):
+                       pg=None, reshard_after_forward: bool = True,
+                       param_offload: bool = False):
     """Apply FSDP2 per decoder layer + root.  Works for world_size 1..N on
     nccl(RCCL) and gloo alike.  ``pg`` restricts sharding to a subgroup
     (disaggregated split: FSDP over the trainer ranks only).
     ``reshard_after_forward=False`` keeps gathered params resident across
     the step's many micro passes — ONE all-gather per step instead of one
     per micro fwd/bwd, paid with +full-param memory (288 GB affords it for
-    the 8B benchmark config)."""
+    the 8B benchmark config).  ``param_offload`` puts sharded params (and
+    their optimizer state) in pinned host memory between uses — the 70B
+    PPO memory lever (actor+critic+engine beyond 288 GB)."""
     if not (dist.is_available() and dist.is_initialized()):
         return model
     from torch.distributed.fsdp import MixedPrecisionPolicy, fully_shard
@@ -47,6 +50,9 @@ def _maybe_fully_shard(model: nn.Module, mixed_precision: bool = True,
                                   reduce_dtype=torch.float32)
     kwargs = {"mp_policy": mp} if mp else {}
     kwargs["reshard_after_forward"] = reshard_after_forward
+    if param_offload and torch.cuda.is_available():
+        from torch.distributed.fsdp import CPUOffloadPolicy
+        kwargs["offload_policy"] = CPUOffloadPolicy()
     if pg is not None:
         from torch.distributed.device_mesh import DeviceMesh
         dev_type = "cuda" if torch.cuda.is_available() else "cpu"
@@ -68,6 +74,24 @@ def _maybe_fully_shard(model: nn.Module, mixed_precision: bool = True,
 def _build_optimizer(params, cfg: OptimConfig):
     return torch.optim.AdamW(params, lr=cfg.lr, betas=tuple(cfg.betas),
                              eps=cfg.eps, weight_decay=cfg.weight_decay)
+
+
+def offload_optimizer_state(optimizer: torch.optim.Optimizer) -> None:
+    """Move optimizer state tensors to host between steps (verl's
+    offload_fsdp_optimizer capability — the fsdp.optimizer_offload knob).
+    Paid with an H2D/D2H pass around each optimizer step; buys ~2 x 4 bytes
+    x params of HBM for 70B-scale PPO."""
+    for st in optimizer.state.values():
+        for k, v in st.items():
+            if torch.is_tensor(v) and v.device.type != "cpu":
+                st[k] = v.to("cpu", non_blocking=True)
+
+
+def load_optimizer_state(optimizer: torch.optim.Optimizer, device) -> None:
+    for st in optimizer.state.values():
+        for k, v in st.items():
+            if torch.is_tensor(v) and str(v.device) != str(device):
+                st[k] = v.to(device, non_blocking=True)
 
 
 def _build_lr_scheduler(optimizer, cfg: OptimConfig):
@@ -181,7 +205,8 @@ class ActorWorker:
         self.pg = pg
         self.model = _maybe_fully_shard(
             model, pg=pg,
-            reshard_after_forward=cfg.fsdp.reshard_after_forward)
+            reshard_after_forward=cfg.fsdp.reshard_after_forward,
+            param_offload=cfg.fsdp.param_offload)
         if pg is not None and cfg.ulysses_sequence_parallel_size > 1:
             # disagg: SP groups over the trainer subgroup must have been
             # pre-built (a world collective) before the role branch —
@@ -342,7 +367,12 @@ class ActorWorker:
         if not math.isfinite(gn):
             self.optimizer.zero_grad()
             return gn
+        offl = self.cfg.fsdp.optimizer_offload and self.device != "cpu"
+        if offl:
+            load_optimizer_state(self.optimizer, self.device)
         self.optimizer.step()
+        if offl:
+            offload_optimizer_state(self.optimizer)
         self.optimizer.zero_grad()
         return gn
 
@@ -357,7 +387,8 @@ class CriticWorker:
         self.pg = pg
         self.model = _maybe_fully_shard(
             model, pg=pg,
-            reshard_after_forward=cfg.fsdp.reshard_after_forward)
+            reshard_after_forward=cfg.fsdp.reshard_after_forward,
+            param_offload=cfg.fsdp.param_offload)
         if pg is not None and \
                 getattr(cfg, "ulysses_sequence_parallel_size", 1) > 1:
             assert cfg.ulysses_sequence_parallel_size in _SP_GROUPS, \
@@ -439,7 +470,12 @@ class CriticWorker:
         if is_opt_step:
             gn = _clip_grad_norm(self.model, self.cfg.optim.grad_clip)
             if math.isfinite(gn):
+                offl = self.cfg.fsdp.optimizer_offload and self.device != "cpu"
+                if offl:
+                    load_optimizer_state(self.optimizer, self.device)
                 self.optimizer.step()
+                if offl:
+                    offload_optimizer_state(self.optimizer)
             self.optimizer.zero_grad()
             metrics.setdefault("critic/grad_norm", []).append(gn)
         if is_lr_step:
