@@ -192,6 +192,10 @@ class TrainerConfig:
     val_before_train: bool = False
     device: str = "cuda"
     seed: int = 1
+    # per-step profiling window (reference: global_profiler.tool=nsys with
+    # profile steps, main_stream.py:79-93) — torch.profiler chrome traces
+    profile_steps: List[int] = field(default_factory=list)
+    profile_dir: str = "profiles/torch"
 
 
 @dataclass

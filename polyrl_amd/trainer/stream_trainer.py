@@ -33,6 +33,7 @@ from ..models import create_model, get_model_config
 from ..protocol import TensorBatch
 from ..reward import load_reward_manager
 from ..rollout.engine import Engine, SamplingParams
+from ..utils.profiling import step_profiler
 from ..transfer.weight_transfer import WeightPublisher
 from .checkpoint import CheckpointManager
 from .rollout_coordinator import LocalRolloutCoordinator
@@ -327,10 +328,13 @@ class StreamPPOTrainer:
                 step_in_run += 1
                 metrics: Dict[str, float] = {}
                 timing: Dict[str, float] = {}
-                with marked_timer("step", timing):
-                    batch_metrics = self._run_step(
-                        global_batch, n, local_bs, local_stream, mini_local,
-                        local_total, timing)
+                with step_profiler(self.global_step in
+                                   c.trainer.profile_steps,
+                                   c.trainer.profile_dir, self.global_step):
+                    with marked_timer("step", timing):
+                        batch_metrics = self._run_step(
+                            global_batch, n, local_bs, local_stream,
+                            mini_local, local_total, timing)
                 metrics.update(batch_metrics)
                 metrics.update(compute_timing_metrics(
                     self._last_full_batch, timing))

@@ -174,3 +174,16 @@ def test_remax_greedy_baseline_path(tmp_path):
     mask = full["response_mask"].float()
     expect = (scores - base).unsqueeze(-1) * mask
     assert torch.allclose(full["advantages"], expect, atol=1e-6)
+
+
+def test_profile_steps_emit_chrome_trace(tmp_path):
+    """trainer.profile_steps wraps those steps in a torch.profiler window
+    and writes a chrome trace (reference: per-step nsys start/stop hooks,
+    main_stream.py:79-93 capability)."""
+    cfg = tiny_config(tmp_path)
+    cfg.trainer.profile_steps = [2]
+    cfg.trainer.profile_dir = str(tmp_path / "prof")
+    trainer = StreamPPOTrainer(cfg, reward_fn=load_reward_manager("random"))
+    trainer.fit(max_steps=2)
+    assert (tmp_path / "prof" / "step_2.json").exists()
+    assert not (tmp_path / "prof" / "step_1.json").exists()
