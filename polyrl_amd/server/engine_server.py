@@ -142,6 +142,31 @@ def create_app(engine: Engine, runner: Optional[EngineRunner] = None):
             None, lambda: runner.update_weights(sd, version))
         return {"success": True, "message": f"weights at version {version}"}
 
+    @app.post("/update_weights_from_tensor")
+    async def update_weights_from_tensor(request: Request):
+        """Reference-adapter surface (sglang_http_async_engine.py:155-299
+        capability): JSON {"version": V, "tensors": {name: {"data": b64,
+        "shape": [...], "dtype": "bfloat16"}}} — base64 raw tensor bytes.
+        The agent/TCP routes are the efficient paths; this one exists for
+        drop-in client compatibility."""
+        import base64
+
+        import numpy as np
+        import torch
+        body = await request.json()
+        version = int(body.get("version", runner.weight_version + 1))
+        sd = {}
+        for name, t in body["tensors"].items():
+            dt = getattr(torch, t.get("dtype", "float32"))
+            raw = base64.b64decode(t["data"])
+            # frombuffer has no bf16: view as uint8 then reinterpret
+            ten = torch.frombuffer(bytearray(raw), dtype=torch.uint8) \
+                .view(dt).reshape(t["shape"])
+            sd[name] = ten
+        await asyncio.get_running_loop().run_in_executor(
+            None, lambda: runner.update_weights(sd, version))
+        return {"success": True, "message": f"weights at version {version}"}
+
     @app.post("/weights_handshake")
     async def weights_handshake(request: Request):
         """Arm a TCP bulk receive (the reference's receiver-agent bootstrap,

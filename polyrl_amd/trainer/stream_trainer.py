@@ -345,8 +345,9 @@ class StreamPPOTrainer:
                 metrics["training/global_step"] = self.global_step
                 if self.rank == 0:
                     self.tracking.log(metrics, self.global_step)
-                if c.trainer.save_freq > 0 and \
-                        self.global_step % c.trainer.save_freq == 0:
+                if (c.trainer.save_freq > 0 and
+                        self.global_step % c.trainer.save_freq == 0) or \
+                        self._should_save_esi(timing.get("step", 0.0)):
                     self.save_checkpoint()
                 if c.trainer.test_freq > 0 and \
                         self.global_step % c.trainer.test_freq == 0:
@@ -558,6 +559,29 @@ class StreamPPOTrainer:
             ibatch["advantages"] = adv
             ibatch["returns"] = ret
         return ibatch
+
+    def _should_save_esi(self, last_step_s: float) -> bool:
+        """Spot/ESI expiration-aware checkpointing (the reference's
+        should_save_ckpt_esi, stream_ray_trainer.py:604-623): when
+        POLYRL_ESI_EXPIRE_AT (epoch seconds) is set and another step +
+        save would not fit before expiry, save NOW.  Rank-0's decision is
+        broadcast so every rank enters the (collective) save together."""
+        import os as _os
+        import time as _time
+        save = False
+        if self.rank == 0:
+            exp = _os.environ.get("POLYRL_ESI_EXPIRE_AT")
+            if exp and not getattr(self, "_esi_saved", False):
+                remaining = float(exp) - _time.time()
+                margin = max(2.0 * last_step_s, 30.0)
+                save = remaining < margin
+        if dist.is_initialized() and dist.get_world_size(self.pg) > 1:
+            box = [save]
+            dist.broadcast_object_list(box, src=0, group=self.pg)
+            save = box[0]
+        if save:
+            self._esi_saved = True
+        return save
 
     # ------------------------------------------------------------ checkpoint
     def save_checkpoint(self):

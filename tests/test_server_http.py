@@ -178,3 +178,31 @@ def test_tcp_weight_push_into_served_engine(served):
     name = "model.embed_tokens.weight"
     assert torch.allclose(eng.model._name_map[name], sd[name])
     server.should_exit = True
+
+
+def test_update_weights_from_tensor_route(served):
+    """Reference-adapter compatible route: base64 tensor payload installs
+    weights under the step lock (patches.py:498-566 HTTP surface)."""
+    import base64
+
+    import httpx
+    cfg, model, eng, app = served
+    name, tref = next(iter(eng.model._name_map.items()))
+    new = torch.full_like(tref, 0.125)
+    payload = {"version": 41, "tensors": {
+        name: {"data": base64.b64encode(
+                   new.view(torch.uint8).numpy().tobytes()
+                   if new.dtype == torch.bfloat16
+                   else new.numpy().tobytes()).decode(),
+               "shape": list(new.shape),
+               "dtype": str(new.dtype).split(".")[-1]}}}
+
+    async def go():
+        async with httpx.AsyncClient(transport=_transport(app),
+                                     base_url="http://t") as c:
+            r = await c.post("/update_weights_from_tensor", json=payload)
+            assert r.status_code == 200, r.text
+            assert r.json()["success"]
+    asyncio.run(go())
+    assert torch.equal(eng.model._name_map[name], new)
+    assert app.state.runner.weight_version == 41

@@ -187,3 +187,18 @@ def test_profile_steps_emit_chrome_trace(tmp_path):
     trainer.fit(max_steps=2)
     assert (tmp_path / "prof" / "step_2.json").exists()
     assert not (tmp_path / "prof" / "step_1.json").exists()
+
+
+def test_esi_expiry_forces_checkpoint(tmp_path, monkeypatch):
+    """POLYRL_ESI_EXPIRE_AT near in the future forces a save even with
+    save_freq disabled (should_save_ckpt_esi capability,
+    stream_ray_trainer.py:604-623)."""
+    import time
+
+    cfg = tiny_config(tmp_path)
+    cfg.trainer.save_freq = -1
+    monkeypatch.setenv("POLYRL_ESI_EXPIRE_AT", str(time.time() + 5))
+    trainer = StreamPPOTrainer(cfg, reward_fn=load_reward_manager("random"))
+    trainer.fit(max_steps=1)
+    ckpts = list((tmp_path / "ckpt").glob("global_step_*"))
+    assert ckpts, "ESI expiry did not trigger a checkpoint"
