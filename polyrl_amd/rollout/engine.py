@@ -22,7 +22,7 @@ import torch
 
 import polyrl_amd.ops as ops
 from ..models.registry import DecoderConfig
-from .kv_cache import PagedKVCache
+from .kv_cache import PagedKVCache, RadixCache
 
 
 @dataclass
@@ -332,11 +332,13 @@ class Engine:
                  max_model_len: Optional[int] = None,
                  decode_chunk_size: int = 16,
                  tp_ctx=None,
-                 seed: int = 0):
+                 seed: int = 0,
+                 enable_radix_cache: bool = False):
         self.cfg = cfg
         self.device = device
         self.dtype = dtype
         self.model = InferenceModel(cfg, device, dtype, tp_ctx=tp_ctx)
+        self.model._engine_owner = self   # weight publishers flush the radix
         self.tp = self.model.tp
         self.max_running = max_running_requests
         self.max_batched_tokens = max_num_batched_tokens
@@ -363,6 +365,9 @@ class Engine:
         self._gen = torch.Generator().manual_seed(seed)  # CPU ref path RNG
         self.decode_chunk_size = max(decode_chunk_size, 1)
         self.enable_prefix_sharing = True
+        # cross-request KV prefix reuse (SGLang radix-cache capability);
+        # deterministic trie ops, so TP ranks stay in lockstep
+        self.radix = RadixCache(self.kv) if enable_radix_cache else None
         # hipGraph-captured decode iteration (one replay per token):
         # removes ~300 kernel-launch round-trips per decode step.  Disabled
         # under TP (capturing RCCL collectives in a graph is unvalidated).
@@ -420,6 +425,8 @@ class Engine:
         self._kv_shape = (self.kv.num_layers, self.kv.num_kv_heads,
                           self.kv.head_dim, self.kv.num_pages,
                           self.kv.page_size)
+        if self.radix is not None:
+            self.radix.reset()
         self.kv.k_cache = []
         self.kv.v_cache = []
         self._kv_released = True
@@ -432,6 +439,8 @@ class Engine:
         L, Hk, D, P, ps = self._kv_shape
         self.kv = PagedKVCache(L, Hk, D, P, ps, dtype=self.dtype,
                                device=self.device)
+        if self.radix is not None:
+            self.radix = RadixCache(self.kv)
         self._kv_released = False
 
     def num_queued(self) -> int:
@@ -469,14 +478,27 @@ class Engine:
         while self.waiting and len(self.running) < self.max_running \
                 and admit_budget > 0:
             nxt = self.waiting[0]
-            need = min(len(nxt.input_ids) - nxt.prefill_pos, admit_budget)
-            if not self.kv.can_allocate(need):
+            radix_pages: List[int] = []
+            radix_len = 0
+            if self.radix is not None and nxt.prefill_pos == 0:
+                radix_pages, radix_len = self.radix.match(nxt.input_ids)
+            need = min(len(nxt.input_ids) - nxt.prefill_pos - radix_len,
+                       admit_budget)
+            if not self._ensure_pages(need):
+                if radix_pages:
+                    self.kv.unref_pages(radix_pages)
                 break
             admit_budget -= need
             self.waiting.pop(0)
             sid = self._seq_counter
             self._seq_counter += 1
             self._seq_ids[nxt.rid] = sid
+            if radix_len:
+                # cached prompt pages seed the page table; prefill resumes
+                # at the page-aligned boundary (chunked-prefill path
+                # already attends to the full cached history)
+                self.kv.seed_seq(sid, radix_pages, radix_len)
+                nxt.prefill_pos = radix_len
             self.running.append(nxt)
             candidates.append(nxt)
         for r in candidates:
@@ -486,7 +508,8 @@ class Engine:
             take = min(need, token_budget)
             if take <= 0:
                 continue
-            if not self.kv.allocate(self._seq_ids[r.rid], take):
+            if not self._ensure_pages(take) or \
+                    not self.kv.allocate(self._seq_ids[r.rid], take):
                 break
             prefill_reqs.append(r)
             prefill_lens.append(take)
@@ -500,7 +523,8 @@ class Engine:
         if decode_reqs:
             ok = []
             for r in decode_reqs:
-                if self.kv.allocate(self._seq_ids[r.rid], 1):
+                if self._ensure_pages(1) and \
+                        self.kv.allocate(self._seq_ids[r.rid], 1):
                     ok.append(r)
                 else:
                     # out of KV pages: abort (scheduler continues elsewhere)
@@ -533,9 +557,29 @@ class Engine:
         return [outs[f"{rid_prefix}-{i}"] for i in range(len(prompts))]
 
     # -------------------------------------------------------------- internals
+    def _ensure_pages(self, num_tokens: int) -> bool:
+        """Free-list check with radix-cache LRU eviction under pressure."""
+        if self.kv.can_allocate(num_tokens):
+            return True
+        if self.radix is not None:
+            need = (num_tokens + self.kv.page_size - 1) // self.kv.page_size \
+                - self.kv.free_pages
+            self.radix.evict(need)
+        return self.kv.can_allocate(num_tokens)
+
+    def flush_radix(self):
+        """Drop cached prefixes (stale after a weight update — the
+        reference flushes SGLang's cache post-update, patches.py:360-387)."""
+        if self.radix is not None:
+            self.radix.reset()
+
     def _emit(self, r: Request) -> RequestOutput:
         sid = self._seq_ids.pop(r.rid, None)
         if sid is not None:
+            if self.radix is not None:
+                written = self.kv.seq_len(sid)   # last sampled token has no KV
+                toks = (r.input_ids + r.output_ids)[:written]
+                self.radix.insert(toks, self.kv._seq_pages.get(sid, []))
             self.kv.free_seq(sid)
         if not r.finish_reason:
             r.finish_reason = "length"

@@ -81,6 +81,24 @@ class PagedKVCache:
                 self._free.append(p)
         self._seq_len.pop(seq_id, None)
 
+    def seed_seq(self, seq_id: int, pages: List[int], length: int):
+        """Start a sequence whose first ``length`` (page-aligned) tokens
+        live in already-filled shared pages (radix-cache hit).  Takes
+        OWNERSHIP of one pre-taken ref per page (RadixCache.match)."""
+        assert seq_id not in self._seq_pages and length % self.page_size == 0
+        self._seq_pages[seq_id] = list(pages)
+        self._seq_len[seq_id] = length
+
+    def unref_pages(self, pages: List[int]):
+        """Drop refs taken by RadixCache.match on an admission bail-out."""
+        for p in pages:
+            r = self._ref.get(p, 1) - 1
+            if r <= 0:
+                self._ref.pop(p, None)
+                self._free.append(p)
+            else:
+                self._ref[p] = r
+
     def fork_seq(self, parent_id: int, child_ids: List[int],
                  prefix_len: int) -> Optional[List[Tuple[int, int]]]:
         """Prefix sharing: every child references the parent's FULL pages
@@ -135,3 +153,140 @@ class PagedKVCache:
             pages = self._seq_pages[s]
             pt[i, :len(pages)] = torch.tensor(pages, dtype=torch.int32)
         return pt
+
+
+class _RadixNode:
+    __slots__ = ("chunk", "page", "children", "parent", "t")
+
+    def __init__(self, chunk, page, parent):
+        self.chunk = chunk          # tuple of page_size token ids
+        self.page = page            # page id this node pins
+        self.children: Dict[tuple, "_RadixNode"] = {}
+        self.parent = parent
+        self.t = 0                  # LRU clock
+
+
+class RadixCache:
+    """Cross-request KV prefix cache (the SGLang radix-cache capability the
+    reference relies on for shared system prompts / multi-turn reuse —
+    SURVEY.md §2.2.2 kernel-suite row, §2.4.3).
+
+    Page-granular trie: each node is exactly ``page_size`` tokens pinning one
+    KV page (+1 on the allocator refcount).  A new request walks the trie
+    over its prompt in page chunks; matched pages seed its page table
+    (skipping their prefill entirely) and only the tail is computed.
+    Finished sequences donate their full pages back into the trie.  Under
+    page pressure the engine evicts LRU leaves (pages still referenced by
+    running sequences survive the node's removal via the refcount).
+
+    MI355X sizing note: at 288 GB HBM the KV pool is large enough that
+    eviction is rare — the tree mostly just grows, which is the cheap path.
+    """
+
+    def __init__(self, kv: PagedKVCache):
+        self.kv = kv
+        self.ps = kv.page_size
+        self.root: Dict[tuple, _RadixNode] = {}
+        self._clock = 0
+        self._nodes = 0
+        self.hit_tokens = 0
+        self.query_tokens = 0
+
+    # ---------------------------------------------------------------- match
+    def match(self, tokens: List[int]) -> Tuple[List[int], int]:
+        """Longest page-aligned cached prefix of ``tokens`` that still
+        leaves >= 1 token to prefill.  Takes ONE allocator ref per returned
+        page (the caller hands them to ``seed_seq`` — which takes ownership
+        of those refs — or must ``unref_pages`` on bail-out)."""
+        self._clock += 1
+        self.query_tokens += len(tokens)
+        max_match = ((len(tokens) - 1) // self.ps) * self.ps
+        pages: List[int] = []
+        level = self.root
+        off = 0
+        while off < max_match:
+            chunk = tuple(tokens[off:off + self.ps])
+            node = level.get(chunk)
+            if node is None:
+                break
+            node.t = self._clock
+            pages.append(node.page)
+            level = node.children
+            off += self.ps
+        for p in pages:
+            self.kv._ref[p] = self.kv._ref.get(p, 0) + 1
+        self.hit_tokens += off
+        return pages, off
+
+    # --------------------------------------------------------------- insert
+    def insert(self, tokens: List[int], pages: List[int]) -> int:
+        """Absorb a finished sequence's FULL pages into the trie.  ``pages``
+        must align 1:1 with page-sized chunks of ``tokens``.  Existing nodes
+        are just touched (the donor's duplicate page stays with the donor);
+        new nodes take +1 ref on the donated page.  Returns nodes added."""
+        self._clock += 1
+        added = 0
+        level = self.root
+        for i in range(len(tokens) // self.ps):
+            chunk = tuple(tokens[i * self.ps:(i + 1) * self.ps])
+            node = level.get(chunk)
+            if node is None:
+                node = _RadixNode(chunk, pages[i], None)
+                node.t = self._clock
+                level[chunk] = node
+                self.kv._ref[pages[i]] = self.kv._ref.get(pages[i], 0) + 1
+                self._nodes += 1
+                added += 1
+            else:
+                node.t = self._clock
+            level = node.children
+        return added
+
+    # --------------------------------------------------------------- evict
+    def _leaves(self):
+        out = []
+        stack = [(self.root, None)]
+        while stack:
+            level, parent = stack.pop()
+            for chunk, node in level.items():
+                if node.children:
+                    stack.append((node.children, node))
+                else:
+                    out.append((node, level, chunk))
+        return out
+
+    def evict(self, need_pages: int) -> int:
+        """Drop LRU leaves until >= need_pages returned to the free list (or
+        the tree is empty).  Only pages whose last reference was the cache's
+        actually free; others (still used by running seqs) just unpin."""
+        freed = 0
+        while freed < need_pages:
+            leaves = self._leaves()
+            if not leaves:
+                break
+            leaves.sort(key=lambda x: x[0].t)
+            progressed = False
+            for node, level, chunk in leaves:
+                if freed >= need_pages:
+                    break
+                del level[chunk]
+                self._nodes -= 1
+                progressed = True
+                r = self.kv._ref.get(node.page, 1) - 1
+                if r <= 0:
+                    self.kv._ref.pop(node.page, None)
+                    self.kv._free.append(node.page)
+                    freed += 1
+                else:
+                    self.kv._ref[node.page] = r
+            if not progressed:
+                break
+        return freed
+
+    def reset(self):
+        """Drop the whole tree (weight update => cached KV is stale)."""
+        self.evict(1 << 62)
+
+    @property
+    def num_nodes(self) -> int:
+        return self._nodes

@@ -143,5 +143,6 @@ class EngineRunner:
             self.engine.model.load_state_dict(state_dict, strict=strict)
             if abort_in_flight and self.engine.has_work():
                 self.engine.abort_request(abort_all=True)
+            self.engine.flush_radix()   # cached KV is stale on new weights
             self.weight_version = version
         self._wake.set()

@@ -105,7 +105,9 @@ def create_app(engine: Engine, runner: Optional[EngineRunner] = None):
 
     @app.post("/flush_cache")
     async def flush_cache():
-        # paged KV frees per request; full flush = drop any retained state
+        # paged KV frees per request; the retained state is the radix
+        # prefix cache — drop it
+        engine.flush_radix()
         return {"status": "ok"}
 
     @app.post("/update_weights_from_agent")

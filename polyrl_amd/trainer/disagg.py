@@ -113,6 +113,7 @@ def rollout_serve_loop(cfg, model_cfg, rank: int, device: str, dtype,
         if op == "publish":
             with runner.lock:  # excludes in-flight generation during swap
                 version = plane.receive(apply_weight)
+                engine.flush_radix()
             runner.weight_version = version
         elif op == "exit":
             break

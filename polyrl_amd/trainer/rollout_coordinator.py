@@ -196,3 +196,4 @@ class LocalRolloutCoordinator:
     # -------------------------------------------------------------- weights
     def update_weights(self, state_dict):
         self.engine.model.load_state_dict(state_dict, strict=False)
+        self.engine.flush_radix()

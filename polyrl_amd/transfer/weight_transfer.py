@@ -67,6 +67,10 @@ class WeightPublisher:
                 if emb is not None:
                     full = emb.full_tensor() if hasattr(emb, "full_tensor") else emb
                     c.update_named("lm_head.weight", full)
+        for c in self.consumers:
+            eng = getattr(c, "_engine_owner", None)
+            if eng is not None:
+                eng.flush_radix()   # cached KV prefixes are stale now
         if dist.is_available() and dist.is_initialized():
             dist.barrier()
         self.last_publish_s = time.perf_counter() - t0

@@ -225,3 +225,103 @@ def test_release_resume_memory(setup):
     outs = eng.generate([prompt], SamplingParams(temperature=0.0,
                                                  max_new_tokens=3), "post")
     assert len(outs[0].output_ids) == 3
+
+
+# ------------------------------------------------------- radix prefix cache
+
+
+def _fresh_engine(cfg, model, kv_bytes=8 << 20, radix=True, **kw):
+    e = Engine(cfg, device="cpu", dtype=torch.float32,
+               kv_bytes_budget=kv_bytes, enable_radix_cache=radix, **kw)
+    e.model.load_state_dict(model.state_dict())
+    return e
+
+
+def test_radix_cache_hits_and_greedy_equality(setup):
+    """Cross-request prefix reuse (SGLang radix-cache capability,
+    SURVEY.md §2.2.2): a shared 40-token system prompt is prefilled once;
+    later requests seed their page table from the trie and skip it.
+    Greedy outputs must equal the no-cache engine's exactly."""
+    cfg, model, _ = setup
+    sys_prefix = list(range(100, 140))
+    p1, p2 = sys_prefix + [7, 8, 9], sys_prefix + [21, 22]
+    sp = SamplingParams(temperature=0.0, max_new_tokens=6)
+    e0 = _fresh_engine(cfg, model, radix=False)
+    ref = {i: e0.generate([p], sp, f"r{i}")[0].output_ids
+           for i, p in enumerate((p1, p2))}
+    e = _fresh_engine(cfg, model)
+    o1 = e.generate([p1], sp, "a")[0]
+    assert e.radix.num_nodes > 0          # finished seq donated its pages
+    h0 = e.radix.hit_tokens
+    o2 = e.generate([p2], sp, "b")[0]
+    assert e.radix.hit_tokens - h0 >= 32  # 2 full pages of the shared prefix
+    assert o1.output_ids == ref[0] and o2.output_ids == ref[1]
+    # generated tokens are cached too (multi-turn / continuation reuse):
+    # re-asking p1 hits its full cached pages
+    h1 = e.radix.hit_tokens
+    assert e.generate([p1], sp, "c")[0].output_ids == ref[0]
+    assert e.radix.hit_tokens - h1 >= 32
+    e.flush_radix()
+    assert e.radix.num_nodes == 0
+    assert e.kv.free_pages == e.kv.num_pages   # accounting closes
+
+
+def test_radix_eviction_under_page_pressure(setup):
+    """A tiny KV pool forces LRU leaf eviction: new work must still
+    complete, and accounting must close when the tree is flushed."""
+    cfg, model, _ = setup
+    # ~24 pages total: each 40+6-token request needs 3; cache grows until
+    # the allocator runs dry and eviction kicks in
+    from polyrl_amd.rollout.kv_cache import PagedKVCache
+    bt = PagedKVCache.bytes_per_token(cfg.num_hidden_layers,
+                                      cfg.num_key_value_heads, cfg.head_dim)
+    e = _fresh_engine(cfg, model, kv_bytes=bt * 16 * 24)
+    assert e.kv.num_pages <= 32
+    sp = SamplingParams(temperature=0.0, max_new_tokens=6)
+    e0 = _fresh_engine(cfg, model, radix=False, kv_bytes=bt * 16 * 24)
+    for i in range(12):
+        p = [(1000 + 37 * i + j) % cfg.vocab_size for j in range(40)]
+        out = e.generate([p], sp, f"q{i}")[0]
+        refo = e0.generate([p], sp, f"q{i}")[0]
+        assert out.output_ids == refo.output_ids, i
+        assert out.finish_reason == "length"
+    e.flush_radix()
+    assert e.kv.free_pages == e.kv.num_pages
+
+
+def test_radix_composes_with_group_fork(setup):
+    """Group prefix-sharing (n samples) on top of a radix-cached system
+    prompt: greedy children equal the separate-request reference."""
+    cfg, model, _ = setup
+    sys_prefix = list(range(200, 240))
+    prompt = sys_prefix + [3, 4, 5]
+    sp = SamplingParams(temperature=0.0, max_new_tokens=5)
+    e0 = _fresh_engine(cfg, model, radix=False)
+    ref = e0.generate([prompt], sp, "r")[0].output_ids
+    e = _fresh_engine(cfg, model)
+    e.generate([sys_prefix + [99]], sp, "warm")      # populate the trie
+    h0 = e.radix.hit_tokens
+    e.add_request_group("grp", prompt, sp, 3)
+    outs = {}
+    while e.has_work():
+        for o in e.step():
+            outs[o.rid] = o
+    assert e.radix.hit_tokens > h0                   # parent hit the prefix
+    for s in range(3):
+        assert outs[f"grp-s{s}"].output_ids == ref
+    assert not e.has_work()
+
+
+def test_radix_flushed_on_weight_update(setup):
+    """Weight swap invalidates cached KV (the reference flushes SGLang's
+    cache after /update_weights — patches.py:360-387)."""
+    from polyrl_amd.rollout.runner import EngineRunner
+    cfg, model, _ = setup
+    e = _fresh_engine(cfg, model)
+    sp = SamplingParams(temperature=0.0, max_new_tokens=4)
+    e.generate([list(range(300, 340))], sp, "w")
+    assert e.radix.num_nodes > 0
+    runner = EngineRunner(e)
+    sd = {k: v + 0.01 for k, v in model.state_dict().items()}
+    runner.update_weights(sd, version=1)
+    assert e.radix.num_nodes == 0

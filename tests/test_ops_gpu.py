@@ -547,3 +547,44 @@ def test_group_prefix_sharing_gpu():
         assert outs[f"grp-s{s}"].output_ids == sep[s].output_ids, \
             (s, outs[f"grp-s{s}"].output_ids, sep[s].output_ids)
     assert eng.kv.free_pages == eng.kv.num_pages
+
+
+@pytest.mark.gpu
+def test_engine_radix_cache_greedy_equality_gpu():
+    """Radix prefix cache on the HIP kernel path: a request hitting a
+    cached system prompt must produce the same greedy tokens as a fresh
+    no-cache engine (cached KV pages == recomputed KV pages)."""
+    from polyrl_amd.models import create_model
+    from polyrl_amd.models.registry import DecoderConfig
+    from polyrl_amd.rollout.engine import Engine, SamplingParams
+    cfg = DecoderConfig(arch="llama", vocab_size=512, hidden_size=256,
+                        intermediate_size=512, num_hidden_layers=2,
+                        num_attention_heads=2, num_key_value_heads=1,
+                        head_dim=128, max_position_embeddings=256,
+                        rope_theta=10000.0, rms_norm_eps=1e-6)
+    torch.manual_seed(60)
+    model = create_model(cfg, kind="actor", dtype="bfloat16", device=DEV)
+    sys_prefix = [int(t) for t in
+                  torch.randint(0, cfg.vocab_size, (40,))]
+    p1 = sys_prefix + [7, 8, 9]
+    p2 = sys_prefix + [21, 22]
+    sp = SamplingParams(temperature=0.0, max_new_tokens=6)
+
+    e0 = Engine(cfg, device=DEV, dtype=torch.bfloat16,
+                kv_bytes_budget=32 << 20)
+    e0.model.load_state_dict(model.state_dict())
+    ref1 = e0.generate([p1], sp, "r1")[0].output_ids
+    ref2 = e0.generate([p2], sp, "r2")[0].output_ids
+
+    e = Engine(cfg, device=DEV, dtype=torch.bfloat16,
+               kv_bytes_budget=32 << 20, enable_radix_cache=True)
+    e.model.load_state_dict(model.state_dict())
+    o1 = e.generate([p1], sp, "a")[0].output_ids
+    assert e.radix.num_nodes > 0
+    h0 = e.radix.hit_tokens
+    o2 = e.generate([p2], sp, "b")[0].output_ids
+    assert e.radix.hit_tokens - h0 >= 32, "no radix hit on shared prefix"
+    assert o1 == ref1, (o1, ref1)
+    assert o2 == ref2, (o2, ref2)
+    e.flush_radix()
+    assert e.kv.free_pages == e.kv.num_pages
