@@ -130,3 +130,32 @@ def test_tp2_rollout_fsdp_trainer_world2(tmp_path):
     ])
     assert r.returncode == 0, \
         f"stdout:\n{r.stdout[-3000:]}\nstderr:\n{r.stderr[-3000:]}"
+
+
+@pytest.mark.timeout(600)
+def test_world4_grpo_stream(tmp_path):
+    """World-size 4 on gloo (SURVEY.md §4 tier 3: step parity at world
+    sizes beyond 2): streamed GRPO step + per-rank sharded checkpoint."""
+    r = _run_torchrun([
+        "actor_rollout_ref.model.path=llama-debug-cpu",
+        "actor_rollout_ref.model.dtype=float32",
+        "actor_rollout_ref.model.enable_gradient_checkpointing=false",
+        "actor_rollout_ref.actor.ppo_mini_batch_size=8",
+        "actor_rollout_ref.actor.ppo_max_token_len_per_gpu=512",
+        "actor_rollout_ref.rollout.sampling.n=2",
+        "actor_rollout_ref.rollout.response_length=8",
+        "actor_rollout_ref.rollout.min_stream_batch_size=4",
+        "data.train_batch_size=8",
+        "data.max_prompt_length=16",
+        "data.synthetic_num_prompts=32",
+        f"trainer.default_local_dir={tmp_path}/ckpt",
+        "trainer.save_freq=1",
+        "trainer.resume_mode=disable",
+        "reward=random",
+        "max_steps=1",
+    ], nproc=4)
+    assert r.returncode == 0, f"stdout:\n{r.stdout[-3000:]}\nstderr:\n{r.stderr[-3000:]}"
+    d = tmp_path / "ckpt" / "global_step_1" / "actor"
+    files = sorted(os.listdir(d))
+    for rk in range(4):
+        assert f"model_world_size_4_rank_{rk}.pt" in files
