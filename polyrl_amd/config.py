@@ -119,6 +119,10 @@ class RolloutConfig:
     max_running_requests: int = 256
     page_size: int = 16                  # KV tokens per page
     decode_chunk_size: int = 16          # device-resident decode chunk
+    # cross-request KV prefix reuse (radix trie).  Off by default in RL
+    # training: weights change every step and the cache is flushed on each
+    # install; turn on for serving / multi-turn / shared system prompts.
+    enable_radix_cache: bool = False
     # disaggregated split (BASELINE config #4): the LAST num_rollout_ranks
     # ranks of the world serve rollout; 0 = co-located
     num_rollout_ranks: int = 0

@@ -280,13 +280,19 @@ def main():
                         "the interface that routes to the manager)")
     p.add_argument("--load", default=None,
                    help="safetensors checkpoint to load (else random init)")
+    p.add_argument("--radix-cache", action="store_true", default=True,
+                   help="cross-request KV prefix reuse (default on for "
+                        "serving; RL trainers flush it every publish)")
+    p.add_argument("--no-radix-cache", dest="radix_cache",
+                   action="store_false")
     args = p.parse_args()
 
     cfg = get_config(args.model)
     device = "cuda" if torch.cuda.is_available() else "cpu"
     dtype = getattr(torch, args.dtype) if device == "cuda" else torch.float32
     engine = Engine(cfg, device=device, dtype=dtype,
-                    kv_bytes_budget=int(args.kv_gb * (1 << 30)))
+                    kv_bytes_budget=int(args.kv_gb * (1 << 30)),
+                    enable_radix_cache=args.radix_cache)
     if args.load:
         from safetensors.torch import load_file
         engine.model.load_state_dict(load_file(args.load), strict=False)

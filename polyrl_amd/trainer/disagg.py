@@ -88,6 +88,7 @@ def rollout_serve_loop(cfg, model_cfg, rank: int, device: str, dtype,
                     max_num_batched_tokens=ro.max_num_batched_tokens,
                     max_model_len=ro.prompt_length + ro.response_length,
                     decode_chunk_size=ro.decode_chunk_size,
+                    enable_radix_cache=ro.enable_radix_cache,
                     seed=cfg.trainer.seed)
     runner = EngineRunner(engine)
     app = create_app(engine, runner)

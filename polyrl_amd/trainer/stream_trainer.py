@@ -176,6 +176,7 @@ class StreamPPOTrainer:
                                  max_num_batched_tokens=ro.max_num_batched_tokens,
                                  max_model_len=ro.prompt_length + ro.response_length,
                                  decode_chunk_size=ro.decode_chunk_size,
+                                 enable_radix_cache=ro.enable_radix_cache,
                                  tp_ctx=tp_ctx,
                                  seed=eng_seed)
             if self.elastic:
