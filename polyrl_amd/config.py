@@ -123,6 +123,9 @@ class RolloutConfig:
     # training: weights change every step and the cache is flushed on each
     # install; turn on for serving / multi-turn / shared system prompts.
     enable_radix_cache: bool = False
+    # multi-turn rollouts: declared for config parity (the reference's
+    # RolloutConfig carries it; the stream path doesn't exercise it)
+    multi_turn: bool = False
     # disaggregated split (BASELINE config #4): the LAST num_rollout_ranks
     # ranks of the world serve rollout; 0 = co-located
     num_rollout_ranks: int = 0

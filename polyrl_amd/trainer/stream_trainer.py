@@ -301,6 +301,15 @@ class StreamPPOTrainer:
         sp = c.actor_rollout_ref.actor.ulysses_sequence_parallel_size
         if sp > 1:
             assert world % sp == 0, f"world {world} % sp {sp} != 0"
+        # reference fences kept 1:1 (stream_dp_actor.py:145-146, multi-turn)
+        if c.actor_rollout_ref.actor.ppo_epochs != 1:
+            raise NotImplementedError(
+                "ppo_epochs != 1 is fenced (matches the reference: "
+                "stream_dp_actor.py:145-146 — streamed minibatches are "
+                "consumed once)")
+        if ro.multi_turn:
+            raise NotImplementedError("multi-turn rollouts are declared for "
+                                      "config parity but not implemented")
 
     # ------------------------------------------------------------------- fit
     def fit(self, max_steps: Optional[int] = None):

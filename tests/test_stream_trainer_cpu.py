@@ -202,3 +202,16 @@ def test_esi_expiry_forces_checkpoint(tmp_path, monkeypatch):
     trainer.fit(max_steps=1)
     ckpts = list((tmp_path / "ckpt").glob("global_step_*"))
     assert ckpts, "ESI expiry did not trigger a checkpoint"
+
+
+def test_reference_fences(tmp_path):
+    """Config fences the reference also has: ppo_epochs != 1 and multi-turn
+    raise NotImplementedError at construction."""
+    cfg = tiny_config(tmp_path)
+    cfg.actor_rollout_ref.actor.ppo_epochs = 2
+    with pytest.raises(NotImplementedError):
+        StreamPPOTrainer(cfg, reward_fn=load_reward_manager("random"))
+    cfg2 = tiny_config(tmp_path)
+    cfg2.actor_rollout_ref.rollout.multi_turn = True
+    with pytest.raises(NotImplementedError):
+        StreamPPOTrainer(cfg2, reward_fn=load_reward_manager("random"))
