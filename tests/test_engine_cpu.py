@@ -325,3 +325,42 @@ def test_radix_flushed_on_weight_update(setup):
     sd = {k: v + 0.01 for k, v in model.state_dict().items()}
     runner.update_weights(sd, version=1)
     assert e.radix.num_nodes == 0
+
+
+def test_radix_hit_with_chunked_prefill_tail(setup):
+    """Radix-seeded request whose uncached tail spans multiple prefill
+    chunks (max_num_batched_tokens < tail length): greedy equality holds
+    (the chunked-prefill path attends to cached + freshly-written history)."""
+    cfg, model, _ = setup
+    sys_prefix = list(range(50, 82))          # 2 full pages
+    tail = [(7 * j + 3) % cfg.vocab_size for j in range(40)]
+    prompt = sys_prefix + tail
+    sp = SamplingParams(temperature=0.0, max_new_tokens=5)
+    e0 = _fresh_engine(cfg, model, radix=False)
+    ref = e0.generate([prompt], sp, "r")[0].output_ids
+    # chunk budget 16 => the 40-token tail prefills over 3 chunks
+    e = _fresh_engine(cfg, model, max_num_batched_tokens=16)
+    e.generate([sys_prefix + [1, 2]], sp, "warm")
+    h0 = e.radix.hit_tokens
+    out = e.generate([prompt], sp, "q")[0]
+    assert e.radix.hit_tokens - h0 >= 32
+    assert out.output_ids == ref
+
+
+def test_radix_abort_mid_prefill_accounting(setup):
+    """Abort a radix-seeded request before its prefill completes: pages
+    (cached + fresh) must all return / stay cache-owned — accounting closes
+    after a flush."""
+    cfg, model, _ = setup
+    sys_prefix = list(range(50, 82))
+    prompt = sys_prefix + [(11 * j + 1) % cfg.vocab_size for j in range(40)]
+    sp = SamplingParams(temperature=0.0, max_new_tokens=5)
+    e = _fresh_engine(cfg, model, max_num_batched_tokens=16)
+    e.generate([sys_prefix + [9]], sp, "warm")
+    e.add_request("victim", prompt, sp)
+    e.step()                                   # admit + first chunk only
+    e.abort_request("victim")
+    while e.has_work():
+        e.step()
+    e.flush_radix()
+    assert e.kv.free_pages == e.kv.num_pages
