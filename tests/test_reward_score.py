@@ -86,3 +86,28 @@ def test_parquet_ground_truth_roundtrip(tmp_path):
     assert list(b.non_tensors["ground_truth"]) == ["3", "4"]
     assert list(b.non_tensors["data_source"]) == ["gsm8k", "gsm8k"]
     assert b["input_ids"].shape == (2, 8)
+
+
+def test_gsm8k_preprocess_to_dataset_roundtrip(tmp_path):
+    """examples/data_preprocess/gsm8k.py writes the schema
+    ParquetRLHFDataset + the gsm8k scorer consume (reference:
+    examples/data_preprocess/openr1.py capability)."""
+    import subprocess
+    import sys
+
+    out = tmp_path / "g.parquet"
+    r = subprocess.run(
+        [sys.executable, "examples/data_preprocess/gsm8k.py",
+         "--synthetic", "6", "--out", str(out)],
+        capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stderr[-2000:]
+    from polyrl_amd.data import ParquetRLHFDataset
+    from polyrl_amd.reward_score import default_compute_score
+    ds = ParquetRLHFDataset([str(out)], max_prompt_length=64,
+                            tokenizer=lambda s: [ord(c) % 512 for c in s][:64])
+    b = ds.batch([0, 1, 2])
+    gts = b.non_tensors["ground_truth"]
+    assert default_compute_score("openai/gsm8k", f"x #### {gts[0]}",
+                                 str(gts[0])) == 1.0
+    assert default_compute_score("openai/gsm8k", "x #### 999999",
+                                 str(gts[0])) == 0.0
