@@ -20,9 +20,9 @@ import torch
 import torch.distributed as dist
 
 
-def _world_rank():
+def _world_rank(pg=None):
     if dist.is_available() and dist.is_initialized():
-        return dist.get_world_size(), dist.get_rank()
+        return dist.get_world_size(pg), dist.get_rank(pg)
     return 1, 0
 
 
@@ -50,22 +50,40 @@ def _copy_into(dst_sd: Dict[str, Any], src: Dict[str, Any]):
 
 
 def _load_optim_state(optimizer, src: Dict[str, Any]):
-    """Copy tensor state into DTensor-aware optimizer state in place after a
-    structural load."""
+    """Structural load, then re-wrap plain local shards as DTensors for
+    DTensor (FSDP2) params — Adam kernels refuse mixed Tensor/DTensor
+    state (aten.lerp_ "mixed torch.Tensor and DTensor")."""
     optimizer.load_state_dict(src)
+    try:
+        from torch.distributed.tensor import DTensor
+    except ImportError:
+        return
+    for group in optimizer.param_groups:
+        for prm in group["params"]:
+            st = optimizer.state.get(prm)
+            if not st or not isinstance(prm, DTensor):
+                continue
+            for k, v in st.items():
+                if torch.is_tensor(v) and v.dim() > 0 \
+                        and not isinstance(v, DTensor):
+                    st[k] = DTensor.from_local(
+                        v.to(prm.device), prm.device_mesh, prm.placements)
 
 
 class CheckpointManager:
-    def __init__(self, root: str, role: str = "actor"):
+    def __init__(self, root: str, role: str = "actor", pg=None):
         self.root = root
         self.role = role
+        # collective scope: the trainer subgroup in disaggregated mode —
+        # a WORLD barrier would deadlock (rollout ranks never save)
+        self.pg = pg
 
     def _dir(self, step: int) -> str:
         return os.path.join(self.root, f"global_step_{step}", self.role)
 
     def save(self, step: int, model, optimizer=None, lr_scheduler=None,
              extra: Optional[Dict[str, Any]] = None):
-        world, rank = _world_rank()
+        world, rank = _world_rank(self.pg)
         d = self._dir(step)
         os.makedirs(d, exist_ok=True)
         torch.save(_localize(model.state_dict()),
@@ -86,7 +104,7 @@ class CheckpointManager:
                                    "latest_checkpointed_iteration.txt"), "w") as f:
                 f.write(str(step))
         if dist.is_available() and dist.is_initialized():
-            dist.barrier()
+            dist.barrier(group=self.pg)
 
     def load(self, model, optimizer=None, lr_scheduler=None,
              step: Optional[int] = None) -> Optional[Dict[str, Any]]:
@@ -94,7 +112,7 @@ class CheckpointManager:
             step = self.latest_step()
             if step is None:
                 return None
-        world, rank = _world_rank()
+        world, rank = _world_rank(self.pg)
         d = self._dir(step)
         msd = torch.load(os.path.join(
             d, f"model_world_size_{world}_rank_{rank}.pt"),

@@ -205,9 +205,10 @@ class StreamPPOTrainer:
         self.reward_fn = reward_fn or load_reward_manager("constant")
 
         self.ckpt_actor = CheckpointManager(config.trainer.default_local_dir,
-                                            "actor")
-        self.ckpt_critic = CheckpointManager(config.trainer.default_local_dir,
-                                             "critic") if self.use_critic else None
+                                            "actor", pg=self.pg)
+        self.ckpt_critic = CheckpointManager(
+            config.trainer.default_local_dir, "critic",
+            pg=self.pg) if self.use_critic else None
         self.tracking = Tracking(config.trainer.project_name,
                                  config.trainer.experiment_name,
                                  config.trainer.logger if self.rank == 0 else [],

@@ -198,6 +198,7 @@ def test_disagg_with_ulysses_sp2(tmp_path):
          "actor_rollout_ref.rollout.min_stream_batch_size=4",
          "actor_rollout_ref.rollout.num_rollout_ranks=1",
          "actor_rollout_ref.rollout.rollout_port_base=31830",
+         "trainer.save_freq=1",
          "data.train_batch_size=8",
          "data.max_prompt_length=16",
          "data.synthetic_num_prompts=32",

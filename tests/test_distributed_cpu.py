@@ -159,3 +159,34 @@ def test_world4_grpo_stream(tmp_path):
     files = sorted(os.listdir(d))
     for rk in range(4):
         assert f"model_world_size_4_rank_{rk}.pt" in files
+
+
+@pytest.mark.timeout(600)
+def test_world2_resume_with_optimizer_state(tmp_path):
+    """FSDP2 sharded checkpoint round-trip across PROCESS restarts at
+    world 2: save at step 2, auto-resume, train one more step (exercises
+    plain-shard -> DTensor optimizer state load)."""
+    base = [
+        "actor_rollout_ref.model.path=llama-debug-cpu",
+        "actor_rollout_ref.model.dtype=float32",
+        "actor_rollout_ref.model.enable_gradient_checkpointing=false",
+        "actor_rollout_ref.actor.ppo_mini_batch_size=8",
+        "actor_rollout_ref.actor.ppo_max_token_len_per_gpu=512",
+        "actor_rollout_ref.rollout.sampling.n=2",
+        "actor_rollout_ref.rollout.response_length=8",
+        "actor_rollout_ref.rollout.min_stream_batch_size=4",
+        "data.train_batch_size=8",
+        "data.max_prompt_length=16",
+        "data.synthetic_num_prompts=32",
+        f"trainer.default_local_dir={tmp_path}/ckpt",
+        "trainer.save_freq=2",
+        "reward=random",
+    ]
+    r = _run_torchrun(base + ["trainer.resume_mode=disable", "max_steps=2"])
+    assert r.returncode == 0, f"{r.stdout[-3000:]}\n{r.stderr[-3000:]}"
+    assert (tmp_path / "ckpt" / "global_step_2").is_dir()
+    r = _run_torchrun(base + ["trainer.resume_mode=auto", "max_steps=1"])
+    assert r.returncode == 0, f"{r.stdout[-3000:]}\n{r.stderr[-3000:]}"
+    # resumed run continued from step 2 -> saved step 4? (save_freq=2 and
+    # one more step lands on global_step 3: no new dir).  The load itself
+    # succeeding (incl. optimizer state into DTensor params) is the assert.
