@@ -46,6 +46,102 @@ def create_manager_app(scheduler, loop: Optional[asyncio.AbstractEventLoop]
         return {"status": "registered", "instance_id": inst.instance_id,
                 "weight_version": scheduler.latest_weight_version}
 
+    @app.post("/register_local_rollout_instances")
+    async def register_local_rollout_instances(request: Request):
+        """Batch-register co-located engines by address
+        (handlers.rs route; local = trainer-GPU engines subject to the
+        local time-box)."""
+        body = await request.json()
+        from ..server import HttpInstance
+        ids = []
+        for addr in body["addrs"]:
+            inst = HttpInstance(addr, is_local=True,
+                                weight_state_fn=remote_weight_state_fn)
+            await _run(scheduler.register_instance(inst))
+            ids.append(inst.instance_id)
+        return {"status": "registered", "instance_ids": ids}
+
+    def _parse_group(gid, body) -> "GroupRequest":
+        sp = body.get("sampling_params", {})
+        from .types import GroupRequest, SamplingSpec
+        return GroupRequest(
+            gid=int(gid), input_ids=list(body["input_ids"]),
+            n=int(sp.get("n", 1)),
+            sampling=SamplingSpec(
+                temperature=float(sp.get("temperature", 1.0)),
+                top_k=int(sp.get("top_k", -1)),
+                top_p=float(sp.get("top_p", 1.0)),
+                max_new_tokens=int(sp.get("max_new_tokens", 128)),
+                stop_token_ids=tuple(sp.get("stop_token_ids", ()))),
+            return_logprob=bool(body.get("return_logprob", True)))
+
+    def _sample_json(smp):
+        return {"output_ids": smp.output_ids,
+                "output_logprobs": smp.output_logprobs,
+                "finish_reason": smp.finish_reason,
+                "completion_tokens": smp.completion_tokens,
+                "num_migrations": smp.num_migrations}
+
+    @app.post("/generate")
+    async def generate(request: Request):
+        """Relay one prompt group through the scheduler (continuation /
+        RR / version gating apply — handlers.rs:330-418)."""
+        body = await request.json()
+        req = _parse_group(body.get("gid", 0), body)
+        res = await _run(scheduler.process_group(req))
+        return {"gid": res.gid, "instance_ids": res.instance_ids,
+                "samples": [_sample_json(x) for x in res.samples]}
+
+    @app.post("/batch_generate_requests")
+    async def batch_generate_requests(request: Request):
+        """The reference's NDJSON streaming contract
+        (handlers.rs:442-564): body [[gid, generate_request], ...] (+
+        optional max_local_gen_s); response streams one JSON line per
+        completed group as it finishes, first line = submit notifier."""
+        import json as _json
+
+        from fastapi.responses import StreamingResponse
+        body = await request.json()
+        pairs = body if isinstance(body, list) else body["requests"]
+        window = None if isinstance(body, list) \
+            else body.get("max_local_gen_s")
+        groups = [_parse_group(gid, g) for gid, g in pairs]
+
+        import queue as _q
+        out_q: _q.Queue = _q.Queue()
+
+        async def pump():
+            try:
+                async for item in scheduler.submit_batch(
+                        groups, max_local_gen_s=window):
+                    if isinstance(item, dict):
+                        out_q.put(_json.dumps(item))
+                    else:
+                        out_q.put(_json.dumps(
+                            {"gid": item.gid,
+                             "instance_ids": item.instance_ids,
+                             "samples": [_sample_json(x)
+                                         for x in item.samples]}))
+            finally:
+                out_q.put(None)
+
+        if loop is None or loop is asyncio.get_running_loop():
+            asyncio.ensure_future(pump())
+        else:
+            asyncio.run_coroutine_threadsafe(pump(), loop)
+
+        async def lines():
+            while True:
+                try:
+                    item = out_q.get_nowait()
+                except _q.Empty:
+                    await asyncio.sleep(0.005)
+                    continue
+                if item is None:
+                    break
+                yield item + "\n"
+        return StreamingResponse(lines(), media_type="application/x-ndjson")
+
     @app.get("/get_instances_status")
     async def get_instances_status():
         out = []
