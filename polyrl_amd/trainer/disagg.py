@@ -320,6 +320,18 @@ class DisaggCoordinator:
         assert got % self.n_trainer == 0, \
             f"stream round of {got} samples !% {self.n_trainer} trainer ranks"
         per = got // self.n_trainer
+        sizes = [len(res.samples) for _, res in groups]
+        if len(set(sizes)) == 1 and len(groups) % self.n_trainer == 0:
+            # token-balanced shards (the reference's _balance_batch DP
+            # seqlen balance, stream_ray_trainer.py:406-410): equal group
+            # COUNTS per rank, response tokens spread karmarkar-karp style
+            toks = [sum(len(x.output_ids) for x in res.samples)
+                    for _, res in groups]
+            from ..core.seqlen import get_seqlen_balanced_partitions
+            parts = get_seqlen_balanced_partitions(
+                toks, self.n_trainer, equal_size=True)
+            return [[self._make_batch([groups[i] for i in sorted(pt)])
+                     for pt in parts]]
         out = []
         gi = 0
         for _ in range(self.n_trainer):
