@@ -68,6 +68,9 @@ class ActorConfig:
     entropy_coeff: float = 0.0
     use_kl_loss: bool = False
     kl_loss_coef: float = 0.001
+    # truncated importance sampling vs the ROLLOUT policy's logprobs
+    # (off-policy correction for streamed/stale rollouts); 0 disables
+    tis_imp_ratio_cap: float = 0.0
     kl_loss_type: str = "low_var_kl"
     policy_loss_type: str = "vanilla"
     ulysses_sequence_parallel_size: int = 1

@@ -268,8 +268,15 @@ def compute_policy_loss_vanilla(
     clip_ratio_high: Optional[float] = None,
     clip_ratio_c: float = 3.0,
     loss_agg_mode: str = "token-mean",
+    importance_weights: Optional[torch.Tensor] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor, torch.Tensor]:
     """PPO clipped surrogate with dual-clip (verl 'vanilla' loss).
+
+    ``importance_weights``: per-token truncated importance-sampling weights
+    correcting for rollouts generated on slightly stale weights (the
+    reference stream actor's importance-weight capability,
+    stream_dp_actor.py:153-224): detached, multiplied into the per-token
+    loss before aggregation.
 
     Returns (pg_loss, pg_clipfrac, ppo_kl, pg_clipfrac_lower).
     """
@@ -293,6 +300,8 @@ def compute_policy_loss_vanilla(
         torch.gt(clip_pg_losses1, pg_losses3).float() * (advantages < 0).float(),
         response_mask)
     pg_losses = torch.where(advantages < 0, clip_pg_losses2, clip_pg_losses1)
+    if importance_weights is not None:
+        pg_losses = pg_losses * importance_weights.detach()
     pg_loss = agg_loss(pg_losses, response_mask, loss_agg_mode)
     return pg_loss, pg_clipfrac, ppo_kl, pg_clipfrac_lower
 

@@ -326,13 +326,24 @@ class ActorWorker:
             advantages = mb["advantages"].to(self.device)
             log_prob, ent = self._forward_logprobs(
                 mb, want_entropy=bool(self.cfg.entropy_coeff), grad=True)
+            extra = {}
+            if self.cfg.tis_imp_ratio_cap > 0 and \
+                    "rollout_log_probs" in mb.tensors and \
+                    self.cfg.policy_loss_type == "vanilla":
+                # truncated importance sampling: rollout happened on the
+                # engine's (possibly stale) weights; reweight each token by
+                # min(pi_old/pi_rollout, cap) — stream_dp_actor.py:153-224
+                rollout_lp = mb["rollout_log_probs"].to(self.device)
+                extra["importance_weights"] = torch.exp(
+                    (old_log_prob - rollout_lp).clamp(-20, 20)
+                ).clamp(max=self.cfg.tis_imp_ratio_cap)
             pg_loss, pg_clipfrac, ppo_kl, pg_clipfrac_lower = loss_fn(
                 old_log_prob=old_log_prob, log_prob=log_prob,
                 advantages=advantages, response_mask=response_mask,
                 clip_ratio=self.cfg.clip_ratio,
                 clip_ratio_low=self.cfg.clip_ratio_low,
                 clip_ratio_high=self.cfg.clip_ratio_high,
-                loss_agg_mode=self.cfg.loss_agg_mode)
+                loss_agg_mode=self.cfg.loss_agg_mode, **extra)
             loss = pg_loss
             if self.cfg.entropy_coeff:
                 loss = loss - self.cfg.entropy_coeff * algos.agg_loss(

@@ -218,3 +218,26 @@ def test_flops_counter_mfu():
     f = fc.train_step_flops(65536, 512)
     mfu = fc.mfu(f, seconds=7.0, n_gpus=1)
     assert 0.0 < mfu < 1.0
+
+
+def test_vanilla_loss_importance_weights():
+    """Truncated importance sampling (stream_dp_actor.py:153-224 capability):
+    per-token weights scale the clipped surrogate before aggregation."""
+    import torch
+
+    from polyrl_amd.core import algos
+    torch.manual_seed(0)
+    B, L = 4, 6
+    old = torch.randn(B, L) * 0.1
+    new = old + torch.randn(B, L) * 0.05
+    adv = torch.randn(B, L)
+    mask = torch.ones(B, L)
+    base, *_ = algos.compute_policy_loss_vanilla(old, new, adv, mask)
+    iw = torch.full((B, L), 2.0)
+    scaled, *_ = algos.compute_policy_loss_vanilla(
+        old, new, adv, mask, importance_weights=iw)
+    assert torch.allclose(scaled, base * 2.0, atol=1e-6)
+    # cap semantics at the caller: exp(old - rollout).clamp(max=cap)
+    rollout = old - 10.0                     # wildly off-policy rollout
+    w = torch.exp((old - rollout).clamp(-20, 20)).clamp(max=3.0)
+    assert torch.all(w == 3.0)

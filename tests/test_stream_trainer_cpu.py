@@ -215,3 +215,15 @@ def test_reference_fences(tmp_path):
     cfg2.actor_rollout_ref.rollout.multi_turn = True
     with pytest.raises(NotImplementedError):
         StreamPPOTrainer(cfg2, reward_fn=load_reward_manager("random"))
+
+
+def test_tis_importance_weight_path(tmp_path):
+    """actor.tis_imp_ratio_cap reweights tokens by pi_old/pi_rollout
+    (capped): the run completes and trains."""
+    cfg = tiny_config(tmp_path)
+    cfg.actor_rollout_ref.actor.tis_imp_ratio_cap = 2.0
+    trainer = StreamPPOTrainer(cfg, reward_fn=load_reward_manager("random"))
+    before = snapshot(trainer.actor.model)
+    trainer.fit(max_steps=1)
+    after = snapshot(trainer.actor.model)
+    assert any(not torch.equal(before[k], after[k]) for k in before)
