@@ -1,0 +1,118 @@
+"""Property-based tests (hypothesis) over core invariants: the batch
+protocol, seqlen-balanced partitioning, and the radix cache's page
+accounting.  These are the load-bearing invariants the rest of the stack
+assumes (the new build must do better than the reference's
+zero tests)."""
+from __future__ import annotations
+
+import numpy as np
+import pytest
+import torch
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+from polyrl_amd.core.seqlen import get_seqlen_balanced_partitions
+from polyrl_amd.protocol import TensorBatch
+from polyrl_amd.rollout.kv_cache import PagedKVCache, RadixCache
+
+
+def mk_batch(n: int, L: int = 4) -> TensorBatch:
+    return TensorBatch.from_dict(
+        tensors={"x": torch.arange(n * L).reshape(n, L),
+                 "y": torch.randn(n)},
+        non_tensors={"uid": np.array([f"u{i}" for i in range(n)],
+                                     dtype=object)})
+
+
+@settings(max_examples=50, deadline=None)
+@given(n=st.integers(1, 40), k=st.integers(1, 12))
+def test_split_concat_roundtrip(n, k):
+    b = mk_batch(n)
+    parts = b.split(k)
+    assert sum(len(p) for p in parts) == n
+    rt = TensorBatch.concat(parts)
+    assert torch.equal(rt["x"], b["x"]) and torch.equal(rt["y"], b["y"])
+    assert list(rt["uid"]) == list(b["uid"])
+
+
+@settings(max_examples=50, deadline=None)
+@given(n=st.integers(1, 40), div=st.integers(1, 16))
+def test_pad_unpad_inverse(n, div):
+    b = mk_batch(n)
+    p = b.pad_to_divisor(div)
+    assert len(p) % div == 0
+    u = p.unpad()
+    assert len(u) == n
+    assert torch.equal(u["x"], b["x"])
+    assert list(u["uid"]) == list(b["uid"])
+
+
+@settings(max_examples=50, deadline=None)
+@given(seqlens=st.lists(st.integers(1, 2048), min_size=1, max_size=64),
+       k=st.integers(1, 8))
+def test_balanced_partitions_are_partitions(seqlens, k):
+    k = min(k, len(seqlens))
+    parts = get_seqlen_balanced_partitions(seqlens, k, equal_size=False)
+    flat = sorted(i for p in parts for i in p)
+    assert flat == list(range(len(seqlens)))          # exact partition
+    assert len(parts) == k and all(p for p in parts)  # none empty
+
+
+@settings(max_examples=30, deadline=None)
+@given(groups=st.integers(1, 8), per=st.integers(1, 6),
+       seed=st.integers(0, 10_000))
+def test_balanced_equal_size_counts(groups, per, seed):
+    rng = np.random.default_rng(seed)
+    n = groups * per
+    seqlens = [int(x) for x in rng.integers(1, 1024, size=n)]
+    parts = get_seqlen_balanced_partitions(seqlens, groups, equal_size=True)
+    assert all(len(p) == per for p in parts)
+    flat = sorted(i for p in parts for i in p)
+    assert flat == list(range(n))
+    # balance is no worse than the contiguous split
+    sums = [sum(seqlens[i] for i in p) for p in parts]
+    contig = [sum(seqlens[g * per:(g + 1) * per]) for g in range(groups)]
+    assert max(sums) <= max(contig)
+
+
+@settings(max_examples=25, deadline=None)
+@given(seed=st.integers(0, 10_000), num_pages=st.integers(8, 40))
+def test_radix_page_accounting_closes(seed, num_pages):
+    """Random interleaving of insert / match+seed / free / evict keeps the
+    allocator's books: after freeing every sequence and flushing the tree,
+    every page is back on the free list."""
+    rng = np.random.default_rng(seed)
+    kv = PagedKVCache(num_layers=1, num_kv_heads=1, head_dim=8,
+                      num_pages=num_pages, page_size=4, device="cpu")
+    radix = RadixCache(kv)
+    live = {}           # seq_id -> token list
+    next_sid = 0
+    for _ in range(40):
+        op = rng.integers(0, 3)
+        if op == 0:     # new seq: match, seed, allocate tail, maybe insert
+            toks = [int(t) for t in rng.integers(0, 5, size=rng.integers(1, 20))]
+            pages, mlen = radix.match(toks)
+            sid = next_sid
+            next_sid += 1
+            need = len(toks) - mlen
+            if not kv.can_allocate(need):
+                radix.evict((need + 3) // 4)
+            if not kv.can_allocate(need):
+                kv.unref_pages(pages)
+                continue
+            kv.seed_seq(sid, pages, mlen)
+            assert kv.allocate(sid, need)
+            live[sid] = toks
+        elif op == 1 and live:   # finish a seq: donate + free
+            sid = list(live)[int(rng.integers(0, len(live)))]
+            toks = live.pop(sid)
+            radix.insert(toks, kv._seq_pages[sid])
+            kv.free_seq(sid)
+        elif op == 2:   # pressure eviction
+            radix.evict(int(rng.integers(1, 4)))
+    for sid in list(live):
+        kv.free_seq(sid)
+    radix.reset()
+    assert kv.free_pages == kv.num_pages, \
+        f"leak: {kv.num_pages - kv.free_pages} pages ({kv._ref})"
+    assert radix.num_nodes == 0
