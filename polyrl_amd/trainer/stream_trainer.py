@@ -402,6 +402,14 @@ class StreamPPOTrainer:
             "val/response_length/mean": float(lens.mean()),
             "val/n": float(len(full)),
         }
+        # per-data-source breakdown (the reference logs val scores by
+        # source via the reward-manager dispatch)
+        ds = full.non_tensors.get("data_source")
+        if ds is not None:
+            import numpy as _np
+            for src in sorted({str(x) for x in ds}):
+                m = torch.from_numpy(_np.array([str(x) == src for x in ds]))
+                out[f"val/score/{src}/mean"] = float(seq_scores[m].mean())
         if dist.is_available() and dist.is_initialized():
             t = torch.tensor([out["val/score/mean"], out["val/n"]])
             dist.all_reduce(t, group=self.pg)
