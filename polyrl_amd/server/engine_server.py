@@ -182,7 +182,8 @@ def create_app(engine: Engine, runner: Optional[EngineRunner] = None):
         metas = body["metas"]            # [(name, shape, dtype_str), ...]
         num_streams = int(body.get("num_streams", 4))
         total = 0
-        for _, shape, dts in metas:
+        for meta in metas:
+            shape, dts = meta[1], meta[2]
             n = 1
             for s in shape:
                 n *= s
@@ -219,13 +220,18 @@ def create_app(engine: Engine, runner: Optional[EngineRunner] = None):
                                  "message": "tcp receive timed out"}, 500)
         sd = {}
         off = 0
-        for name, shape, dts in st.wt_metas:
+        for meta in st.wt_metas:
+            name, shape, dts = meta[0], meta[1], meta[2]
+            scale = float(meta[3]) if len(meta) > 3 else None
             dt = getattr(torch, dts)
             n = 1
             for s in shape:
                 n *= s
             nbytes = n * torch.tensor([], dtype=dt).element_size()
-            sd[name] = st.wt_buffer[off:off + nbytes].view(dt).view(shape)
+            view = st.wt_buffer[off:off + nbytes].view(dt).view(shape)
+            if scale is not None:      # fp8-compressed: dequantize
+                view = view.to(torch.float32) * scale
+            sd[name] = view
             off += nbytes
         await asyncio.get_running_loop().run_in_executor(
             None, lambda: runner.update_weights(sd, version))
