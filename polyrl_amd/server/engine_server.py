@@ -117,9 +117,14 @@ def create_app(engine: Engine, runner: Optional[EngineRunner] = None):
         safetensors bytes with the version in X-Weight-Version header."""
         import torch
         ctype = request.headers.get("content-type", "")
+        unlock = False
         if ctype.startswith("application/json"):
             body = await request.json()
             version = int(body["version"])
+            # off-policy mode (reference roadmap 'unlock weight update of
+            # rollout engines'): keep in-flight generation running across
+            # the swap; the trainer's TIS reweighting corrects for it
+            unlock = bool(body.get("unlock", False))
             if body.get("ack_only"):
                 # weights were delivered out of band (collective broadcast
                 # over xGMI, transfer/collective.py): version acknowledgment
@@ -141,7 +146,8 @@ def create_app(engine: Engine, runner: Optional[EngineRunner] = None):
         # run the blocking swap off the event loop; takes the step lock so
         # it excludes in-flight generation
         await asyncio.get_running_loop().run_in_executor(
-            None, lambda: runner.update_weights(sd, version))
+            None, lambda: runner.update_weights(
+                sd, version, abort_in_flight=not unlock))
         return {"success": True, "message": f"weights at version {version}"}
 
     @app.post("/update_weights_from_tensor")
@@ -157,6 +163,7 @@ def create_app(engine: Engine, runner: Optional[EngineRunner] = None):
         import torch
         body = await request.json()
         version = int(body.get("version", runner.weight_version + 1))
+        unlock = bool(body.get("unlock", False))
         sd = {}
         for name, t in body["tensors"].items():
             dt = getattr(torch, t.get("dtype", "float32"))
@@ -166,7 +173,8 @@ def create_app(engine: Engine, runner: Optional[EngineRunner] = None):
                 .view(dt).reshape(t["shape"])
             sd[name] = ten
         await asyncio.get_running_loop().run_in_executor(
-            None, lambda: runner.update_weights(sd, version))
+            None, lambda: runner.update_weights(
+                sd, version, abort_in_flight=not unlock))
         return {"success": True, "message": f"weights at version {version}"}
 
     @app.post("/weights_handshake")
@@ -234,7 +242,9 @@ def create_app(engine: Engine, runner: Optional[EngineRunner] = None):
             sd[name] = view
             off += nbytes
         await asyncio.get_running_loop().run_in_executor(
-            None, lambda: runner.update_weights(sd, version))
+            None, lambda: runner.update_weights(
+                sd, version,
+                abort_in_flight=not bool(body.get("unlock", False))))
         st.wt_rx.close()
         st.wt_rx = None
         return {"success": True, "message": f"weights at version {version}"}

@@ -248,3 +248,44 @@ def test_tcp_weight_push_fp8_compressed(served):
     assert not torch.equal(got, ref)      # really went through quantization
     assert runner.weight_version == 5
     server.should_exit = True
+
+
+def test_unlocked_weight_update_keeps_generation_running(served):
+    """Off-policy mode (reference roadmap: 'unlock weight update of rollout
+    engines', unchecked there): update with {"unlock": true} swaps weights
+    WITHOUT aborting in-flight requests — they run to completion across
+    versions (the trainer's TIS reweighting corrects the mix)."""
+    import base64
+
+    import httpx
+    cfg, model, eng, app = served
+    runner = app.state.runner
+
+    async def go():
+        async with httpx.AsyncClient(transport=_transport(app),
+                                     base_url="http://t") as c:
+            gen = asyncio.ensure_future(c.post("/generate", json={
+                "input_ids": [5, 9, 2], "stream": False,
+                "sampling_params": {"n": 1, "max_new_tokens": 40,
+                                    "temperature": 1.0}}))
+            await asyncio.sleep(0.05)        # let decode start
+            name, tref = next(iter(eng.model._name_map.items()))
+            new = torch.full_like(tref, 0.0625)
+            payload = {"version": runner.weight_version + 1,
+                       "unlock": True,
+                       "tensors": {name: {
+                           "data": base64.b64encode(
+                               new.numpy().tobytes()).decode(),
+                           "shape": list(new.shape),
+                           "dtype": str(new.dtype).split(".")[-1]}}}
+            r = await c.post("/update_weights_from_tensor", json=payload)
+            assert r.status_code == 200
+            resp = await gen
+            assert resp.status_code == 200
+            out = resp.json()
+            # NOT aborted: ran to its full token budget across the swap
+            fr = out[0]["meta_info"]["finish_reason"]["type"] \
+                if isinstance(out, list) else \
+                out["meta_info"]["finish_reason"]["type"]
+            assert fr in ("length", "stop"), fr
+    asyncio.run(go())
