@@ -20,12 +20,19 @@ from ..rollout.engine import Engine, SamplingParams
 from ..rollout.runner import EngineRunner
 
 
-def create_app(engine: Engine, runner: Optional[EngineRunner] = None):
+def create_app(engine: Engine, runner: Optional[EngineRunner] = None,
+               allowed_sender_cidrs: Optional[List[str]] = None):
 
     app = FastAPI(title="polyrl-amd rollout instance")
     runner = runner or EngineRunner(engine)
     app.state.runner = runner
     app.state.engine = engine
+    app.state.allowed_sender_cidrs = allowed_sender_cidrs
+
+    def _sender_ok(request: Request) -> bool:
+        from ..transfer.tcp_engine import addr_allowed
+        host = request.client.host if request.client else None
+        return addr_allowed(host, app.state.allowed_sender_cidrs)
 
     def _sp(d: dict) -> SamplingParams:
         return SamplingParams(
@@ -112,6 +119,10 @@ def create_app(engine: Engine, runner: Optional[EngineRunner] = None):
 
     @app.post("/update_weights_from_agent")
     async def update_weights_from_agent(request: Request):
+        if not _sender_ok(request):
+            return JSONResponse({"success": False,
+                                 "message": "sender not in allowed CIDRs"},
+                                403)
         """Install new weights.  Accepts either
         {"version": V, "path": "/node/local/file.safetensors"} or raw
         safetensors bytes with the version in X-Weight-Version header."""
@@ -179,6 +190,10 @@ def create_app(engine: Engine, runner: Optional[EngineRunner] = None):
 
     @app.post("/weights_handshake")
     async def weights_handshake(request: Request):
+        if not _sender_ok(request):
+            return JSONResponse({"success": False,
+                                 "message": "sender not in allowed CIDRs"},
+                                403)
         """Arm a TCP bulk receive (the reference's receiver-agent bootstrap,
         receiver_agent.py:184-240): allocate/reuse a registered CPU buffer
         sized to the incoming state dict, listen on N stream ports, return
@@ -301,6 +316,9 @@ def main():
                         "serving; RL trainers flush it every publish)")
     p.add_argument("--no-radix-cache", dest="radix_cache",
                    action="store_false")
+    p.add_argument("--allowed-sender-cidrs", default=None,
+                   help="comma-separated CIDRs allowed to push weights "
+                        "(reference: allowed_sender_ips)")
     args = p.parse_args()
 
     cfg = get_config(args.model)
@@ -315,7 +333,9 @@ def main():
     else:
         for _, t in engine.model._name_map.items():
             t.normal_(0, 0.02)
-    app = create_app(engine)
+    cidrs = args.allowed_sender_cidrs.split(",") \
+        if args.allowed_sender_cidrs else None
+    app = create_app(engine, allowed_sender_cidrs=cidrs)
 
     if args.manager:
         import requests

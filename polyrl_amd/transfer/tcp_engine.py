@@ -22,6 +22,26 @@ from typing import Dict, List, Optional, Tuple
 import torch
 
 _HDR = struct.Struct("<QQ")          # offset, length
+
+
+def addr_allowed(host: Optional[str], cidrs: Optional[List[str]]) -> bool:
+    """CIDR allow-list for transfer peers (the reference's
+    allowed_sender_ips filter, utils.rs:303-339 / weight_transfer
+    utils.py:12-53).  No list (or unknown host) => allow."""
+    if not cidrs or host is None:
+        return True
+    import ipaddress
+    try:
+        ip = ipaddress.ip_address(host)
+    except ValueError:
+        return False
+    for c in cidrs:
+        try:
+            if ip in ipaddress.ip_network(c, strict=False):
+                return True
+        except ValueError:
+            continue
+    return False
 SOCK_BUF = 16 << 20                  # 16 MB socket buffers (config.toml tune)
 CHUNK = 4 << 20
 

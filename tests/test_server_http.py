@@ -289,3 +289,36 @@ def test_unlocked_weight_update_keeps_generation_running(served):
                 out["meta_info"]["finish_reason"]["type"]
             assert fr in ("length", "stop"), fr
     asyncio.run(go())
+
+
+def test_sender_cidr_allowlist():
+    """allowed_sender_cidrs gates the weight-delivery routes (reference:
+    allowed_sender_ips CIDR filter, utils.rs:303-339)."""
+    import httpx
+
+    from polyrl_amd.transfer.tcp_engine import addr_allowed
+    assert addr_allowed("192.168.1.7", ["192.168.0.0/16"])
+    assert not addr_allowed("10.0.0.1", ["192.168.0.0/16"])
+    assert addr_allowed("10.0.0.1", None)          # no list => open
+    assert addr_allowed(None, ["192.168.0.0/16"])  # unknown host => allow
+
+    torch.manual_seed(0)
+    cfg = get_model_config("llama-debug-cpu")
+    eng = Engine(cfg, device="cpu", dtype=torch.float32,
+                 kv_bytes_budget=8 << 20)
+    app = create_app(eng, allowed_sender_cidrs=["192.168.0.0/16"])
+
+    async def go():
+        # ASGITransport presents client 127.0.0.1 -> not in the list
+        async with httpx.AsyncClient(transport=_transport(app),
+                                     base_url="http://t") as c:
+            r = await c.post("/update_weights_from_agent",
+                             json={"version": 1, "ack_only": True})
+            assert r.status_code == 403, r.text
+            r = await c.post("/weights_handshake",
+                             json={"metas": [], "num_streams": 1})
+            assert r.status_code == 403
+            # generation routes unaffected
+            r = await c.get("/health")
+            assert r.status_code == 200
+    asyncio.run(go())
