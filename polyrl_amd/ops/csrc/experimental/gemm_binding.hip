@@ -8,6 +8,8 @@
 typedef __hip_bfloat16 bf16;
 extern "C" __global__ void gemm_bf16_nt(const bf16*, const bf16*, bf16*,
                                         int, int, int);
+extern "C" __global__ void gemm_decode_bf16(const bf16*, const bf16*,
+                                            float*, int, int);
 
 torch::Tensor gemm_nt(torch::Tensor A, torch::Tensor W) {
   TORCH_CHECK(A.is_cuda() && W.is_cuda());
@@ -34,4 +36,33 @@ torch::Tensor gemm_nt(torch::Tensor A, torch::Tensor W) {
   return C;
 }
 
-PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) { m.def("gemm_nt", &gemm_nt); }
+torch::Tensor gemm_decode(torch::Tensor x, torch::Tensor W, int ksplit) {
+  TORCH_CHECK(x.is_cuda() && W.is_cuda());
+  TORCH_CHECK(x.dtype() == torch::kBFloat16 && W.dtype() == torch::kBFloat16);
+  TORCH_CHECK(x.is_contiguous() && W.is_contiguous());
+  const int M = x.size(0), K = x.size(1), N = W.size(0);
+  TORCH_CHECK(M == 256 && W.size(1) == K);
+  TORCH_CHECK(N % 64 == 0 && K % (64 * ksplit) == 0);
+  auto out = ksplit == 1
+      ? torch::empty({M, N}, x.options().dtype(torch::kFloat32))
+      : torch::zeros({M, N}, x.options().dtype(torch::kFloat32));
+  const int lds_bytes = 2 * (256 * 64 + 64 * 64) * 2;   // 80 KiB
+  static bool attr2 = false;
+  if (!attr2) {
+    (void)hipFuncSetAttribute((const void*)gemm_decode_bf16,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              lds_bytes);
+    attr2 = true;
+  }
+  dim3 grid(N / 64, ksplit);
+  hipLaunchKernelGGL(gemm_decode_bf16, grid, dim3(512), lds_bytes,
+                     c10::hip::getCurrentHIPStream().stream(),
+                     (const bf16*)x.data_ptr(), (const bf16*)W.data_ptr(),
+                     out.data_ptr<float>(), N, K);
+  return out;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("gemm_nt", &gemm_nt);
+  m.def("gemm_decode", &gemm_decode);
+}
