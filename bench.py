@@ -111,6 +111,11 @@ def main():
     # POLYRL_BENCH_LOG=1 prints per-step phase timings (gen/prep/update/...)
     cfg.trainer.logger = (["console"]
                           if os.environ.get("POLYRL_BENCH_LOG") else [])
+    # POLYRL_BENCH_PROFILE=1 wraps the LAST warmup step in torch.profiler
+    # (chrome trace) without touching the timed region
+    if os.environ.get("POLYRL_BENCH_PROFILE") and rank == 0:
+        cfg.trainer.profile_steps = [args.warmup]
+        cfg.trainer.profile_dir = "gpurun_out/torch_prof"
     cfg.trainer.resume_mode = "disable"
     cfg.trainer.default_local_dir = "/tmp/polyrl_bench_ckpt"
 
