@@ -38,6 +38,11 @@ from .types import (GroupRequest, GroupResult, MetricsUpdate, SampleResult,
 @dataclass
 class SchedulerConfig:
     max_assigned_batches_per_stats_check: int = 4   # config.toml default
+    # pluggable dispatch policy over the eligible set (the reference's
+    # 'interface of algorithm-driven request scheduling' roadmap line):
+    # "zero_queue_rr" (state.rs default) | "least_loaded" | a callable
+    # (eligible_ids, states, rr_counter) -> instance_id
+    scheduling_policy: object = "zero_queue_rr"
     stats_interval_s: float = 1.0                   # instance_manager.rs:39
     max_retries: int = 5                            # handlers.rs retry cap
     initial_max_local_gen_s: float = 150.0          # state.rs:79
@@ -180,8 +185,7 @@ class RolloutScheduler:
                     continue
                 eligible.append(iid)
             if eligible:
-                self._rr = (self._rr + 1) % len(eligible)
-                iid = eligible[self._rr]
+                iid = self._pick(eligible)
                 self._states[iid].assigned_batches += 1
                 return self._states[iid].inst
             async with cond:
@@ -191,6 +195,21 @@ class RolloutScheduler:
                 except asyncio.TimeoutError:
                     pass
                 self.refresh_stats()
+
+    def _pick(self, eligible: List[str]) -> str:
+        """Dispatch policy over the eligible set (algorithm-driven
+        scheduling interface)."""
+        pol = self.cfg.scheduling_policy
+        if callable(pol):
+            return pol(eligible, self._states, self._rr)
+        if pol == "least_loaded":
+            return min(eligible,
+                       key=lambda i: (self._states[i].stats.num_running
+                                      + self._states[i].stats.num_queued,
+                                      self._states[i].assigned_batches, i))
+        # zero_queue_rr (default)
+        self._rr = (self._rr + 1) % len(eligible)
+        return eligible[self._rr]
 
     # ------------------------------------------------- single group w/ retry
     async def process_group(self, req: GroupRequest) -> GroupResult:

@@ -286,3 +286,33 @@ def test_http_stream_failure_continuation():
         assert "http://flaky" not in [i.instance_id
                                       for i in sched.instances()]  # evicted
     asyncio.run(go())
+
+
+def test_scheduling_policy_interface():
+    """Pluggable dispatch policy (the reference's 'algorithm-driven request
+    scheduling' interface): least_loaded prefers the idle instance; a
+    custom callable is honored verbatim."""
+    async def go():
+        # least_loaded: instance with fewer running samples wins
+        sched = RolloutScheduler(SchedulerConfig(
+            scheduling_policy="least_loaded"))
+        a = FakeInstance("a")
+        b = FakeInstance("b")
+        await sched.register_instance(a, skip_health_check=True)
+        await sched.register_instance(b, skip_health_check=True)
+        sched._states["a"].stats.num_running = 5
+        sched._states["b"].stats.num_running = 0
+        picked = await sched.next_instance()
+        assert picked.instance_id == "b"
+
+        # custom callable
+        sched2 = RolloutScheduler(SchedulerConfig(
+            scheduling_policy=lambda elig, states, rr: sorted(elig)[-1]))
+        await sched2.register_instance(FakeInstance("x"),
+                                       skip_health_check=True)
+        await sched2.register_instance(FakeInstance("z"),
+                                       skip_health_check=True)
+        assert (await sched2.next_instance()).instance_id == "z"
+        await sched.close()
+        await sched2.close()
+    run(go())
