@@ -20,6 +20,9 @@ import yaml
 @dataclass
 class ModelConfig:
     path: str = "llama3-1b"            # registry name or HF-style dir
+    # optional safetensors weights (file or HF model dir) loaded by name
+    # into the freshly built model before FSDP sharding
+    load_weights: str = ""
     dtype: str = "bfloat16"
     override_config: dict = field(default_factory=dict)
     enable_gradient_checkpointing: bool = True

@@ -82,6 +82,15 @@ class StreamPPOTrainer:
         torch.manual_seed(config.trainer.seed)
         actor_model = create_model(model_cfg, kind="actor", dtype=dtype,
                                    device=device)
+        if arr.model.load_weights:
+            from ..models.hf_loader import load_hf_checkpoint
+            miss, unexp = load_hf_checkpoint(actor_model,
+                                             arr.model.load_weights)
+            if self.rank == 0:
+                print(f"[trainer] loaded HF weights from "
+                      f"{arr.model.load_weights} "
+                      f"(missing {len(miss)}, unexpected {len(unexp)})",
+                      flush=True)
         if arr.model.enable_gradient_checkpointing:
             actor_model.gradient_checkpointing_enable()
         if arr.model.lora_rank > 0:

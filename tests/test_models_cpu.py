@@ -163,3 +163,35 @@ def test_lora_backward_only_adapters():
             assert p.grad is not None and torch.isfinite(p.grad).all(), n
         elif "value_head" not in n:
             assert p.grad is None, n
+
+
+def test_hf_checkpoint_loader_roundtrip(tmp_path):
+    """Save a model's state dict as sharded safetensors (HF layout) and
+    load it into a fresh model by name; tied lm_head fallback covered."""
+    import torch
+    from safetensors.torch import save_file
+
+    from polyrl_amd.models import create_model, get_model_config
+    from polyrl_amd.models.hf_loader import load_hf_checkpoint
+
+    cfg = get_model_config("llama-debug-cpu")
+    torch.manual_seed(0)
+    src = create_model(cfg, kind="actor", dtype="float32", device="cpu")
+    sd = {k: v.clone().contiguous() for k, v in src.state_dict().items()}
+    # shard into 2 files like HF repos; drop lm_head to force the tied
+    # fallback path
+    keys = [k for k in sd if k != "lm_head.weight"]
+    half = len(keys) // 2
+    save_file({k: sd[k] for k in keys[:half]},
+              str(tmp_path / "model-00001-of-00002.safetensors"))
+    save_file({k: sd[k] for k in keys[half:]},
+              str(tmp_path / "model-00002-of-00002.safetensors"))
+
+    torch.manual_seed(1)
+    dst = create_model(cfg, kind="actor", dtype="float32", device="cpu")
+    missing, unexpected = load_hf_checkpoint(dst, str(tmp_path))
+    assert missing == [] and unexpected == []
+    for k in keys:
+        assert torch.equal(dst.state_dict()[k], sd[k]), k
+    assert torch.equal(dst.state_dict()["lm_head.weight"],
+                       sd["model.embed_tokens.weight"])
