@@ -328,8 +328,21 @@ def main():
                     kv_bytes_budget=int(args.kv_gb * (1 << 30)),
                     enable_radix_cache=args.radix_cache)
     if args.load:
-        from safetensors.torch import load_file
-        engine.model.load_state_dict(load_file(args.load), strict=False)
+        import os as _os
+        if _os.path.isdir(args.load):
+            # HF model dir (sharded safetensors): the engine's fused
+            # buffers ingest per-name via load_state_dict
+            import glob as _glob
+
+            from safetensors.torch import load_file
+            sd = {}
+            for f in sorted(_glob.glob(_os.path.join(args.load,
+                                                     "*.safetensors"))):
+                sd.update(load_file(f))
+            engine.model.load_state_dict(sd, strict=False)
+        else:
+            from safetensors.torch import load_file
+            engine.model.load_state_dict(load_file(args.load), strict=False)
     else:
         for _, t in engine.model._name_map.items():
             t.normal_(0, 0.02)
