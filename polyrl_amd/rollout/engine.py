@@ -774,7 +774,13 @@ class Engine:
                 else:
                     ok.append(False)
             if not all(ok):
-                # roll back to single-token decode for this round
+                # roll back to single-token decode for this round — and
+                # roll back the successful growths too, or their _seq_len
+                # would count slots that never get written (the radix
+                # cache would donate garbage KV for those positions)
+                for o, sid in zip(ok, seq_ids):
+                    if o:
+                        self.kv.shrink_seq(sid, C - 1)
                 C = 1
 
         tokens = torch.tensor([r.output_ids[-1] for r in reqs],

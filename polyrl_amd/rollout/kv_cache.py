@@ -130,6 +130,26 @@ class PagedKVCache:
             self._seq_len[cid] = prefix_len
         return copies
 
+    def shrink_seq(self, seq_id: int, num_tokens: int):
+        """Roll back the last ``num_tokens`` of an allocation (chunked-decode
+        growth that was abandoned): trailing pages no longer needed are
+        released (decref — shared pages survive).  The seq-len bookkeeping
+        must stay exact: RadixCache.insert donates pages for exactly the
+        WRITTEN tokens, so an inflated length would cache garbage KV."""
+        cur = self._seq_len[seq_id] - num_tokens
+        assert cur >= 0
+        pages = self._seq_pages[seq_id]
+        need = (cur + self.page_size - 1) // self.page_size
+        while len(pages) > max(need, 0):
+            p = pages.pop()
+            r = self._ref.get(p, 1) - 1
+            if r <= 0:
+                self._ref.pop(p, None)
+                self._free.append(p)
+            else:
+                self._ref[p] = r
+        self._seq_len[seq_id] = cur
+
     def seq_len(self, seq_id: int) -> int:
         return self._seq_len.get(seq_id, 0)
 

@@ -364,3 +364,51 @@ def test_radix_abort_mid_prefill_accounting(setup):
         e.step()
     e.flush_radix()
     assert e.kv.free_pages == e.kv.num_pages
+
+
+def test_kv_shrink_seq_rollback_accounting(setup):
+    """shrink_seq rolls back abandoned chunk growth exactly: length and
+    trailing pages return; shared pages survive via refcount."""
+    from polyrl_amd.rollout.kv_cache import PagedKVCache
+    kv = PagedKVCache(1, 1, 8, num_pages=8, page_size=4, device="cpu")
+    assert kv.allocate(0, 6)           # 2 pages, len 6
+    free0 = kv.free_pages
+    assert kv.allocate(0, 7)           # grow to 13 -> 4 pages
+    kv.shrink_seq(0, 7)                # roll the growth back
+    assert kv.seq_len(0) == 6
+    assert kv.free_pages == free0
+    kv.free_seq(0)
+    assert kv.free_pages == kv.num_pages
+
+
+def test_radix_correct_after_mixed_chunk_rollback(setup):
+    """Page pressure during chunked decode triggers the mixed-allocation
+    rollback; with the radix cache on, later identical prompts must still
+    decode EXACTLY like a fresh engine (an inflated _seq_len would donate
+    never-written KV)."""
+    cfg, model, _ = setup
+    from polyrl_amd.rollout.kv_cache import PagedKVCache
+    bt = PagedKVCache.bytes_per_token(cfg.num_hidden_layers,
+                                      cfg.num_key_value_heads, cfg.head_dim)
+    sp = SamplingParams(temperature=0.0, max_new_tokens=24)
+    prompts = [[(59 * i + j) % cfg.vocab_size for j in range(20)]
+               for i in range(6)]
+    e0 = _fresh_engine(cfg, model, radix=False, kv_bytes=bt * 16 * 1024)
+    refs = [e0.generate([p], sp, f"r{i}")[0].output_ids
+            for i, p in enumerate(prompts)]
+    # tiny pool + chunked decode: concurrent requests exhaust pages mid-chunk
+    e = _fresh_engine(cfg, model, kv_bytes=bt * 16 * 22,
+                      decode_chunk_size=8, max_num_batched_tokens=64)
+    for rep in range(2):               # second round hits the radix
+        for i, p in enumerate(prompts):
+            e.add_request(f"x{rep}-{i}", p, sp)
+        outs = {}
+        while e.has_work():
+            for o in e.step():
+                outs[o.rid] = o
+        for i in range(6):
+            o = outs[f"x{rep}-{i}"]
+            if o.finish_reason == "length":       # survived the pressure
+                assert o.output_ids == refs[i], (rep, i)
+    e.flush_radix()
+    assert e.kv.free_pages == e.kv.num_pages
