@@ -221,12 +221,14 @@ class ActorRolloutRefConfig:
 
 @dataclass
 class SchedulerConfig:
-    """In-process rollout scheduler (replaces the Rust rollout-manager)."""
+    """In-process rollout scheduler knobs (replaces the Rust
+    rollout-manager's config.toml; plumbed into scheduler/manager.py by the
+    disagg/elastic coordinators).  The HTTP facade port lives on
+    RolloutConfig.rollout_manager_port."""
     max_assigned_batches_per_stats_check: int = 4
     health_check_interval_s: float = 1.0
     max_retries: int = 5
-    enable_http: bool = False            # HTTP facade for remote elastic instances
-    http_port: int = 5000
+    scheduling_policy: str = "zero_queue_rr"   # | least_loaded
 
 
 @dataclass

@@ -191,7 +191,8 @@ class DisaggCoordinator:
                  pad_token_id: int = 0, device="cpu",
                  max_local_gen_s: float = 0.0,
                  manager_port: int = 0,
-                 remote_weight_state_fn=None):
+                 remote_weight_state_fn=None,
+                 sched_cfg=None):
         """rollout_urls entries are either plain url strings (remote) or
         (url, is_local) tuples.  ``manager_port`` > 0 serves the manager
         HTTP facade (scheduler/http_api.py) on rank 0 so elastic remote
@@ -214,8 +215,16 @@ class DisaggCoordinator:
             self.loop = asyncio.new_event_loop()
             t = threading.Thread(target=self.loop.run_forever, daemon=True)
             t.start()
-            self.scheduler = RolloutScheduler(SchedulerConfig(
-                stats_interval_s=0.2))
+            sc = SchedulerConfig(stats_interval_s=0.2)
+            if sched_cfg is not None:      # PPOConfig.scheduler knobs
+                sc.max_assigned_batches_per_stats_check = \
+                    sched_cfg.max_assigned_batches_per_stats_check
+                sc.health_check_interval_s = sched_cfg.health_check_interval_s
+                sc.max_retries = sched_cfg.max_retries
+                sc.scheduling_policy = getattr(sched_cfg,
+                                               "scheduling_policy",
+                                               sc.scheduling_policy)
+            self.scheduler = RolloutScheduler(sc)
 
             async def _register():
                 from ..server import HttpInstance

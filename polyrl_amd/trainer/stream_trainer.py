@@ -158,7 +158,8 @@ class StreamPPOTrainer:
             self.coordinator = DisaggCoordinator(
                 ro.response_length, self.pg, urls, rank=self.rank,
                 n_trainer=self.world, pad_token_id=0, device="cpu",
-                max_local_gen_s=ro.max_local_gen_s)
+                max_local_gen_s=ro.max_local_gen_s,
+                sched_cfg=self.config.scheduler)
             self.publisher = DisaggPublisher(
                 self.actor.model, device,
                 self.coordinator.scheduler, self.coordinator.loop,
@@ -257,7 +258,8 @@ class StreamPPOTrainer:
             n_trainer=self.world, pad_token_id=0, device="cpu",
             max_local_gen_s=ro.max_local_gen_s,
             manager_port=ro.rollout_manager_port if self.rank == 0 else 0,
-            remote_weight_state_fn=lambda v: self.publisher.snapshot_cache())
+            remote_weight_state_fn=lambda v: self.publisher.snapshot_cache(),
+            sched_cfg=self.config.scheduler)
         self.publisher = ElasticPublisher(
             self.actor.model, self.engine.model, self.coordinator,
             tie=model_cfg.tie_word_embeddings, trainer_group=self.pg)
