@@ -259,6 +259,8 @@ class DisaggCoordinator:
             self._gid += 1
             self._meta[gid] = {
                 "uid": str(uids[g]),
+                "extras": {k: v[g] for k, v in prompts.non_tensors.items()
+                           if k != "uid"},
                 "prompt_ids": ids[g].cpu(),
                 "prompt_mask": mask[g].cpu(),
             }
@@ -366,7 +368,9 @@ class DisaggCoordinator:
         outputs = [[_O(s) for s in res.samples] for _, res in pairs]
         return postprocess_groups(prompt_ids, prompt_mask, uids, outputs,
                                   self.response_length, self.pad,
-                                  self.device)
+                                  self.device,
+                                  group_extras=[m.get("extras")
+                                                for m, _ in pairs])
 
 
 class ElasticPublisher:

@@ -12,7 +12,7 @@ interface.
 """
 from __future__ import annotations
 
-from typing import Dict, Iterator, List
+from typing import Dict, Iterator, List, Optional
 
 import numpy as np
 import torch
@@ -29,8 +29,12 @@ def postprocess_groups(
     response_length: int,
     pad_token_id: int = 0,
     device="cpu",
+    group_extras: Optional[List[dict]] = None,
 ) -> TensorBatch:
-    """Build the training batch for a set of finished prompt groups."""
+    """Build the training batch for a set of finished prompt groups.
+    ``group_extras``: per-group dict of extra per-row values (data_source,
+    ground_truth, ...) carried through from the dataset so reward managers
+    can score real data."""
     G = len(outputs)
     n = len(outputs[0])
     Lp = prompt_ids.shape[1]
@@ -52,6 +56,15 @@ def postprocess_groups(
     position_ids = torch.clamp(torch.cumsum(attention_mask, dim=1) - 1, min=0)
     uids = np.array([group_uids[g] for g in range(G) for _ in range(n)],
                     dtype=object)
+    non_tensors = {"uid": uids}
+    if group_extras and any(group_extras):
+        keys = set()
+        for e in group_extras:
+            keys.update((e or {}).keys())
+        for k in sorted(keys):
+            non_tensors[k] = np.array(
+                [(group_extras[g] or {}).get(k) for g in range(G)
+                 for _ in range(n)], dtype=object)
     return TensorBatch.from_dict(
         tensors={
             "prompts": prompts,
@@ -62,7 +75,7 @@ def postprocess_groups(
             "position_ids": position_ids,
             "rollout_log_probs": rollout_lp,
         },
-        non_tensors={"uid": uids},
+        non_tensors=non_tensors,
     ).to(device)
 
 
@@ -101,6 +114,8 @@ class LocalRolloutCoordinator:
             raw = ids[g][mask[g].bool()].tolist()
             self._groups[gid] = {
                 "uid": str(uids[g]),
+                "extras": {k: v[g] for k, v in prompts.non_tensors.items()
+                           if k != "uid"},
                 "prompt_ids": ids[g].cpu(),
                 "prompt_mask": mask[g].cpu(),
                 "outputs": [None] * n,
@@ -191,7 +206,9 @@ class LocalRolloutCoordinator:
         outputs = [g["outputs"] for g in groups]
         return postprocess_groups(prompt_ids, prompt_mask, uids, outputs,
                                   self.response_length, self.pad_token_id,
-                                  self.device)
+                                  self.device,
+                                  group_extras=[g.get("extras")
+                                                for g in groups])
 
     # -------------------------------------------------------------- weights
     def update_weights(self, state_dict):

@@ -206,12 +206,33 @@ class StreamPPOTrainer:
         dcfg = config.data
         if dataset is not None:
             self.dataset = dataset
+        elif dcfg.train_files:
+            # tokenized parquet prompts (data.py schema; see
+            # examples/data_preprocess/gsm8k.py for the preprocessor)
+            from ..data import ParquetRLHFDataset
+            from ..utils.tokenizer import get_tokenizer
+            tok = get_tokenizer(config.actor_rollout_ref.model.path)
+            self.dataset = ParquetRLHFDataset(
+                list(dcfg.train_files), dcfg.max_prompt_length,
+                prompt_key=dcfg.prompt_key, tokenizer=tok,
+                input_ids_key="input_ids",
+                filter_overlong=dcfg.filter_overlong_prompts)
         else:
             self.dataset = SyntheticPromptDataset(
                 num_prompts=dcfg.synthetic_num_prompts,
                 vocab_size=model_cfg.vocab_size,
                 max_prompt_length=dcfg.max_prompt_length,
                 seed=dcfg.seed)
+        self.val_dataset = None
+        if dcfg.val_files:
+            from ..data import ParquetRLHFDataset
+            from ..utils.tokenizer import get_tokenizer
+            self.val_dataset = ParquetRLHFDataset(
+                list(dcfg.val_files), dcfg.max_prompt_length,
+                prompt_key=dcfg.prompt_key,
+                tokenizer=get_tokenizer(config.actor_rollout_ref.model.path),
+                input_ids_key="input_ids",
+                filter_overlong=dcfg.filter_overlong_prompts)
         self.reward_fn = reward_fn or load_reward_manager("constant")
 
         self.ckpt_actor = CheckpointManager(config.trainer.default_local_dir,
@@ -326,6 +347,12 @@ class StreamPPOTrainer:
     # ------------------------------------------------------------------- fit
     def fit(self, max_steps: Optional[int] = None):
         c = self.config
+        if c.trainer.val_before_train and self.global_step == 0 \
+                and not getattr(self, "_pre_validated", False):
+            self._pre_validated = True
+            val = self.validate()
+            if self.rank == 0 and val:
+                self.tracking.log(val, 0)
         total_steps = max_steps or c.trainer.total_training_steps or 1
         ro = c.actor_rollout_ref.rollout
         n = ro.sampling.n
@@ -386,11 +413,13 @@ class StreamPPOTrainer:
         sglang_rollout_remote.py:184-196)."""
         c = self.config
         ro = c.actor_rollout_ref.rollout
+        ds = self.val_dataset if getattr(self, "val_dataset", None) \
+            is not None else self.dataset
         nval = num_prompts or max(c.data.train_batch_size // self.world, 1)
         if self.sched_coordinated or self.tp_rollout:
             nval = max(nval - nval % self.world, self.world)
-        idx = list(range(min(nval, len(self.dataset))))
-        batch = self.dataset.batch(idx)
+        idx = list(range(min(nval, len(ds))))
+        batch = ds.batch(idx)
         sampling = SamplingParams(temperature=0.0,
                                   max_new_tokens=ro.response_length)
         self.publisher.publish()
@@ -626,10 +655,20 @@ class StreamPPOTrainer:
                                   self.critic.lr_scheduler, extra)
 
     def _maybe_resume(self):
-        if self.config.trainer.resume_mode == "disable":
+        tr = self.config.trainer
+        if tr.resume_mode == "disable":
             return
+        step = None
+        if tr.resume_mode == "resume_path":
+            # explicit checkpoint dir: .../global_step_N
+            import re as _re
+            assert tr.resume_from_path, \
+                "resume_mode=resume_path needs trainer.resume_from_path"
+            m = _re.search(r"global_step_(\d+)", tr.resume_from_path)
+            assert m, f"no global_step_N in {tr.resume_from_path!r}"
+            step = int(m.group(1))
         ex = self.ckpt_actor.load(self.actor.model, self.actor.optimizer,
-                                  self.actor.lr_scheduler)
+                                  self.actor.lr_scheduler, step=step)
         if ex is not None:
             self.global_step = int(ex.get("global_step", 0))
             if self.use_critic:

@@ -227,3 +227,42 @@ def test_tis_importance_weight_path(tmp_path):
     trainer.fit(max_steps=1)
     after = snapshot(trainer.actor.model)
     assert any(not torch.equal(before[k], after[k]) for k in before)
+
+
+def test_parquet_train_files_and_val_before_train(tmp_path):
+    """data.train_files/val_files load tokenized parquet prompts;
+    trainer.val_before_train runs a validation pass at step 0."""
+    import numpy as np
+    import pandas as pd
+
+    rows = [{"prompt": f"q{i}", "input_ids": list(range(1, 6 + i % 3)),
+             "data_source": "openai/gsm8k", "ground_truth": str(i)}
+            for i in range(16)]
+    f = tmp_path / "train.parquet"
+    pd.DataFrame(rows).to_parquet(f)
+    cfg = tiny_config(tmp_path)
+    cfg.data.train_files = [str(f)]
+    cfg.data.val_files = [str(f)]
+    cfg.data.train_batch_size = 8
+    cfg.trainer.val_before_train = True
+    trainer = StreamPPOTrainer(cfg, reward_fn=load_reward_manager("random"))
+    assert len(trainer.dataset) == 16
+    assert trainer.val_dataset is not None
+    trainer.fit(max_steps=1)
+    assert trainer.global_step == 1
+    full = trainer._last_full_batch
+    assert "ground_truth" in full.non_tensors       # passthrough for rewards
+
+
+def test_resume_from_explicit_path(tmp_path):
+    """trainer.resume_mode=resume_path restores the named global_step_N."""
+    cfg = tiny_config(tmp_path)
+    cfg.trainer.save_freq = 1
+    trainer = StreamPPOTrainer(cfg, reward_fn=load_reward_manager("random"))
+    trainer.fit(max_steps=2)               # saves steps 1 and 2
+    cfg2 = tiny_config(tmp_path)
+    cfg2.trainer.resume_mode = "resume_path"
+    cfg2.trainer.resume_from_path = str(
+        tmp_path / "ckpt" / "global_step_1")
+    t2 = StreamPPOTrainer(cfg2, reward_fn=load_reward_manager("random"))
+    assert t2.global_step == 1             # the EARLIER step, not latest
