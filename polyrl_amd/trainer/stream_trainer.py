@@ -646,13 +646,16 @@ class StreamPPOTrainer:
     # ------------------------------------------------------------ checkpoint
     def save_checkpoint(self):
         extra = {"global_step": self.global_step}
-        self.ckpt_actor.save(self.global_step, self.actor.model,
-                             self.actor.optimizer, self.actor.lr_scheduler,
+        contents = self.config.actor_rollout_ref.actor.checkpoint_contents
+        opt = self.actor.optimizer if "optimizer" in contents else None
+        lr = self.actor.lr_scheduler if "extra" in contents else None
+        self.ckpt_actor.save(self.global_step, self.actor.model, opt, lr,
                              extra)
         if self.use_critic:
+            copt = self.critic.optimizer if "optimizer" in contents else None
+            clr = self.critic.lr_scheduler if "extra" in contents else None
             self.ckpt_critic.save(self.global_step, self.critic.model,
-                                  self.critic.optimizer,
-                                  self.critic.lr_scheduler, extra)
+                                  copt, clr, extra)
 
     def _maybe_resume(self):
         tr = self.config.trainer

@@ -31,7 +31,9 @@ from ..protocol import TensorBatch
 
 def _maybe_fully_shard(model: nn.Module, mixed_precision: bool = True,
                        pg=None, reshard_after_forward: bool = True,
-                       param_offload: bool = False):
+                       param_offload: bool = False,
+                       param_dtype: str = "bfloat16",
+                       reduce_dtype: str = "float32"):
     """Apply FSDP2 per decoder layer + root.  Works for world_size 1..N on
     nccl(RCCL) and gloo alike.  ``pg`` restricts sharding to a subgroup
     (disaggregated split: FSDP over the trainer ranks only).
@@ -46,8 +48,8 @@ def _maybe_fully_shard(model: nn.Module, mixed_precision: bool = True,
     from torch.distributed.fsdp import MixedPrecisionPolicy, fully_shard
     mp = None
     if mixed_precision and next(model.parameters()).dtype == torch.bfloat16:
-        mp = MixedPrecisionPolicy(param_dtype=torch.bfloat16,
-                                  reduce_dtype=torch.float32)
+        mp = MixedPrecisionPolicy(param_dtype=getattr(torch, param_dtype),
+                                  reduce_dtype=getattr(torch, reduce_dtype))
     kwargs = {"mp_policy": mp} if mp else {}
     kwargs["reshard_after_forward"] = reshard_after_forward
     if param_offload and torch.cuda.is_available():
@@ -206,7 +208,9 @@ class ActorWorker:
         self.model = _maybe_fully_shard(
             model, pg=pg,
             reshard_after_forward=cfg.fsdp.reshard_after_forward,
-            param_offload=cfg.fsdp.param_offload)
+            param_offload=cfg.fsdp.param_offload,
+            param_dtype=cfg.fsdp.mixed_precision_dtype,
+            reduce_dtype=cfg.fsdp.reduce_dtype)
         if pg is not None and cfg.ulysses_sequence_parallel_size > 1:
             # disagg: SP groups over the trainer subgroup must have been
             # pre-built (a world collective) before the role branch —
@@ -399,7 +403,9 @@ class CriticWorker:
         self.model = _maybe_fully_shard(
             model, pg=pg,
             reshard_after_forward=cfg.fsdp.reshard_after_forward,
-            param_offload=cfg.fsdp.param_offload)
+            param_offload=cfg.fsdp.param_offload,
+            param_dtype=cfg.fsdp.mixed_precision_dtype,
+            reduce_dtype=cfg.fsdp.reduce_dtype)
         if pg is not None and \
                 getattr(cfg, "ulysses_sequence_parallel_size", 1) > 1:
             assert cfg.ulysses_sequence_parallel_size in _SP_GROUPS, \

@@ -266,3 +266,19 @@ def test_resume_from_explicit_path(tmp_path):
         tmp_path / "ckpt" / "global_step_1")
     t2 = StreamPPOTrainer(cfg2, reward_fn=load_reward_manager("random"))
     assert t2.global_step == 1             # the EARLIER step, not latest
+
+
+def test_checkpoint_contents_model_only(tmp_path):
+    """actor.checkpoint_contents=[model] writes no optimizer shards
+    (configurable save contents, stream_fsdp_workers.py:357-376)."""
+    import os
+
+    cfg = tiny_config(tmp_path)
+    cfg.actor_rollout_ref.actor.checkpoint_contents = ["model"]
+    cfg.trainer.save_freq = 1
+    trainer = StreamPPOTrainer(cfg, reward_fn=load_reward_manager("random"))
+    trainer.fit(max_steps=1)
+    d = tmp_path / "ckpt" / "global_step_1" / "actor"
+    files = os.listdir(d)
+    assert any(f.startswith("model_") for f in files)
+    assert not any(f.startswith("optim_") for f in files)
