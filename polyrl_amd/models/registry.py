@@ -96,7 +96,11 @@ MODEL_CONFIGS = {
 }
 
 
-def get_model_config(name: str) -> DecoderConfig:
+def get_model_config(name: str, override: dict = None) -> DecoderConfig:
     if name not in MODEL_CONFIGS:
         raise KeyError(f"unknown model {name!r}; known: {sorted(MODEL_CONFIGS)}")
-    return MODEL_CONFIGS[name]
+    cfg = MODEL_CONFIGS[name]
+    if override:
+        import dataclasses
+        cfg = dataclasses.replace(cfg, **override)   # registry stays pristine
+    return cfg

@@ -66,7 +66,8 @@ class StreamPPOTrainer:
         self._validate_config()
 
         arr = config.actor_rollout_ref
-        model_cfg = get_model_config(arr.model.path)
+        model_cfg = get_model_config(arr.model.path,
+                                     override=arr.model.override_config)
         self.model_cfg = model_cfg
         dtype = arr.model.dtype
 
@@ -464,6 +465,12 @@ class StreamPPOTrainer:
         c = self.config
         ro = c.actor_rollout_ref.rollout
 
+        # free_cache_engine: the reference's rollout_mode/trainer_mode
+        # memory dance (stream_fsdp_workers.py:467-492) — resume the KV
+        # pool for generation, release it once the stream drains.  Off by
+        # default on MI355X (288 GB fits both resident).
+        if ro.free_cache_engine and self.engine is not None:
+            self.engine.resume_memory()
         # 1. publish current weights to the rollout plane (every iteration,
         #    incl. bootstrap — §3.3)
         with marked_timer("weight_sync", timing):
@@ -524,6 +531,9 @@ class StreamPPOTrainer:
                             all_metrics.setdefault(k, []).extend(v)
         full = TensorBatch.concat(ibatches) if ibatches else TensorBatch()
         self._last_full_batch = full
+        if ro.free_cache_engine and self.engine is not None \
+                and not self.engine.has_work():
+            self.engine.release_memory()
         metrics = reduce_metrics(all_metrics)
         metrics.update(compute_data_metrics(full, self.use_critic))
         # feedback to the scheduler's adaptive local-gen time-box

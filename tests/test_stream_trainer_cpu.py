@@ -282,3 +282,20 @@ def test_checkpoint_contents_model_only(tmp_path):
     files = os.listdir(d)
     assert any(f.startswith("model_") for f in files)
     assert not any(f.startswith("optim_") for f in files)
+
+
+def test_free_cache_engine_and_override_config(tmp_path):
+    """rollout.free_cache_engine releases the KV pool between steps
+    (rollout_mode/trainer_mode dance); model.override_config patches the
+    registry geometry without mutating it."""
+    from polyrl_amd.models import get_model_config
+
+    cfg = tiny_config(tmp_path)
+    cfg.actor_rollout_ref.rollout.free_cache_engine = True
+    cfg.actor_rollout_ref.model.override_config = {"num_hidden_layers": 1}
+    trainer = StreamPPOTrainer(cfg, reward_fn=load_reward_manager("random"))
+    assert len(trainer.actor.model.model.layers) == 1
+    assert get_model_config("llama-debug-cpu").num_hidden_layers != 1
+    trainer.fit(max_steps=2)
+    # KV tensors are released after the step's stream drained
+    assert trainer.engine.kv.k_cache == []
