@@ -423,6 +423,8 @@ class StreamPPOTrainer:
         batch = ds.batch(idx)
         sampling = SamplingParams(temperature=0.0,
                                   max_new_tokens=ro.response_length)
+        if self.engine is not None:
+            self.engine.resume_memory()      # no-op unless released
         self.publisher.publish()
         self.coordinator.submit(batch, sampling, 1)
         stream = (len(idx) // self.world

@@ -293,6 +293,7 @@ def test_free_cache_engine_and_override_config(tmp_path):
     cfg = tiny_config(tmp_path)
     cfg.actor_rollout_ref.rollout.free_cache_engine = True
     cfg.actor_rollout_ref.model.override_config = {"num_hidden_layers": 1}
+    cfg.trainer.test_freq = 2            # validate() on a released engine
     trainer = StreamPPOTrainer(cfg, reward_fn=load_reward_manager("random"))
     assert len(trainer.actor.model.model.layers) == 1
     assert get_model_config("llama-debug-cpu").num_hidden_layers != 1
