@@ -458,6 +458,9 @@ class StreamPPOTrainer:
             dist.all_reduce(t, group=self.pg)
             out["val/score/mean"] = float(t[0] / self.world)
             out["val/n"] = float(t[1])
+        if ro.free_cache_engine and self.engine is not None \
+                and not self.engine.has_work():
+            self.engine.release_memory()     # back to trainer_mode
         return out
 
     # ------------------------------------------------------------- one step
