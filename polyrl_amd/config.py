@@ -209,6 +209,8 @@ class TrainerConfig:
     # profile steps, main_stream.py:79-93) — torch.profiler chrome traces
     profile_steps: List[int] = field(default_factory=list)
     profile_dir: str = "profiles/torch"
+    # dump per-step rollout samples (uid, response ids, score) as jsonl
+    rollout_data_dir: str = ""
 
 
 @dataclass

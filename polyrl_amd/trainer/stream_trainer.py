@@ -541,6 +541,28 @@ class StreamPPOTrainer:
             self.engine.release_memory()
         metrics = reduce_metrics(all_metrics)
         metrics.update(compute_data_metrics(full, self.use_critic))
+        if torch.cuda.is_available():
+            metrics["perf/max_memory_allocated_gb"] = \
+                torch.cuda.max_memory_allocated() / (1 << 30)
+            metrics["perf/max_memory_reserved_gb"] = \
+                torch.cuda.max_memory_reserved() / (1 << 30)
+        # rollout data dump (stream_ray_trainer.py:585-587 capability)
+        dump = getattr(c.trainer, "rollout_data_dir", "")
+        if dump and self.rank == 0 and len(full):
+            import json as _json
+            import os as _os
+            _os.makedirs(dump, exist_ok=True)
+            with open(_os.path.join(
+                    dump, f"step_{self.global_step}.jsonl"), "w") as f:
+                scores = full["token_level_scores"].sum(-1)
+                rm = full["response_mask"]
+                for i in range(len(full)):
+                    f.write(_json.dumps({
+                        "uid": str(full["uid"][i]),
+                        "response_ids":
+                            full["responses"][i][rm[i].bool()].tolist(),
+                        "score": float(scores[i]),
+                    }) + "\n")
         # feedback to the scheduler's adaptive local-gen time-box
         # (stream_ray_trainer.py:691-704 capability)
         if self.sched_coordinated and self.rank == 0 and \

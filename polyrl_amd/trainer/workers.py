@@ -437,7 +437,9 @@ class CriticWorker:
 
     def _split(self, batch: TensorBatch):
         if self.cfg.use_dynamic_bsz:
-            return prepare_dynamic_batch(batch, self.cfg.ppo_max_token_len_per_gpu)
+            budget = self.cfg.ppo_max_token_len_per_gpu * \
+                getattr(self.cfg, "ulysses_sequence_parallel_size", 1)
+            return prepare_dynamic_batch(batch, budget)
         mbs = self.cfg.ppo_micro_batch_size_per_gpu or len(batch)
         return fixed_micro_batches(batch, mbs)
 

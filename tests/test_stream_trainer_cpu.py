@@ -300,3 +300,19 @@ def test_free_cache_engine_and_override_config(tmp_path):
     trainer.fit(max_steps=2)
     # KV tensors are released after the step's stream drained
     assert trainer.engine.kv.k_cache == []
+
+
+def test_rollout_data_dump(tmp_path):
+    """trainer.rollout_data_dir dumps each step's samples as jsonl
+    (stream_ray_trainer.py:585-587 capability)."""
+    import json
+
+    cfg = tiny_config(tmp_path)
+    cfg.trainer.rollout_data_dir = str(tmp_path / "dump")
+    trainer = StreamPPOTrainer(cfg, reward_fn=load_reward_manager("random"))
+    trainer.fit(max_steps=1)
+    f = tmp_path / "dump" / "step_1.jsonl"
+    rows = [json.loads(line) for line in f.read_text().splitlines()]
+    assert len(rows) == 16                # 8 prompts x n=2
+    assert all("uid" in r and "score" in r and r["response_ids"]
+               for r in rows)
