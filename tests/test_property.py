@@ -118,3 +118,64 @@ def test_radix_page_accounting_closes(seed, num_pages):
     assert kv.free_pages == kv.num_pages, \
         f"leak: {kv.num_pages - kv.free_pages} pages ({kv._ref})"
     assert radix.num_nodes == 0
+
+
+@settings(max_examples=30, deadline=None)
+@given(g=st.integers(1, 6), n=st.integers(2, 6), L=st.integers(1, 10),
+       seed=st.integers(0, 9999))
+def test_grpo_group_advantage_invariants(g, n, L, seed):
+    """GRPO: within each prompt group the (unnormalized) advantage sums to
+    ~0 and is constant across each sample's tokens."""
+    from polyrl_amd.core import algos
+    torch.manual_seed(seed)
+    B = g * n
+    rewards = torch.zeros(B, L)
+    rewards[:, -1] = torch.randn(B)                 # outcome reward
+    mask = torch.ones(B, L)
+    index = np.array([f"p{i // n}" for i in range(B)], dtype=object)
+    adv, ret = algos.compute_grpo_outcome_advantage(
+        rewards, mask, index, norm_adv_by_std_in_grpo=False)
+    assert torch.equal(adv, ret)
+    # constant over tokens of a sample
+    assert torch.allclose(adv, adv[:, :1].expand(-1, L), atol=1e-6)
+    # zero group mean
+    for grp in range(g):
+        sl = adv[grp * n:(grp + 1) * n, 0]
+        assert abs(float(sl.mean())) < 1e-5
+
+
+@settings(max_examples=30, deadline=None)
+@given(B=st.integers(1, 8), L=st.integers(2, 12), seed=st.integers(0, 9999))
+def test_gae_lam1_gamma1_is_reward_to_go(B, L, seed):
+    """GAE(gamma=1, lam=1): returns = reward-to-go; adv = returns - values
+    (whitened adv aside, the raw relation holds via returns)."""
+    from polyrl_amd.core import algos
+    torch.manual_seed(seed)
+    rewards = torch.randn(B, L)
+    values = torch.randn(B, L)
+    mask = torch.ones(B, L)
+    adv, ret = algos.compute_gae_advantage_return(rewards, values, mask,
+                                                  gamma=1.0, lam=1.0)
+    rtg = torch.flip(torch.cumsum(torch.flip(rewards, [1]), 1), [1])
+    assert torch.allclose(ret, rtg, atol=1e-4)
+
+
+@settings(max_examples=30, deadline=None)
+@given(g=st.integers(1, 5), n=st.integers(2, 5), seed=st.integers(0, 9999))
+def test_rloo_leave_one_out_property(g, n, seed):
+    """RLOO: advantage_i = score_i - mean(scores of the OTHER n-1)."""
+    from polyrl_amd.core import algos
+    torch.manual_seed(seed)
+    B = g * n
+    L = 3
+    rewards = torch.zeros(B, L)
+    rewards[:, -1] = torch.randn(B)
+    mask = torch.ones(B, L)
+    index = np.array([f"p{i // n}" for i in range(B)], dtype=object)
+    adv, _ = algos.compute_rloo_outcome_advantage(rewards, mask, index)
+    scores = rewards.sum(-1)
+    for i in range(B):
+        grp = i // n
+        others = [j for j in range(grp * n, (grp + 1) * n) if j != i]
+        expect = scores[i] - scores[others].mean()
+        assert abs(float(adv[i, 0]) - float(expect)) < 1e-5
