@@ -179,3 +179,47 @@ def test_rloo_leave_one_out_property(g, n, seed):
         others = [j for j in range(grp * n, (grp + 1) * n) if j != i]
         expect = scores[i] - scores[others].mean()
         assert abs(float(adv[i, 0]) - float(expect)) < 1e-5
+
+
+@settings(max_examples=40, deadline=None)
+@given(lens=st.lists(st.integers(0, 8), min_size=1, max_size=6),
+       total=st.integers(4, 40))
+def test_continuation_chain_token_exact(lens, total):
+    """A chain of partial generations stitched via continuation_request +
+    merge_sample reconstructs EXACTLY the tokens a single uninterrupted
+    generation would produce: prompt extension and max_new shrinkage are
+    token-exact at every hop (handlers.rs:330-418 contract)."""
+    from polyrl_amd.scheduler.types import (GroupRequest, SampleResult,
+                                            SamplingSpec, continuation_request,
+                                            merge_sample)
+    req = GroupRequest(gid=0, input_ids=[1, 2, 3], n=1,
+                       sampling=SamplingSpec(max_new_tokens=total))
+    # the "true" stream of tokens an uninterrupted instance would emit
+    stream = [100 + t for t in range(total)]
+    merged = SampleResult()
+    consumed = 0
+    for piece in lens:
+        piece = min(piece, total - consumed)
+        part = SampleResult(
+            output_ids=stream[consumed:consumed + piece],
+            output_logprobs=[-0.5] * piece,
+            finish_reason="abort", completion_tokens=piece)
+        consumed += piece
+        merged = merge_sample(merged, part) if merged.output_ids or \
+            merged.num_migrations else part
+        if consumed >= total:
+            break
+        # the manager always continues from the ORIGINAL request + the
+        # full accumulated sample (manager.py enqueue_continuation)
+        cont = continuation_request(req, 0, merged)
+        assert cont.input_ids == [1, 2, 3] + stream[:consumed]
+        assert cont.sampling.max_new_tokens == total - consumed
+    # finish the tail on the last instance
+    if consumed < total:
+        tail = SampleResult(output_ids=stream[consumed:total],
+                            output_logprobs=[-0.5] * (total - consumed),
+                            finish_reason="length",
+                            completion_tokens=total - consumed)
+        merged = merge_sample(merged, tail)
+    assert merged.output_ids == stream
+    assert len(merged.output_logprobs) == total
