@@ -49,3 +49,22 @@ def test_weight_state_dict_roundtrip():
         off += n
     for k in sd:
         assert torch.equal(out[k], sd[k].float())
+
+
+def test_fp8_quantization_error_bound():
+    """Per-tensor e4m3 + scale: worst-case relative error for normal
+    weights stays under e4m3's ~6% mantissa step (the compress="fp8"
+    payload math, transfer/tcp_engine.py)."""
+    import torch
+    torch.manual_seed(0)
+    for shape in [(64, 64), (3, 1000), (17,)]:
+        t = torch.randn(*shape) * 0.02          # weight-scale magnitudes
+        scale = float(t.abs().amax().clamp(min=1e-12)) / 448.0
+        q = (t.float() / scale).clamp(-448.0, 448.0).to(torch.float8_e4m3fn)
+        back = q.to(torch.float32) * scale
+        # elements above 1% of amax: bounded relative error
+        big = t.abs() > 0.01 * t.abs().amax()
+        rel = ((back - t).abs()[big] / t.abs()[big]).max().item()
+        assert rel < 0.07, rel
+        # bytes really halve vs bf16
+        assert q.element_size() == 1
