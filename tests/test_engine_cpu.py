@@ -412,3 +412,19 @@ def test_radix_correct_after_mixed_chunk_rollback(setup):
                 assert o.output_ids == refs[i], (rep, i)
     e.flush_radix()
     assert e.kv.free_pages == e.kv.num_pages
+
+
+def test_max_model_len_fence(setup):
+    """A request whose prompt+response would exceed max_model_len stops at
+    the fence with finish_reason='length' (never writes past the KV
+    geometry the hipGraph page tables are sized for)."""
+    cfg, model, _ = setup
+    e = Engine(cfg, device="cpu", dtype=torch.float32,
+               kv_bytes_budget=8 << 20, max_model_len=24)
+    e.model.load_state_dict(model.state_dict())
+    out = e.generate([list(range(7, 27))],          # 20-token prompt
+                     SamplingParams(temperature=0.0, max_new_tokens=50),
+                     "fence")[0]
+    assert out.finish_reason == "length"
+    assert len(out.output_ids) <= 24 - 20 + 1
+    assert e.kv.free_pages == e.kv.num_pages

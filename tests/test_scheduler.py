@@ -316,3 +316,25 @@ def test_scheduling_policy_interface():
         await sched.close()
         await sched2.close()
     run(go())
+
+
+def test_health_gate_rejects_dead_instance():
+    """register_instance health-gates joins: an instance that never
+    responds healthy raises after the (short) timeout and never enters
+    the pool (instance_manager.rs:5-37)."""
+    async def go():
+        sched = RolloutScheduler(SchedulerConfig(
+            health_check_timeout_s=0.2, health_check_interval_s=0.05))
+        dead = FakeInstance("dead")
+        dead.healthy = False
+
+        import pytest as _pytest
+        with _pytest.raises(TimeoutError):
+            await sched.register_instance(dead)
+        assert sched.num_active() == 0
+        # a healthy one joins through the same gate
+        ok = FakeInstance("ok")
+        await sched.register_instance(ok)
+        assert sched.num_active() == 1
+        await sched.close()
+    run(go())
