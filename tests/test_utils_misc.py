@@ -79,3 +79,22 @@ def test_flops_counter_mfu_math():
     assert "perf/update_mfu" in m
     assert 0.0 < m["perf/update_mfu"] < 1.0
     assert math.isfinite(m["perf/throughput_tokens_per_s_all_gpus"])
+
+
+def test_reduce_and_data_metrics():
+    from polyrl_amd.core.metrics import (compute_data_metrics,
+                                         reduce_metrics)
+    r = reduce_metrics({"a/loss": [1.0, 3.0], "b/x": [2.0]})
+    assert r["a/loss"] == 2.0 and r["b/x"] == 2.0
+    b = TensorBatch.from_dict(tensors={
+        "token_level_scores": torch.tensor([[0., 1.], [0., 3.]]),
+        "advantages": torch.randn(2, 2),
+        "returns": torch.randn(2, 2),
+        "response_mask": torch.ones(2, 2, dtype=torch.long),
+        "attention_mask": torch.ones(2, 4, dtype=torch.long),
+        "responses": torch.zeros(2, 2, dtype=torch.long),
+        "prompts": torch.zeros(2, 2, dtype=torch.long),
+    })
+    m = compute_data_metrics(b, use_critic=False)
+    assert m["critic/score/mean"] == 2.0
+    assert m["response_length/mean"] == 2.0
