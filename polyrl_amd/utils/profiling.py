@@ -57,10 +57,18 @@ def log_gpu_memory(tag: str, rank: int = 0):
 
 
 class GPUMemoryLogger:
-    """Context manager measuring peak memory of a region."""
+    """Context manager OR decorator measuring peak memory of a region
+    (the reference uses the decorator form — stream_dp_actor.py:84)."""
 
     def __init__(self, tag: str):
         self.tag = tag
+
+    def __call__(self, fn):
+        @functools.wraps(fn)
+        def wrapper(*a, **kw):
+            with GPUMemoryLogger(self.tag):
+                return fn(*a, **kw)
+        return wrapper
 
     def __enter__(self):
         if torch.cuda.is_available():

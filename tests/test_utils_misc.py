@@ -98,3 +98,47 @@ def test_reduce_and_data_metrics():
     m = compute_data_metrics(b, use_critic=False)
     assert m["critic/score/mean"] == 2.0
     assert m["response_length/mean"] == 2.0
+
+
+def test_main_ppo_sync_entry(tmp_path):
+    """main_ppo: the synchronous A/B baseline — forces the stream size to
+    the whole batch (one update wave per full generation)."""
+    from polyrl_amd.trainer.main_ppo import main as ppo_main
+    ppo_main([
+        "actor_rollout_ref.model.path=llama-debug-cpu",
+        "actor_rollout_ref.model.dtype=float32",
+        "actor_rollout_ref.model.enable_gradient_checkpointing=false",
+        "actor_rollout_ref.actor.ppo_mini_batch_size=8",
+        "actor_rollout_ref.actor.ppo_max_token_len_per_gpu=512",
+        "actor_rollout_ref.rollout.sampling.n=2",
+        "actor_rollout_ref.rollout.response_length=8",
+        "data.train_batch_size=8",
+        "data.max_prompt_length=16",
+        "data.synthetic_num_prompts=16",
+        f"trainer.default_local_dir={tmp_path}/ckpt",
+        "trainer.resume_mode=disable",
+        "reward=random",
+        "max_steps=1",
+    ])
+
+
+def test_profiling_helpers_cpu_noop():
+    """roctx_range / annotate / GPUMemoryLogger degrade to no-ops on CPU
+    and preserve the wrapped function's behavior."""
+    from polyrl_amd.utils.profiling import (GPUMemoryLogger, annotate,
+                                            log_gpu_memory, roctx_range)
+    with roctx_range("x"):
+        pass
+
+    @annotate("phase")
+    def f(a, b=2):
+        return a + b
+
+    assert f(1) == 3
+
+    @GPUMemoryLogger("role")
+    def g(x):
+        return x * 2
+
+    assert g(4) == 8
+    log_gpu_memory("tag")     # must not raise without CUDA
