@@ -111,3 +111,25 @@ def test_gsm8k_preprocess_to_dataset_roundtrip(tmp_path):
                                  str(gts[0])) == 1.0
     assert default_compute_score("openai/gsm8k", "x #### 999999",
                                  str(gts[0])) == 0.0
+
+
+def test_custom_reward_function_from_file(tmp_path):
+    """reward manager 'custom' loads a scoring fn from a python file
+    (the reference's custom_reward_function config)."""
+    import torch
+
+    from polyrl_amd.protocol import TensorBatch
+    from polyrl_amd.reward import load_reward_manager
+
+    f = tmp_path / "score.py"
+    f.write_text(
+        "def compute_score(sample, bonus=0.0):\n"
+        "    return float(sample['responses'].sum()) + bonus\n")
+    rm = load_reward_manager("custom", path=str(f), bonus=1.0)
+    b = TensorBatch.from_dict(tensors={
+        "responses": torch.tensor([[1, 2], [3, 4]]),
+        "response_mask": torch.ones(2, 2, dtype=torch.long)})
+    scores = rm(b)
+    # sequence score lands on the last response token
+    assert scores.shape == (2, 2)
+    assert float(scores[0].sum()) == 4.0 and float(scores[1].sum()) == 8.0
