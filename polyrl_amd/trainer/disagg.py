@@ -366,11 +366,15 @@ class DisaggCoordinator:
                 self.output_logprobs = s.output_logprobs
                 self.finish_reason = s.finish_reason
         outputs = [[_O(s) for s in res.samples] for _, res in pairs]
-        return postprocess_groups(prompt_ids, prompt_mask, uids, outputs,
-                                  self.response_length, self.pad,
-                                  self.device,
-                                  group_extras=[m.get("extras")
-                                                for m, _ in pairs])
+        out = postprocess_groups(prompt_ids, prompt_mask, uids, outputs,
+                                 self.response_length, self.pad,
+                                 self.device,
+                                 group_extras=[m.get("extras")
+                                               for m, _ in pairs])
+        # fault-tolerance observability: continuation hops in this slice
+        out.meta_info["num_migrations"] = int(sum(
+            s.num_migrations for _, res in pairs for s in res.samples))
+        return out
 
 
 class ElasticPublisher:

@@ -577,6 +577,8 @@ class StreamPPOTrainer:
             metrics["training/max_local_gen_s"] = new_window
             metrics["training/num_rollout_instances"] = float(
                 len(self.coordinator.scheduler.instances()))
+            metrics["training/num_migrations"] = float(sum(
+                b.meta_info.get("num_migrations", 0) for b in ibatches))
         return metrics
 
     def _collect_remax_baselines(self, submit_batch: TensorBatch,
