@@ -430,6 +430,13 @@ class Engine:
         self.kv.k_cache = []
         self.kv.v_cache = []
         self._kv_released = True
+        # Captured decode graphs hold device pointers into the freed KV
+        # tensors; replaying them after a release/resume cycle would read
+        # and write stale memory.  Drop them so they re-capture against the
+        # new PagedKVCache (the pool is dropped too — it owns the old
+        # allocations).
+        self._graphs = {}
+        self._graph_pool = None
         if str(self.device).startswith("cuda"):
             torch.cuda.empty_cache()
 
@@ -576,7 +583,12 @@ class Engine:
     def _emit(self, r: Request) -> RequestOutput:
         sid = self._seq_ids.pop(r.rid, None)
         if sid is not None:
-            if self.radix is not None:
+            # Aborted requests (weight-update time-box) must NOT donate KV
+            # to the radix cache: they are emitted on the step AFTER
+            # flush_radix(), so their pages hold OLD-weight KV and a
+            # token-exact continuation of the same request would radix-match
+            # that stale prefix and skip recomputing it under new weights.
+            if self.radix is not None and r.finish_reason != "abort":
                 written = self.kv.seq_len(sid)   # last sampled token has no KV
                 toks = (r.input_ids + r.output_ids)[:written]
                 self.radix.insert(toks, self.kv._seq_pages.get(sid, []))

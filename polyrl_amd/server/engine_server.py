@@ -168,6 +168,10 @@ def create_app(engine: Engine, runner: Optional[EngineRunner] = None,
         "shape": [...], "dtype": "bfloat16"}}} — base64 raw tensor bytes.
         The agent/TCP routes are the efficient paths; this one exists for
         drop-in client compatibility."""
+        if not _sender_ok(request):
+            return JSONResponse({"success": False,
+                                 "message": "sender not in allowed CIDRs"},
+                                403)
         import base64
 
         import numpy as np
@@ -230,6 +234,10 @@ def create_app(engine: Engine, runner: Optional[EngineRunner] = None,
         """Install weights from the armed TCP receive: wait for all spans,
         reconstruct named views (patches.py:205-215 capability), swap under
         the step lock."""
+        if not _sender_ok(request):
+            return JSONResponse({"success": False,
+                                 "message": "sender not in allowed CIDRs"},
+                                403)
         import torch
         body = await request.json()
         version = int(body["version"])

@@ -65,6 +65,7 @@ class TcpWeightReceiver:
         self._lock = threading.Lock()
         self._done = threading.Event()
         self._expected: Optional[int] = None
+        self._failed: Optional[str] = None
         for _ in range(num_streams):
             s = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
             s.setsockopt(socket.SOL_SOCKET, socket.SO_RCVBUF, SOCK_BUF)
@@ -77,6 +78,7 @@ class TcpWeightReceiver:
         """Arm for one transfer of ``total_bytes`` (default: whole buffer)."""
         with self._lock:
             self._received = 0
+            self._failed = None
         self._done.clear()
         self._expected = total_bytes if total_bytes is not None \
             else self.buffer.numel()
@@ -98,6 +100,15 @@ class TcpWeightReceiver:
                 off, length = _HDR.unpack(hdr)
                 if length == 0:      # stream end marker
                     return
+                if off < 0 or length < 0 or off + length > len(self._mv):
+                    # Fail fast instead of hanging until the install
+                    # timeout: flag the transfer failed and wake wait().
+                    with self._lock:
+                        self._failed = (
+                            f"span ({off}, {length}) exceeds buffer "
+                            f"size {len(self._mv)}")
+                    self._done.set()
+                    raise ConnectionError(self._failed)
                 got = 0
                 while got < length:
                     n = conn.recv_into(self._mv[off + got: off + length],
@@ -125,6 +136,8 @@ class TcpWeightReceiver:
 
     def wait(self, timeout: float = 300.0) -> bool:
         ok = self._done.wait(timeout)
+        if self._failed is not None:
+            raise RuntimeError(f"tcp weight receive failed: {self._failed}")
         for t in self._threads:
             t.join(timeout=5.0)
         return ok
