@@ -234,6 +234,36 @@ class SchedulerConfig:
 
 
 @dataclass
+class SandboxFusionConfig:
+    """Remote code-execution sandbox (reference:
+    reward_model.sandbox_fusion {url, max_concurrent, memory_limit_mb},
+    trainer/ppo/reward.py:128-141)."""
+    url: Optional[str] = None
+    max_concurrent: int = 64
+    memory_limit_mb: int = 1024
+
+
+@dataclass
+class CustomRewardFunctionConfig:
+    """Custom scoring fn loaded from a python file (reference:
+    custom_reward_function {path, name}, reward.py:60-93)."""
+    path: Optional[str] = None
+    name: str = "compute_score"
+
+
+@dataclass
+class RewardModelConfig:
+    """Reward-manager selection (reference: config.reward_model tree,
+    reward.py:95-150 — naive | prime | batch | dapo registry)."""
+    reward_manager: str = "naive"
+    sandbox_fusion: SandboxFusionConfig = field(
+        default_factory=SandboxFusionConfig)
+    # dapo manager knobs
+    overlong_buffer_len: int = 0
+    overlong_penalty_factor: float = 1.0
+
+
+@dataclass
 class PPOConfig:
     data: DataConfig = field(default_factory=DataConfig)
     actor_rollout_ref: ActorRolloutRefConfig = field(default_factory=ActorRolloutRefConfig)
@@ -241,6 +271,9 @@ class PPOConfig:
     algorithm: AlgorithmConfig = field(default_factory=AlgorithmConfig)
     trainer: TrainerConfig = field(default_factory=TrainerConfig)
     scheduler: SchedulerConfig = field(default_factory=SchedulerConfig)
+    reward_model: RewardModelConfig = field(default_factory=RewardModelConfig)
+    custom_reward_function: CustomRewardFunctionConfig = field(
+        default_factory=CustomRewardFunctionConfig)
 
 
 # ---------------------------------------------------------------- overrides

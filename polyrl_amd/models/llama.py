@@ -21,6 +21,15 @@ import torch.nn.functional as F
 from .registry import DecoderConfig
 
 
+def _pack_align() -> int:
+    """GEMM M-alignment for the packed varlen path (0/1 disables)."""
+    import os
+    try:
+        return max(int(os.environ.get("POLYRL_PACK_ALIGN", "256")), 1)
+    except ValueError:
+        return 256
+
+
 class RMSNorm(nn.Module):
     def __init__(self, dim: int, eps: float):
         super().__init__()
@@ -199,6 +208,22 @@ class DecoderModel(nn.Module):
         torch.cumsum(seqlens, 0, out=cu[1:])
         ids_p = input_ids[valid]                     # (total,)
         pos_p = position_ids[valid]
+        # Pad the packed token count to a multiple of PACK_ALIGN: dynamic
+        # token-budget micros produce arbitrary totals (6472, 7216, ...)
+        # and hipBLASLt loses ~35-45% on the [M,14336]x[14336,4096] trunk
+        # GEMMs at unaligned M (measured: profiles/PROFILES.md round 2).
+        # The pad rows form one trailing fake sequence so attention output
+        # stays finite (their grads are exactly zero — the scatter below
+        # never reads them).
+        T = int(ids_p.shape[0])
+        align = _pack_align()
+        Tp = -(-T // align) * align if align > 1 else T
+        if Tp != T:
+            padn = Tp - T
+            ids_p = torch.cat([ids_p, ids_p.new_zeros(padn)])
+            pos_p = torch.cat([pos_p, torch.arange(
+                padn, device=pos_p.device, dtype=pos_p.dtype)])
+            cu = torch.cat([cu, cu.new_full((1,), Tp)])
         cos, sin = self.rotary.get(pos_p)            # (total, D/2)
         x = self.embed_tokens(ids_p)
         # tile tables host-built ONCE per forward, shared by all layers
@@ -217,7 +242,7 @@ class DecoderModel(nn.Module):
                 x = layer(x, cos, sin, None, None, cu_pack)
         x = self.norm(x)
         out = torch.zeros(B, L, x.shape[-1], dtype=x.dtype, device=x.device)
-        out[valid] = x
+        out[valid] = x[:T]
         return out
 
     def forward(self, input_ids, attention_mask=None, position_ids=None):

@@ -122,20 +122,161 @@ class NaiveRewardManager:
         return _place_scores(batch, scores)
 
 
+class PrimeRewardManager(NaiveRewardManager):
+    """Parallel-verification manager (reference: verl's PrimeRewardManager,
+    selected by reward_model.reward_manager='prime', reward.py:112-120):
+    scores every sample concurrently in a thread pool — the scorers that
+    matter here (code execution, sandboxed HTTP) block on I/O, so threads
+    give real overlap; a scorer exception marks that sample 0."""
+
+    def __init__(self, tokenizer=None, compute_score=None,
+                 max_workers: int = 16, **_):
+        super().__init__(tokenizer=tokenizer, compute_score=compute_score)
+        self.max_workers = max_workers
+
+    def __call__(self, batch: TensorBatch) -> torch.Tensor:
+        from concurrent.futures import ThreadPoolExecutor
+        assert self.tokenizer is not None, "prime reward needs a tokenizer"
+        resp = batch["responses"]
+        mask = batch["response_mask"]
+        ds = batch.non_tensors.get("data_source")
+        gt = batch.non_tensors.get("ground_truth")
+        assert ds is not None and gt is not None
+
+        def score_one(i: int) -> float:
+            ids = resp[i][mask[i].bool()].tolist()
+            text = self.tokenizer.decode(ids)
+            try:
+                return float(self.compute_score(str(ds[i]), text,
+                                                str(gt[i])))
+            except Exception:                      # noqa: BLE001
+                return 0.0
+
+        with ThreadPoolExecutor(max_workers=self.max_workers) as ex:
+            scores = torch.tensor(list(ex.map(score_one,
+                                              range(len(batch)))))
+        return _place_scores(batch, scores)
+
+
+class BatchRewardManager(NaiveRewardManager):
+    """Batched-API manager (reference: verl's BatchRewardManager): calls
+    compute_score ONCE with lists (data_sources, solution_strs,
+    ground_truths) — for scorers that amortize over the batch (an RM
+    forward, a batched sandbox call)."""
+
+    def __call__(self, batch: TensorBatch) -> torch.Tensor:
+        assert self.tokenizer is not None, "batch reward needs a tokenizer"
+        resp = batch["responses"]
+        mask = batch["response_mask"]
+        ds = batch.non_tensors.get("data_source")
+        gt = batch.non_tensors.get("ground_truth")
+        assert ds is not None and gt is not None
+        texts = [self.tokenizer.decode(resp[i][mask[i].bool()].tolist())
+                 for i in range(len(batch))]
+        try:
+            res = self.compute_score([str(d) for d in ds], texts,
+                                     [str(g) for g in gt])
+            scores = torch.as_tensor([float(r) for r in res],
+                                     dtype=torch.float32)
+        except TypeError:
+            # scorer has the scalar signature: fall back to a loop
+            scores = torch.tensor([
+                float(self.compute_score(str(ds[i]), texts[i], str(gt[i])))
+                for i in range(len(batch))])
+        return _place_scores(batch, scores)
+
+
+class DAPORewardManager(NaiveRewardManager):
+    """DAPO manager (reference: verl's DAPORewardManager, the style the
+    math_dapo/aime recipes train with): rule score plus a soft overlong
+    penalty — responses longer than (max_response_length -
+    overlong_buffer_len) lose up to penalty_factor linearly with the
+    overflow, discouraging truncation-length collapse."""
+
+    def __init__(self, tokenizer=None, compute_score=None,
+                 overlong_buffer_len: int = 0,
+                 overlong_penalty_factor: float = 1.0,
+                 max_response_length: Optional[int] = None, **_):
+        super().__init__(tokenizer=tokenizer, compute_score=compute_score)
+        self.overlong_buffer_len = overlong_buffer_len
+        self.overlong_penalty_factor = overlong_penalty_factor
+        self.max_response_length = max_response_length
+
+    def __call__(self, batch: TensorBatch) -> torch.Tensor:
+        out = super().__call__(batch)
+        if not self.overlong_buffer_len:
+            return out
+        resp_mask = batch["response_mask"]
+        max_len = self.max_response_length or resp_mask.shape[1]
+        lens = resp_mask.sum(-1).long()
+        expected = max_len - self.overlong_buffer_len
+        for i in range(len(batch)):
+            over = int(lens[i]) - expected
+            if over > 0 and lens[i] > 0:
+                pen = min(over / self.overlong_buffer_len, 1.0) \
+                    * self.overlong_penalty_factor
+                out[i, lens[i] - 1] -= pen
+        return out
+
+
 _REGISTRY: Dict[str, type] = {
     "constant": ConstantReward,
     "random": RandomReward,
     "length": LengthReward,
     "naive": NaiveRewardManager,
+    "prime": PrimeRewardManager,
+    "batch": BatchRewardManager,
+    "dapo": DAPORewardManager,
 }
 
 
-def load_reward_manager(name: str = "constant", **kwargs):
+def load_reward_manager(name: str = "constant", *,
+                        sandbox_fusion_url: Optional[str] = None,
+                        sandbox_max_concurrent: int = 64,
+                        sandbox_memory_limit_mb: int = 1024,
+                        **kwargs):
+    """Build a reward manager by name (reference surface:
+    trainer/ppo/reward.py:95-150 — naive/prime/batch/dapo registry +
+    custom fn from file + sandbox-fusion url gated by a concurrency
+    semaphore)."""
     if name == "custom":
         return FunctionReward(**kwargs)
     if name not in _REGISTRY:
         raise KeyError(f"unknown reward manager {name!r}: {sorted(_REGISTRY)}")
+    if sandbox_fusion_url and "compute_score" not in kwargs:
+        import functools
+        import threading
+
+        from .reward_score import default_compute_score
+        sem = threading.Semaphore(sandbox_max_concurrent)
+        kwargs["compute_score"] = functools.partial(
+            default_compute_score, sandbox_fusion_url=sandbox_fusion_url,
+            concurrent_semaphore=sem,
+            memory_limit_mb=sandbox_memory_limit_mb)
     return _REGISTRY[name](**kwargs)
+
+
+def load_reward_manager_from_config(cfg, tokenizer=None):
+    """Reference precedence (trainer/ppo/reward.py:95-150): a custom
+    reward function file wins; else the data_source dispatch, wrapped with
+    the sandbox-fusion url + semaphore when configured; manager style from
+    reward_model.reward_manager."""
+    crf = getattr(cfg, "custom_reward_function", None)
+    if crf is not None and crf.path:
+        return FunctionReward(path=crf.path, name=crf.name)
+    rm = cfg.reward_model
+    sb = rm.sandbox_fusion
+    kwargs = {}
+    if rm.reward_manager == "dapo":
+        kwargs.update(
+            overlong_buffer_len=rm.overlong_buffer_len,
+            overlong_penalty_factor=rm.overlong_penalty_factor,
+            max_response_length=cfg.data.max_response_length)
+    return load_reward_manager(
+        rm.reward_manager, tokenizer=tokenizer,
+        sandbox_fusion_url=sb.url,
+        sandbox_max_concurrent=sb.max_concurrent,
+        sandbox_memory_limit_mb=sb.memory_limit_mb, **kwargs)
 
 
 def compute_reward(batch: TensorBatch, reward_fn) -> torch.Tensor:

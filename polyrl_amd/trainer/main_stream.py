@@ -56,7 +56,15 @@ def main(argv=None):
         print(f"[main_stream] world={world} device={device_type} "
               f"model={cfg.actor_rollout_ref.model.path} "
               f"adv={cfg.algorithm.adv_estimator}", flush=True)
-    reward_fn = load_reward_manager(reward_name)
+    if reward_name == "config":
+        # reference precedence: custom fn file > sandbox-wrapped dispatch,
+        # manager style from reward_model.reward_manager
+        from ..reward import load_reward_manager_from_config
+        from ..utils.tokenizer import get_tokenizer
+        reward_fn = load_reward_manager_from_config(
+            cfg, tokenizer=get_tokenizer(cfg.actor_rollout_ref.model.path))
+    else:
+        reward_fn = load_reward_manager(reward_name)
 
     ro = cfg.actor_rollout_ref.rollout
     if ro.num_rollout_ranks > 0:

@@ -42,10 +42,42 @@ def bench(fn, iters=20, warmup=5):
     return s.elapsed_time(e) / iters * 1e3   # us
 
 
+def m_sweep():
+    """Does M%256 alignment alone recover the in-situ GEMM losses?
+    In-trainer micros measured M in {6472, 7000, 7216, 7428} at 0.88-0.92
+    PF on the [M,14336]x[14336,4096] slots; this sweeps raw vs 256-aligned
+    M over the two hot layouts."""
+    dev = "cuda"
+    shapes = [6472, 6656, 7000, 7168, 7216, 7424, 7428, 7680, 8192]
+    N, K = 4096, 14336        # down fwd (nt) / gate dgrad (nn) slot
+    print(f"{'M':>6} {'M%256':>6} | {'nt_us':>8} {'nt_TF':>7} | "
+          f"{'nn_us':>8} {'nn_TF':>7} | {'tn_us':>8} {'tn_TF':>7}")
+    for M in shapes:
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev)
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=dev)
+        dy = torch.randn(M, N, dtype=torch.bfloat16, device=dev)
+        fl = 2.0 * M * N * K
+        t_nt = bench(lambda: x @ w.t())
+        t_nn = bench(lambda: dy @ w)
+        t_tn = bench(lambda: dy.t() @ x)
+        print(f"{M:>6} {M % 256:>6} | {t_nt:8.1f} {fl/t_nt/1e6:7.0f} | "
+              f"{t_nn:8.1f} {fl/t_nn/1e6:7.0f} | "
+              f"{t_tn:8.1f} {fl/t_tn/1e6:7.0f}")
+        # reverse slot: gate fwd (nt) at [M,4096] -> 14336
+        w2 = torch.randn(K, N, dtype=torch.bfloat16, device=dev)
+        dy2 = torch.randn(M, K, dtype=torch.bfloat16, device=dev)
+        x2 = torch.randn(M, N, dtype=torch.bfloat16, device=dev)
+        t2 = bench(lambda: x2 @ w2.t())
+        t2n = bench(lambda: dy2 @ w2)
+        print(f"{'':>6} {'rev':>6} | {t2:8.1f} {fl/t2/1e6:7.0f} | "
+              f"{t2n:8.1f} {fl/t2n/1e6:7.0f} |")
+
+
 def main():
     assert torch.cuda.is_available()
     dev = "cuda"
     torch.manual_seed(0)
+    m_sweep()
     print(f"{'layer':>8} {'M':>6} {'N':>7} {'K':>6} | "
           f"{'nt_us':>8} {'nt_TF':>7} | {'nn_us':>8} {'nn_TF':>7} | "
           f"{'tn_us':>8} {'tn_TF':>7}")

@@ -133,3 +133,160 @@ def test_custom_reward_function_from_file(tmp_path):
     # sequence score lands on the last response token
     assert scores.shape == (2, 2)
     assert float(scores[0].sum()) == 4.0 and float(scores[1].sum()) == 8.0
+
+
+# ---------------------------------------------------------------- round 2:
+# full data_source dispatch parity (reference reward_score/__init__.py:19-117)
+
+def test_math_dapo_scorer():
+    from polyrl_amd.reward_score import default_compute_score, math_dapo
+    r = math_dapo.compute_score("thus \\boxed{42}", "42")
+    assert r["score"] == 1.0 and r["acc"]
+    r = math_dapo.compute_score("thus \\boxed{41}", "42")
+    assert r["score"] == -1.0 and not r["acc"]
+    # bare "Answer: x" fallback
+    assert math_dapo.compute_score("The final answer is 7", "7")["acc"]
+    # dispatch collapses the dict to its score; aime* routes here
+    assert default_compute_score("math_dapo", "\\boxed{5}", "5") == 1.0
+    assert default_compute_score("aime_2024", "\\boxed{5}", "6") == -1.0
+
+
+def test_geo3k_scorer():
+    from polyrl_amd.reward_score import default_compute_score
+    assert default_compute_score("hiyouga/geometry3k",
+                                 "area is \\boxed{12.5}", "12.5") == 1.0
+    # numeric tolerance, not string match
+    assert default_compute_score("hiyouga/geometry3k",
+                                 "\\boxed{12.50001}", "12.5") == 1.0
+    assert default_compute_score("hiyouga/geometry3k",
+                                 "\\boxed{13}", "12.5") == 0.0
+
+
+def test_search_r1_scorer():
+    from polyrl_amd.reward_score import default_compute_score
+    sol = "<think>...</think><answer>the Eiffel Tower</answer>"
+    assert default_compute_score("searchR1_nq", sol, "eiffel tower") == 1.0
+    assert default_compute_score("searchR1_hotpotqa", sol,
+                                 ["Eiffel Tower", "paris tower"]) == 1.0
+    assert default_compute_score("searchR1_nq", "no tags here", "x") == 0.0
+
+
+def test_code_exec_scorer_local():
+    """prime_code-style local execution: run the program on stdin/stdout
+    test cases (no sandbox URL -> subprocess fallback, reference
+    __init__.py:88-95)."""
+    import json
+
+    from polyrl_amd.reward_score import default_compute_score
+    sol = "```python\nx = int(input())\nprint(x * 2)\n```"
+    gt = json.dumps({"inputs": ["3\n", "10\n"], "outputs": ["6\n", "20\n"]})
+    assert default_compute_score("codecontests", sol, gt) == 1.0
+    half = json.dumps({"inputs": ["3\n", "10\n"], "outputs": ["6\n", "99\n"]})
+    assert default_compute_score("apps", sol, half) == 0.5
+    assert default_compute_score("taco", "no code at all &%", gt) == 0.0
+
+
+def test_unknown_source_raises():
+    import pytest as _pt
+
+    from polyrl_amd.reward_score import default_compute_score
+    with _pt.raises(KeyError):
+        default_compute_score("not_a_dataset", "x", "y")
+
+
+def _mk_scored_batch(texts, sources, gts, Lr=16):
+    tok = ToyTok()
+    B = len(texts)
+    resp = torch.zeros(B, Lr, dtype=torch.long)
+    mask = torch.zeros(B, Lr, dtype=torch.long)
+    for i, t in enumerate(texts):
+        ids = [ord(c) - 48 for c in t][:Lr]
+        resp[i, :len(ids)] = torch.tensor(ids)
+        mask[i, :len(ids)] = 1
+    return TensorBatch(
+        tensors={"responses": resp, "response_mask": mask},
+        non_tensors={"data_source": np.array(sources, dtype=object),
+                     "ground_truth": np.array(gts, dtype=object)})
+
+
+def test_prime_manager_parallel_scoring():
+    from polyrl_amd.reward import load_reward_manager
+    b = _mk_scored_batch(["#### 3", "#### 4", "#### 3"],
+                         ["gsm8k"] * 3, ["3"] * 3)
+    rm = load_reward_manager("prime", tokenizer=ToyTok(), max_workers=3)
+    s = rm(b)
+    assert s.sum() == 2.0
+
+
+def test_batch_manager_list_api():
+    from polyrl_amd.reward import load_reward_manager
+
+    calls = []
+
+    def batched(sources, sols, gts):
+        calls.append(len(sols))
+        return [1.0 if g in s else 0.0 for s, g in zip(sols, gts)]
+
+    b = _mk_scored_batch(["#### 3", "#### 4"], ["gsm8k"] * 2, ["3", "9"])
+    rm = load_reward_manager("batch", tokenizer=ToyTok(),
+                             compute_score=batched)
+    s = rm(b)
+    assert calls == [2]
+    assert s[0].sum() == 1.0 and s[1].sum() == 0.0
+
+
+def test_dapo_manager_overlong_penalty():
+    from polyrl_amd.reward import load_reward_manager
+    Lr = 16
+    # correct answer but response length 14 > (16 - 4) => penalized
+    long_text = "#### 3" + "x" * 8          # 14 chars
+    short_text = "#### 3"                    # 6 chars, under the threshold
+    b = _mk_scored_batch([long_text, short_text], ["gsm8k"] * 2,
+                         ["3"] * 2, Lr=Lr)
+    rm = load_reward_manager("dapo", tokenizer=ToyTok(),
+                             overlong_buffer_len=4,
+                             overlong_penalty_factor=1.0,
+                             max_response_length=Lr)
+    s = rm(b)
+    # long: 1.0 - min(2/4, 1)*1.0 = 0.5 ; short: untouched 1.0
+    assert abs(float(s[0].sum()) - 0.5) < 1e-6
+    assert float(s[1].sum()) == 1.0
+
+
+def test_sandbox_semaphore_wrapping():
+    """load_reward_manager with a sandbox url builds a semaphore-gated
+    compute_score partial (reference reward.py:128-141); we verify the
+    partial wiring without a live sandbox by hitting an unroutable URL —
+    sandbox errors score 0, not raise."""
+    from polyrl_amd.reward import load_reward_manager
+    rm = load_reward_manager("naive", tokenizer=ToyTok(),
+                             sandbox_fusion_url="http://127.0.0.1:1",
+                             sandbox_max_concurrent=2)
+    import functools
+    assert isinstance(rm.compute_score, functools.partial)
+    assert rm.compute_score.keywords["sandbox_fusion_url"] \
+        == "http://127.0.0.1:1"
+    b = _mk_scored_batch(["some code"], ["codecontests"], ["{}"])
+    s = rm(b)
+    assert float(s.sum()) == 0.0
+
+
+def test_load_reward_manager_from_config():
+    from polyrl_amd.config import PPOConfig
+    from polyrl_amd.reward import (DAPORewardManager, FunctionReward,
+                                   load_reward_manager_from_config)
+    cfg = PPOConfig()
+    cfg.reward_model.reward_manager = "dapo"
+    cfg.reward_model.overlong_buffer_len = 8
+    rm = load_reward_manager_from_config(cfg, tokenizer=ToyTok())
+    assert isinstance(rm, DAPORewardManager)
+    assert rm.overlong_buffer_len == 8
+    assert rm.max_response_length == cfg.data.max_response_length
+    # custom fn file wins over everything
+    import tempfile
+    with tempfile.NamedTemporaryFile("w", suffix=".py", delete=False) as f:
+        f.write("def compute_score(sample):\n    return 1.0\n")
+        path = f.name
+    cfg.custom_reward_function.path = path
+    rm2 = load_reward_manager_from_config(cfg)
+    assert isinstance(rm2, FunctionReward)
