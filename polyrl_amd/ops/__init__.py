@@ -247,20 +247,34 @@ class _FlashAttnVarlen(torch.autograd.Function):
         ext.varlen_prefill_attention(out, qc, kc, vc, cu_seqlens, cu_seqlens,
                                      t64_seq, t64_q0, scale, causal, lse)
         ctx.save_for_backward(qc, kc, vc, out, lse, cu_seqlens,
-                              t32_seq, t32_k0)
+                              t32_seq, t32_k0, t64_seq, t64_q0)
         ctx.scale = scale
         ctx.causal = causal
         return out
 
     @staticmethod
     def backward(ctx, dout):
-        q, k, v, out, lse, cu, t32_seq, t32_k0 = ctx.saved_tensors
+        q, k, v, out, lse, cu, t32_seq, t32_k0, t64_seq, t64_q0 = \
+            ctx.saved_tensors
         ext = _require_ext()
+        import os as _os
+        ver = _os.environ.get("POLYRL_ATTN_BWD", "v3")
+        if ver == "v3":
+            # atomic-free dkv+dq split: every output row written exactly
+            # once, so the grad buffers need no zero-fill
+            dq = torch.empty(q.shape, dtype=torch.float32, device=q.device)
+            dk = torch.empty(k.shape, dtype=torch.float32, device=q.device)
+            dv = torch.empty(v.shape, dtype=torch.float32, device=q.device)
+            ext.varlen_attention_backward_v3(
+                dq, dk, dv, q, k, v, out, dout.contiguous(), lse,
+                cu, cu, t64_seq, t64_q0, ctx.scale, ctx.causal)
+            return (dq.to(q.dtype), dk.to(q.dtype), dv.to(q.dtype),
+                    None, None, None, None, None, None, None)
         dq = torch.zeros(q.shape, dtype=torch.float32, device=q.device)
         dk = torch.zeros(k.shape, dtype=torch.float32, device=q.device)
         dv = torch.zeros(v.shape, dtype=torch.float32, device=q.device)
-        import os as _os
-        use_v2 = _os.environ.get("POLYRL_ATTN_BWD_V2", "0") == "1"
+        use_v2 = ver == "v2" or \
+            _os.environ.get("POLYRL_ATTN_BWD_V2", "0") == "1"
         ext.varlen_attention_backward(
             dq, dk, dv, q, k, v, out, dout.contiguous(), lse,
             cu, cu, t32_seq, t32_k0, ctx.scale, ctx.causal, use_v2)

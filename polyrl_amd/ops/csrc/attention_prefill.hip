@@ -1,4 +1,4 @@
-// Varlen causal prefill attention, bf16, head_dim 128, GQA — MFMA forward.
+// Varlen causal prefill attention, bf16, head_dim 64/128, GQA — MFMA forward.
 //
 // Reference capability: chunked-prefill flash attention (SURVEY.md §2.4.3
 // row 2).  MI355X-first structure (flash-style online softmax, no S matrix
@@ -29,20 +29,21 @@ typedef __attribute__((ext_vector_type(4))) float f32x4_t;
 #define QROWS_PER_WAVE 16
 #define NWAVE 4
 #define QTILE (QROWS_PER_WAVE * NWAVE)  // 64
-#define HEAD_DIM 128
 
 // K-tile swizzle: byte_col ^= (row & 7) << 4  (16-B granularity, preserves
 // ds_read_b128 alignment; spreads a 16-lane group over 8 slots)
+template <int D>
 DEV_INLINE int kswz(int row, int byte_col) {
-  return row * (HEAD_DIM * 2) + (byte_col ^ ((row & 7) << 4));
+  return row * (D * 2) + (byte_col ^ ((row & 7) << 4));
 }
 
+template <int HEAD_DIM>
 __global__ __launch_bounds__(256) void prefill_attn_kernel(
-    bf16_t* __restrict__ out,        // (total_q, Hq, 128)
+    bf16_t* __restrict__ out,        // (total_q, Hq, D)
     float* __restrict__ lse,         // (total_q, Hq) log-sum-exp, or null
-    const bf16_t* __restrict__ q,    // (total_q, Hq, 128)
-    const bf16_t* __restrict__ k,    // (total_k, Hk, 128)
-    const bf16_t* __restrict__ v,    // (total_k, Hk, 128)
+    const bf16_t* __restrict__ q,    // (total_q, Hq, D)
+    const bf16_t* __restrict__ k,    // (total_k, Hk, D)
+    const bf16_t* __restrict__ v,    // (total_k, Hk, D)
     const int* __restrict__ cu_q,    // (B+1,)
     const int* __restrict__ cu_k,
     const int* __restrict__ tile_seq,  // (ntiles,)
@@ -66,32 +67,34 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
   const int wid = threadIdx.x >> 6;
   const int l15 = lane & 15;
   const int lhi = lane >> 4;                       // 0..3
+  constexpr int KS = HEAD_DIM / 32;                // A/B k-steps over d
+  constexpr int NDT = HEAD_DIM / 16;               // 16-wide d-tiles of O
 
   // this wave's q rows: local q0 + wid*16 + (row index 0..15)
   const int wq0 = q0 + wid * QROWS_PER_WAVE;
 
   // ---- Q fragments (A-layout): row i = l15, k-elem = lhi*8 + j ----
   // 4 k-steps cover d = 0..127.  Rows past Lq load row 0 (masked later).
-  bf16x8_t qfrag[4];
+  bf16x8_t qfrag[KS];
   {
     const int qrow_l = wq0 + l15;
     const long grow = (long)qbeg + ((qrow_l < Lq) ? qrow_l : 0);
     const bf16_t* qp = q + grow * ldq + (long)hq * HEAD_DIM + lhi * 8;
 #pragma unroll
-    for (int ks = 0; ks < 4; ++ks)
+    for (int ks = 0; ks < KS; ++ks)
       qfrag[ks] = *reinterpret_cast<const bf16x8_t*>(qp + ks * 32);
   }
 
   // online-softmax state: rows (lhi*4 + r) for r=0..3 (C-layout rows)
   float m_run[4], l_run[4];
-  f32x4_t o_acc[8];  // 8 d-tiles of 16; reg r = row lhi*4+r
+  f32x4_t o_acc[NDT];  // d-tiles of 16; reg r = row lhi*4+r
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     m_run[r] = -1e30f;
     l_run[r] = 0.f;
   }
 #pragma unroll
-  for (int nt = 0; nt < 8; ++nt) o_acc[nt] = {0.f, 0.f, 0.f, 0.f};
+  for (int nt = 0; nt < NDT; ++nt) o_acc[nt] = {0.f, 0.f, 0.f, 0.f};
 
   // causal kv extent for this block
   const int max_qpos = qk_off + min(q0 + QTILE - 1, Lq - 1);
@@ -103,7 +106,7 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
     {
       const int tid = threadIdx.x;
 #pragma unroll
-      for (int it = 0; it < 2; ++it) {
+      for (int it = 0; it < KVBLK * HEAD_DIM / (256 * 8); ++it) {
         const int flat = (tid + it * 256) * 8;
         const int row = flat / HEAD_DIM;
         const int col = flat % HEAD_DIM;
@@ -118,7 +121,7 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
           for (int j = 0; j < 8; ++j) kv8.v[j] = f2bf(0.f), vv8.v[j] = f2bf(0.f);
         }
         *reinterpret_cast<bf16x8*>(
-            reinterpret_cast<char*>(Ks) + kswz(row, col * 2)) = kv8;
+            reinterpret_cast<char*>(Ks) + kswz<HEAD_DIM>(row, col * 2)) = kv8;
         *reinterpret_cast<bf16x8*>(Vs + row * HEAD_DIM + col) = vv8;
       }
     }
@@ -131,9 +134,10 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
     for (int nt = 0; nt < 2; ++nt) {
       const int krow = nt * 16 + l15;
 #pragma unroll
-      for (int ks = 0; ks < 4; ++ks) {
+      for (int ks = 0; ks < KS; ++ks) {
         const bf16x8_t bf = *reinterpret_cast<const bf16x8_t*>(
-            reinterpret_cast<char*>(Ks) + kswz(krow, (ks * 32 + lhi * 8) * 2));
+            reinterpret_cast<char*>(Ks) +
+            kswz<HEAD_DIM>(krow, (ks * 32 + lhi * 8) * 2));
         c[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[ks], bf, c[nt],
                                                         0, 0, 0);
       }
@@ -178,7 +182,7 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
     }
     // rescale O
 #pragma unroll
-    for (int nt = 0; nt < 8; ++nt)
+    for (int nt = 0; nt < NDT; ++nt)
 #pragma unroll
       for (int r = 0; r < 4; ++r) o_acc[nt][r] *= alpha[r];
 
@@ -196,7 +200,7 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
     // ---------------- P @ V ------------------------------------------------
     // B-frag (key x d): lane needs V[key = lhi*8 + j][d = nt*16 + l15]
 #pragma unroll
-    for (int nt = 0; nt < 8; ++nt) {
+    for (int nt = 0; nt < NDT; ++nt) {
       bf16x8_t bv;
 #pragma unroll
       for (int j = 0; j < 8; ++j)
@@ -216,7 +220,7 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
     const float inv = (l_run[r] > 0.f) ? 1.0f / l_run[r] : 0.f;
     bf16_t* op = out + (((long)qbeg + qrow_l) * Hq + hq) * HEAD_DIM;
 #pragma unroll
-    for (int nt = 0; nt < 8; ++nt)
+    for (int nt = 0; nt < NDT; ++nt)
       op[nt * 16 + l15] = f2bf(o_acc[nt][r] * inv);
     if (lse != nullptr && l15 == 0)
       lse[((long)qbeg + qrow_l) * Hq + hq] =
@@ -246,11 +250,12 @@ void varlen_prefill_attention(torch::Tensor out, torch::Tensor q,
   TORCH_CHECK(cu_seqlens_q.dtype() == torch::kInt32);
   TORCH_CHECK(tile_seq.dtype() == torch::kInt32 && tile_q0.dtype() == torch::kInt32);
   const int Hq = q.size(1), D = q.size(2), Hk = k.size(1);
-  TORCH_CHECK(D == 128, "prefill kernel supports head_dim 128");
+  TORCH_CHECK(D == 128 || D == 64, "prefill kernel supports head_dim 64/128");
   TORCH_CHECK(Hq % Hk == 0);
   const int ntiles = tile_seq.size(0);
   auto stream = at::hip::getCurrentHIPStream();
-  prefill_attn_kernel<<<dim3(ntiles, Hq), dim3(256), 0, stream>>>(
+  auto kern = (D == 128) ? prefill_attn_kernel<128> : prefill_attn_kernel<64>;
+  kern<<<dim3(ntiles, Hq), dim3(256), 0, stream>>>(
       (bf16_t*)out.data_ptr(),
       want_lse ? lse.data_ptr<float>() : nullptr,
       (const bf16_t*)q.data_ptr(),

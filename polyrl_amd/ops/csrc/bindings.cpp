@@ -35,6 +35,13 @@ void varlen_attention_backward(
     torch::Tensor cu_seqlens_q, torch::Tensor cu_seqlens_k,
     torch::Tensor tile_seq, torch::Tensor tile_k0,
     double scale, bool causal, bool use_v2);
+void varlen_attention_backward_v3(
+    torch::Tensor dq, torch::Tensor dk, torch::Tensor dv,
+    torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    torch::Tensor out, torch::Tensor dout, torch::Tensor lse,
+    torch::Tensor cu_seqlens_q, torch::Tensor cu_seqlens_k,
+    torch::Tensor t64_seq, torch::Tensor t64_q0,
+    double scale, bool causal);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, CDNA4)");
@@ -53,4 +60,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "varlen causal prefill attention (MFMA, bf16; optional LSE out)");
   m.def("varlen_attention_backward", &varlen_attention_backward,
         "varlen causal flash-attention backward (MFMA, bf16)");
+  m.def("varlen_attention_backward_v3", &varlen_attention_backward_v3,
+        "varlen flash-attention backward v3 (atomic-free dkv+dq split, "
+        "head_dim 64/128)");
 }

@@ -72,12 +72,13 @@ class StreamPPOTrainer:
         dtype = arr.model.dtype
 
         def _set_remove_padding(m, mc, enabled):
-            # packed varlen path: GPU kernels need head_dim 128 (llama/qwen
-            # families); the CPU tier runs the torch reference at any dim
+            # packed varlen path: GPU kernels cover head_dim 64 and 128
+            # (llama/qwen families); the CPU tier runs the torch reference
+            # at any dim
             trunk = getattr(m, "model", None)
             if trunk is None or not hasattr(trunk, "use_remove_padding"):
                 return
-            if enabled and (device == "cpu" or mc.head_dim == 128):
+            if enabled and (device == "cpu" or mc.head_dim in (64, 128)):
                 trunk.use_remove_padding = True
 
         torch.manual_seed(config.trainer.seed)

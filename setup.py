@@ -32,6 +32,7 @@ sources = [
         "attention_decode.hip",
         "attention_prefill.hip",
         "attention_backward.hip",
+        "attention_backward_v3.hip",
     )
 ]
 

@@ -588,3 +588,83 @@ def test_engine_radix_cache_greedy_equality_gpu():
     assert o2 == ref2, (o2, ref2)
     e.flush_radix()
     assert e.kv.free_pages == e.kv.num_pages
+
+
+# ------------------------------------- flash-attn backward v3 (atomic-free)
+
+
+@pytest.mark.parametrize("D,Hq,Hk", [(128, 8, 2), (64, 8, 8), (64, 16, 4)])
+def test_flash_attn_backward_v3_vs_ref(D, Hq, Hk, monkeypatch):
+    """v3 dkv+dq split vs torch-autograd fp32 reference, D in {64, 128},
+    with GQA and ragged sequence lengths."""
+    monkeypatch.setenv("POLYRL_ATTN_BWD", "v3")
+    torch.manual_seed(42)
+    lens = [100, 64, 31, 1, 130]
+    total = sum(lens)
+    cu = torch.tensor([0] + list(torch.tensor(lens).cumsum(0)),
+                      dtype=torch.int32, device=DEV)
+    scale = 1.0 / math.sqrt(D)
+    q = (torch.randn(total, Hq, D, device=DEV) / 4).bfloat16().requires_grad_()
+    k = (torch.randn(total, Hk, D, device=DEV) / 4).bfloat16().requires_grad_()
+    v = (torch.randn(total, Hk, D, device=DEV) / 4).bfloat16().requires_grad_()
+    w = torch.randn(total, Hq, D, device=DEV)
+    out = ops.flash_attn_varlen(q, k, v, cu, scale, causal=True)
+    (out.float() * w).sum().backward()
+    dq, dk, dv = q.grad.clone(), k.grad.clone(), v.grad.clone()
+
+    q2 = q.detach().cpu().float().requires_grad_()
+    k2 = k.detach().cpu().float().requires_grad_()
+    v2 = v.detach().cpu().float().requires_grad_()
+    out2 = ref.varlen_prefill_attention(q2, k2, v2, cu.cpu(), cu.cpu(),
+                                        scale, True)
+    (out2 * w.cpu()).sum().backward()
+    assert rel_err(out.cpu(), out2.detach()) < 2e-2
+    assert rel_err(dq.cpu(), q2.grad) < 3e-2, rel_err(dq.cpu(), q2.grad)
+    assert rel_err(dk.cpu(), k2.grad) < 3e-2, rel_err(dk.cpu(), k2.grad)
+    assert rel_err(dv.cpu(), v2.grad) < 3e-2, rel_err(dv.cpu(), v2.grad)
+
+
+def test_flash_attn_backward_v3_matches_v1(monkeypatch):
+    """v3 and v1 compute the same math from identical saved tensors —
+    agreement should be much tighter than either-vs-fp32."""
+    torch.manual_seed(43)
+    Hq, Hk, D = 8, 2, 128
+    lens = [512, 200, 64]
+    total = sum(lens)
+    cu = torch.tensor([0] + list(torch.tensor(lens).cumsum(0)),
+                      dtype=torch.int32, device=DEV)
+    scale = 1.0 / math.sqrt(D)
+    qv = (torch.randn(total, Hq, D, device=DEV) / 4).bfloat16()
+    kv_ = (torch.randn(total, Hk, D, device=DEV) / 4).bfloat16()
+    vv = (torch.randn(total, Hk, D, device=DEV) / 4).bfloat16()
+    w = torch.randn(total, Hq, D, device=DEV)
+    grads = {}
+    for ver in ("v1", "v3"):
+        monkeypatch.setenv("POLYRL_ATTN_BWD", ver)
+        q = qv.clone().requires_grad_()
+        k = kv_.clone().requires_grad_()
+        v = vv.clone().requires_grad_()
+        out = ops.flash_attn_varlen(q, k, v, cu, scale, causal=True)
+        (out.float() * w).sum().backward()
+        grads[ver] = (q.grad.clone(), k.grad.clone(), v.grad.clone())
+    for a, b in zip(grads["v1"], grads["v3"]):
+        assert rel_err(a, b) < 5e-3, rel_err(a, b)
+
+
+def test_prefill_d64_forward():
+    """head_dim-64 forward vs fp32 reference (qwen-small / llama3-1b class
+    models take the packed path too — VERDICT round-1 gap #3)."""
+    torch.manual_seed(44)
+    Hq, Hk, D = 12, 4, 64
+    lens = [70, 128, 5]
+    total = sum(lens)
+    cu = torch.tensor([0] + list(torch.tensor(lens).cumsum(0)),
+                      dtype=torch.int32, device=DEV)
+    scale = 1.0 / math.sqrt(D)
+    q = (torch.randn(total, Hq, D, device=DEV) / 4).bfloat16()
+    k = (torch.randn(total, Hk, D, device=DEV) / 4).bfloat16()
+    v = (torch.randn(total, Hk, D, device=DEV) / 4).bfloat16()
+    out = ops.varlen_prefill_attention(q, k, v, cu, cu, scale)
+    expect = ref.varlen_prefill_attention(q.cpu(), k.cpu(), v.cpu(),
+                                          cu.cpu(), cu.cpu(), scale)
+    assert rel_err(out.cpu(), expect) < 2e-2
