@@ -279,7 +279,12 @@ class ActorWorker:
     def _forward_logprobs_sp(self, ids, am, pos, Lr, want_entropy):
         """Ulysses path: full inputs on every SP rank; per-shard logits ->
         per-shard logprobs -> autograd-aware sequence gather -> response
-        slice (verl's gather_outputs_and_unpad capability)."""
+        slice (verl's gather_outputs_and_unpad capability).
+
+        When the trunk takes the packed x SP path (use_remove_padding with
+        heads % sp == 0) logits come back as the rank's (Tp/sp, V) PACKED
+        token shard; labels are packed the same way and the gathered
+        logprobs are scattered onto the (B, L) grid at the end."""
         from ..parallel.ulysses import gather_seq, pad_to_multiple, slice_for_rank
         sp = self.sp_size
         B, L = ids.shape
@@ -287,10 +292,35 @@ class ActorWorker:
         am_p = pad_to_multiple(am, sp, 1)
         pos_p = pad_to_multiple(pos, sp, 1)
         logits = self.model(ids_p, attention_mask=am_p,
-                            position_ids=pos_p).float()   # (B, Lp/sp, V)
+                            position_ids=pos_p).float()
         labels_p = torch.cat(
             [ids_p[:, 1:], torch.zeros(B, 1, dtype=ids.dtype,
                                        device=ids.device)], dim=1)
+        trunk = getattr(self.model, "model", None)
+        meta = getattr(trunk, "pack_sp_meta", None)
+        if meta is not None:                       # packed x SP
+            labels_pk = labels_p[meta["valid"]]
+            pad = meta["Tp"] - meta["T"]
+            if pad:
+                labels_pk = torch.cat([labels_pk, labels_pk.new_zeros(pad)])
+            r = torch.distributed.get_rank(self.sp_group)
+            sh = meta["shard"]
+            labels_shard = labels_pk[r * sh:(r + 1) * sh]
+            lp_shard = algos.logprobs_from_logits(logits, labels_shard)
+            lp_pk = gather_seq(lp_shard, 0, self.sp_group)[:meta["T"]]
+            lp_grid = torch.zeros(B, ids_p.shape[1], dtype=lp_pk.dtype,
+                                  device=lp_pk.device)
+            lp_grid[meta["valid"]] = lp_pk
+            lp = lp_grid[:, L - Lr - 1:L - 1]
+            if want_entropy:
+                ent_shard = algos.entropy_from_logits(logits)
+                ent_pk = gather_seq(ent_shard, 0, self.sp_group)[:meta["T"]]
+                ent_grid = torch.zeros_like(lp_grid)
+                ent_grid[meta["valid"]] = ent_pk
+                ent = ent_grid[:, L - Lr - 1:L - 1]
+            else:
+                ent = torch.zeros_like(lp)
+            return lp, ent
         labels_shard = slice_for_rank(labels_p, 1, self.sp_group)
         lp_shard = algos.logprobs_from_logits(logits, labels_shard)
         lp = gather_seq(lp_shard, 1, self.sp_group)[:, :L]
@@ -429,9 +459,19 @@ class CriticWorker:
         from ..parallel.ulysses import gather_seq, pad_to_multiple
         sp = self.sp_size
         L = ids.shape[1]
-        v_shard = self.model(pad_to_multiple(ids, sp, 1),
+        ids_p = pad_to_multiple(ids, sp, 1)
+        v_shard = self.model(ids_p,
                              attention_mask=pad_to_multiple(am, sp, 1),
                              position_ids=pad_to_multiple(pos, sp, 1))
+        trunk = getattr(self.model, "model", None)
+        meta = getattr(trunk, "pack_sp_meta", None)
+        if meta is not None:                       # packed x SP token shard
+            v_pk = gather_seq(v_shard.float(), 0,
+                              self.sp_group)[:meta["T"]]
+            v_grid = torch.zeros(ids_p.shape, dtype=v_pk.dtype,
+                                 device=v_pk.device)
+            v_grid[meta["valid"]] = v_pk
+            return v_grid[:, L - Lr - 1:L - 1]
         v = gather_seq(v_shard.float(), 1, self.sp_group)[:, :L]
         return v[:, L - Lr - 1:L - 1]
 

