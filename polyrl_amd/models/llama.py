@@ -30,6 +30,23 @@ def _pack_align() -> int:
         return 256
 
 
+def _pack_pad_to(model) -> int:
+    """Fixed-M padding target for the packed path (0 disables).  Measured
+    (profiles/PROFILES.md round 2): hipBLASLt loses 20-45% at mid-size M
+    from CU-grid tail quantization — M=8192 fills 256 CUs in exactly 2
+    waves (1.6 PF nt) while M~7000 strands a partial wave (0.88-1.27 PF),
+    and 256-alignment alone does NOT recover it.  Padding every micro to
+    the token budget makes all trunk GEMM shapes fixed AND optimal."""
+    import os
+    env = os.environ.get("POLYRL_PACK_PAD_TO")
+    if env is not None:
+        try:
+            return max(int(env), 0)
+        except ValueError:
+            return 0
+    return int(getattr(model, "pack_pad_to", 0) or 0)
+
+
 class RMSNorm(nn.Module):
     def __init__(self, dim: int, eps: float):
         super().__init__()
@@ -237,7 +254,11 @@ class DecoderModel(nn.Module):
         pos_p = position_ids[valid]
         T = int(ids_p.shape[0])
         chunk = sp * _pack_align()
-        Tp = -(-T // chunk) * chunk
+        pad_to = _pack_pad_to(self)
+        if pad_to and T <= pad_to:
+            Tp = -(-pad_to // chunk) * chunk
+        else:
+            Tp = -(-T // chunk) * chunk
         if Tp != T:
             padn = Tp - T
             ids_p = torch.cat([ids_p, ids_p.new_zeros(padn)])
@@ -289,7 +310,11 @@ class DecoderModel(nn.Module):
         # never reads them).
         T = int(ids_p.shape[0])
         align = _pack_align()
-        Tp = -(-T // align) * align if align > 1 else T
+        pad_to = _pack_pad_to(self)
+        if pad_to and T <= pad_to:
+            Tp = pad_to
+        else:
+            Tp = -(-T // align) * align if align > 1 else T
         if Tp != T:
             padn = Tp - T
             ids_p = torch.cat([ids_p, ids_p.new_zeros(padn)])

@@ -71,7 +71,7 @@ class StreamPPOTrainer:
         self.model_cfg = model_cfg
         dtype = arr.model.dtype
 
-        def _set_remove_padding(m, mc, enabled):
+        def _set_remove_padding(m, mc, enabled, pad_to=0):
             # packed varlen path: GPU kernels cover head_dim 64 and 128
             # (llama/qwen families); the CPU tier runs the torch reference
             # at any dim
@@ -80,6 +80,10 @@ class StreamPPOTrainer:
                 return
             if enabled and (device == "cpu" or mc.head_dim in (64, 128)):
                 trunk.use_remove_padding = True
+                # fixed-M padding: every dynamic-budget micro lands on the
+                # same (optimal) GEMM shapes (models/llama.py _pack_pad_to)
+                if pad_to and device != "cpu":
+                    trunk.pack_pad_to = int(pad_to)
 
         torch.manual_seed(config.trainer.seed)
         actor_model = create_model(model_cfg, kind="actor", dtype=dtype,
@@ -99,7 +103,11 @@ class StreamPPOTrainer:
             from ..models.lora import apply_lora
             apply_lora(actor_model, arr.model.lora_rank,
                        alpha=arr.model.lora_alpha)
-        _set_remove_padding(actor_model, model_cfg, arr.model.use_remove_padding)
+        actor_pad_to = (arr.actor.ppo_max_token_len_per_gpu *
+                        arr.actor.ulysses_sequence_parallel_size
+                        if arr.actor.use_dynamic_bsz else 0)
+        _set_remove_padding(actor_model, model_cfg,
+                            arr.model.use_remove_padding, actor_pad_to)
         self.actor = ActorWorker(actor_model, arr.actor, device=device,
                                  pg=self.pg)
 
@@ -114,7 +122,7 @@ class StreamPPOTrainer:
             for p in ref_model.parameters():
                 p.requires_grad_(False)
             _set_remove_padding(ref_model, model_cfg,
-                                arr.model.use_remove_padding)
+                                arr.model.use_remove_padding, actor_pad_to)
             self.ref = ActorWorker(ref_model, arr.actor, device=device,
                                    is_ref=True, pg=self.pg)
 
@@ -130,8 +138,13 @@ class StreamPPOTrainer:
                                         dtype=dtype, device=device)
             if critic_cfg.model.enable_gradient_checkpointing:
                 critic_model.gradient_checkpointing_enable()
+            critic_pad_to = (critic_cfg.ppo_max_token_len_per_gpu *
+                             getattr(critic_cfg,
+                                     "ulysses_sequence_parallel_size", 1)
+                             if critic_cfg.use_dynamic_bsz else 0)
             _set_remove_padding(critic_model, critic_model_cfg,
-                                critic_cfg.model.use_remove_padding)
+                                critic_cfg.model.use_remove_padding,
+                                critic_pad_to)
             self.critic = CriticWorker(critic_model, critic_cfg,
                                        device=device, pg=self.pg)
 
