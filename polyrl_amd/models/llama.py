@@ -21,6 +21,16 @@ import torch.nn.functional as F
 from .registry import DecoderConfig
 
 
+def _lin(mod: nn.Module, x: torch.Tensor) -> torch.Tensor:
+    """Route trainer linears through the algo-pinned hipBLASLt path
+    (ops.tuned_linear); identical math to mod(x).  Wrapped modules (LoRA)
+    keep their own forward."""
+    if type(mod) is not nn.Linear:
+        return mod(x)
+    import polyrl_amd.ops as pops
+    return pops.tuned_linear(x, mod.weight, mod.bias)
+
+
 def _pack_align() -> int:
     """GEMM M-alignment for the packed varlen path (0/1 disables)."""
     import os
@@ -114,9 +124,9 @@ class Attention(nn.Module):
         the reverse all-to-all restores the token shard."""
         import polyrl_amd.ops as pops
         T = x.shape[0]
-        q = self.q_proj(x).view(T, self.num_heads, self.head_dim)
-        k = self.k_proj(x).view(T, self.num_kv_heads, self.head_dim)
-        v = self.v_proj(x).view(T, self.num_kv_heads, self.head_dim)
+        q = _lin(self.q_proj, x).view(T, self.num_heads, self.head_dim)
+        k = _lin(self.k_proj, x).view(T, self.num_kv_heads, self.head_dim)
+        v = _lin(self.v_proj, x).view(T, self.num_kv_heads, self.head_dim)
         cos = cos.to(q.dtype).unsqueeze(1)         # (T, 1, D/2)
         sin = sin.to(q.dtype).unsqueeze(1)
         d = self.head_dim // 2
@@ -142,7 +152,7 @@ class Attention(nn.Module):
             from ..parallel.ulysses import all_to_all_4d
             # (1, T, H/sp, D) -> scatter tokens, gather heads -> (1, T/sp, H, D)
             o = all_to_all_4d(o.unsqueeze(0), 1, 2, sp_group).squeeze(0)
-        return self.o_proj(o.reshape(x.shape[0], -1))
+        return _lin(self.o_proj, o.reshape(x.shape[0], -1))
 
     def forward(self, x, cos, sin, attn_bias_mask: Optional[torch.Tensor],
                 sp_group=None):
@@ -151,9 +161,9 @@ class Attention(nn.Module):
         all-to-all trades it for a head shard over the full sequence
         (parallel/ulysses.py)."""
         B, Ls, _ = x.shape
-        q = self.q_proj(x).view(B, Ls, self.num_heads, self.head_dim)
-        k = self.k_proj(x).view(B, Ls, self.num_kv_heads, self.head_dim)
-        v = self.v_proj(x).view(B, Ls, self.num_kv_heads, self.head_dim)
+        q = _lin(self.q_proj, x).view(B, Ls, self.num_heads, self.head_dim)
+        k = _lin(self.k_proj, x).view(B, Ls, self.num_kv_heads, self.head_dim)
+        v = _lin(self.v_proj, x).view(B, Ls, self.num_kv_heads, self.head_dim)
         if sp_group is not None:
             from ..parallel.ulysses import all_to_all_4d
             import torch.distributed as _dist
@@ -197,7 +207,8 @@ class MLP(nn.Module):
         self.down_proj = nn.Linear(i, h, bias=cfg.mlp_bias)
 
     def forward(self, x):
-        return self.down_proj(F.silu(self.gate_proj(x)) * self.up_proj(x))
+        return _lin(self.down_proj,
+                    F.silu(_lin(self.gate_proj, x)) * _lin(self.up_proj, x))
 
 
 class DecoderLayer(nn.Module):
@@ -432,7 +443,7 @@ class CausalLM(nn.Module):
         sp_on = self.model.ulysses is not None and self.model.ulysses.enabled
         if logits_slice is not None and not sp_on:
             hidden = hidden[:, logits_slice]
-        return self.lm_head(hidden)
+        return _lin(self.lm_head, hidden.contiguous())
 
     @torch.no_grad()
     def num_params(self) -> int:

@@ -282,6 +282,47 @@ class _FlashAttnVarlen(torch.autograd.Function):
                 None, None, None, None, None, None, None)
 
 
+class _TunedLinear(torch.autograd.Function):
+    """nn.Linear matmuls through the per-shape hipBLASLt algo search
+    (gemm_tuned.cpp): fwd nt, dgrad nn, wgrad tn — the trainer's three hot
+    layouts, each pinned to its measured-fastest algorithm in-process."""
+
+    @staticmethod
+    def forward(ctx, x, w):
+        ext = _require_ext()
+        ctx.save_for_backward(x, w)
+        return ext.tuned_linear_fwd(x, w)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        ext = _require_ext()
+        dy = dy.contiguous()
+        dx = ext.tuned_linear_dgrad(dy, w) if ctx.needs_input_grad[0] else None
+        dw = ext.tuned_linear_wgrad(dy, x) if ctx.needs_input_grad[1] else None
+        return dx, dw
+
+
+def _tuned_gemm_enabled() -> bool:
+    import os
+    return os.environ.get("POLYRL_TUNED_GEMM", "1") == "1"
+
+
+def tuned_linear(x: torch.Tensor, w: torch.Tensor,
+                 b: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """F.linear with per-shape algo pinning on GPU bf16 (training trunk
+    shapes); falls through to F.linear elsewhere (CPU tier, tiny dims,
+    non-bf16)."""
+    if (x.is_cuda and x.dtype == torch.bfloat16
+            and w.dtype == torch.bfloat16 and w.shape[0] >= 64
+            and w.shape[1] >= 64 and _tuned_gemm_enabled()):
+        shp = x.shape
+        y = _TunedLinear.apply(x.reshape(-1, shp[-1]).contiguous(), w)
+        y = y.view(*shp[:-1], w.shape[0])
+        return y if b is None else y + b
+    return torch.nn.functional.linear(x, w, b)
+
+
 def build_varlen_tiles(cu_seqlens_cpu: torch.Tensor, device):
     """Precompute the (64-row forward, 32-key backward) tile tables once per
     packed batch; reused by every layer's flash_attn_varlen call."""

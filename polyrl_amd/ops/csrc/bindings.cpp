@@ -35,6 +35,9 @@ void varlen_attention_backward(
     torch::Tensor cu_seqlens_q, torch::Tensor cu_seqlens_k,
     torch::Tensor tile_seq, torch::Tensor tile_k0,
     double scale, bool causal, bool use_v2);
+torch::Tensor tuned_linear_fwd(torch::Tensor x, torch::Tensor w);
+torch::Tensor tuned_linear_dgrad(torch::Tensor dy, torch::Tensor w);
+torch::Tensor tuned_linear_wgrad(torch::Tensor dy, torch::Tensor x);
 void varlen_attention_backward_v3(
     torch::Tensor dq, torch::Tensor dk, torch::Tensor dv,
     torch::Tensor q, torch::Tensor k, torch::Tensor v,
@@ -60,6 +63,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "varlen causal prefill attention (MFMA, bf16; optional LSE out)");
   m.def("varlen_attention_backward", &varlen_attention_backward,
         "varlen causal flash-attention backward (MFMA, bf16)");
+  m.def("tuned_linear_fwd", &tuned_linear_fwd,
+        "Y = X W^T via hipBLASLt with per-shape in-process algo search");
+  m.def("tuned_linear_dgrad", &tuned_linear_dgrad, "dX = dY W (algo-pinned)");
+  m.def("tuned_linear_wgrad", &tuned_linear_wgrad,
+        "dW = dY^T X (algo-pinned)");
   m.def("varlen_attention_backward_v3", &varlen_attention_backward_v3,
         "varlen flash-attention backward v3 (atomic-free dkv+dq split, "
         "head_dim 64/128)");

@@ -1,0 +1,229 @@
+// Per-shape hipBLASLt algo search for the trainer's hot GEMMs (host code).
+//
+// Why: hipBLASLt's default heuristic picks 20-45%-off algorithms for the
+// trainer's in-situ shapes (measured, profiles/PROFILES.md round 2: nt at
+// M~7000 runs 0.88-1.27 PF vs 1.6 PF for the best shape; committed
+// TunableOp tables proved NON-transferable across boxes in round 1).  This
+// is the robust in-process alternative: on the first call for a (layout,
+// M, N, K) key, request a heuristic candidate list, time each candidate on
+// the live stream, and cache the winner for the process lifetime.  No
+// files, no cross-box reuse, bounded cost (~32 algos x 4 runs per shape,
+// once).
+//
+// Reference capability: the update-loop GEMM throughput the reference gets
+// from cuBLAS/flash-attn inside torch (stream_dp_actor.py:153-224).
+//
+// Layout math (row-major tensors on a column-major BLAS):
+//   fwd  (nt): Y[M,N] = X[M,K] @ W[N,K]^T  ->  C(N,M) = op_T(W) op_N(X)
+//   dgrad(nn): dX[M,K] = dY[M,N] @ W[N,K]  ->  C(K,M) = op_N(W) op_N(dY)
+//   wgrad(tn): dW[N,K] = dY[M,N]^T @ X[M,K]->  C(K,N) = op_N(X) op_T(dY)
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hipblaslt/hipblaslt.h>
+
+#include <map>
+#include <mutex>
+#include <tuple>
+#include <vector>
+
+#define HIPBLASLT_CHECK(expr)                                             \
+  do {                                                                    \
+    hipblasStatus_t s_ = (expr);                                          \
+    TORCH_CHECK(s_ == HIPBLAS_STATUS_SUCCESS, "hipblaslt error ", (int)s_, \
+                " at " #expr);                                            \
+  } while (0)
+
+namespace {
+
+constexpr size_t kWorkspaceBytes = 128ull << 20;
+constexpr int kMaxAlgos = 48;
+constexpr int kTimedIters = 3;
+
+struct LtState {
+  hipblasLtHandle_t handle = nullptr;
+  void* workspace = nullptr;
+  std::mutex mu;
+  // key: (mode, M, N, K) -> winning algo
+  std::map<std::tuple<int, int64_t, int64_t, int64_t>, hipblasLtMatmulAlgo_t>
+      cache;
+};
+
+LtState& lt() {
+  static LtState s;
+  static std::once_flag once;
+  std::call_once(once, [] {
+    HIPBLASLT_CHECK(hipblasLtCreate(&s.handle));
+    TORCH_CHECK(hipMalloc(&s.workspace, kWorkspaceBytes) == hipSuccess,
+                "hipblaslt workspace alloc failed");
+  });
+  return s;
+}
+
+struct Plan {
+  hipblasLtMatmulDesc_t op = nullptr;
+  hipblasLtMatrixLayout_t la = nullptr, lb = nullptr, lc = nullptr;
+  ~Plan() {
+    if (op) hipblasLtMatmulDescDestroy(op);
+    if (la) hipblasLtMatrixLayoutDestroy(la);
+    if (lb) hipblasLtMatrixLayoutDestroy(lb);
+    if (lc) hipblasLtMatrixLayoutDestroy(lc);
+  }
+};
+
+// column-major problem: C(m,n) = opA(A) * opB(B)
+void build_plan(Plan& p, hipblasOperation_t opA, hipblasOperation_t opB,
+                int64_t m, int64_t n, int64_t k, int64_t lda, int64_t ldb,
+                int64_t ldc) {
+  HIPBLASLT_CHECK(hipblasLtMatmulDescCreate(&p.op, HIPBLAS_COMPUTE_32F,
+                                            HIP_R_32F));
+  HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+      p.op, HIPBLASLT_MATMUL_DESC_TRANSA, &opA, sizeof(opA)));
+  HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+      p.op, HIPBLASLT_MATMUL_DESC_TRANSB, &opB, sizeof(opB)));
+  const int64_t ra = (opA == HIPBLAS_OP_N) ? m : k;
+  const int64_t ca = (opA == HIPBLAS_OP_N) ? k : m;
+  const int64_t rb = (opB == HIPBLAS_OP_N) ? k : n;
+  const int64_t cb = (opB == HIPBLAS_OP_N) ? n : k;
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&p.la, HIP_R_16BF, ra, ca, lda));
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&p.lb, HIP_R_16BF, rb, cb, ldb));
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&p.lc, HIP_R_16BF, m, n, ldc));
+}
+
+// mode: 0 = fwd nt, 1 = dgrad nn, 2 = wgrad tn
+void tuned_mm(int mode, const torch::Tensor& A_row, const torch::Tensor& B_row,
+              torch::Tensor& C_row, int64_t M, int64_t N, int64_t K) {
+  auto& s = lt();
+  auto stream = at::hip::getCurrentHIPStream();
+
+  hipblasOperation_t opA, opB;
+  int64_t m, n, k, lda, ldb, ldc;
+  const void *Aptr, *Bptr;
+  void* Cptr = C_row.data_ptr();
+  if (mode == 0) {           // C(N,M) = W^T(cm K,N) X(cm K,M)
+    opA = HIPBLAS_OP_T; opB = HIPBLAS_OP_N;
+    m = N; n = M; k = K; lda = K; ldb = K; ldc = N;
+    Aptr = B_row.data_ptr();  // W
+    Bptr = A_row.data_ptr();  // X
+  } else if (mode == 1) {    // C(K,M) = W(cm K,N) dY(cm N,M)
+    opA = HIPBLAS_OP_N; opB = HIPBLAS_OP_N;
+    m = K; n = M; k = N; lda = K; ldb = N; ldc = K;
+    Aptr = B_row.data_ptr();  // W
+    Bptr = A_row.data_ptr();  // dY
+  } else {                   // C(K,N) = X(cm K,M) dY^T(cm N,M)
+    opA = HIPBLAS_OP_N; opB = HIPBLAS_OP_T;
+    m = K; n = N; k = M; lda = K; ldb = N; ldc = K;
+    Aptr = B_row.data_ptr();  // X
+    Bptr = A_row.data_ptr();  // dY
+  }
+
+  Plan p;
+  build_plan(p, opA, opB, m, n, k, lda, ldb, ldc);
+  const float alpha = 1.f, beta = 0.f;
+
+  hipblasLtMatmulAlgo_t algo;
+  bool have_algo = false;
+  {
+    std::lock_guard<std::mutex> g(s.mu);
+    auto it = s.cache.find({mode, M, N, K});
+    if (it != s.cache.end()) {
+      algo = it->second;
+      have_algo = true;
+    }
+  }
+
+  if (!have_algo) {
+    hipblasLtMatmulPreference_t pref;
+    HIPBLASLT_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
+    size_t ws = kWorkspaceBytes;
+    HIPBLASLT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
+        pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &ws, sizeof(ws)));
+    std::vector<hipblasLtMatmulHeuristicResult_t> results(kMaxAlgos);
+    int found = 0;
+    HIPBLASLT_CHECK(hipblasLtMatmulAlgoGetHeuristic(
+        s.handle, p.op, p.la, p.lb, p.lc, p.lc, pref, kMaxAlgos,
+        results.data(), &found));
+    hipblasLtMatmulPreferenceDestroy(pref);
+    TORCH_CHECK(found > 0, "hipblaslt: no algo for shape mode=", mode,
+                " M=", M, " N=", N, " K=", K);
+
+    hipEvent_t ev0, ev1;
+    hipEventCreate(&ev0);
+    hipEventCreate(&ev1);
+    float best_ms = 1e30f;
+    int best = 0;
+    for (int i = 0; i < found; ++i) {
+      // warmup (also validates the algo actually runs)
+      hipblasStatus_t st = hipblasLtMatmul(
+          s.handle, p.op, &alpha, Aptr, p.la, Bptr, p.lb, &beta, Cptr, p.lc,
+          Cptr, p.lc, &results[i].algo, s.workspace, kWorkspaceBytes, stream);
+      if (st != HIPBLAS_STATUS_SUCCESS) continue;
+      hipEventRecord(ev0, stream);
+      for (int it2 = 0; it2 < kTimedIters; ++it2)
+        hipblasLtMatmul(s.handle, p.op, &alpha, Aptr, p.la, Bptr, p.lb,
+                        &beta, Cptr, p.lc, Cptr, p.lc, &results[i].algo,
+                        s.workspace, kWorkspaceBytes, stream);
+      hipEventRecord(ev1, stream);
+      hipEventSynchronize(ev1);
+      float ms = 1e30f;
+      hipEventElapsedTime(&ms, ev0, ev1);
+      if (ms < best_ms) {
+        best_ms = ms;
+        best = i;
+      }
+    }
+    hipEventDestroy(ev0);
+    hipEventDestroy(ev1);
+    TORCH_CHECK(best_ms < 1e29f, "hipblaslt: every candidate failed, mode=",
+                mode, " M=", M, " N=", N, " K=", K);
+    algo = results[best].algo;
+    {
+      std::lock_guard<std::mutex> g(s.mu);
+      s.cache[{mode, M, N, K}] = algo;
+    }
+  }
+
+  HIPBLASLT_CHECK(hipblasLtMatmul(
+      s.handle, p.op, &alpha, Aptr, p.la, Bptr, p.lb, &beta, Cptr, p.lc,
+      Cptr, p.lc, &algo, s.workspace, kWorkspaceBytes, stream));
+}
+
+void check2d(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.dim() == 2 && t.is_contiguous() &&
+                  t.dtype() == torch::kBFloat16,
+              name, " must be contiguous 2-D bf16 on GPU");
+}
+
+}  // namespace
+
+// Y[M,N] = X[M,K] @ W[N,K]^T
+torch::Tensor tuned_linear_fwd(torch::Tensor x, torch::Tensor w) {
+  check2d(x, "x");
+  check2d(w, "w");
+  const int64_t M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K, "shape mismatch");
+  auto y = torch::empty({M, N}, x.options());
+  tuned_mm(0, x, w, y, M, N, K);
+  return y;
+}
+
+// dX[M,K] = dY[M,N] @ W[N,K]
+torch::Tensor tuned_linear_dgrad(torch::Tensor dy, torch::Tensor w) {
+  check2d(dy, "dy");
+  check2d(w, "w");
+  const int64_t M = dy.size(0), N = dy.size(1), K = w.size(1);
+  TORCH_CHECK(w.size(0) == N, "shape mismatch");
+  auto dx = torch::empty({M, K}, dy.options());
+  tuned_mm(1, dy, w, dx, M, N, K);
+  return dx;
+}
+
+// dW[N,K] = dY[M,N]^T @ X[M,K]
+torch::Tensor tuned_linear_wgrad(torch::Tensor dy, torch::Tensor x) {
+  check2d(dy, "dy");
+  check2d(x, "x");
+  const int64_t M = dy.size(0), N = dy.size(1), K = x.size(1);
+  TORCH_CHECK(x.size(0) == M, "shape mismatch");
+  auto dw = torch::empty({N, K}, dy.options());
+  tuned_mm(2, dy, x, dw, M, N, K);
+  return dw;
+}

@@ -25,6 +25,7 @@ sources = [
     os.path.join(CSRC, f)
     for f in (
         "bindings.cpp",
+        "gemm_tuned.cpp",
         "elementwise.hip",
         "logprobs.hip",
         "sampling.hip",
@@ -42,6 +43,7 @@ setup(
         CUDAExtension(
             name="polyrl_amd._hip",
             sources=sources,
+            libraries=["hipblaslt"],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],

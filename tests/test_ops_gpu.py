@@ -668,3 +668,34 @@ def test_prefill_d64_forward():
     expect = ref.varlen_prefill_attention(q.cpu(), k.cpu(), v.cpu(),
                                           cu.cpu(), cu.cpu(), scale)
     assert rel_err(out.cpu(), expect) < 2e-2
+
+
+# -------------------------------------------- algo-pinned hipBLASLt linears
+
+
+def test_tuned_linear_numerics():
+    """tuned_linear fwd/dgrad/wgrad vs fp32 reference (gemm_tuned.cpp does
+    an in-process algo search on first use per shape)."""
+    torch.manual_seed(50)
+    M, N, K = 1000, 512, 256     # deliberately odd M
+    x = (torch.randn(M, K, device=DEV) / 4).bfloat16().requires_grad_()
+    w = (torch.randn(N, K, device=DEV) / 4).bfloat16().requires_grad_()
+    b = torch.randn(N, device=DEV).bfloat16().requires_grad_()
+    y = ops.tuned_linear(x, w, b)
+    g = torch.randn(M, N, device=DEV).bfloat16()
+    y.backward(g)
+
+    x2 = x.detach().float().requires_grad_()
+    w2 = w.detach().float().requires_grad_()
+    b2 = b.detach().float().requires_grad_()
+    y2 = torch.nn.functional.linear(x2, w2, b2)
+    y2.backward(g.float())
+    assert rel_err(y.float(), y2.detach()) < 1e-2
+    assert rel_err(x.grad.float(), x2.grad) < 1e-2
+    assert rel_err(w.grad.float(), w2.grad) < 1e-2
+    assert rel_err(b.grad.float(), b2.grad) < 1e-2
+    # 3-D input path (padded layout)
+    x3 = (torch.randn(4, 32, K, device=DEV) / 4).bfloat16()
+    y3 = ops.tuned_linear(x3, w.detach())
+    y3r = torch.nn.functional.linear(x3.float(), w.detach().float())
+    assert rel_err(y3.float(), y3r) < 1e-2
