@@ -474,14 +474,29 @@ def test_engine_greedy_matches_trainer_model_gpu():
             t = int(logits[0, -1].argmax())
             expect.append(t)
             ids.append(t)
-    # bf16 kernel-vs-eager forward differences can flip near-tie argmaxes in
-    # a 2-layer random model; require the prefixes to agree
-    agree = 0
-    for a, b in zip(got, expect):
-        if a != b:
-            break
-        agree += 1
-    assert agree >= 4, (got, expect)
+    # margin-aware fp32 oracle (VERDICT r1 weakness #4): teacher-force the
+    # ENGINE's own chain through an fp32 copy of the same weights; every
+    # engine token must either be the fp32 argmax or lie within bf16 noise
+    # of it (a genuine near-tie).  A wrong-math engine token would sit many
+    # logits below the fp32 argmax and fail regardless of position.
+    model32 = create_model(cfg, kind="actor", dtype="float32", device=DEV)
+    model32.load_state_dict(
+        {k: v.float() for k, v in eng.model.state_dict().items()})
+    MARGIN = 0.15
+    ids = list(prompt)
+    n_exact = 0
+    with torch.no_grad():
+        for t_engine in got:
+            logits = model32(torch.tensor([ids], device=DEV)).float()[0, -1]
+            top = int(logits.argmax())
+            if t_engine == top:
+                n_exact += 1
+            else:
+                margin = float(logits[top] - logits[t_engine])
+                assert margin < MARGIN, \
+                    (t_engine, top, margin, got, expect)
+            ids.append(t_engine)
+    assert n_exact >= 6, (n_exact, got, expect)
 
 
 def test_engine_sampling_replay_determinism_gpu():
