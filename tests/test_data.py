@@ -54,3 +54,31 @@ def test_parquet_text_tokenization_and_overlong_filter(tmp_path):
     b = ds.batch([0, 1])
     assert list(b["ground_truth"]) == ["1", "3"]
     assert b["input_ids"].shape[1] == 16
+
+
+def test_openr1_preprocessor_roundtrip(tmp_path):
+    """OpenR1 preprocessor (reference examples/data_preprocess/openr1.py
+    parity): synthetic rows -> train/test parquet -> dataset -> math scorer
+    accepts the ground truth."""
+    import subprocess
+    import sys
+
+    import os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "examples/data_preprocess/openr1.py",
+         "--synthetic", "20", "--out-dir", str(tmp_path)],
+        capture_output=True, text=True, timeout=120, cwd=repo)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert (tmp_path / "train.parquet").exists()
+    assert (tmp_path / "test.parquet").exists()
+    from polyrl_amd.data import ParquetRLHFDataset
+    from polyrl_amd.reward_score import default_compute_score
+    ds = ParquetRLHFDataset([str(tmp_path / "train.parquet")],
+                            max_prompt_length=96,
+                            tokenizer=lambda s: [ord(c) % 512 for c in s][:96])
+    b = ds.batch([0, 1])
+    assert b.non_tensors["data_source"][0] == "open-r1/OpenR1-Math-220k"
+    gt = str(b.non_tensors["ground_truth"][0])
+    assert default_compute_score("open-r1/OpenR1-Math-220k",
+                                 f"\\boxed{{{gt}}}", gt) == 1.0
