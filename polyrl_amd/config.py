@@ -111,6 +111,20 @@ class SamplingConfig:
 
 
 @dataclass
+class MultiTurnConfig:
+    """Reference parity: verl_stream/workers/config/rollout.py:44-58."""
+    enable: bool = False
+    max_assistant_turns: int = 2
+    max_user_turns: int = 8
+    max_tool_response_length: int = 256
+    # python file + fn implementing the interaction:
+    #   fn(prompt_ids: list[int], response_ids: list[int]) ->
+    #       (user_ids: list[int] | None, done: bool)
+    interaction_path: Optional[str] = None
+    interaction_name: str = "generate_turn"
+
+
+@dataclass
 class RolloutConfig:
     name: str = "native"                 # in-process MI355X decoder
     prompt_length: int = 512
@@ -129,9 +143,11 @@ class RolloutConfig:
     # training: weights change every step and the cache is flushed on each
     # install; turn on for serving / multi-turn / shared system prompts.
     enable_radix_cache: bool = False
-    # multi-turn rollouts: declared for config parity (the reference's
-    # RolloutConfig carries it; the stream path doesn't exercise it)
-    multi_turn: bool = False
+    # multi-turn rollouts (reference: MultiTurnConfig, config/rollout.py:44
+    # — an interaction/environment generates the next user turn between
+    # assistant turns; only assistant tokens carry loss)
+    multi_turn: "MultiTurnConfig" = field(
+        default_factory=lambda: MultiTurnConfig())
     # disaggregated split (BASELINE config #4): the LAST num_rollout_ranks
     # ranks of the world serve rollout; 0 = co-located
     num_rollout_ranks: int = 0

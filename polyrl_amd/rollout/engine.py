@@ -61,6 +61,9 @@ class RequestOutput:
     output_ids: List[int]
     output_logprobs: List[float]
     finish_reason: str
+    # multi-turn: 1 = assistant token (carries loss), 0 = user/tool token;
+    # None = single-turn (all tokens are assistant)
+    loss_mask: Optional[List[int]] = None
 
 
 class LayerWeights:

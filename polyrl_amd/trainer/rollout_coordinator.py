@@ -41,18 +41,25 @@ def postprocess_groups(
     B = G * n
     resp = torch.full((B, response_length), pad_token_id, dtype=torch.long)
     resp_mask = torch.zeros(B, response_length, dtype=torch.long)
+    presence = torch.zeros(B, response_length, dtype=torch.long)
     rollout_lp = torch.zeros(B, response_length, dtype=torch.float32)
     for g in range(G):
         for s, out in enumerate(outputs[g]):
             i = g * n + s
             L = min(len(out.output_ids), response_length)
             resp[i, :L] = torch.tensor(out.output_ids[:L], dtype=torch.long)
-            resp_mask[i, :L] = 1
+            presence[i, :L] = 1
+            if out.loss_mask is not None:
+                # multi-turn: user/tool tokens attend but carry no loss
+                resp_mask[i, :L] = torch.tensor(out.loss_mask[:L],
+                                                dtype=torch.long)
+            else:
+                resp_mask[i, :L] = 1
             rollout_lp[i, :L] = torch.tensor(out.output_logprobs[:L])
     prompts = prompt_ids.repeat_interleave(n, dim=0)
     pmask = prompt_mask.repeat_interleave(n, dim=0)
     input_ids = torch.cat([prompts, resp], dim=1)
-    attention_mask = torch.cat([pmask, resp_mask], dim=1)
+    attention_mask = torch.cat([pmask, presence], dim=1)
     position_ids = torch.clamp(torch.cumsum(attention_mask, dim=1) - 1, min=0)
     uids = np.array([group_uids[g] for g in range(G) for _ in range(n)],
                     dtype=object)
@@ -79,6 +86,25 @@ def postprocess_groups(
     ).to(device)
 
 
+def load_interaction(mt_cfg) -> Optional[dict]:
+    """Build the multi-turn kwargs dict from a MultiTurnConfig: loads the
+    interaction fn from the configured python file (same pattern as the
+    custom reward loader; reference: interaction_config_path,
+    config/rollout.py:54)."""
+    if mt_cfg is None or not getattr(mt_cfg, "enable", False):
+        return None
+    import importlib.util
+    spec = importlib.util.spec_from_file_location(
+        "polyrl_interaction", mt_cfg.interaction_path)
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    fn = getattr(mod, mt_cfg.interaction_name)
+    return {"interaction": fn,
+            "max_assistant_turns": mt_cfg.max_assistant_turns,
+            "max_user_turns": mt_cfg.max_user_turns,
+            "max_tool_response_length": mt_cfg.max_tool_response_length}
+
+
 class LocalRolloutCoordinator:
     """One in-process engine per trainer rank (co-located mode).
 
@@ -90,12 +116,16 @@ class LocalRolloutCoordinator:
 
     def __init__(self, engine: Engine, response_length: int,
                  pad_token_id: int = 0, device="cpu",
-                 shard=None):
+                 shard=None, multi_turn=None):
         self.engine = engine
         self.response_length = response_length
         self.pad_token_id = pad_token_id
         self.device = device
         self.shard = shard
+        # multi-turn (reference MultiTurnConfig capability): dict with
+        # interaction fn + turn caps; see load_interaction()
+        self.multi_turn = multi_turn
+        self._mt_state: Dict[str, dict] = {}
         self._groups: Dict[str, dict] = {}
         self._submit_counter = 0
 
@@ -118,6 +148,8 @@ class LocalRolloutCoordinator:
                            if k != "uid"},
                 "prompt_ids": ids[g].cpu(),
                 "prompt_mask": mask[g].cpu(),
+                "raw": raw,
+                "sampling": sampling,
                 "outputs": [None] * n,
                 "done": 0,
                 "n": n,
@@ -143,11 +175,62 @@ class LocalRolloutCoordinator:
                 grp = self._groups.get(gid)
                 if grp is None:
                     continue
+                if self.multi_turn is not None:
+                    out = self._multi_turn_step(grp, out)
+                    if out is None:       # resubmitted for the next turn
+                        continue
                 grp["outputs"][int(s)] = out
                 grp["done"] += 1
                 if grp["done"] == grp["n"]:
                     done.append(self._groups.pop(gid))
         return done
+
+    def _multi_turn_step(self, grp: dict, out: RequestOutput
+                         ) -> Optional[RequestOutput]:
+        """Absorb one finished assistant turn; either resubmit the request
+        with the interaction's next user turn appended (returns None) or
+        finalize the sample with a loss mask over assistant tokens only
+        (reference capability: MultiTurnConfig + interaction, SURVEY.md
+        §2.1 RolloutConfig row)."""
+        mt = self.multi_turn
+        st = self._mt_state.get(out.rid)
+        if st is None:
+            st = {"a_turns": 0, "u_turns": 0, "ids": [], "lp": [], "loss": []}
+            self._mt_state[out.rid] = st
+        st["ids"] += out.output_ids
+        st["lp"] += out.output_logprobs
+        st["loss"] += [1] * len(out.output_ids)
+        st["a_turns"] += 1
+
+        def finalize(reason: str) -> RequestOutput:
+            self._mt_state.pop(out.rid, None)
+            L = self.response_length
+            return RequestOutput(rid=out.rid, output_ids=st["ids"][:L],
+                                 output_logprobs=st["lp"][:L],
+                                 finish_reason=reason,
+                                 loss_mask=st["loss"][:L])
+
+        budget = self.response_length - len(st["ids"])
+        if (out.finish_reason == "abort"
+                or st["a_turns"] >= mt["max_assistant_turns"]
+                or st["u_turns"] >= mt["max_user_turns"]
+                or budget <= 1):
+            return finalize(out.finish_reason)
+        user_ids, done = mt["interaction"](grp["raw"], list(st["ids"]))
+        if done or not user_ids:
+            return finalize("stop")
+        user_ids = list(user_ids)[: mt.get("max_tool_response_length", 256)]
+        if len(user_ids) >= budget:       # no room left to answer
+            return finalize("length")
+        st["ids"] += user_ids
+        st["lp"] += [0.0] * len(user_ids)
+        st["loss"] += [0] * len(user_ids)
+        st["u_turns"] += 1
+        import copy
+        samp = copy.copy(grp["sampling"])
+        samp.max_new_tokens = self.response_length - len(st["ids"])
+        self.engine.add_request(out.rid, grp["raw"] + st["ids"], samp)
+        return None
 
     def stream_batches(self, stream_size: int) -> Iterator[TensorBatch]:
         """Yield TensorBatches of EXACTLY stream_size samples (whole groups;

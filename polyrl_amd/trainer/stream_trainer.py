@@ -210,9 +210,11 @@ class StreamPPOTrainer:
                 shard = None
                 if self.tp_rollout:
                     shard = (self.rank, self.world)
+                from .rollout_coordinator import load_interaction
                 self.coordinator = LocalRolloutCoordinator(
                     self.engine, ro.response_length, pad_token_id=0,
-                    device="cpu", shard=shard)
+                    device="cpu", shard=shard,
+                    multi_turn=load_interaction(ro.multi_turn))
                 self.publisher = WeightPublisher(
                     self.actor.model, [self.engine.model],
                     tie_word_embeddings=model_cfg.tie_word_embeddings)
@@ -355,9 +357,9 @@ class StreamPPOTrainer:
                 "ppo_epochs != 1 is fenced (matches the reference: "
                 "stream_dp_actor.py:145-146 — streamed minibatches are "
                 "consumed once)")
-        if ro.multi_turn:
-            raise NotImplementedError("multi-turn rollouts are declared for "
-                                      "config parity but not implemented")
+        if ro.multi_turn.enable and ro.multi_turn.interaction_path is None:
+            raise ValueError("multi_turn.enable needs an interaction_path "
+                             "(python file with the turn-generator fn)")
 
     # ------------------------------------------------------------------- fit
     def fit(self, max_steps: Optional[int] = None):

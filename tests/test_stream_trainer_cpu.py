@@ -205,15 +205,16 @@ def test_esi_expiry_forces_checkpoint(tmp_path, monkeypatch):
 
 
 def test_reference_fences(tmp_path):
-    """Config fences the reference also has: ppo_epochs != 1 and multi-turn
-    raise NotImplementedError at construction."""
+    """Config fences the reference also has: ppo_epochs != 1 raises at
+    construction; multi-turn without an interaction file is a config
+    error."""
     cfg = tiny_config(tmp_path)
     cfg.actor_rollout_ref.actor.ppo_epochs = 2
     with pytest.raises(NotImplementedError):
         StreamPPOTrainer(cfg, reward_fn=load_reward_manager("random"))
     cfg2 = tiny_config(tmp_path)
-    cfg2.actor_rollout_ref.rollout.multi_turn = True
-    with pytest.raises(NotImplementedError):
+    cfg2.actor_rollout_ref.rollout.multi_turn.enable = True
+    with pytest.raises(ValueError):
         StreamPPOTrainer(cfg2, reward_fn=load_reward_manager("random"))
 
 
@@ -316,3 +317,51 @@ def test_rollout_data_dump(tmp_path):
     assert len(rows) == 16                # 8 prompts x n=2
     assert all("uid" in r and "score" in r and r["response_ids"]
                for r in rows)
+
+
+def test_multi_turn_rollout(tmp_path):
+    """Multi-turn rollouts (reference MultiTurnConfig capability): an
+    interaction file generates the next user turn; user tokens attend but
+    carry no loss; training runs end-to-end."""
+    inter = tmp_path / "interaction.py"
+    inter.write_text(
+        "def generate_turn(prompt_ids, response_ids):\n"
+        "    # one user follow-up of 3 tokens, then done\n"
+        "    if sum(1 for _ in response_ids) >= 10:\n"
+        "        return None, True\n"
+        "    return [7, 8, 9], False\n")
+    cfg = tiny_config(tmp_path)
+    cfg.actor_rollout_ref.rollout.response_length = 16   # room for 2 turns
+    mt = cfg.actor_rollout_ref.rollout.multi_turn
+    mt.enable = True
+    mt.interaction_path = str(inter)
+    mt.max_assistant_turns = 2
+    trainer = StreamPPOTrainer(cfg, reward_fn=load_reward_manager("random"))
+    assert trainer.coordinator.multi_turn is not None
+    trainer.fit(max_steps=1)
+
+    # unit-level: the loss mask marks assistant tokens only
+    import torch
+
+    from polyrl_amd.protocol import TensorBatch
+    from polyrl_amd.rollout.engine import SamplingParams
+    coord = trainer.coordinator
+    prompts = TensorBatch(
+        tensors={"input_ids": torch.tensor([[11, 12, 13]]),
+                 "attention_mask": torch.ones(1, 3, dtype=torch.long)},
+        non_tensors={"uid": __import__("numpy").array(["u0"], dtype=object)})
+    coord.submit(prompts, SamplingParams(temperature=1.0, max_new_tokens=5),
+                 n=1)
+    batches = list(coord.stream_batches(1))
+    assert len(batches) == 1
+    b = batches[0]
+    resp_mask = b["response_mask"][0]
+    attn = b["attention_mask"][0][3:]      # response presence
+    # assistant turn 1 (5 toks) + user (3 toks, masked) + assistant turn 2
+    n_present = int(attn.sum())
+    n_loss = int(resp_mask.sum())
+    assert n_present > n_loss >= 5, (n_present, n_loss)
+    # the user tokens [7,8,9] sit in the response with loss 0
+    resp = b["responses"][0][:n_present].tolist()
+    assert [7, 8, 9] == resp[5:8]
+    assert resp_mask[5:8].sum() == 0
