@@ -49,10 +49,10 @@ def postprocess_groups(
             L = min(len(out.output_ids), response_length)
             resp[i, :L] = torch.tensor(out.output_ids[:L], dtype=torch.long)
             presence[i, :L] = 1
-            if out.loss_mask is not None:
+            lm = getattr(out, "loss_mask", None)
+            if lm is not None:
                 # multi-turn: user/tool tokens attend but carry no loss
-                resp_mask[i, :L] = torch.tensor(out.loss_mask[:L],
-                                                dtype=torch.long)
+                resp_mask[i, :L] = torch.tensor(lm[:L], dtype=torch.long)
             else:
                 resp_mask[i, :L] = 1
             rollout_lp[i, :L] = torch.tensor(out.output_logprobs[:L])
