@@ -522,7 +522,14 @@ class StreamPPOTrainer:
         ibatches: List[TensorBatch] = []
         cum = 0                       # cumulative local samples trained
         warmup = c.trainer.critic_warmup
-        for ibatch in self.coordinator.stream_batches(local_stream):
+        stream_it = iter(self.coordinator.stream_batches(local_stream))
+        while True:
+            # engine decode runs inside the generator pull: time it so the
+            # step breakdown attributes rollout-vs-update wall explicitly
+            with marked_timer("gen_wait", timing):
+                ibatch = next(stream_it, None)
+            if ibatch is None:
+                break
             with marked_timer("prep", timing):
                 ibatch = self._prepare_ibatch(ibatch, timing)
             ibatches.append(ibatch)
@@ -586,7 +593,8 @@ class StreamPPOTrainer:
             step_t = timing.get("step", 0.0)
             # "prep" already contains the reward/adv/logprob sub-timers
             busy = sum(timing.get(k, 0.0) for k in
-                       ("update", "prep", "weight_sync", "gen_baseline"))
+                       ("update", "prep", "weight_sync", "gen_baseline",
+                        "gen_wait"))
             bubble = max(step_t - busy, 0.0)
             thr = len(full) / step_t if step_t > 0 else 0.0
             new_window = self.coordinator.update_metrics(step_t, bubble, thr)
