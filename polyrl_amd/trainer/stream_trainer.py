@@ -593,8 +593,7 @@ class StreamPPOTrainer:
             step_t = timing.get("step", 0.0)
             # "prep" already contains the reward/adv/logprob sub-timers
             busy = sum(timing.get(k, 0.0) for k in
-                       ("update", "prep", "weight_sync", "gen_baseline",
-                        "gen_wait"))
+                       ("update", "prep", "weight_sync", "gen_baseline"))
             bubble = max(step_t - busy, 0.0)
             thr = len(full) / step_t if step_t > 0 else 0.0
             new_window = self.coordinator.update_metrics(step_t, bubble, thr)
