@@ -21,6 +21,15 @@ import torch.nn.functional as F
 from .registry import DecoderConfig
 
 
+def _norm(mod, x: torch.Tensor) -> torch.Tensor:
+    """RMSNorm through the fused trainer kernel on GPU bf16; eager
+    reference elsewhere (CPU tier / fp32)."""
+    import polyrl_amd.ops as pops
+    if x.is_cuda and x.dtype == torch.bfloat16             and mod.weight.dtype == torch.bfloat16             and pops._fused_norm_enabled():
+        return pops.rmsnorm_train(x, mod.weight, mod.variance_epsilon)
+    return mod(x)
+
+
 def _lin(mod: nn.Module, x: torch.Tensor) -> torch.Tensor:
     """Route trainer linears through the algo-pinned hipBLASLt path
     (ops.tuned_linear); identical math to mod(x).  Wrapped modules (LoRA)
@@ -222,11 +231,22 @@ class DecoderLayer(nn.Module):
     def forward(self, x, cos, sin, attn_bias_mask, sp_group=None,
                 cu_seqlens=None):
         if cu_seqlens is not None:
-            x = x + self.self_attn.forward_packed(
-                self.input_layernorm(x), cos, sin, cu_seqlens, sp_group)
-        else:
-            x = x + self.self_attn(self.input_layernorm(x), cos, sin,
-                                   attn_bias_mask, sp_group)
+            attn_out = self.self_attn.forward_packed(
+                _norm(self.input_layernorm, x), cos, sin, cu_seqlens,
+                sp_group)
+            import polyrl_amd.ops as pops
+            if attn_out.is_cuda and attn_out.dtype == torch.bfloat16                     and pops._fused_norm_enabled():
+                # fused residual-add + norm: h = x + attn_out computed
+                # inside the norm kernel (one pass; rmsnorm_train.hip)
+                m_in, h = pops.fused_add_rmsnorm_train(
+                    attn_out, x, self.post_attention_layernorm.weight,
+                    self.post_attention_layernorm.variance_epsilon)
+            else:
+                h = x + attn_out
+                m_in = self.post_attention_layernorm(h)
+            return h + self.mlp(m_in)
+        x = x + self.self_attn(self.input_layernorm(x), cos, sin,
+                               attn_bias_mask, sp_group)
         x = x + self.mlp(self.post_attention_layernorm(x))
         return x
 
@@ -294,7 +314,7 @@ class DecoderModel(nn.Module):
                     use_reentrant=False)
             else:
                 x = layer(x, cos, sin, None, grp, cu_pack)
-        x = self.norm(x)
+        x = _norm(self.norm, x)
         self.pack_sp_meta = {"valid": valid, "T": T, "Tp": Tp,
                              "shard": shard}
         return x                                    # (Tp/sp, H)
@@ -348,7 +368,7 @@ class DecoderModel(nn.Module):
                     use_reentrant=False)
             else:
                 x = layer(x, cos, sin, None, None, cu_pack)
-        x = self.norm(x)
+        x = _norm(self.norm, x)
         out = torch.zeros(B, L, x.shape[-1], dtype=x.dtype, device=x.device)
         out[valid] = x[:T]
         return out

@@ -282,6 +282,48 @@ class _FlashAttnVarlen(torch.autograd.Function):
                 None, None, None, None, None, None, None)
 
 
+class _RMSNormTrain(torch.autograd.Function):
+    """Trainer RMSNorm with optional fused residual add — hand-written
+    CDNA4 fwd+bwd (rmsnorm_train.hip).  Returns (y, h) where h = x + res
+    (the next residual); grads flow through both outputs."""
+
+    @staticmethod
+    def forward(ctx, x, res, w, eps):
+        ext = _require_ext()
+        res_t = res if res is not None else torch.Tensor()
+        y, h, rstd = ext.rmsnorm_train_fwd(x.contiguous(), res_t, w, eps)
+        ctx.save_for_backward(h, w, rstd)
+        ctx.has_res = res is not None
+        return y, h
+
+    @staticmethod
+    def backward(ctx, dy, dh_out):
+        h, w, rstd = ctx.saved_tensors
+        ext = _require_ext()
+        dx, dw = ext.rmsnorm_train_bwd(dy, h, w, rstd)
+        if dh_out is not None:
+            dx = dx + dh_out
+        return (dx, dx if ctx.has_res else None, dw.to(w.dtype), None)
+
+
+def _fused_norm_enabled() -> bool:
+    import os
+    return os.environ.get("POLYRL_FUSED_NORM", "1") == "1"
+
+
+def rmsnorm_train(x: torch.Tensor, w: torch.Tensor, eps: float
+                  ) -> torch.Tensor:
+    """Differentiable RMSNorm (trainer path)."""
+    y, _ = _RMSNormTrain.apply(x, None, w, eps)
+    return y
+
+
+def fused_add_rmsnorm_train(x: torch.Tensor, res: torch.Tensor,
+                            w: torch.Tensor, eps: float):
+    """(norm(x+res), x+res) — one kernel for the residual add + norm."""
+    return _RMSNormTrain.apply(x, res, w, eps)
+
+
 class _TunedLinear(torch.autograd.Function):
     """nn.Linear matmuls through the per-shape hipBLASLt algo search
     (gemm_tuned.cpp): fwd nt, dgrad nn, wgrad tn — the trainer's three hot
@@ -306,6 +348,15 @@ class _TunedLinear(torch.autograd.Function):
 def _tuned_gemm_enabled() -> bool:
     import os
     return os.environ.get("POLYRL_TUNED_GEMM", "1") == "1"
+
+
+def tuned_mm_nt(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """Inference-path y = x @ w.T with per-shape algo pinning (no autograd).
+    Used by the engine's decode projections: skinny-M GEMMs measured up to
+    7x off the weight-read bound on the heuristic pick
+    (profiles/PROFILES.md r02 kernel stats)."""
+    ext = _require_ext()
+    return ext.tuned_linear_fwd(x, w)
 
 
 def tuned_linear(x: torch.Tensor, w: torch.Tensor,

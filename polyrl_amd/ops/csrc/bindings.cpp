@@ -1,5 +1,6 @@
 // pybind bindings for the polyrl_amd CDNA4 kernel suite.
 #include <torch/extension.h>
+#include <vector>
 
 void rmsnorm(torch::Tensor out, torch::Tensor x, torch::Tensor weight,
              double eps);
@@ -35,6 +36,13 @@ void varlen_attention_backward(
     torch::Tensor cu_seqlens_q, torch::Tensor cu_seqlens_k,
     torch::Tensor tile_seq, torch::Tensor tile_k0,
     double scale, bool causal, bool use_v2);
+std::vector<torch::Tensor> rmsnorm_train_fwd(torch::Tensor x,
+                                             torch::Tensor res,
+                                             torch::Tensor w, double eps);
+std::vector<torch::Tensor> rmsnorm_train_bwd(torch::Tensor dy,
+                                             torch::Tensor h,
+                                             torch::Tensor w,
+                                             torch::Tensor rstd);
 torch::Tensor tuned_linear_fwd(torch::Tensor x, torch::Tensor w);
 torch::Tensor tuned_linear_dgrad(torch::Tensor dy, torch::Tensor w);
 torch::Tensor tuned_linear_wgrad(torch::Tensor dy, torch::Tensor x);
@@ -63,6 +71,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "varlen causal prefill attention (MFMA, bf16; optional LSE out)");
   m.def("varlen_attention_backward", &varlen_attention_backward,
         "varlen causal flash-attention backward (MFMA, bf16)");
+  m.def("rmsnorm_train_fwd", &rmsnorm_train_fwd,
+        "trainer RMSNorm fwd (+optional fused residual): y, h, rstd");
+  m.def("rmsnorm_train_bwd", &rmsnorm_train_bwd,
+        "trainer RMSNorm bwd: dx, dw(fp32)");
   m.def("tuned_linear_fwd", &tuned_linear_fwd,
         "Y = X W^T via hipBLASLt with per-shape in-process algo search");
   m.def("tuned_linear_dgrad", &tuned_linear_dgrad, "dX = dY W (algo-pinned)");

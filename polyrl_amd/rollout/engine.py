@@ -25,6 +25,23 @@ from ..models.registry import DecoderConfig
 from .kv_cache import PagedKVCache, RadixCache
 
 
+def _mm_nt(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """y = x @ w.T; decode-sized GEMMs (M <= 512) go through the
+    algo-pinned hipBLASLt path (ops.tuned_mm_nt) — the heuristic pick runs
+    skinny decode projections far off the weight-read bound.  Prefill Ms
+    are ragged (chunk sums), so they keep the torch path to bound the
+    number of shape searches."""
+    if x.is_cuda and x.dtype == torch.bfloat16 and x.shape[0] <= 512 \
+            and _tuned_decode_enabled():
+        return ops.tuned_mm_nt(x, w)
+    return x @ w.t()
+
+
+def _tuned_decode_enabled() -> bool:
+    import os
+    return os.environ.get("POLYRL_TUNED_DECODE", "1") == "1"
+
+
 @dataclass
 class SamplingParams:
     temperature: float = 1.0
@@ -266,7 +283,7 @@ class InferenceModel:
             else:
                 hidden, residual = ops.fused_add_rmsnorm(
                     hidden, residual, lw.input_ln, cfg.rms_norm_eps)
-            qkv = hidden @ lw.wqkv.t()
+            qkv = _mm_nt(hidden, lw.wqkv)
             if lw.bqkv is not None:
                 qkv = qkv + lw.bqkv
             # strided head views into the fused qkv row (no copies: the
@@ -278,14 +295,15 @@ class InferenceModel:
             ops.kv_cache_append(kv.k_cache[li], kv.v_cache[li], k, v,
                                 slot_mapping)
             attn_out = attn_fn(li, q, k, v)           # (N, Hq, D)
-            hidden = attn_out.view(-1, Hq * D) @ lw.wo.t()
+            hidden = _mm_nt(attn_out.reshape(-1, Hq * D), lw.wo)
             self.tp.all_reduce_(hidden)               # col-parallel o_proj
             hidden, residual = ops.fused_add_rmsnorm(
                 hidden, residual, lw.post_ln, cfg.rms_norm_eps)
-            gate_up = hidden @ lw.w_gate_up.t()
+            gate_up = _mm_nt(hidden, lw.w_gate_up)
             i_sz = cfg.intermediate_size // tp
-            hidden = ops.silu_mul(gate_up.narrow(1, 0, i_sz),
-                                  gate_up.narrow(1, i_sz, i_sz)) @ lw.w_down.t()
+            hidden = _mm_nt(ops.silu_mul(gate_up.narrow(1, 0, i_sz),
+                                         gate_up.narrow(1, i_sz, i_sz)),
+                            lw.w_down)
             self.tp.all_reduce_(hidden)               # col-parallel down_proj
         # final residual add + norm (fused; residual buffer is dead after)
         normed, _ = ops.fused_add_rmsnorm(hidden, residual, self.final_norm,
@@ -320,7 +338,7 @@ class InferenceModel:
 
     @torch.no_grad()
     def logits(self, hidden: torch.Tensor) -> torch.Tensor:
-        local = hidden @ self.lm_head.t()
+        local = _mm_nt(hidden, self.lm_head)
         return self.tp.all_gather_cat(local, dim=-1)
 
 

@@ -27,6 +27,7 @@ sources = [
         "bindings.cpp",
         "gemm_tuned.cpp",
         "elementwise.hip",
+        "rmsnorm_train.hip",
         "logprobs.hip",
         "sampling.hip",
         "kv_cache.hip",

@@ -714,3 +714,50 @@ def test_tuned_linear_numerics():
     y3 = ops.tuned_linear(x3, w.detach())
     y3r = torch.nn.functional.linear(x3.float(), w.detach().float())
     assert rel_err(y3.float(), y3r) < 1e-2
+
+
+def test_rmsnorm_train_fwd_bwd():
+    """Fused trainer RMSNorm (+residual) fwd/bwd vs fp32 autograd ref."""
+    torch.manual_seed(55)
+    T, H = 777, 512
+    x = (torch.randn(T, H, device=DEV) / 2).bfloat16().requires_grad_()
+    res = (torch.randn(T, H, device=DEV) / 2).bfloat16().requires_grad_()
+    w = torch.randn(H, device=DEV).bfloat16().requires_grad_()
+    eps = 1e-6
+    y, h = ops.fused_add_rmsnorm_train(x, res, w, eps)
+    g = torch.randn(T, H, device=DEV)
+    gh = torch.randn(T, H, device=DEV)
+    (y.float() * g + h.float() * gh).sum().backward()
+
+    x2 = x.detach().float().requires_grad_()
+    r2 = res.detach().float().requires_grad_()
+    w2 = w.detach().float().requires_grad_()
+    h2 = x2 + r2
+    y2 = torch.nn.functional.rms_norm(h2, (H,), None, eps) * w2
+    (y2 * g + h2 * gh).sum().backward()
+    assert rel_err(y.float(), y2.detach()) < 1e-2
+    assert rel_err(h.float(), h2.detach()) < 1e-2
+    assert rel_err(x.grad.float(), x2.grad) < 2e-2
+    assert rel_err(res.grad.float(), r2.grad) < 2e-2
+    assert rel_err(w.grad.float(), w2.grad) < 2e-2
+
+    # no-residual variant
+    xx = (torch.randn(T, H, device=DEV) / 2).bfloat16().requires_grad_()
+    yy = ops.rmsnorm_train(xx, w.detach(), eps)
+    yy.float().mul(g).sum().backward()
+    xr = xx.detach().float().requires_grad_()
+    yr = torch.nn.functional.rms_norm(xr, (H,), None, eps) * w.detach().float()
+    yr.mul(g).sum().backward()
+    assert rel_err(yy.float(), yr.detach()) < 1e-2
+    assert rel_err(xx.grad.float(), xr.grad) < 2e-2
+
+
+def test_tuned_mm_nt_decode_shapes():
+    """Engine decode projection path: algo-pinned nt matmul at skinny M."""
+    torch.manual_seed(56)
+    for M, N, K in ((32, 6144, 4096), (128, 4096, 4096), (64, 512, 256)):
+        x = (torch.randn(M, K, device=DEV) / 4).bfloat16()
+        w = (torch.randn(N, K, device=DEV) / 4).bfloat16()
+        y = ops.tuned_mm_nt(x, w)
+        yr = (x.float() @ w.float().t())
+        assert rel_err(y.float(), yr) < 1e-2, (M, N, K)
