@@ -82,8 +82,10 @@ def main():
     cfg = PPOConfig()
     cfg.actor_rollout_ref.model.path = args.model
     cfg.actor_rollout_ref.model.dtype = "bfloat16" if device == "cuda" else "float32"
-    cfg.actor_rollout_ref.model.enable_gradient_checkpointing = False
-    cfg.critic.model.enable_gradient_checkpointing = False
+    # long-response configs may need activation headroom for KV
+    ckpt = os.environ.get("POLYRL_BENCH_GRAD_CKPT") == "1"
+    cfg.actor_rollout_ref.model.enable_gradient_checkpointing = ckpt
+    cfg.critic.model.enable_gradient_checkpointing = ckpt
     # keep gathered params resident across the step's micro passes (one
     # all-gather per step; +full-param memory, fine at 288 GB)
     cfg.actor_rollout_ref.actor.fsdp.reshard_after_forward = False
