@@ -83,9 +83,9 @@ def gpu_pair():
 def test_timebox_continuation_over_real_http(gpu_pair):
     """Slow local engine is aborted by the time-box; the sample continues
     token-exactly on the fast remote engine over real HTTP."""
-    from polyrl_amd.scheduler.manager import RolloutScheduler
-    from polyrl_amd.scheduler.types import (GroupRequest, SamplingSpec,
-                                            SchedulerConfig)
+    from polyrl_amd.scheduler.manager import (RolloutScheduler,
+                                              SchedulerConfig)
+    from polyrl_amd.scheduler.types import GroupRequest, SamplingSpec
     from polyrl_amd.server import HttpInstance
     cfg, model, engines, ports = gpu_pair
 
