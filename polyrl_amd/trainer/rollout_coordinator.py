@@ -116,7 +116,7 @@ class LocalRolloutCoordinator:
 
     def __init__(self, engine: Engine, response_length: int,
                  pad_token_id: int = 0, device="cpu",
-                 shard=None, multi_turn=None):
+                 shard=None, multi_turn=None, async_decode: bool = True):
         self.engine = engine
         self.response_length = response_length
         self.pad_token_id = pad_token_id
@@ -128,6 +128,56 @@ class LocalRolloutCoordinator:
         self._mt_state: Dict[str, dict] = {}
         self._groups: Dict[str, dict] = {}
         self._submit_counter = 0
+        # --- async decode pump (the in-process analog of the reference's
+        # separate SGLang server process): a side thread steps the engine
+        # on its own HIP stream so decode overlaps the trainer's update
+        # kernels on the default stream.  Semantics are IDENTICAL to
+        # serial decode: the engine holds its own weight buffers at the
+        # step's version, and publish happens only between steps when the
+        # pump is idle (stream_batches drains everything first).
+        import os
+        self.async_decode = (async_decode and not self.shard
+                             and os.environ.get("POLYRL_ASYNC_DECODE", "1")
+                             == "1")
+        self._pump = None
+        self._pump_err = None
+        self._ready_q = None
+        self._side_stream = None
+
+    # ---------------------------------------------------------- decode pump
+    def _pump_loop(self):
+        import torch as _t
+        try:
+            use_stream = str(self.engine.device).startswith("cuda")
+            if use_stream:
+                if self._side_stream is None:
+                    self._side_stream = _t.cuda.Stream()
+                # order: step-start weight publish (default stream) must be
+                # visible before this step's decode kernels
+                self._side_stream.wait_stream(_t.cuda.current_stream())
+                ctx = _t.cuda.stream(self._side_stream)
+            else:
+                import contextlib
+                ctx = contextlib.nullcontext()
+            with ctx:
+                while self.engine.has_work():
+                    for grp in self.poll():
+                        self._ready_q.put(grp)
+            if use_stream:
+                # make emitted work visible to the default stream consumers
+                _t.cuda.current_stream().synchronize()
+        except BaseException as e:                 # noqa: BLE001
+            self._pump_err = e
+        finally:
+            self._ready_q.put(None)               # end-of-stream sentinel
+
+    def _start_pump(self):
+        import queue
+        import threading
+        self._ready_q = queue.Queue()
+        self._pump_err = None
+        self._pump = threading.Thread(target=self._pump_loop, daemon=True)
+        self._pump.start()
 
     # --------------------------------------------------------------- submit
     def submit(self, prompts: TensorBatch, sampling: SamplingParams, n: int):
@@ -243,6 +293,9 @@ class LocalRolloutCoordinator:
         yield rank r's group-slice (deterministic, identical on all ranks —
         no communication)."""
         round_size = stream_size * (self.shard[1] if self.shard else 1)
+        if self.async_decode and self._groups:
+            yield from self._stream_batches_async(round_size)
+            return
         ready: List[dict] = []
         ready_samples = 0
         while self._groups or ready:
@@ -262,6 +315,41 @@ class LocalRolloutCoordinator:
                 yield self._emit(ready)
                 ready = []
                 ready_samples = 0
+
+    def _stream_batches_async(self, round_size: int):
+        """Pump-thread variant: decode runs on a side thread/stream while
+        the caller (trainer) updates between yields."""
+        self._start_pump()
+        ready: List[dict] = []
+        ready_samples = 0
+        done = False
+        try:
+            while not done or ready:
+                if not done:
+                    grp = self._ready_q.get()
+                    if grp is None:
+                        done = True
+                        if self._pump_err is not None:
+                            raise self._pump_err
+                    else:
+                        ready.append(grp)
+                        ready_samples += grp["n"]
+                while ready_samples >= round_size and ready:
+                    take, taken = [], 0
+                    while ready and taken < round_size:
+                        g = ready.pop(0)
+                        take.append(g)
+                        taken += g["n"]
+                    ready_samples -= taken
+                    yield self._emit(take)
+                if done and ready:               # tail
+                    yield self._emit(ready)
+                    ready = []
+                    ready_samples = 0
+        finally:
+            if self._pump is not None:
+                self._pump.join(timeout=600.0)
+                self._pump = None
 
     def _emit(self, groups: List[dict]) -> TensorBatch:
         if not self.shard:

@@ -365,3 +365,25 @@ def test_multi_turn_rollout(tmp_path):
     resp = b["responses"][0][:n_present].tolist()
     assert [7, 8, 9] == resp[5:8]
     assert resp_mask[5:8].sum() == 0
+
+
+def test_async_decode_matches_serial(tmp_path, monkeypatch):
+    """The async decode pump must produce EXACTLY the serial rollouts (the
+    engine's counter-based RNG is deterministic; the pump only moves the
+    same engine.step() calls to a side thread)."""
+    def run(async_on):
+        monkeypatch.setenv("POLYRL_ASYNC_DECODE", "1" if async_on else "0")
+        cfg = tiny_config(tmp_path / ("a" if async_on else "s"))
+        trainer = StreamPPOTrainer(cfg,
+                                   reward_fn=load_reward_manager("random"))
+        trainer.fit(max_steps=1)
+        full = trainer._last_full_batch
+        return (full["responses"].clone(), full["rollout_log_probs"].clone(),
+                snapshot(trainer.actor.model))
+
+    r_async, lp_async, w_async = run(True)
+    r_sync, lp_sync, w_sync = run(False)
+    assert torch.equal(r_async, r_sync)
+    assert torch.equal(lp_async, lp_sync)
+    for k in w_sync:
+        assert torch.allclose(w_async[k], w_sync[k], atol=1e-6), k
