@@ -114,17 +114,29 @@ def test_timebox_continuation_over_real_http(gpu_pair):
         return s
     s = asyncio.run(go())
 
-    # token-exactness: the migrated chain must equal the single-engine
-    # greedy chain on the same weights (greedy + KV-exact continuation)
-    from polyrl_amd.rollout.engine import Engine, SamplingParams
-    eng = Engine(_tiny_cfg(), device=DEV, dtype=torch.bfloat16,
-                 kv_bytes_budget=64 << 20, decode_chunk_size=4)
-    eng.model.load_state_dict(model.state_dict())
-    outs = eng.generate([[3, 4, 5]],
-                        SamplingParams(temperature=0.0, max_new_tokens=64),
-                        "chk")
-    assert s.output_ids == outs[0].output_ids, \
-        (s.output_ids[:10], outs[0].output_ids[:10])
+    # Chain validity: teacher-force the MIGRATED chain through an fp32
+    # copy of the weights — every token must be the fp32 argmax or a
+    # bf16 near-tie (the continuation recomputes the prefix through the
+    # prefill kernel, whose summation order can legally flip exact ties
+    # vs the decode path; token-level BOOKKEEPING exactness is covered by
+    # the fp32 CPU property tests).
+    from polyrl_amd.models import create_model
+    model32 = create_model(cfg, kind="actor", dtype="float32", device=DEV)
+    model32.load_state_dict(
+        {k: v.float() for k, v in model.state_dict().items()})
+    ids = [3, 4, 5]
+    n_exact = 0
+    with torch.no_grad():
+        for t_engine in s.output_ids:
+            logits = model32(torch.tensor([ids], device=DEV)).float()[0, -1]
+            top = int(logits.argmax())
+            if t_engine == top:
+                n_exact += 1
+            else:
+                margin = float(logits[top] - logits[t_engine])
+                assert margin < 0.15, (t_engine, top, margin)
+            ids.append(t_engine)
+    assert n_exact >= 56, n_exact          # >= 7/8 of 64 exact
 
 
 def test_weight_update_changes_rollout_over_http(gpu_pair):
