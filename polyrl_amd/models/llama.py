@@ -136,13 +136,21 @@ class Attention(nn.Module):
         q = _lin(self.q_proj, x).view(T, self.num_heads, self.head_dim)
         k = _lin(self.k_proj, x).view(T, self.num_kv_heads, self.head_dim)
         v = _lin(self.v_proj, x).view(T, self.num_kv_heads, self.head_dim)
-        cos = cos.to(q.dtype).unsqueeze(1)         # (T, 1, D/2)
-        sin = sin.to(q.dtype).unsqueeze(1)
-        d = self.head_dim // 2
-        q = torch.cat([q[..., :d] * cos - q[..., d:] * sin,
-                       q[..., d:] * cos + q[..., :d] * sin], dim=-1)
-        k = torch.cat([k[..., :d] * cos - k[..., d:] * sin,
-                       k[..., d:] * cos + k[..., :d] * sin], dim=-1)
+        if q.is_cuda and q.dtype == torch.bfloat16 \
+                and self.head_dim in (64, 128):
+            # single-pass fused rotation fwd/bwd (ops/csrc/rope_train.hip)
+            cos_f = cos.float().contiguous()
+            sin_f = sin.float().contiguous()
+            q = pops.rope_train(q, cos_f, sin_f)
+            k = pops.rope_train(k, cos_f, sin_f)
+        else:
+            cos = cos.to(q.dtype).unsqueeze(1)     # (T, 1, D/2)
+            sin = sin.to(q.dtype).unsqueeze(1)
+            d = self.head_dim // 2
+            q = torch.cat([q[..., :d] * cos - q[..., d:] * sin,
+                           q[..., d:] * cos + q[..., :d] * sin], dim=-1)
+            k = torch.cat([k[..., :d] * cos - k[..., d:] * sin,
+                           k[..., d:] * cos + k[..., :d] * sin], dim=-1)
         if sp_group is not None:
             from ..parallel.ulysses import all_to_all_4d
             import torch.distributed as _dist

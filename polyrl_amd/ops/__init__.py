@@ -324,6 +324,29 @@ def fused_add_rmsnorm_train(x: torch.Tensor, res: torch.Tensor,
     return _RMSNormTrain.apply(x, res, w, eps)
 
 
+class _RopeTrain(torch.autograd.Function):
+    """Trainer NEOX RoPE in one kernel pass each way (rope_train.hip)."""
+
+    @staticmethod
+    def forward(ctx, x, cos_t, sin_t):
+        ext = _require_ext()
+        ctx.save_for_backward(cos_t, sin_t)
+        return ext.rope_train_apply(x.contiguous(), cos_t, sin_t, False)
+
+    @staticmethod
+    def backward(ctx, dy):
+        cos_t, sin_t = ctx.saved_tensors
+        ext = _require_ext()
+        return (ext.rope_train_apply(dy.contiguous(), cos_t, sin_t, True),
+                None, None)
+
+
+def rope_train(x: torch.Tensor, cos_t: torch.Tensor, sin_t: torch.Tensor
+               ) -> torch.Tensor:
+    """(T, H, D) bf16 with (T, D/2) fp32 tables — differentiable."""
+    return _RopeTrain.apply(x, cos_t, sin_t)
+
+
 class _TunedLinear(torch.autograd.Function):
     """nn.Linear matmuls through the per-shape hipBLASLt algo search
     (gemm_tuned.cpp): fwd nt, dgrad nn, wgrad tn — the trainer's three hot

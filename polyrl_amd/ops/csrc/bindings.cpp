@@ -43,6 +43,8 @@ std::vector<torch::Tensor> rmsnorm_train_bwd(torch::Tensor dy,
                                              torch::Tensor h,
                                              torch::Tensor w,
                                              torch::Tensor rstd);
+torch::Tensor rope_train_apply(torch::Tensor x, torch::Tensor cos_t,
+                               torch::Tensor sin_t, bool backward);
 torch::Tensor tuned_linear_fwd(torch::Tensor x, torch::Tensor w);
 torch::Tensor tuned_linear_dgrad(torch::Tensor dy, torch::Tensor w);
 torch::Tensor tuned_linear_wgrad(torch::Tensor dy, torch::Tensor x);
@@ -75,6 +77,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "trainer RMSNorm fwd (+optional fused residual): y, h, rstd");
   m.def("rmsnorm_train_bwd", &rmsnorm_train_bwd,
         "trainer RMSNorm bwd: dx, dw(fp32)");
+  m.def("rope_train_apply", &rope_train_apply,
+        "trainer NEOX RoPE (one pass; backward = rotate by -theta)");
   m.def("tuned_linear_fwd", &tuned_linear_fwd,
         "Y = X W^T via hipBLASLt with per-shape in-process algo search");
   m.def("tuned_linear_dgrad", &tuned_linear_dgrad, "dX = dY W (algo-pinned)");

@@ -28,6 +28,7 @@ sources = [
         "gemm_tuned.cpp",
         "elementwise.hip",
         "rmsnorm_train.hip",
+        "rope_train.hip",
         "logprobs.hip",
         "sampling.hip",
         "kv_cache.hip",

@@ -761,3 +761,28 @@ def test_tuned_mm_nt_decode_shapes():
         y = ops.tuned_mm_nt(x, w)
         yr = (x.float() @ w.float().t())
         assert rel_err(y.float(), yr) < 1e-2, (M, N, K)
+
+
+def test_rope_train_fwd_bwd():
+    """Fused trainer RoPE vs the eager rotate-half composition + autograd."""
+    torch.manual_seed(57)
+    T, H, D = 333, 8, 128
+    x = (torch.randn(T, H, D, device=DEV) / 2).bfloat16().requires_grad_()
+    pos = torch.randint(0, 500, (T,), device=DEV)
+    inv = 1.0 / (10000.0 ** (torch.arange(0, D, 2, device=DEV,
+                                          dtype=torch.float32) / D))
+    freqs = pos.float()[:, None] * inv[None, :]
+    cos, sin = freqs.cos().contiguous(), freqs.sin().contiguous()
+    y = ops.rope_train(x, cos, sin)
+    g = torch.randn(T, H, D, device=DEV)
+    (y.float() * g).sum().backward()
+
+    x2 = x.detach().float().requires_grad_()
+    c2 = cos.unsqueeze(1)
+    s2 = sin.unsqueeze(1)
+    d = D // 2
+    y2 = torch.cat([x2[..., :d] * c2 - x2[..., d:] * s2,
+                    x2[..., d:] * c2 + x2[..., :d] * s2], dim=-1)
+    (y2 * g).sum().backward()
+    assert rel_err(y.float(), y2.detach()) < 1e-2
+    assert rel_err(x.grad.float(), x2.grad) < 1e-2
