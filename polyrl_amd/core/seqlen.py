@@ -68,9 +68,37 @@ def balance_batch_indices(attention_mask: torch.Tensor, world_size: int) -> torc
     return torch.tensor(idx, dtype=torch.long)
 
 
-def prepare_dynamic_batch(batch: TensorBatch, max_token_len: int
+def _dense_partitions(seqlens: List[int], budget: int) -> List[List[int]]:
+    """First-fit-decreasing bin packing: fill micros to ~the budget, leave
+    ONE small tail instead of k equally-underfull micros.  Measured on
+    MI355X (profiles/PROFILES.md r2): balanced partitions at budget 8192
+    produce 4 x ~6960-token micros that fixed-M padding rounds up to 8192
+    each — ~15% padded GEMM work; dense packing makes that pad ~free."""
+    order = sorted(range(len(seqlens)), key=lambda i: -seqlens[i])
+    bins: List[List[int]] = []
+    loads: List[int] = []
+    for idx in order:
+        s = seqlens[idx]
+        placed = False
+        for b in range(len(bins)):
+            if loads[b] + s <= budget:
+                bins[b].append(idx)
+                loads[b] += s
+                placed = True
+                break
+        if not placed:
+            bins.append([idx])
+            loads.append(s)
+    return [sorted(b) for b in bins]
+
+
+def prepare_dynamic_batch(batch: TensorBatch, max_token_len: int,
+                          packing: str = "dense"
                           ) -> Tuple[List[TensorBatch], List[List[int]]]:
-    """Split a batch into micro-batches bounded by a per-micro-batch token budget.
+    """Split a batch into micro-batches bounded by a per-micro-batch token
+    budget.  ``packing``: "dense" (default — first-fit-decreasing, micros
+    ~full + one tail) or "balanced" (near-equal micro sizes, the reference
+    seqlen_balancing behavior).
 
     Returns (micro_batches, index_lists); restore_dynamic_batch inverts it.
     """
@@ -79,19 +107,21 @@ def prepare_dynamic_batch(batch: TensorBatch, max_token_len: int
     max_seq = max(seqlens) if seqlens else 0
     assert max_seq <= max_token_len, \
         f"one sample has {max_seq} tokens > budget {max_token_len}"
-    total = sum(seqlens)
-    k = max(1, -(-total // max_token_len))  # ceil
-    # grow k until every partition fits the budget
-    while k <= len(seqlens):
-        if k > len(seqlens):
-            break
+    if packing == "dense":
+        parts = _dense_partitions(seqlens, max_token_len)
+    else:
+        total = sum(seqlens)
+        k = max(1, -(-total // max_token_len))  # ceil
+        # grow k until every partition fits the budget
+        while k <= len(seqlens):
+            parts = get_seqlen_balanced_partitions(
+                seqlens, min(k, len(seqlens)), equal_size=False)
+            if all(sum(seqlens[i] for i in p) <= max_token_len
+                   for p in parts):
+                break
+            k += 1
         parts = get_seqlen_balanced_partitions(seqlens, min(k, len(seqlens)),
                                                equal_size=False)
-        if all(sum(seqlens[i] for i in p) <= max_token_len for p in parts):
-            break
-        k += 1
-    parts = get_seqlen_balanced_partitions(seqlens, min(k, len(seqlens)),
-                                           equal_size=False)
     micro = [batch.slice(torch.tensor(p, dtype=torch.long)) for p in parts]
     return micro, parts
 

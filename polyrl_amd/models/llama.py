@@ -294,7 +294,7 @@ class DecoderModel(nn.Module):
         T = int(ids_p.shape[0])
         chunk = sp * _pack_align()
         pad_to = _pack_pad_to(self)
-        if pad_to and T <= pad_to:
+        if pad_to and 0.75 * pad_to <= T <= pad_to:
             Tp = -(-pad_to // chunk) * chunk
         else:
             Tp = -(-T // chunk) * chunk
@@ -350,7 +350,10 @@ class DecoderModel(nn.Module):
         T = int(ids_p.shape[0])
         align = _pack_align()
         pad_to = _pack_pad_to(self)
-        if pad_to and T <= pad_to:
+        # fixed-M padding only pays off when the micro is nearly full
+        # (dense packing leaves one small tail micro — padding THAT to the
+        # whole budget would waste more GEMM work than the shape buys)
+        if pad_to and 0.75 * pad_to <= T <= pad_to:
             Tp = pad_to
         else:
             Tp = -(-T // align) * align if align > 1 else T

@@ -135,9 +135,15 @@ class LocalRolloutCoordinator:
         # serial decode: the engine holds its own weight buffers at the
         # step's version, and publish happens only between steps when the
         # pump is idle (stream_batches drains everything first).
+        # Default OFF: measured ZERO overlap on uniform-length synthetic
+        # benches (every sample finishes decode at the same iteration, so
+        # updates cannot start earlier) at ~1.5% thread/stream overhead.
+        # Turn ON (POLYRL_ASYNC_DECODE=1 / rollout.async_decode) for real
+        # EOS-terminated workloads, where staggered completions let decode
+        # of the tail overlap updates on the head.
         import os
         self.async_decode = (async_decode and not self.shard
-                             and os.environ.get("POLYRL_ASYNC_DECODE", "1")
+                             and os.environ.get("POLYRL_ASYNC_DECODE", "0")
                              == "1")
         self._pump = None
         self._pump_err = None

@@ -69,7 +69,7 @@ def gpu_pair():
             real_step = eng.step
 
             def slow_step(*a, _rs=real_step, **kw):
-                time.sleep(0.02)       # ~50 tok/s => the time-box wins
+                time.sleep(0.03)       # ~130 tok/s incl. chunks => box wins
                 return _rs(*a, **kw)
             eng.step = slow_step
         runner = EngineRunner(eng)
@@ -102,7 +102,7 @@ def test_timebox_continuation_over_real_http(gpu_pair):
                            sampling=SamplingSpec(max_new_tokens=64,
                                                  temperature=0.0))
         items = []
-        async for it in sched.submit_batch([req], max_local_gen_s=0.4):
+        async for it in sched.submit_batch([req], max_local_gen_s=0.25):
             items.append(it)
         await sched.close()
         assert items[0]["type"] == "notifier"

@@ -59,3 +59,36 @@ def test_fixed_micro_batches():
     b = make_batch([4] * 10)
     micro, parts = fixed_micro_batches(b, 4)
     assert [len(m) for m in micro] == [4, 4, 2]
+
+
+def test_dense_partitions_fill_budget():
+    """Dense packing (default): micros ~full + one small tail, vs balanced
+    k equally-underfull micros; both respect the budget and partition."""
+    import torch
+
+    from polyrl_amd.core.seqlen import prepare_dynamic_batch
+    from polyrl_amd.protocol import TensorBatch
+    torch.manual_seed(0)
+    B, L = 64, 512
+    lens = torch.randint(300, 512, (B,))
+    am = torch.zeros(B, L, dtype=torch.long)
+    for i in range(B):
+        am[i, :lens[i]] = 1
+    batch = TensorBatch(tensors={"attention_mask": am,
+                                 "input_ids": torch.zeros(B, L, dtype=torch.long)})
+    budget = 8192
+    micro_d, parts_d = prepare_dynamic_batch(batch, budget, packing="dense")
+    micro_b, parts_b = prepare_dynamic_batch(batch, budget,
+                                             packing="balanced")
+    for parts in (parts_d, parts_b):
+        got = sorted(i for p in parts for i in p)
+        assert got == list(range(B))
+        for p in parts:
+            assert sum(int(lens[i]) for i in p) <= budget
+    # dense: all but (at most) one micro filled to >= 75% of the budget --
+    # the fixed-M padding threshold
+    fills = sorted((sum(int(lens[i]) for i in p) for p in parts_d),
+                   reverse=True)
+    assert all(f >= 0.75 * budget for f in fills[:-1]), fills
+    # and dense never uses more micros than balanced
+    assert len(parts_d) <= len(parts_b) + 1
