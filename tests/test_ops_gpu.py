@@ -786,3 +786,54 @@ def test_rope_train_fwd_bwd():
     (y2 * g).sum().backward()
     assert rel_err(y.float(), y2.detach()) < 1e-2
     assert rel_err(x.grad.float(), x2.grad) < 1e-2
+
+
+def test_fused_trainer_paths_update_equivalence(monkeypatch):
+    """One actor update step with ALL fused trainer paths (norm, rope,
+    tuned linears) vs the eager composition: logprobs and grad-norm must
+    agree within bf16 noise — guards against silent training corruption
+    from any fused kernel."""
+    from polyrl_amd.config import ActorConfig
+    from polyrl_amd.models import create_model
+    from polyrl_amd.models.registry import DecoderConfig
+    from polyrl_amd.protocol import TensorBatch
+    from polyrl_amd.trainer.workers import ActorWorker
+
+    cfg = DecoderConfig(arch="llama", vocab_size=512, hidden_size=256,
+                        intermediate_size=512, num_hidden_layers=2,
+                        num_attention_heads=4, num_key_value_heads=2,
+                        head_dim=64, max_position_embeddings=256,
+                        rope_theta=10000.0, rms_norm_eps=1e-6)
+    torch.manual_seed(80)
+    B, Lp, Lr = 6, 10, 8
+    ids = torch.randint(0, cfg.vocab_size, (B, Lp + Lr))
+    am = torch.ones(B, Lp + Lr, dtype=torch.long)
+    am[0, :4] = 0
+    pos = (torch.cumsum(am, 1) - 1).clamp(min=0)
+    batch = TensorBatch(tensors={
+        "input_ids": ids, "attention_mask": am, "position_ids": pos,
+        "responses": ids[:, Lp:],
+        "response_mask": torch.ones(B, Lr),
+        "old_log_probs": torch.zeros(B, Lr),
+        "advantages": torch.randn(B, Lr),
+    })
+
+    def run(fused):
+        monkeypatch.setenv("POLYRL_FUSED_NORM", "1" if fused else "0")
+        monkeypatch.setenv("POLYRL_TUNED_GEMM", "1" if fused else "0")
+        torch.manual_seed(81)
+        model = create_model(cfg, kind="actor", dtype="bfloat16", device=DEV)
+        model.model.use_remove_padding = True
+        acfg = ActorConfig()
+        acfg.use_dynamic_bsz = False
+        acfg.ppo_micro_batch_size_per_gpu = B
+        w = ActorWorker(model, acfg, device=DEV)
+        lp, _ = w.compute_log_prob(batch)
+        m = w.update_policy_stream(batch, is_opt_step=True, is_lr_step=True,
+                                   accum_scale=1.0)
+        return lp, m["actor/grad_norm"][0]
+
+    lp_f, gn_f = run(True)
+    lp_e, gn_e = run(False)
+    assert rel_err(lp_f, lp_e) < 5e-2, rel_err(lp_f, lp_e)
+    assert abs(gn_f - gn_e) / max(gn_e, 1e-6) < 0.1, (gn_f, gn_e)
