@@ -97,13 +97,23 @@ def test_timebox_continuation_over_real_http(gpu_pair):
         sched = RolloutScheduler(SchedulerConfig(stats_interval_s=0.05))
         await sched.register_instance(local, skip_health_check=True)
         await sched.register_instance(remote, skip_health_check=True)
-        sched._states["remote-fast"].assigned_batches = 10 ** 9
+        # force first dispatch to the slow local: take the remote out of
+        # the active pool until the local is mid-generation (an admission
+        # counter alone is reset by the 1 Hz stats worker)
+        sched._active.remove("remote-fast")
+
+        async def readd():
+            await asyncio.sleep(0.1)
+            if "remote-fast" not in sched._active:
+                sched._active.append("remote-fast")
+        task = asyncio.get_running_loop().create_task(readd())
         req = GroupRequest(gid=0, input_ids=[3, 4, 5], n=1,
                            sampling=SamplingSpec(max_new_tokens=64,
                                                  temperature=0.0))
         items = []
         async for it in sched.submit_batch([req], max_local_gen_s=0.25):
             items.append(it)
+        await task
         await sched.close()
         assert items[0]["type"] == "notifier"
         s = items[1].samples[0]

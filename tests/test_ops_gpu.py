@@ -481,7 +481,7 @@ def test_engine_greedy_matches_trainer_model_gpu():
     # logits below the fp32 argmax and fail regardless of position.
     model32 = create_model(cfg, kind="actor", dtype="float32", device=DEV)
     model32.load_state_dict(
-        {k: v.float() for k, v in eng.model.state_dict().items()})
+        {k: v.float() for k, v in model.state_dict().items()})
     MARGIN = 0.15
     ids = list(prompt)
     n_exact = 0
