@@ -25,8 +25,12 @@ def _norm(mod, x: torch.Tensor) -> torch.Tensor:
     """RMSNorm through the fused trainer kernel on GPU bf16; eager
     reference elsewhere (CPU tier / fp32)."""
     import polyrl_amd.ops as pops
-    if x.is_cuda and x.dtype == torch.bfloat16             and mod.weight.dtype == torch.bfloat16             and pops._fused_norm_enabled():
-        return pops.rmsnorm_train(x, mod.weight, mod.variance_epsilon)
+    w = mod.weight
+    if x.is_cuda and x.dtype == torch.bfloat16 \
+            and w.dtype == torch.bfloat16 \
+            and not hasattr(w, "placements") \
+            and pops._fused_norm_enabled():
+        return pops.rmsnorm_train(x, w, mod.variance_epsilon)
     return mod(x)
 
 
@@ -243,7 +247,10 @@ class DecoderLayer(nn.Module):
                 _norm(self.input_layernorm, x), cos, sin, cu_seqlens,
                 sp_group)
             import polyrl_amd.ops as pops
-            if attn_out.is_cuda and attn_out.dtype == torch.bfloat16                     and pops._fused_norm_enabled():
+            _w2 = self.post_attention_layernorm.weight
+            if attn_out.is_cuda and attn_out.dtype == torch.bfloat16 \
+                    and not hasattr(_w2, "placements") \
+                    and pops._fused_norm_enabled():
                 # fused residual-add + norm: h = x + attn_out computed
                 # inside the norm kernel (one pass; rmsnorm_train.hip)
                 m_in, h = pops.fused_add_rmsnorm_train(

@@ -386,10 +386,12 @@ def tuned_linear(x: torch.Tensor, w: torch.Tensor,
                  b: Optional[torch.Tensor] = None) -> torch.Tensor:
     """F.linear with per-shape algo pinning on GPU bf16 (training trunk
     shapes); falls through to F.linear elsewhere (CPU tier, tiny dims,
-    non-bf16)."""
+    non-bf16, DTensor-wrapped FSDP params)."""
     if (x.is_cuda and x.dtype == torch.bfloat16
             and w.dtype == torch.bfloat16 and w.shape[0] >= 64
-            and w.shape[1] >= 64 and _tuned_gemm_enabled()):
+            and w.shape[1] >= 64 and _tuned_gemm_enabled()
+            and not hasattr(w, "placements")
+            and not hasattr(x, "placements")):
         shp = x.shape
         y = _TunedLinear.apply(x.reshape(-1, shp[-1]).contiguous(), w)
         y = y.view(*shp[:-1], w.shape[0])

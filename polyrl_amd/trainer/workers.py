@@ -79,7 +79,11 @@ def _build_optimizer(params, cfg: OptimConfig):
     the foreach default's extra kernel launches and intermediate reads
     are measurable at 8B params."""
     params = list(params)
-    fused = any(p.is_cuda for p in params if isinstance(p, torch.Tensor))
+    # fused only for plain CUDA tensors: DTensor (FSDP2 world>1) support
+    # for the fused multi-tensor kernel is not validated on this stack
+    fused = bool(params) and all(
+        isinstance(p, torch.Tensor) and p.is_cuda
+        and not hasattr(p, "placements") for p in params)
     try:
         return torch.optim.AdamW(params, lr=cfg.lr, betas=tuple(cfg.betas),
                                  eps=cfg.eps, weight_decay=cfg.weight_decay,
