@@ -25,7 +25,7 @@
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
 typedef __attribute__((ext_vector_type(4))) float f32x4_t;
 
-#define KVBLK 32
+#define KVBLK 64
 #define QROWS_PER_WAVE 16
 #define NWAVE 4
 #define QTILE (QROWS_PER_WAVE * NWAVE)  // 64
@@ -127,11 +127,14 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
     }
     __syncthreads();
 
-    // ---------------- QK^T: S[16 q][32 keys] = Q @ K^T --------------------
+    // ---------------- QK^T: S[16 q][KVBLK keys] = Q @ K^T -----------------
     // B-frag (d x key): lane reads K[key = nt*16 + l15][d = ks*32 + lhi*8 ..]
-    f32x4_t c[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+    constexpr int NKT = KVBLK / 16;
+    f32x4_t c[NKT];
 #pragma unroll
-    for (int nt = 0; nt < 2; ++nt) {
+    for (int nt = 0; nt < NKT; ++nt) c[nt] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int nt = 0; nt < NKT; ++nt) {
       const int krow = nt * 16 + l15;
 #pragma unroll
       for (int ks = 0; ks < KS; ++ks) {
@@ -145,9 +148,9 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
 
     // ---------------- scale + causal mask ---------------------------------
     // C layout: row = lhi*4 + r, col = l15
-    float s[2][4];
+    float s[NKT][4];
 #pragma unroll
-    for (int nt = 0; nt < 2; ++nt) {
+    for (int nt = 0; nt < NKT; ++nt) {
       const int kpos = t0 + nt * 16 + l15;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
@@ -160,11 +163,13 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
     }
 
     // ---------------- online softmax (per C-row) ---------------------------
-    float p[2][4];
+    float p[NKT][4];
     float alpha[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      float rm = fmaxf(s[0][r], s[1][r]);
+      float rm = s[0][r];
+#pragma unroll
+      for (int nt = 1; nt < NKT; ++nt) rm = fmaxf(rm, s[nt][r]);
 #pragma unroll
       for (int off = 1; off < 16; off <<= 1) rm = fmaxf(rm, __shfl_xor(rm, off, 64));
       const float mn = fmaxf(m_run[r], rm);
@@ -172,7 +177,7 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
       m_run[r] = mn;
       float rs = 0.f;
 #pragma unroll
-      for (int nt = 0; nt < 2; ++nt) {
+      for (int nt = 0; nt < NKT; ++nt) {
         p[nt][r] = __expf(s[nt][r] - mn);
         rs += p[nt][r];
       }
@@ -189,25 +194,29 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
     // ---------------- P: C-layout -> A-layout through wave-private LDS ----
     bf16_t* ps = Ps[wid];
 #pragma unroll
-    for (int nt = 0; nt < 2; ++nt)
+    for (int nt = 0; nt < NKT; ++nt)
 #pragma unroll
       for (int r = 0; r < 4; ++r)
         ps[(lhi * 4 + r) * KVBLK + nt * 16 + l15] = f2bf(p[nt][r]);
     // wave-private tile: the compiler orders ds_write -> ds_read by lgkmcnt
-    const bf16x8_t pa = *reinterpret_cast<const bf16x8_t*>(
-        ps + l15 * KVBLK + lhi * 8);
 
     // ---------------- P @ V ------------------------------------------------
-    // B-frag (key x d): lane needs V[key = lhi*8 + j][d = nt*16 + l15]
+    // k runs over the KVBLK keys in 32-deep steps (A-frag k = kq*32+lhi*8)
 #pragma unroll
-    for (int nt = 0; nt < NDT; ++nt) {
-      bf16x8_t bv;
+    for (int kq = 0; kq < KVBLK / 32; ++kq) {
+      const bf16x8_t pa = *reinterpret_cast<const bf16x8_t*>(
+          ps + l15 * KVBLK + kq * 32 + lhi * 8);
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        bv[j] = *reinterpret_cast<const __bf16*>(
-            Vs + (lhi * 8 + j) * HEAD_DIM + nt * 16 + l15);
-      o_acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, bv, o_acc[nt],
-                                                          0, 0, 0);
+      for (int nt = 0; nt < NDT; ++nt) {
+        bf16x8_t bv;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          bv[j] = *reinterpret_cast<const __bf16*>(
+              Vs + (kq * 32 + lhi * 8 + j) * HEAD_DIM + nt * 16 + l15);
+        o_acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, bv,
+                                                            o_acc[nt],
+                                                            0, 0, 0);
+      }
     }
     __syncthreads();  // all waves done with Ks/Vs before restage
   }

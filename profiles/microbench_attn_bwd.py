@@ -70,7 +70,33 @@ def probe(nseq=16, L=512, Hq=32, Hk=32, D=128):
 
 if __name__ == "__main__":
     assert torch.cuda.is_available()
+    probe_fwd(16, 512, 32, 8, 128)
+    probe_fwd(4, 2048, 32, 8, 128)
     probe(16, 512, 32, 32, 128)
     probe(16, 512, 32, 8, 128)     # llama3-8b GQA shape
     probe(4, 2048, 32, 8, 128)     # long-seq shape
     probe(16, 512, 16, 16, 64)     # head_dim 64 (v3 only vs itself)
+
+
+def probe_fwd(nseq=16, L=512, Hq=32, Hk=8, D=128, iters=30):
+    import polyrl_amd.ops as ops
+    torch.manual_seed(4)
+    total = nseq * L
+    cu = torch.arange(0, nseq + 1, dtype=torch.int32, device="cuda") * L
+    scale = 1.0 / math.sqrt(D)
+    q = (torch.randn(total, Hq, D, device="cuda") / 4).bfloat16()
+    k = (torch.randn(total, Hk, D, device="cuda") / 4).bfloat16()
+    v = (torch.randn(total, Hk, D, device="cuda") / 4).bfloat16()
+    with torch.no_grad():
+        for _ in range(5):
+            o = ops.flash_attn_varlen(q, k, v, cu, scale, causal=True)
+        torch.cuda.synchronize()
+        s = torch.cuda.Event(True); e = torch.cuda.Event(True)
+        s.record()
+        for _ in range(iters):
+            o = ops.flash_attn_varlen(q, k, v, cu, scale, causal=True)
+        e.record(); torch.cuda.synchronize()
+    fl = 2 * 2 * nseq * (L * L / 2) * Hq * D
+    ms = s.elapsed_time(e) / iters
+    print(f"fwd probe {nseq}x{L} Hq={Hq} Hk={Hk}: {ms:.3f} ms "
+          f"({fl/ms/1e9:.0f} TF/s)")
