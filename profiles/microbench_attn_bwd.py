@@ -68,16 +68,6 @@ def probe(nseq=16, L=512, Hq=32, Hk=32, D=128):
             print(f"  {name}: max abs diff {err:.3e} (max |v1|={den:.3e})")
 
 
-if __name__ == "__main__":
-    assert torch.cuda.is_available()
-    probe_fwd(16, 512, 32, 8, 128)
-    probe_fwd(4, 2048, 32, 8, 128)
-    probe(16, 512, 32, 32, 128)
-    probe(16, 512, 32, 8, 128)     # llama3-8b GQA shape
-    probe(4, 2048, 32, 8, 128)     # long-seq shape
-    probe(16, 512, 16, 16, 64)     # head_dim 64 (v3 only vs itself)
-
-
 def probe_fwd(nseq=16, L=512, Hq=32, Hk=8, D=128, iters=30):
     import polyrl_amd.ops as ops
     torch.manual_seed(4)
@@ -100,3 +90,14 @@ def probe_fwd(nseq=16, L=512, Hq=32, Hk=8, D=128, iters=30):
     ms = s.elapsed_time(e) / iters
     print(f"fwd probe {nseq}x{L} Hq={Hq} Hk={Hk}: {ms:.3f} ms "
           f"({fl/ms/1e9:.0f} TF/s)")
+
+
+if __name__ == "__main__":
+    assert torch.cuda.is_available()
+    probe_fwd(16, 512, 32, 8, 128)
+    probe_fwd(4, 2048, 32, 8, 128)
+    probe(16, 512, 32, 32, 128)
+    probe(16, 512, 32, 8, 128)     # llama3-8b GQA shape
+    probe(4, 2048, 32, 8, 128)     # long-seq shape
+    probe(16, 512, 16, 16, 64)     # head_dim 64 (v3 only vs itself)
+
