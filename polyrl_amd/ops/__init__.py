@@ -324,6 +324,35 @@ def fused_add_rmsnorm_train(x: torch.Tensor, res: torch.Tensor,
     return _RMSNormTrain.apply(x, res, w, eps)
 
 
+class _GatherLogprobsTrain(torch.autograd.Function):
+    """Differentiable per-token logprob gather over the vocab: fwd saves
+    only (bf16 logits, labels, lse) instead of autograd's fp32 log_softmax
+    output; bwd is one streaming pass (logprobs.hip)."""
+
+    @staticmethod
+    def forward(ctx, logits, labels):
+        ext = _require_ext()
+        N = logits.size(0)
+        lp = torch.empty(N, dtype=torch.float32, device=logits.device)
+        lse = torch.empty(N, dtype=torch.float32, device=logits.device)
+        ext.gather_logprobs_train_fwd(lp, lse, logits, labels)
+        ctx.save_for_backward(logits, labels, lse)
+        return lp
+
+    @staticmethod
+    def backward(ctx, dlp):
+        logits, labels, lse = ctx.saved_tensors
+        ext = _require_ext()
+        return ext.gather_logprobs_train_bwd(logits, labels, lse,
+                                             dlp.float()), None
+
+
+def gather_logprobs_train(logits: torch.Tensor, labels: torch.Tensor
+                          ) -> torch.Tensor:
+    """(N, V) bf16 contiguous + (N,) int64 -> (N,) fp32, differentiable."""
+    return _GatherLogprobsTrain.apply(logits.contiguous(), labels.long())
+
+
 class _RopeTrain(torch.autograd.Function):
     """Trainer NEOX RoPE in one kernel pass each way (rope_train.hip)."""
 

@@ -43,6 +43,12 @@ std::vector<torch::Tensor> rmsnorm_train_bwd(torch::Tensor dy,
                                              torch::Tensor h,
                                              torch::Tensor w,
                                              torch::Tensor rstd);
+void gather_logprobs_train_fwd(torch::Tensor out_lp, torch::Tensor out_lse,
+                               torch::Tensor logits, torch::Tensor labels);
+torch::Tensor gather_logprobs_train_bwd(torch::Tensor logits,
+                                        torch::Tensor labels,
+                                        torch::Tensor lse,
+                                        torch::Tensor dlp);
 torch::Tensor rope_train_apply(torch::Tensor x, torch::Tensor cos_t,
                                torch::Tensor sin_t, bool backward);
 torch::Tensor tuned_linear_fwd(torch::Tensor x, torch::Tensor w);
@@ -77,6 +83,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "trainer RMSNorm fwd (+optional fused residual): y, h, rstd");
   m.def("rmsnorm_train_bwd", &rmsnorm_train_bwd,
         "trainer RMSNorm bwd: dx, dw(fp32)");
+  m.def("gather_logprobs_train_fwd", &gather_logprobs_train_fwd,
+        "training logprob gather: lp + logsumexp (bf16 logits)");
+  m.def("gather_logprobs_train_bwd", &gather_logprobs_train_bwd,
+        "dlogits = dlp * (onehot - softmax), one pass");
   m.def("rope_train_apply", &rope_train_apply,
         "trainer NEOX RoPE (one pass; backward = rotate by -theta)");
   m.def("tuned_linear_fwd", &tuned_linear_fwd,

@@ -53,6 +53,41 @@ __global__ void gather_logprobs_kernel(float* __restrict__ out_lp,
   }
 }
 
+// ---------------------------------------------------- training backward
+// d lp / d logits = onehot(label) - softmax(logits); chain with dlp.
+// One streaming pass per row using the forward-saved logsumexp.
+__global__ void gather_logprobs_bwd_kernel(
+    bf16_t* __restrict__ dlogits,
+    const bf16_t* __restrict__ logits,
+    const int64_t* __restrict__ labels,
+    const float* __restrict__ lse,
+    const float* __restrict__ dlp,
+    int V) {
+  const long row = blockIdx.x;
+  const bf16_t* x = logits + row * V;
+  bf16_t* dx = dlogits + row * V;
+  const float g = dlp[row];
+  const float l = lse[row];
+  const long lab = labels[row];
+  const bf16x8* xp = reinterpret_cast<const bf16x8*>(x);
+  bf16x8* dp = reinterpret_cast<bf16x8*>(dx);
+  const int n8 = V / 8;
+  for (int i = threadIdx.x; i < n8; i += blockDim.x) {
+    bf16x8 v = xp[i], o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o.v[j] = f2bf(-g * __expf(bf2f(v.v[j]) - l));
+    dp[i] = o;
+  }
+  // tail when V % 8 != 0
+  for (int i = n8 * 8 + threadIdx.x; i < V; i += blockDim.x)
+    dx[i] = f2bf(-g * __expf(bf2f(x[i]) - l));
+  // add the onehot term after every softmax write landed
+  __syncthreads();
+  if (threadIdx.x == 0)
+    dx[lab] = f2bf(bf2f(dx[lab]) + g);
+}
+
 template <typename T>
 static void launch_gather(torch::Tensor out_lp, torch::Tensor out_ent,
                           torch::Tensor logits, torch::Tensor labels,
@@ -70,6 +105,67 @@ static void launch_gather(torch::Tensor out_lp, torch::Tensor out_ent,
         labels.data_ptr<int64_t>(), V);
   }
   HIP_CHECK_KERNEL();
+}
+
+// training fwd: lp + logsumexp (for the hand-written backward)
+__global__ void gather_logprobs_lse_kernel(
+    float* __restrict__ out_lp, float* __restrict__ out_lse,
+    const bf16_t* __restrict__ logits, const int64_t* __restrict__ labels,
+    int V) {
+  __shared__ float red[8];
+  const long row = blockIdx.x;
+  const bf16_t* x = logits + row * V;
+  float m = -INFINITY, s = 0.f;
+  for (int i = threadIdx.x; i < V; i += blockDim.x) {
+    float xi = bf2f(x[i]);
+    if (xi > m) {
+      s *= __expf(m - xi);
+      m = xi;
+    }
+    s += __expf(xi - m);
+  }
+  float gm = block_reduce_max<4>(m, red);
+  __syncthreads();
+  s *= (m == -INFINITY) ? 0.f : __expf(m - gm);
+  float gs = block_reduce_sum<4>(s, red);
+  if (threadIdx.x == 0) {
+    float lz = gm + __logf(gs);
+    out_lp[row] = bf2f(x[labels[row]]) - lz;
+    out_lse[row] = lz;
+  }
+}
+
+void gather_logprobs_train_fwd(torch::Tensor out_lp, torch::Tensor out_lse,
+                               torch::Tensor logits, torch::Tensor labels) {
+  TORCH_CHECK(logits.is_cuda() && logits.dim() == 2 &&
+              logits.is_contiguous() &&
+              logits.dtype() == torch::kBFloat16);
+  TORCH_CHECK(labels.dtype() == torch::kInt64);
+  const long N = logits.size(0);
+  const int V = logits.size(1);
+  auto stream = at::hip::getCurrentHIPStream();
+  gather_logprobs_lse_kernel<<<dim3(N), dim3(256), 0, stream>>>(
+      out_lp.data_ptr<float>(), out_lse.data_ptr<float>(),
+      (const bf16_t*)logits.data_ptr(), labels.data_ptr<int64_t>(), V);
+  HIP_CHECK_KERNEL();
+}
+
+torch::Tensor gather_logprobs_train_bwd(torch::Tensor logits,
+                                        torch::Tensor labels,
+                                        torch::Tensor lse,
+                                        torch::Tensor dlp) {
+  TORCH_CHECK(logits.is_cuda() && logits.is_contiguous() &&
+              logits.dtype() == torch::kBFloat16);
+  const long N = logits.size(0);
+  const int V = logits.size(1);
+  auto dlogits = torch::empty_like(logits);
+  auto stream = at::hip::getCurrentHIPStream();
+  gather_logprobs_bwd_kernel<<<dim3(N), dim3(256), 0, stream>>>(
+      (bf16_t*)dlogits.data_ptr(), (const bf16_t*)logits.data_ptr(),
+      labels.data_ptr<int64_t>(), lse.data_ptr<float>(),
+      dlp.contiguous().data_ptr<float>(), V);
+  HIP_CHECK_KERNEL();
+  return dlogits;
 }
 
 void gather_logprobs(torch::Tensor out_lp, torch::Tensor out_ent,

@@ -285,6 +285,27 @@ class ActorWorker:
                                                  want_entropy)
             logits = self.model(ids, attention_mask=am, position_ids=pos,
                                 logits_slice=slice(-Lr - 1, -1))
+            fused_ok = (logits.is_cuda and logits.dtype == torch.bfloat16
+                        and getattr(self.cfg, "use_fused_kernels", True))
+            if fused_ok and not want_entropy:
+                # fused CE-style gather: no fp32 logits materialization,
+                # no saved log_softmax (ops/csrc/logprobs.hip train path)
+                import polyrl_amd.ops as pops
+                B = logits.shape[0]
+                flat = logits.reshape(-1, logits.shape[-1])
+                if grad:
+                    lp = pops.gather_logprobs_train(flat, resp.reshape(-1))
+                else:
+                    lp = pops.gather_logprobs(flat, resp.reshape(-1))
+                lp = lp.view(B, Lr)
+                return lp, torch.zeros_like(lp)
+            if fused_ok and want_entropy and not grad:
+                import polyrl_amd.ops as pops
+                B = logits.shape[0]
+                flat = logits.reshape(-1, logits.shape[-1])
+                lp, ent = pops.gather_logprobs(flat, resp.reshape(-1),
+                                               want_entropy=True)
+                return lp.view(B, Lr), ent.view(B, Lr)
             logits = logits.float()
             lp = algos.logprobs_from_logits(logits, resp)
             ent = algos.entropy_from_logits(logits) if want_entropy else \

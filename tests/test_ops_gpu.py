@@ -837,3 +837,27 @@ def test_fused_trainer_paths_update_equivalence(monkeypatch):
     lp_e, gn_e = run(False)
     assert rel_err(lp_f, lp_e) < 5e-2, rel_err(lp_f, lp_e)
     assert abs(gn_f - gn_e) / max(gn_e, 1e-6) < 0.1, (gn_f, gn_e)
+
+
+def test_gather_logprobs_train_fwd_bwd():
+    """Fused training logprob gather (fwd lp+lse, streaming bwd) vs fp32
+    autograd log_softmax-gather."""
+    torch.manual_seed(58)
+    N, V = 64, 32000
+    logits = (torch.randn(N, V, device=DEV) * 3).bfloat16().requires_grad_()
+    labels = torch.randint(0, V, (N,), device=DEV)
+    lp = ops.gather_logprobs_train(logits, labels)
+    g = torch.randn(N, device=DEV)
+    (lp * g).sum().backward()
+
+    l2 = logits.detach().float().requires_grad_()
+    lp2 = torch.log_softmax(l2, -1).gather(-1, labels[:, None]).squeeze(-1)
+    (lp2 * g).sum().backward()
+    assert torch.allclose(lp, lp2.detach(), atol=1e-2, rtol=1e-3)
+    # grads: bf16 vs fp32 reference — compare at bf16 tolerance, and the
+    # onehot slot must carry the dominant term
+    ge = l2.grad
+    gf = logits.grad.float()
+    assert rel_err(gf, ge) < 2e-2, rel_err(gf, ge)
+    for i in range(0, N, 7):
+        assert abs(gf[i, labels[i]] - ge[i, labels[i]]) < 2e-2
