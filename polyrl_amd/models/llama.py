@@ -228,8 +228,13 @@ class MLP(nn.Module):
         self.down_proj = nn.Linear(i, h, bias=cfg.mlp_bias)
 
     def forward(self, x):
-        return _lin(self.down_proj,
-                    F.silu(_lin(self.gate_proj, x)) * _lin(self.up_proj, x))
+        g = _lin(self.gate_proj, x)
+        u = _lin(self.up_proj, x)
+        if g.is_cuda and g.dtype == torch.bfloat16 \
+                and g.numel() % 8 == 0:
+            import polyrl_amd.ops as pops
+            return _lin(self.down_proj, pops.silu_mul_train(g, u))
+        return _lin(self.down_proj, F.silu(g) * u)
 
 
 class DecoderLayer(nn.Module):

@@ -324,6 +324,33 @@ def fused_add_rmsnorm_train(x: torch.Tensor, res: torch.Tensor,
     return _RMSNormTrain.apply(x, res, w, eps)
 
 
+class _SiluMulTrain(torch.autograd.Function):
+    """Differentiable silu(gate)*up in one kernel each way."""
+
+    @staticmethod
+    def forward(ctx, gate, up):
+        ext = _require_ext()
+        g = gate.contiguous()
+        u = up.contiguous()
+        ctx.save_for_backward(g, u)
+        out = torch.empty_like(g)
+        ext.silu_mul(out, g, u)
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        g, u = ctx.saved_tensors
+        ext = _require_ext()
+        dgate = torch.empty_like(g)
+        dup = torch.empty_like(u)
+        ext.silu_mul_bwd(dgate, dup, dy.contiguous(), g, u)
+        return dgate, dup
+
+
+def silu_mul_train(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
+    return _SiluMulTrain.apply(gate, up)
+
+
 class _GatherLogprobsTrain(torch.autograd.Function):
     """Differentiable per-token logprob gather over the vocab: fwd saves
     only (bf16 logits, labels, lse) instead of autograd's fp32 log_softmax

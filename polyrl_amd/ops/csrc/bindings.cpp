@@ -43,6 +43,8 @@ std::vector<torch::Tensor> rmsnorm_train_bwd(torch::Tensor dy,
                                              torch::Tensor h,
                                              torch::Tensor w,
                                              torch::Tensor rstd);
+void silu_mul_bwd(torch::Tensor dgate, torch::Tensor dup, torch::Tensor dy,
+                  torch::Tensor gate, torch::Tensor up);
 void gather_logprobs_train_fwd(torch::Tensor out_lp, torch::Tensor out_lse,
                                torch::Tensor logits, torch::Tensor labels);
 torch::Tensor gather_logprobs_train_bwd(torch::Tensor logits,
@@ -83,6 +85,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "trainer RMSNorm fwd (+optional fused residual): y, h, rstd");
   m.def("rmsnorm_train_bwd", &rmsnorm_train_bwd,
         "trainer RMSNorm bwd: dx, dw(fp32)");
+  m.def("silu_mul_bwd", &silu_mul_bwd, "silu(gate)*up backward, one pass");
   m.def("gather_logprobs_train_fwd", &gather_logprobs_train_fwd,
         "training logprob gather: lp + logsumexp (bf16 logits)");
   m.def("gather_logprobs_train_bwd", &gather_logprobs_train_bwd,

@@ -112,6 +112,53 @@ __global__ void silu_mul_kernel(bf16_t* __restrict__ out,
   }
 }
 
+// training backward of silu(gate)*up: one pass, contiguous operands
+// dgate = dy * up * (sig + g*sig*(1-sig)); dup = dy * silu(g)
+__global__ void silu_mul_bwd_kernel(bf16_t* __restrict__ dgate,
+                                    bf16_t* __restrict__ dup,
+                                    const bf16_t* __restrict__ dy,
+                                    const bf16_t* __restrict__ gate,
+                                    const bf16_t* __restrict__ up,
+                                    long nvec) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < nvec; i += stride) {
+    bf16x8 d = reinterpret_cast<const bf16x8*>(dy)[i];
+    bf16x8 g = reinterpret_cast<const bf16x8*>(gate)[i];
+    bf16x8 u = reinterpret_cast<const bf16x8*>(up)[i];
+    bf16x8 og, ou;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float gf = bf2f(g.v[j]);
+      const float df = bf2f(d.v[j]);
+      const float sig = 1.f / (1.f + __expf(-gf));
+      const float silu = gf * sig;
+      og.v[j] = f2bf(df * bf2f(u.v[j]) * (sig + gf * sig * (1.f - sig)));
+      ou.v[j] = f2bf(df * silu);
+    }
+    reinterpret_cast<bf16x8*>(dgate)[i] = og;
+    reinterpret_cast<bf16x8*>(dup)[i] = ou;
+  }
+}
+
+void silu_mul_bwd(torch::Tensor dgate, torch::Tensor dup, torch::Tensor dy,
+                  torch::Tensor gate, torch::Tensor up) {
+  TORCH_CHECK(dy.is_cuda() && dy.dtype() == torch::kBFloat16);
+  TORCH_CHECK(dy.is_contiguous() && gate.is_contiguous() &&
+              up.is_contiguous() && dgate.is_contiguous() &&
+              dup.is_contiguous());
+  TORCH_CHECK(gate.numel() % 8 == 0);
+  const long nvec = gate.numel() / 8;
+  const long blocks = std::min<long>((nvec + 255) / 256,
+                                     MAX_RESIDENT_BLOCKS);
+  auto stream = at::hip::getCurrentHIPStream();
+  silu_mul_bwd_kernel<<<dim3(blocks), dim3(256), 0, stream>>>(
+      (bf16_t*)dgate.data_ptr(), (bf16_t*)dup.data_ptr(),
+      (const bf16_t*)dy.data_ptr(), (const bf16_t*)gate.data_ptr(),
+      (const bf16_t*)up.data_ptr(), nvec);
+  HIP_CHECK_KERNEL();
+}
+
 void silu_mul(torch::Tensor out, torch::Tensor gate, torch::Tensor up) {
   TORCH_CHECK(gate.is_cuda() && gate.dtype() == torch::kBFloat16);
   TORCH_CHECK(out.is_contiguous());

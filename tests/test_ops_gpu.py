@@ -861,3 +861,20 @@ def test_gather_logprobs_train_fwd_bwd():
     assert rel_err(gf, ge) < 2e-2, rel_err(gf, ge)
     for i in range(0, N, 7):
         assert abs(gf[i, labels[i]] - ge[i, labels[i]]) < 2e-2
+
+
+def test_silu_mul_train_fwd_bwd():
+    torch.manual_seed(59)
+    T, I = 500, 1024
+    g = (torch.randn(T, I, device=DEV)).bfloat16().requires_grad_()
+    u = (torch.randn(T, I, device=DEV)).bfloat16().requires_grad_()
+    y = ops.silu_mul_train(g, u)
+    w = torch.randn(T, I, device=DEV)
+    (y.float() * w).sum().backward()
+    g2 = g.detach().float().requires_grad_()
+    u2 = u.detach().float().requires_grad_()
+    y2 = torch.nn.functional.silu(g2) * u2
+    (y2 * w).sum().backward()
+    assert rel_err(y.float(), y2.detach()) < 1e-2
+    assert rel_err(g.grad.float(), g2.grad) < 2e-2
+    assert rel_err(u.grad.float(), u2.grad) < 2e-2
