@@ -91,7 +91,11 @@ def main():
     cfg.actor_rollout_ref.actor.fsdp.reshard_after_forward = False
     cfg.critic.fsdp.reshard_after_forward = False
     cfg.actor_rollout_ref.actor.ppo_mini_batch_size = total_samples // 2
-    cfg.actor_rollout_ref.actor.ppo_max_token_len_per_gpu = 8192
+    # micro token budget: 8192 fills the CU grid exactly (2 waves of 256
+    # blocks at MT256); long-response configs need at least one sequence
+    seq_len = args.prompt_len + args.response_len
+    budget = max(8192, seq_len)
+    cfg.actor_rollout_ref.actor.ppo_max_token_len_per_gpu = budget
     cfg.actor_rollout_ref.rollout.sampling.n = args.n_samples
     cfg.actor_rollout_ref.rollout.sampling.temperature = 1.0
     cfg.actor_rollout_ref.rollout.prompt_length = args.prompt_len
@@ -103,7 +107,7 @@ def main():
     cfg.critic.model.path = args.model
     cfg.critic.model.dtype = cfg.actor_rollout_ref.model.dtype
     cfg.critic.ppo_mini_batch_size = total_samples // 2
-    cfg.critic.ppo_max_token_len_per_gpu = 8192
+    cfg.critic.ppo_max_token_len_per_gpu = budget
     cfg.data.train_batch_size = global_batch
     cfg.data.max_prompt_length = args.prompt_len
     cfg.data.synthetic_num_prompts = max(
