@@ -211,7 +211,11 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
         }
       }
 
-      // ---- P, dS -> transposed LDS -------------------------------------
+      // ---- P, dS -> transposed LDS (interior tiles skip the mask:
+      // VALU-bound kernels, PMC 6:1 VALU:MFMA) -----------------------
+      const bool interior =
+          (t0 + KB3 <= Lk) && (qb + wid * 16 + 16 <= Lq) &&
+          (!causal || t0 + KB3 - 1 <= qk_off + qb + wid * 16);
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int qrow_c = qb + wid * 16 + lhi * 4 + r;   // C-layout row
@@ -220,16 +224,20 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
         const float lse_r = qdead ? 0.f : lse[gqc];
         const float del_r = qdead ? 0.f : delta[gqc];
         const int qpos = qk_off + qrow_c;
+        const int qcol = wid * 16 + lhi * 4 + r;
 #pragma unroll
         for (int nt = 0; nt < 4; ++nt) {
-          const int kpos = t0 + nt * 16 + l15;
-          const bool dead = qdead || (kpos >= Lk) ||
-                            (causal && kpos > qpos);
-          const float s = sc[nt][r] * scale;
-          const float p = dead ? 0.f : __expf(s - lse_r);
-          const float ds = p * (dpc[nt][r] - del_r) * scale;
-          const int qcol = wid * 16 + lhi * 4 + r;
           const int key = nt * 16 + l15;
+          float p;
+          if (interior) {
+            p = __expf(sc[nt][r] * scale - lse_r);
+          } else {
+            const int kpos = t0 + key;
+            const bool dead = qdead || (kpos >= Lk) ||
+                              (causal && kpos > qpos);
+            p = dead ? 0.f : __expf(sc[nt][r] * scale - lse_r);
+          }
+          const float ds = p * (dpc[nt][r] - del_r) * scale;
           Pt[trid3(key, qcol, QB3)] = f2bf(p);
           DSt[trid3(key, qcol, QB3)] = f2bf(ds);
         }
@@ -407,7 +415,11 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
       }
     }
 
-    // ---- dS -> wave-private LDS rows (no barrier needed) ---------------
+    // ---- dS -> wave-private LDS rows (no barrier needed; interior
+    // tiles skip the mask — PMC 8.9:1 VALU:MFMA) ----------------------
+    const bool interior =
+        (t0 + KB3 <= Lk) && (qt0 + wid * 16 + 16 <= Lq) &&
+        (!causal || t0 + KB3 - 1 <= qk_off + qt0 + wid * 16);
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int qrow_c = qt0 + wid * 16 + lhi * 4 + r;
@@ -418,10 +430,15 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
       const int qpos = qk_off + qrow_c;
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
-        const int kpos = t0 + nt * 16 + l15;
-        const bool dead = qdead || (kpos >= Lk) || (causal && kpos > qpos);
-        const float s = sc[nt][r] * scale;
-        const float p = dead ? 0.f : __expf(s - lse_r);
+        float p;
+        if (interior) {
+          p = __expf(sc[nt][r] * scale - lse_r);
+        } else {
+          const int kpos = t0 + nt * 16 + l15;
+          const bool dead = qdead || (kpos >= Lk) ||
+                            (causal && kpos > qpos);
+          p = dead ? 0.f : __expf(sc[nt][r] * scale - lse_r);
+        }
         const float ds = p * (dpc[nt][r] - del_r) * scale;
         DSq[trid3(wid * 16 + lhi * 4 + r, nt * 16 + l15, KB3)] = f2bf(ds);
       }

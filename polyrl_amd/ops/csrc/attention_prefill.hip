@@ -147,18 +147,30 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
     }
 
     // ---------------- scale + causal mask ---------------------------------
-    // C layout: row = lhi*4 + r, col = l15
+    // C layout: row = lhi*4 + r, col = l15.  Interior tiles (every (q, k)
+    // pair live: tile fully below the causal diagonal of the wave's
+    // earliest row, all rows/keys in range) skip the per-element mask —
+    // the kernels are VALU-bound (PMC: 16:1 VALU:MFMA), not memory-bound.
     float s[NKT][4];
+    const bool interior = (t0 + KVBLK <= Lk) && (wq0 + 16 <= Lq) &&
+                          (!causal || t0 + KVBLK - 1 <= qk_off + wq0);
+    if (interior) {
 #pragma unroll
-    for (int nt = 0; nt < NKT; ++nt) {
-      const int kpos = t0 + nt * 16 + l15;
+      for (int nt = 0; nt < NKT; ++nt)
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int qrow_l = wq0 + lhi * 4 + r;
-        const int qpos = qk_off + qrow_l;
-        const bool dead = (kpos >= Lk) || (qrow_l >= Lq) ||
-                          (causal && kpos > qpos);
-        s[nt][r] = dead ? -1e30f : c[nt][r] * scale;
+        for (int r = 0; r < 4; ++r) s[nt][r] = c[nt][r] * scale;
+    } else {
+#pragma unroll
+      for (int nt = 0; nt < NKT; ++nt) {
+        const int kpos = t0 + nt * 16 + l15;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int qrow_l = wq0 + lhi * 4 + r;
+          const int qpos = qk_off + qrow_l;
+          const bool dead = (kpos >= Lk) || (qrow_l >= Lq) ||
+                            (causal && kpos > qpos);
+          s[nt][r] = dead ? -1e30f : c[nt][r] * scale;
+        }
       }
     }
 
