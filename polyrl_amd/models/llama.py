@@ -231,7 +231,7 @@ class MLP(nn.Module):
         g = _lin(self.gate_proj, x)
         u = _lin(self.up_proj, x)
         if g.is_cuda and g.dtype == torch.bfloat16 \
-                and g.numel() % 8 == 0:
+                and g.shape[-1] % 8 == 0:
             import polyrl_amd.ops as pops
             return _lin(self.down_proj, pops.silu_mul_train(g, u))
         return _lin(self.down_proj, F.silu(g) * u)

@@ -330,12 +330,14 @@ class _SiluMulTrain(torch.autograd.Function):
     @staticmethod
     def forward(ctx, gate, up):
         ext = _require_ext()
-        g = gate.contiguous()
-        u = up.contiguous()
+        shp = gate.shape
+        g = gate.reshape(-1, shp[-1]).contiguous()
+        u = up.reshape(-1, shp[-1]).contiguous()
         ctx.save_for_backward(g, u)
+        ctx.shp = shp
         out = torch.empty_like(g)
         ext.silu_mul(out, g, u)
-        return out
+        return out.view(shp)
 
     @staticmethod
     def backward(ctx, dy):
@@ -343,8 +345,9 @@ class _SiluMulTrain(torch.autograd.Function):
         ext = _require_ext()
         dgate = torch.empty_like(g)
         dup = torch.empty_like(u)
-        ext.silu_mul_bwd(dgate, dup, dy.contiguous(), g, u)
-        return dgate, dup
+        ext.silu_mul_bwd(dgate, dup,
+                         dy.reshape(-1, dy.shape[-1]).contiguous(), g, u)
+        return dgate.view(ctx.shp), dup.view(ctx.shp)
 
 
 def silu_mul_train(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:

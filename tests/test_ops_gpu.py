@@ -878,3 +878,14 @@ def test_silu_mul_train_fwd_bwd():
     assert rel_err(y.float(), y2.detach()) < 1e-2
     assert rel_err(g.grad.float(), g2.grad) < 2e-2
     assert rel_err(u.grad.float(), u2.grad) < 2e-2
+
+
+def test_silu_mul_train_3d():
+    """Padded-layout (B, L, I) inputs flatten inside the Function."""
+    torch.manual_seed(60)
+    g = torch.randn(3, 17, 344, device=DEV).bfloat16().requires_grad_()
+    u = torch.randn_like(g).requires_grad_()
+    y = ops.silu_mul_train(g, u)
+    assert y.shape == g.shape
+    y.float().sum().backward()
+    assert g.grad.shape == g.shape and torch.isfinite(g.grad.float()).all()
