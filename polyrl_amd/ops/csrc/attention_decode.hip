@@ -23,18 +23,29 @@
 template <int D, int G>
 __global__ __launch_bounds__(256) void paged_decode_attn_kernel(
     bf16_t* __restrict__ out,              // (B, Hq, D)
+    float* __restrict__ partial,           // (B, Hq, S, D+2) when nsplit>1
     const bf16_t* __restrict__ q,          // (B, Hq, D)
     const bf16_t* __restrict__ k_cache,    // (num_pages, page_size, Hk, D)
     const bf16_t* __restrict__ v_cache,
     const int* __restrict__ page_table,    // (B, max_pages)
     const int* __restrict__ context_lens,  // (B,)
-    int Hq, int Hk, int page_size, int max_pages, float scale, long ldq) {
+    int Hq, int Hk, int page_size, int max_pages, float scale, long ldq,
+    int nsplit) {
   constexpr int GL = D / 8;        // lanes per key group
   constexpr int KPW = 64 / GL;     // keys per wave per iteration
   constexpr int NW = 4;            // waves per block
-  const int b = blockIdx.x / Hk;
-  const int hk = blockIdx.x % Hk;
+  // flash-decoding context split (PMC-measured: at B*Hk < ~1024 blocks
+  // the kernel is occupancy/latency-bound — 1 wave/SIMD cannot hide the
+  // online-update chain; splitting the context across nsplit blocks per
+  // (b, hk) restores the wave supply, a tiny merge kernel combines)
+  const int b = blockIdx.x / (Hk * nsplit);
+  const int rem = blockIdx.x % (Hk * nsplit);
+  const int hk = rem / nsplit;
+  const int sp = rem % nsplit;
   const int L = context_lens[b];
+  const int Lc = (L + nsplit - 1) / nsplit;
+  const int cbeg = sp * Lc;
+  const int cend = (cbeg + Lc < L) ? (cbeg + Lc) : L;
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
   const int grp = lane / GL;       // key-group within wave
@@ -67,8 +78,8 @@ __global__ __launch_bounds__(256) void paged_decode_attn_kernel(
   const int kv_stride = Hk * D;  // row stride inside a page slot
   const int* ptab = page_table + (long)b * max_pages;
 
-  // stream keys: token t = iter*NW*KPW + wid*KPW + grp
-  for (int t0 = wid * KPW + grp; t0 < L; t0 += NW * KPW) {
+  // stream keys: token t = cbeg + iter*NW*KPW + wid*KPW + grp
+  for (int t0 = cbeg + wid * KPW + grp; t0 < cend; t0 += NW * KPW) {
     const int page = ptab[t0 / page_size];
     const long slot = (long)page * page_size + (t0 % page_size);
     const bf16_t* kp = k_cache + slot * kv_stride + hk * D + d0;
@@ -150,13 +161,56 @@ __global__ __launch_bounds__(256) void paged_decode_attn_kernel(
 #pragma unroll
         for (int j = 0; j < 8; ++j) go[j] += mrg[w][h][gl][j] * a;
       }
-      const float inv = (gl_sum > 0.f) ? 1.0f / gl_sum : 0.f;
-      bf16x8 ov;
+      if (nsplit == 1) {
+        const float inv = (gl_sum > 0.f) ? 1.0f / gl_sum : 0.f;
+        bf16x8 ov;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) ov.v[j] = f2bf(go[j] * inv);
-      *reinterpret_cast<bf16x8*>(out + (((long)b * Hq + hk * G + h) * D + d0)) = ov;
+        for (int j = 0; j < 8; ++j) ov.v[j] = f2bf(go[j] * inv);
+        *reinterpret_cast<bf16x8*>(
+            out + (((long)b * Hq + hk * G + h) * D + d0)) = ov;
+      } else {
+        float* pp = partial +
+            ((((long)b * Hq + hk * G + h) * nsplit) + sp) * (D + 2);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) pp[d0 + j] = go[j];
+        if (gl == 0) {
+          pp[D] = gm;
+          pp[D + 1] = gl_sum;
+        }
+      }
     }
   }
+}
+
+// combine the nsplit context partials of one (b, hq): sum-exp merge
+template <int D>
+__global__ void paged_decode_merge_kernel(
+    bf16_t* __restrict__ out,              // (B, Hq, D)
+    const float* __restrict__ partial,     // (B*Hq, S, D+2)
+    int nsplit) {
+  constexpr int EPL = D / 64;              // d-elems per lane (64 threads)
+  const long row = blockIdx.x;             // b*Hq + hq
+  const int lane = threadIdx.x;
+  const float* base = partial + row * (long)nsplit * (D + 2);
+  float gm = -1e30f;
+  for (int sp = 0; sp < nsplit; ++sp)
+    gm = fmaxf(gm, base[sp * (D + 2) + D]);
+  float gl_sum = 0.f;
+  float acc[EPL];
+#pragma unroll
+  for (int j = 0; j < EPL; ++j) acc[j] = 0.f;
+  for (int sp = 0; sp < nsplit; ++sp) {
+    const float* pp = base + sp * (D + 2);
+    const float a = __expf(pp[D] - gm);
+    gl_sum += pp[D + 1] * a;
+#pragma unroll
+    for (int j = 0; j < EPL; ++j)
+      acc[j] += pp[lane + j * 64] * a;
+  }
+  const float inv = (gl_sum > 0.f) ? 1.0f / gl_sum : 0.f;
+#pragma unroll
+  for (int j = 0; j < EPL; ++j)
+    out[row * D + lane + j * 64] = f2bf(acc[j] * inv);
 }
 
 template <int D>
@@ -167,13 +221,34 @@ static void dispatch_g(torch::Tensor& out, const torch::Tensor& q,
                        int page_size, int max_pages, float scale,
                        hipStream_t stream) {
   const int G = Hq / Hk;
-  const dim3 grid(q.size(0) * Hk), block(256);
+  const long B = q.size(0);
+  // flash-decoding context split: pick nsplit from the grid size alone
+  // (L-independent => hipGraph-replay-safe).  Target >= 1024 blocks so
+  // the latency-bound online-update chain has wave supply; PMC measured
+  // B=32 x Hk=8 (256 blocks) at 7x the KV-read bound from occupancy.
+  int nsplit = 1;
+  const long base_blocks = B * Hk;
+  if (base_blocks < 1024) {
+    nsplit = (int)((1024 + base_blocks - 1) / base_blocks);
+    if (nsplit > 16) nsplit = 16;
+  }
+  torch::Tensor partial;
+  float* ppart = nullptr;
+  if (nsplit > 1) {
+    partial = torch::empty({B * Hq * (long)nsplit * (D + 2)},
+                           q.options().dtype(torch::kFloat32));
+    ppart = partial.data_ptr<float>();
+  }
+  const dim3 grid(B * Hk * nsplit), block(256);
   auto args = [&](auto kern) {
     kern<<<grid, block, 0, stream>>>(
-        (bf16_t*)out.data_ptr(), (const bf16_t*)q.data_ptr(),
+        (bf16_t*)out.data_ptr(), ppart, (const bf16_t*)q.data_ptr(),
         (const bf16_t*)k_cache.data_ptr(), (const bf16_t*)v_cache.data_ptr(),
         page_table.data_ptr<int>(), context_lens.data_ptr<int>(), Hq, Hk,
-        page_size, max_pages, scale, q.stride(0));
+        page_size, max_pages, scale, q.stride(0), nsplit);
+    if (nsplit > 1)
+      paged_decode_merge_kernel<D><<<dim3(B * Hq), dim3(64), 0, stream>>>(
+          (bf16_t*)out.data_ptr(), ppart, nsplit);
   };
   switch (G) {
     case 1: args(paged_decode_attn_kernel<D, 1>); break;
