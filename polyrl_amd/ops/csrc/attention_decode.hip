@@ -232,12 +232,22 @@ static void dispatch_g(torch::Tensor& out, const torch::Tensor& q,
     nsplit = (int)((1024 + base_blocks - 1) / base_blocks);
     if (nsplit > 16) nsplit = 16;
   }
-  torch::Tensor partial;
   float* ppart = nullptr;
   if (nsplit > 1) {
-    partial = torch::empty({B * Hq * (long)nsplit * (D + 2)},
-                           q.options().dtype(torch::kFloat32));
-    ppart = partial.data_ptr<float>();
+    // Thread-local grow-only workspace: a fresh torch::empty every call
+    // would allocate DURING hipGraph capture (the engine's graphed decode
+    // wraps this op) and can deadlock/fail the capture.  The warmup pass
+    // that precedes every capture (rollout/engine.py::_decode_graphed)
+    // sizes this buffer on the same thread, so the captured call reuses a
+    // stable pointer and replays are correct (partials are written then
+    // merged within one launch sequence).
+    static thread_local torch::Tensor ws;
+    const long need = B * Hq * (long)nsplit * (D + 2);
+    if (!ws.defined() || ws.numel() < need ||
+        ws.device() != q.device()) {
+      ws = torch::empty({need}, q.options().dtype(torch::kFloat32));
+    }
+    ppart = ws.data_ptr<float>();
   }
   const dim3 grid(B * Hk * nsplit), block(256);
   auto args = [&](auto kern) {
