@@ -22,6 +22,12 @@ class EngineRunner:
     def __init__(self, engine: Engine):
         self.engine = engine
         self.lock = threading.Lock()        # step / weight-update mutex
+        # Python locks are not FIFO: the pump re-acquires within
+        # microseconds of releasing, so an abort / weight-update / submit
+        # caller can starve for an ENTIRE generation (measured: time-box
+        # aborts landing after the final token).  Callers raise this count
+        # before acquiring; the pump yields between steps while it is set.
+        self._lock_waiters = 0
         self._futures: Dict[str, tuple] = {}  # rid -> (loop, future)
         self._pump: Optional[threading.Thread] = None
         self._stop = threading.Event()
@@ -52,6 +58,8 @@ class EngineRunner:
                 self._wake.wait(timeout=0.02)
                 self._wake.clear()
                 continue
+            while self._lock_waiters > 0:       # yield to admin callers
+                time.sleep(0.0005)
             with self.lock:
                 outs = self.engine.step()
             if outs:
@@ -87,7 +95,7 @@ class EngineRunner:
             rid = f"r{self._rid_counter}"
         fut = loop.create_future()
         self._futures[rid] = (loop, fut)
-        with self.lock:
+        with self._admin_lock():
             self.engine.add_request(rid, input_ids, sp)
         self.start()
         self._wake.set()
@@ -107,7 +115,7 @@ class EngineRunner:
             fut = loop.create_future()
             self._futures[f"{prefix}-s{s_}"] = (loop, fut)
             futs.append(fut)
-        with self.lock:
+        with self._admin_lock():
             self.engine.add_request_group(prefix, input_ids, sp, n)
         self.start()
         self._wake.set()
@@ -119,8 +127,22 @@ class EngineRunner:
                                                             n)))
 
     # ---------------------------------------------------------------- admin
+    def _admin_lock(self):
+        """Priority acquisition for non-pump callers (see _lock_waiters)."""
+        import contextlib
+
+        @contextlib.contextmanager
+        def cm():
+            self._lock_waiters += 1
+            try:
+                with self.lock:
+                    yield
+            finally:
+                self._lock_waiters -= 1
+        return cm()
+
     def abort(self, rid: Optional[str] = None, abort_all: bool = False):
-        with self.lock:
+        with self._admin_lock():
             self.engine.abort_request(rid=rid, abort_all=abort_all)
         self._wake.set()
 
@@ -139,7 +161,7 @@ class EngineRunner:
         than resumed on different weights; the scheduler continues them
         token-exactly on the new version.  Co-located publication happens
         between steps (no in-flight), so this is a no-op there."""
-        with self.lock:
+        with self._admin_lock():
             self.engine.model.load_state_dict(state_dict, strict=strict)
             if abort_in_flight and self.engine.has_work():
                 self.engine.abort_request(abort_all=True)

@@ -126,7 +126,14 @@ def create_manager_app(scheduler, loop: Optional[asyncio.AbstractEventLoop]
                 out_q.put(None)
 
         if loop is None or loop is asyncio.get_running_loop():
-            asyncio.ensure_future(pump())
+            # keep a strong ref: asyncio tasks are weakly referenced and
+            # an unreferenced pump can be garbage-collected mid-stream
+            t = asyncio.ensure_future(pump())
+            _bg = getattr(app.state, "_bg_tasks", None)
+            if _bg is None:
+                _bg = app.state._bg_tasks = set()
+            _bg.add(t)
+            t.add_done_callback(_bg.discard)
         else:
             asyncio.run_coroutine_threadsafe(pump(), loop)
 

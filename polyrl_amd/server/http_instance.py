@@ -28,6 +28,7 @@ class HttpInstance(RolloutInstance):
                                          base_url=self.base_url,
                                          timeout=timeout)
         self._stats = InstanceStats()
+        self._bg_tasks = set()          # strong refs for fire-and-forget
         # weight delivery (checked in order):
         #  * weight_state_fn: version -> state dict, pushed over the
         #    N-stream TCP plane (sender-agent path, tcp_engine.py)
@@ -148,7 +149,11 @@ class HttpInstance(RolloutInstance):
             return push_state_dict_tcp(sd, c, host, version=version)
 
     def abort_all(self):
-        # fire-and-forget from sync context
+        # fire-and-forget from sync context.  The task MUST be strongly
+        # referenced: asyncio holds only weak refs to tasks, and an
+        # unreferenced abort task can be garbage-collected before it ever
+        # POSTs (observed intermittently: time-box aborts silently lost,
+        # the local engine ran to completion with zero migrations).
         async def _abort():
             try:
                 await self._client.post("/abort_request",
@@ -157,7 +162,9 @@ class HttpInstance(RolloutInstance):
                 pass
         try:
             loop = asyncio.get_running_loop()
-            loop.create_task(_abort())
+            t = loop.create_task(_abort())
+            self._bg_tasks.add(t)
+            t.add_done_callback(self._bg_tasks.discard)
         except RuntimeError:
             asyncio.run(_abort())
 

@@ -322,3 +322,92 @@ def test_sender_cidr_allowlist():
             r = await c.get("/health")
             assert r.status_code == 200
     asyncio.run(go())
+
+
+def test_timebox_abort_lands_mid_generation():
+    """Regression (round-2): the runner pump re-acquired the step lock
+    within microseconds, starving abort/update/submit callers for an
+    entire generation (Python locks are not FIFO) — time-box aborts
+    silently landed AFTER the final token and no migration happened.
+    Real uvicorn servers + HTTP instances + scheduler time-box on CPU."""
+    import threading
+    import time as _time
+
+    import uvicorn
+
+    from polyrl_amd.models import create_model
+    from polyrl_amd.models.registry import DecoderConfig
+    from polyrl_amd.rollout.engine import Engine
+    from polyrl_amd.rollout.runner import EngineRunner
+    from polyrl_amd.scheduler.manager import (RolloutScheduler,
+                                              SchedulerConfig)
+    from polyrl_amd.scheduler.types import GroupRequest, SamplingSpec
+    from polyrl_amd.server import create_app
+
+    cfg = get_model_config("llama-debug-cpu")
+    torch.manual_seed(60)
+    model = create_model(cfg, kind="actor", dtype="float32", device="cpu")
+    servers = []
+    ports = (32230, 32231)
+    engines = []
+    for i, port in enumerate(ports):
+        eng = Engine(cfg, device="cpu", dtype=torch.float32,
+                     kv_bytes_budget=16 << 20, decode_chunk_size=4)
+        eng.model.load_state_dict(model.state_dict())
+        if i == 0:
+            rs = eng.step
+
+            def slow(*a, _rs=rs, **kw):
+                _time.sleep(0.03)
+                return _rs(*a, **kw)
+            eng.step = slow
+        runner = EngineRunner(eng)
+        app = create_app(eng, runner)
+        server = uvicorn.Server(uvicorn.Config(app, host="127.0.0.1",
+                                               port=port, log_level="error"))
+        threading.Thread(target=server.run, daemon=True).start()
+        servers.append(server)
+        engines.append(eng)
+    import httpx
+    for port in ports:
+        for _ in range(100):
+            try:
+                if httpx.get(f"http://127.0.0.1:{port}/health",
+                             timeout=1.0).status_code == 200:
+                    break
+            except Exception:
+                _time.sleep(0.05)
+
+    async def go():
+        local = HttpInstance(f"http://127.0.0.1:{ports[0]}",
+                             instance_id="local-slow", is_local=True)
+        remote = HttpInstance(f"http://127.0.0.1:{ports[1]}",
+                              instance_id="remote-fast", is_local=False)
+        sched = RolloutScheduler(SchedulerConfig(stats_interval_s=0.05))
+        await sched.register_instance(local, skip_health_check=True)
+        await sched.register_instance(remote, skip_health_check=True)
+        sched._active.remove("remote-fast")
+
+        async def readd():
+            await asyncio.sleep(0.1)
+            if "remote-fast" not in sched._active:
+                sched._active.append("remote-fast")
+        task = asyncio.get_running_loop().create_task(readd())
+        req = GroupRequest(gid=0, input_ids=[3, 4, 5], n=1,
+                           sampling=SamplingSpec(max_new_tokens=64,
+                                                 temperature=0.0))
+        items = []
+        async for it in sched.submit_batch([req], max_local_gen_s=0.25):
+            items.append(it)
+        await task
+        await sched.close()
+        return items
+
+    items = asyncio.run(go())
+    s = items[1].samples[0]
+    for srv in servers:
+        srv.should_exit = True
+    assert len(s.output_ids) == 64
+    assert s.num_migrations >= 1, \
+        "time-box abort must interrupt the slow local generation"
+    assert "remote-fast" in items[1].instance_ids
