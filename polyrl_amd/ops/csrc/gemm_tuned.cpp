@@ -41,7 +41,6 @@ constexpr int kTimedIters = 3;
 
 struct LtState {
   hipblasLtHandle_t handle = nullptr;
-  void* workspace = nullptr;
   std::mutex mu;
   // key: (mode, M, N, K) -> winning algo
   std::map<std::tuple<int, int64_t, int64_t, int64_t>, hipblasLtMatmulAlgo_t>
@@ -51,12 +50,19 @@ struct LtState {
 LtState& lt() {
   static LtState s;
   static std::once_flag once;
-  std::call_once(once, [] {
-    HIPBLASLT_CHECK(hipblasLtCreate(&s.handle));
-    TORCH_CHECK(hipMalloc(&s.workspace, kWorkspaceBytes) == hipSuccess,
-                "hipblaslt workspace alloc failed");
-  });
+  std::call_once(once, [] { HIPBLASLT_CHECK(hipblasLtCreate(&s.handle)); });
   return s;
+}
+
+// Workspace is per-THREAD: each thread in this stack owns its stream
+// (trainer main, engine pump, uvicorn handlers), and a shared workspace
+// would be written concurrently by matmuls running on different streams.
+void* tl_workspace() {
+  static thread_local void* ws = nullptr;
+  if (ws == nullptr)
+    TORCH_CHECK(hipMalloc(&ws, kWorkspaceBytes) == hipSuccess,
+                "hipblaslt workspace alloc failed");
+  return ws;
 }
 
 struct Plan {
@@ -93,6 +99,7 @@ void build_plan(Plan& p, hipblasOperation_t opA, hipblasOperation_t opB,
 void tuned_mm(int mode, const torch::Tensor& A_row, const torch::Tensor& B_row,
               torch::Tensor& C_row, int64_t M, int64_t N, int64_t K) {
   auto& s = lt();
+  void* workspace = tl_workspace();
   auto stream = at::hip::getCurrentHIPStream();
 
   hipblasOperation_t opA, opB;
@@ -155,13 +162,13 @@ void tuned_mm(int mode, const torch::Tensor& A_row, const torch::Tensor& B_row,
       // warmup (also validates the algo actually runs)
       hipblasStatus_t st = hipblasLtMatmul(
           s.handle, p.op, &alpha, Aptr, p.la, Bptr, p.lb, &beta, Cptr, p.lc,
-          Cptr, p.lc, &results[i].algo, s.workspace, kWorkspaceBytes, stream);
+          Cptr, p.lc, &results[i].algo, workspace, kWorkspaceBytes, stream);
       if (st != HIPBLAS_STATUS_SUCCESS) continue;
       hipEventRecord(ev0, stream);
       for (int it2 = 0; it2 < kTimedIters; ++it2)
         hipblasLtMatmul(s.handle, p.op, &alpha, Aptr, p.la, Bptr, p.lb,
                         &beta, Cptr, p.lc, Cptr, p.lc, &results[i].algo,
-                        s.workspace, kWorkspaceBytes, stream);
+                        workspace, kWorkspaceBytes, stream);
       hipEventRecord(ev1, stream);
       hipEventSynchronize(ev1);
       float ms = 1e30f;
@@ -184,7 +191,7 @@ void tuned_mm(int mode, const torch::Tensor& A_row, const torch::Tensor& B_row,
 
   HIPBLASLT_CHECK(hipblasLtMatmul(
       s.handle, p.op, &alpha, Aptr, p.la, Bptr, p.lb, &beta, Cptr, p.lc,
-      Cptr, p.lc, &algo, s.workspace, kWorkspaceBytes, stream));
+      Cptr, p.lc, &algo, workspace, kWorkspaceBytes, stream));
 }
 
 void check2d(const torch::Tensor& t, const char* name) {
