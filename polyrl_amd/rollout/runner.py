@@ -28,6 +28,7 @@ class EngineRunner:
         # aborts landing after the final token).  Callers raise this count
         # before acquiring; the pump yields between steps while it is set.
         self._lock_waiters = 0
+        self._waiters_mu = threading.Lock()
         self._futures: Dict[str, tuple] = {}  # rid -> (loop, future)
         self._pump: Optional[threading.Thread] = None
         self._stop = threading.Event()
@@ -133,12 +134,14 @@ class EngineRunner:
 
         @contextlib.contextmanager
         def cm():
-            self._lock_waiters += 1
+            with self._waiters_mu:
+                self._lock_waiters += 1
             try:
                 with self.lock:
                     yield
             finally:
-                self._lock_waiters -= 1
+                with self._waiters_mu:
+                    self._lock_waiters -= 1
         return cm()
 
     def abort(self, rid: Optional[str] = None, abort_all: bool = False):
