@@ -117,6 +117,8 @@ class MultiTurnConfig:
     max_assistant_turns: int = 2
     max_user_turns: int = 8
     max_tool_response_length: int = 256
+    # cap tokens per assistant turn (0 = until EOS / response budget)
+    per_turn_max_tokens: int = 0
     # python file + fn implementing the interaction:
     #   fn(prompt_ids: list[int], response_ids: list[int]) ->
     #       (user_ids: list[int] | None, done: bool)

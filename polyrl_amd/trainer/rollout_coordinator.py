@@ -102,7 +102,8 @@ def load_interaction(mt_cfg) -> Optional[dict]:
     return {"interaction": fn,
             "max_assistant_turns": mt_cfg.max_assistant_turns,
             "max_user_turns": mt_cfg.max_user_turns,
-            "max_tool_response_length": mt_cfg.max_tool_response_length}
+            "max_tool_response_length": mt_cfg.max_tool_response_length,
+            "per_turn_max_tokens": getattr(mt_cfg, "per_turn_max_tokens", 0)}
 
 
 class LocalRolloutCoordinator:
@@ -211,7 +212,15 @@ class LocalRolloutCoordinator:
                 "n": n,
             }
             # shared prompt prefill + shared full KV pages for the group
-            self.engine.add_request_group(gid, raw, sampling, n)
+            samp = sampling
+            if self.multi_turn is not None and \
+                    self.multi_turn.get("per_turn_max_tokens", 0) > 0:
+                import copy as _copy
+                samp = _copy.copy(sampling)
+                samp.max_new_tokens = min(
+                    samp.max_new_tokens,
+                    self.multi_turn["per_turn_max_tokens"])
+            self.engine.add_request_group(gid, raw, samp, n)
 
     # --------------------------------------------------------------- stream
     def pending_groups(self) -> int:
@@ -285,6 +294,9 @@ class LocalRolloutCoordinator:
         import copy
         samp = copy.copy(grp["sampling"])
         samp.max_new_tokens = self.response_length - len(st["ids"])
+        cap = mt.get("per_turn_max_tokens", 0)
+        if cap > 0:
+            samp.max_new_tokens = min(samp.max_new_tokens, cap)
         self.engine.add_request(out.rid, grp["raw"] + st["ids"], samp)
         return None
 

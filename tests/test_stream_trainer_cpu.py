@@ -387,3 +387,38 @@ def test_async_decode_matches_serial(tmp_path, monkeypatch):
     assert torch.equal(lp_async, lp_sync)
     for k in w_sync:
         assert torch.allclose(w_async[k], w_sync[k], atol=1e-6), k
+
+
+def test_async_decode_with_multi_turn(tmp_path, monkeypatch):
+    """The decode pump thread performs the multi-turn interaction calls and
+    resubmissions; results must match the serial path exactly."""
+    inter = tmp_path / "interaction.py"
+    inter.write_text(
+        "def generate_turn(prompt_ids, response_ids):\n"
+        "    if len(response_ids) >= 9:\n"
+        "        return None, True\n"
+        "    return [5, 6], False\n")
+
+    def run(async_on):
+        monkeypatch.setenv("POLYRL_ASYNC_DECODE", "1" if async_on else "0")
+        cfg = tiny_config(tmp_path / ("a2" if async_on else "s2"))
+        cfg.actor_rollout_ref.rollout.response_length = 16
+        mt = cfg.actor_rollout_ref.rollout.multi_turn
+        mt.enable = True
+        mt.interaction_path = str(inter)
+        mt.max_assistant_turns = 3
+        mt.per_turn_max_tokens = 5
+        trainer = StreamPPOTrainer(cfg,
+                                   reward_fn=load_reward_manager("random"))
+        trainer.fit(max_steps=1)
+        f = trainer._last_full_batch
+        return (f["responses"].clone(), f["response_mask"].clone(),
+                f["attention_mask"].clone())
+
+    r_a, m_a, am_a = run(True)
+    r_s, m_s, am_s = run(False)
+    assert torch.equal(r_a, r_s)
+    assert torch.equal(m_a, m_s)
+    assert torch.equal(am_a, am_s)
+    # multi-turn really engaged: some present tokens carry no loss
+    assert int(am_a[:, -16:].sum()) > int(m_a.sum())
