@@ -411,3 +411,49 @@ def test_timebox_abort_lands_mid_generation():
     assert s.num_migrations >= 1, \
         "time-box abort must interrupt the slow local generation"
     assert "remote-fast" in items[1].instance_ids
+
+
+def test_submit_admitted_mid_generation():
+    """Continuous batching: a request submitted while the engine decodes a
+    long one must be admitted promptly (same starvation class as the
+    time-box fix — submits also need the step lock)."""
+    import time as _time
+
+    from polyrl_amd.models import create_model
+    from polyrl_amd.rollout.engine import Engine
+    from polyrl_amd.rollout.runner import EngineRunner
+
+    cfg = get_model_config("llama-debug-cpu")
+    torch.manual_seed(3)
+    model = create_model(cfg, kind="actor", dtype="float32", device="cpu")
+    eng = Engine(cfg, device="cpu", dtype=torch.float32,
+                 kv_bytes_budget=16 << 20, decode_chunk_size=4)
+    eng.model.load_state_dict(model.state_dict())
+    rs = eng.step
+
+    def slow(*a, _rs=rs, **kw):
+        _time.sleep(0.02)
+        return _rs(*a, **kw)
+    eng.step = slow
+    runner = EngineRunner(eng)
+
+    async def go():
+        f1 = runner.submit([1, 2, 3], SamplingParams(temperature=0.0,
+                                                     max_new_tokens=64))
+        await asyncio.sleep(0.15)          # first request mid-generation
+        t0 = _time.monotonic()
+        f2 = runner.submit([4, 5, 6], SamplingParams(temperature=0.0,
+                                                     max_new_tokens=4))
+        o2 = await asyncio.wait_for(f2, timeout=30.0)
+        short_wait = _time.monotonic() - t0
+        o1 = await asyncio.wait_for(f1, timeout=60.0)
+        return o1, o2, short_wait
+
+    o1, o2, short_wait = asyncio.run(go())
+    runner.stop()
+    assert len(o1.output_ids) == 64
+    assert len(o2.output_ids) == 4
+    # the short request must NOT have waited for the long one: 4 tokens at
+    # ~0.02s/4-token chunk ~= a few steps, far below the long request's
+    # remaining ~1s
+    assert short_wait < 0.6, short_wait
