@@ -194,14 +194,14 @@ def create_app(engine: Engine, runner: Optional[EngineRunner] = None,
 
     @app.post("/weights_handshake")
     async def weights_handshake(request: Request):
-        if not _sender_ok(request):
-            return JSONResponse({"success": False,
-                                 "message": "sender not in allowed CIDRs"},
-                                403)
         """Arm a TCP bulk receive (the reference's receiver-agent bootstrap,
         receiver_agent.py:184-240): allocate/reuse a registered CPU buffer
         sized to the incoming state dict, listen on N stream ports, return
         them to the sender."""
+        if not _sender_ok(request):
+            return JSONResponse({"success": False,
+                                 "message": "sender not in allowed CIDRs"},
+                                403)
         import torch
 
         from ..transfer.tcp_engine import TcpWeightReceiver
