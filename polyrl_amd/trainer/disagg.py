@@ -192,7 +192,7 @@ class DisaggCoordinator:
                  max_local_gen_s: float = 0.0,
                  manager_port: int = 0,
                  remote_weight_state_fn=None,
-                 sched_cfg=None):
+                 sched_cfg=None, multi_turn=None):
         """rollout_urls entries are either plain url strings (remote) or
         (url, is_local) tuples.  ``manager_port`` > 0 serves the manager
         HTTP facade (scheduler/http_api.py) on rank 0 so elastic remote
@@ -205,6 +205,10 @@ class DisaggCoordinator:
         self.pad = pad_token_id
         self.device = device
         self.max_local_gen_s = max_local_gen_s
+        # multi-turn through the scheduler path: dict from
+        # rollout_coordinator.load_interaction (reference MultiTurnConfig)
+        self.multi_turn = multi_turn
+        self._sampling_spec = None
         self._iter: Optional[StreamingBatchIterator] = None
         self._meta: Dict[int, dict] = {}
         self._gid = 0
@@ -257,14 +261,15 @@ class DisaggCoordinator:
         for g in range(ids.shape[0]):
             gid = self._gid
             self._gid += 1
+            raw = ids[g][mask[g].bool()].tolist()
             self._meta[gid] = {
                 "uid": str(uids[g]),
                 "extras": {k: v[g] for k, v in prompts.non_tensors.items()
                            if k != "uid"},
                 "prompt_ids": ids[g].cpu(),
                 "prompt_mask": mask[g].cpu(),
+                "raw": raw,
             }
-            raw = ids[g][mask[g].bool()].tolist()
             reqs.append(GroupRequest(
                 gid=gid, input_ids=raw, n=n,
                 sampling=SamplingSpec(
@@ -272,6 +277,7 @@ class DisaggCoordinator:
                     top_k=sampling.top_k, top_p=sampling.top_p,
                     max_new_tokens=sampling.max_new_tokens,
                     stop_token_ids=tuple(sampling.stop_token_ids))))
+        self._sampling_spec = reqs[0].sampling if reqs else None
         self._iter = StreamingBatchIterator(
             self.scheduler, reqs,
             max_local_gen_s=self.max_local_gen_s, loop=self.loop)
@@ -318,6 +324,8 @@ class DisaggCoordinator:
                 continue
             res = item
             meta = self._meta.pop(res.gid)
+            if self.multi_turn is not None:
+                self._run_multi_turns(meta["raw"], res)
             groups.append((meta, res))
             got += len(res.samples)
             if got >= need:
@@ -355,6 +363,59 @@ class DisaggCoordinator:
             out.append(self._make_batch(take))
         return [out]
 
+    def _run_multi_turns(self, raw, res):
+        """Scheduler-path multi-turn (reference MultiTurnConfig capability):
+        after each finished assistant turn, ask the interaction for the
+        next user turn and continue the sample through the scheduler (with
+        its full token-level fault tolerance); user tokens carry no loss."""
+        from dataclasses import replace as _replace
+        mt = self.multi_turn
+        for s in res.samples:
+            ids = list(s.output_ids)
+            lps = list(s.output_logprobs)
+            loss = [1] * len(ids)
+            a_turns, u_turns = 1, 0
+            while (s.finish_reason not in ("abort", "error")
+                   and a_turns < mt["max_assistant_turns"]
+                   and u_turns < mt["max_user_turns"]):
+                budget = self.response_length - len(ids)
+                if budget <= 1:
+                    break
+                user_ids, done = mt["interaction"](list(raw), list(ids))
+                if done or not user_ids:
+                    break
+                user_ids = list(user_ids)[
+                    :mt.get("max_tool_response_length", 256)]
+                if len(user_ids) >= budget:
+                    break
+                ids += user_ids
+                lps += [0.0] * len(user_ids)
+                loss += [0] * len(user_ids)
+                u_turns += 1
+                mx = self.response_length - len(ids)
+                cap = mt.get("per_turn_max_tokens", 0)
+                if cap > 0:
+                    mx = min(mx, cap)
+                cont = GroupRequest(
+                    gid=res.gid, input_ids=list(raw) + ids, n=1,
+                    sampling=_replace(self._sampling_spec,
+                                      max_new_tokens=mx))
+                r2 = asyncio.run_coroutine_threadsafe(
+                    self.scheduler.process_group(cont),
+                    self.loop).result(timeout=600)
+                s2 = r2.samples[0]
+                ids += s2.output_ids
+                lps += s2.output_logprobs
+                loss += [1] * len(s2.output_ids)
+                s.finish_reason = s2.finish_reason
+                s.num_migrations += s2.num_migrations
+                res.instance_ids.extend(r2.instance_ids)
+                a_turns += 1
+            Lr = self.response_length
+            s.output_ids = ids[:Lr]
+            s.output_logprobs = lps[:Lr]
+            s.loss_mask = loss[:Lr]
+
     def _make_batch(self, pairs) -> TensorBatch:
         prompt_ids = torch.stack([m["prompt_ids"] for m, _ in pairs])
         prompt_mask = torch.stack([m["prompt_mask"] for m, _ in pairs])
@@ -365,6 +426,7 @@ class DisaggCoordinator:
                 self.output_ids = s.output_ids
                 self.output_logprobs = s.output_logprobs
                 self.finish_reason = s.finish_reason
+                self.loss_mask = getattr(s, "loss_mask", None)
         outputs = [[_O(s) for s in res.samples] for _, res in pairs]
         out = postprocess_groups(prompt_ids, prompt_mask, uids, outputs,
                                  self.response_length, self.pad,

@@ -170,11 +170,13 @@ class StreamPPOTrainer:
                     f"{rollout_port(r, ro.rollout_port_base)}"
                     for r in rollout_ranks]
             self.engine = None
+            from .rollout_coordinator import load_interaction
             self.coordinator = DisaggCoordinator(
                 ro.response_length, self.pg, urls, rank=self.rank,
                 n_trainer=self.world, pad_token_id=0, device="cpu",
                 max_local_gen_s=ro.max_local_gen_s,
-                sched_cfg=self.config.scheduler)
+                sched_cfg=self.config.scheduler,
+                multi_turn=load_interaction(ro.multi_turn))
             self.publisher = DisaggPublisher(
                 self.actor.model, device,
                 self.coordinator.scheduler, self.coordinator.loop,
@@ -291,13 +293,15 @@ class StreamPPOTrainer:
             dist.barrier(group=self.pg)
         specs = [(f"http://127.0.0.1:{rollout_port(r, ro.rollout_port_base)}",
                   True) for r in range(gworld)]
+        from .rollout_coordinator import load_interaction
         self.coordinator = DisaggCoordinator(
             ro.response_length, self.pg, specs, rank=self.rank,
             n_trainer=self.world, pad_token_id=0, device="cpu",
             max_local_gen_s=ro.max_local_gen_s,
             manager_port=ro.rollout_manager_port if self.rank == 0 else 0,
             remote_weight_state_fn=lambda v: self.publisher.snapshot_cache(),
-            sched_cfg=self.config.scheduler)
+            sched_cfg=self.config.scheduler,
+            multi_turn=load_interaction(ro.multi_turn))
         self.publisher = ElasticPublisher(
             self.actor.model, self.engine.model, self.coordinator,
             tie=model_cfg.tie_word_embeddings, trainer_group=self.pg)

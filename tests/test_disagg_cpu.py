@@ -211,3 +211,52 @@ def test_disagg_with_ulysses_sp2(tmp_path):
     assert r.returncode == 0, \
         f"stdout:\n{r.stdout[-4000:]}\nstderr:\n{r.stderr[-4000:]}"
     assert "fit done" in r.stdout
+
+
+def test_scheduler_path_multi_turn():
+    """Multi-turn through the DisaggCoordinator/scheduler path: the
+    interaction's user turns are inserted with loss 0 and the assistant
+    continuations run through the scheduler (unit-level, FakeInstance)."""
+    import asyncio
+    import time
+
+    import numpy as np
+    import torch
+
+    from polyrl_amd.protocol import TensorBatch
+    from polyrl_amd.rollout.engine import SamplingParams
+    from polyrl_amd.scheduler.instances import FakeInstance
+    from polyrl_amd.trainer.disagg import DisaggCoordinator
+
+    mt = {"interaction":
+          lambda prompt, resp: (None, True) if len(resp) >= 10
+          else ([9, 9], False),
+          "max_assistant_turns": 3, "max_user_turns": 5,
+          "max_tool_response_length": 8, "per_turn_max_tokens": 4}
+    coord = DisaggCoordinator(response_length=16, trainer_group=None,
+                              rollout_urls=[], rank=0, n_trainer=1,
+                              multi_turn=mt)
+
+    async def _reg():
+        await coord.scheduler.register_instance(
+            FakeInstance("f0", is_local=True), skip_health_check=True)
+    asyncio.run_coroutine_threadsafe(_reg(), coord.loop).result(timeout=30)
+
+    prompts = TensorBatch(
+        tensors={"input_ids": torch.tensor([[11, 12, 13]]),
+                 "attention_mask": torch.ones(1, 3, dtype=torch.long)},
+        non_tensors={"uid": np.array(["u0"], dtype=object)})
+    coord.submit(prompts, SamplingParams(temperature=1.0, max_new_tokens=4),
+                 n=1)
+    shards = coord._next_shards(1)    # rank-0 view (no dist in this test)
+    b = shards[0] if not isinstance(shards[0], list) else shards[0][0]
+    attn = b["attention_mask"][0][3:]
+    loss = b["response_mask"][0]
+    n_present = int(attn.sum())
+    n_loss = int(loss.sum())
+    # turn1 (4 assistant) + user [9,9] + turn2 (4) + user + turn3 => user
+    # tokens attend but carry no loss
+    assert n_present > n_loss, (n_present, n_loss)
+    resp = b["responses"][0][:n_present].tolist()
+    assert resp[4:6] == [9, 9]
+    assert loss[4:6].sum() == 0
