@@ -65,9 +65,9 @@ def normalize(ans: str) -> str:
     # 0.5 == 1/2 style: canonicalize plain decimals that equal ints
     try:
         f = float(s)
-        if f == int(f):
+        if f == f and abs(f) != float("inf") and f == int(f):
             return str(int(f))
-    except ValueError:
+    except (ValueError, OverflowError):
         pass
     return s
 
