@@ -228,10 +228,13 @@ def test_continuation_chain_token_exact(lens, total):
 # ------------------------------------------------- round-2 scorer properties
 
 
-@given(st.text(max_size=200), st.text(min_size=1, max_size=30))
+@given(st.text(max_size=200),
+       st.text(min_size=1, max_size=30,
+               alphabet=st.characters(blacklist_characters="{}")))
 @settings(max_examples=60, deadline=None)
 def test_math_dapo_score_domain(noise, gt):
-    """score is always +1/-1; a boxed ground truth always scores +1."""
+    """score is always +1/-1; a boxed (brace-free) ground truth always
+    scores +1 — unbalanced braces are not representable in boxed form."""
     from polyrl_amd.reward_score import math_dapo
     r = math_dapo.compute_score(noise, gt)
     assert r["score"] in (1.0, -1.0)
