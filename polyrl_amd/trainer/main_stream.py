@@ -61,8 +61,20 @@ def main(argv=None):
         # manager style from reward_model.reward_manager
         from ..reward import load_reward_manager_from_config
         from ..utils.tokenizer import get_tokenizer
-        reward_fn = load_reward_manager_from_config(
-            cfg, tokenizer=get_tokenizer(cfg.actor_rollout_ref.model.path))
+        tok = get_tokenizer(cfg.actor_rollout_ref.model.path)
+        if tok is None:
+            # registry-name models carry no HF tokenizer assets: decode to
+            # the space-joined token ids so rule scorers still run
+            # (useful for synthetic plumbing; real runs pass an HF dir)
+            class _IdsTok:
+                def decode(self, ids):
+                    return " ".join(str(int(i)) for i in ids)
+            tok = _IdsTok()
+            if rank == 0:
+                print("[main_stream] no HF tokenizer for "
+                      f"{cfg.actor_rollout_ref.model.path!r}; rule scorers "
+                      "will see space-joined token ids", flush=True)
+        reward_fn = load_reward_manager_from_config(cfg, tokenizer=tok)
     else:
         reward_fn = load_reward_manager(reward_name)
 

@@ -422,3 +422,41 @@ def test_async_decode_with_multi_turn(tmp_path, monkeypatch):
     assert torch.equal(am_a, am_s)
     # multi-turn really engaged: some present tokens carry no loss
     assert int(am_a[:, -16:].sum()) > int(m_a.sum())
+
+
+def test_main_stream_reward_config_dapo(tmp_path):
+    """reward=config entry: reward_model.reward_manager=dapo + parquet
+    ground truth, through the real main_stream entry point (the
+    reference's load_reward_manager precedence, reward.py:95-150)."""
+    import subprocess
+    import sys
+
+    import pandas as pd
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    rows = [{"prompt": f"what is {i}+1?",
+             "input_ids": [7 + i % 11, 3, 5, (i * 13) % 500],
+             "data_source": "openai/gsm8k",
+             "ground_truth": str(i + 1)} for i in range(16)]
+    pd.DataFrame(rows).to_parquet(tmp_path / "g.parquet")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = repo
+    r = subprocess.run(
+        [sys.executable, "-m", "polyrl_amd.trainer.main_stream",
+         "actor_rollout_ref.model.path=llama-debug-cpu",
+         "actor_rollout_ref.model.dtype=float32",
+         "actor_rollout_ref.model.enable_gradient_checkpointing=false",
+         "actor_rollout_ref.actor.ppo_mini_batch_size=8",
+         "actor_rollout_ref.rollout.sampling.n=2",
+         "actor_rollout_ref.rollout.response_length=8",
+         "actor_rollout_ref.rollout.min_stream_batch_size=4",
+         "data.train_batch_size=8",
+         "data.max_prompt_length=24",
+         f"data.train_files=[{tmp_path}/g.parquet]",
+         "reward_model.reward_manager=dapo",
+         "reward_model.overlong_buffer_len=2",
+         f"trainer.default_local_dir={tmp_path}/ckpt",
+         "trainer.resume_mode=disable",
+         "reward=config",
+         "max_steps=1"],
+        capture_output=True, text=True, timeout=420, env=env)
+    assert r.returncode == 0, f"{r.stdout[-2500:]}\n{r.stderr[-2500:]}"
