@@ -48,7 +48,7 @@ def main():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=3)
-    p.add_argument("--warmup", type=int, default=1)
+    p.add_argument("--warmup", type=int, default=2)
     p.add_argument("--model", type=str, default="llama3-8b")
     p.add_argument("--batch-per-gpu", type=int, default=16,
                    help="prompts per GPU per step (weak scaling)")
