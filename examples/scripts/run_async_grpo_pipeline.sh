@@ -19,5 +19,5 @@ python -m torch.distributed.run --nnodes=1 --nproc-per-node=8 \
   data.train_batch_size=128 \
   data.max_prompt_length=512 \
   trainer.save_freq=20 trainer.test_freq=20 \
-  reward=naive \
+  reward=config \
   "$@"
