@@ -230,3 +230,46 @@ def test_pack_align_padding_equivalence(monkeypatch):
     loss_pad = (out_pad.float() ** 2).mean()
     loss_pad.backward()
     assert torch.allclose(g_ref, m.model.embed_tokens.weight.grad, atol=1e-5)
+
+
+def test_pack_pad_to_guards(monkeypatch):
+    """pack_pad_to applies only to >=75%-full micros (the dense-packing
+    tail must not blow up to the whole budget); env overrides the model
+    attribute; POLYRL_PACK_PAD_TO=0 disables."""
+    import torch
+
+    from polyrl_amd.models import create_model, get_model_config
+
+    cfg = get_model_config("llama-tiny")
+    m = create_model(cfg, kind="actor", dtype="float32", device="cpu")
+    trunk = m.model
+    trunk.use_remove_padding = True
+    trunk.pack_pad_to = 64
+    monkeypatch.setenv("POLYRL_PACK_ALIGN", "8")
+
+    # nearly-full micro (48/64 = 75%): padded to 64
+    trunk.pack_pad_to = 64
+    B, L = 1, 48
+    ids = torch.randint(0, cfg.vocab_size, (B, L))
+    am = torch.ones(B, L, dtype=torch.long)
+    pos = torch.arange(L).expand(B, L)
+    # capture Tp via the embed input size
+    sizes = []
+    orig_embed = trunk.embed_tokens.forward
+
+    def spy_embed(x):
+        sizes.append(x.shape[0])
+        return orig_embed(x)
+    trunk.embed_tokens.forward = spy_embed
+    m(ids, attention_mask=am, position_ids=pos)
+    assert sizes[-1] == 64, sizes            # padded to budget
+    # small tail micro (16/64 < 75%): align-8 only
+    ids2 = ids[:, :16]
+    am2 = am[:, :16]
+    pos2 = pos[:, :16]
+    m(ids2, attention_mask=am2, position_ids=pos2)
+    assert sizes[-1] == 16, sizes
+    # env disable
+    monkeypatch.setenv("POLYRL_PACK_PAD_TO", "0")
+    m(ids, attention_mask=am, position_ids=pos)
+    assert sizes[-1] == 48, sizes
