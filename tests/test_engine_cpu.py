@@ -428,3 +428,58 @@ def test_max_model_len_fence(setup):
     assert out.finish_reason == "length"
     assert len(out.output_ids) <= 24 - 20 + 1
     assert e.kv.free_pages == e.kv.num_pages
+
+
+# ----------------------------------------------- round-2 ADVICE regressions
+
+
+def test_release_memory_drops_captured_graphs(setup):
+    """captured decode graphs hold device pointers into
+    the KV tensors; release_memory()/resume_memory() must drop them so
+    replays never touch freed memory."""
+    cfg, model, _ = setup
+    eng = _fresh_engine(cfg, model, radix=False)
+    eng._graphs[4] = {"fake": "capture state"}
+    eng._graph_pool = object()
+    eng.release_memory()
+    assert eng._graphs == {}
+    assert eng._graph_pool is None
+    eng.resume_memory()
+    outs = eng.generate([[1, 2, 3]], SamplingParams(temperature=0.0,
+                                                    max_new_tokens=4), "rr")
+    assert len(outs[0].output_ids) == 4
+
+
+def test_aborted_request_does_not_donate_radix(setup):
+    """an aborted request's KV pages hold PRE-update
+    weights' KV (it is emitted on the step AFTER flush_radix), so it must
+    not insert into the radix trie — a token-exact continuation would
+    otherwise radix-match stale KV and skip recomputing under the new
+    weights."""
+    cfg, model, _ = setup
+    eng = _fresh_engine(cfg, model, radix=True, decode_chunk_size=4)
+    prompt = list(range(40, 80))           # 40 tokens, page-aligned span
+    eng.add_request("a0", prompt, SamplingParams(temperature=0.0,
+                                                 max_new_tokens=32))
+    for _ in range(3):                     # prefill + 2 decode chunks
+        eng.step()
+    assert eng.num_running() == 1
+    eng.abort_request(rid="a0")
+    outs = []
+    while eng.has_work():
+        outs.extend(eng.step())
+    assert outs and outs[0].finish_reason == "abort"
+    assert len(outs[0].output_ids) > 0     # partial output captured
+    # the aborted prefix must NOT be radix-matchable
+    pages, matched = eng.radix.match(prompt + outs[0].output_ids)
+    assert matched == 0, matched
+    if pages:
+        eng.kv.unref_pages(pages)
+    # control: a NORMALLY finished request does donate
+    eng.add_request("b0", prompt, SamplingParams(temperature=0.0,
+                                                 max_new_tokens=8))
+    while eng.has_work():
+        eng.step()
+    pages2, matched2 = eng.radix.match(prompt + [0])
+    assert matched2 > 0
+    eng.kv.unref_pages(pages2)
