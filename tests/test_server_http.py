@@ -318,6 +318,14 @@ def test_sender_cidr_allowlist():
             r = await c.post("/weights_handshake",
                              json={"metas": [], "num_streams": 1})
             assert r.status_code == 403
+            # the compat/tcp install routes are gated
+            # too — the allow-list must not be bypassable through them
+            r = await c.post("/update_weights_from_tensor",
+                             json={"version": 1, "tensors": {}})
+            assert r.status_code == 403, r.text
+            r = await c.post("/update_weights_from_tcp",
+                             json={"version": 1})
+            assert r.status_code == 403, r.text
             # generation routes unaffected
             r = await c.get("/health")
             assert r.status_code == 200
