@@ -68,3 +68,28 @@ def test_fp8_quantization_error_bound():
         assert rel < 0.07, rel
         # bytes really halve vs bf16
         assert q.element_size() == 1
+
+
+def test_oversized_span_fails_fast():
+    """a sender span beyond the registered buffer must
+    surface a transfer error from wait() immediately instead of hanging
+    until the install timeout."""
+    import socket
+    import struct
+
+    import pytest
+    import torch
+
+    from polyrl_amd.transfer.tcp_engine import TcpWeightReceiver
+
+    buf = torch.zeros(1024, dtype=torch.uint8)
+    rx = TcpWeightReceiver(buf, num_streams=1)
+    rx.expect(1024)
+    s = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+    s.connect(("127.0.0.1", rx.ports[0]))
+    # header claims a span past the end of the buffer
+    s.sendall(struct.Struct("<QQ").pack(512, 4096))
+    with pytest.raises(RuntimeError, match="exceeds buffer"):
+        rx.wait(timeout=10.0)
+    s.close()
+    rx.close()
