@@ -260,3 +260,48 @@ def test_scheduler_path_multi_turn():
     resp = b["responses"][0][:n_present].tolist()
     assert resp[4:6] == [9, 9]
     assert loss[4:6].sum() == 0
+
+
+@pytest.mark.timeout(600)
+def test_disagg_multi_turn_world3(tmp_path):
+    """Full-stack: disaggregated split (2 trainer + 1 rollout) WITH
+    multi-turn interactions — user turns inserted by the trainer-side
+    coordinator, continuations through the scheduler/HTTP engines."""
+    inter = tmp_path / "interaction.py"
+    inter.write_text(
+        "def generate_turn(prompt_ids, response_ids):\n"
+        "    if len(response_ids) >= 8:\n"
+        "        return None, True\n"
+        "    return [3, 4], False\n")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node=3", "--master-addr", "127.0.0.1",
+         "--master-port", "29676",
+         "-m", "polyrl_amd.trainer.main_stream",
+         "actor_rollout_ref.model.path=llama-debug-cpu",
+         "actor_rollout_ref.model.dtype=float32",
+         "actor_rollout_ref.model.enable_gradient_checkpointing=false",
+         "actor_rollout_ref.actor.ppo_mini_batch_size=8",
+         "actor_rollout_ref.actor.ppo_max_token_len_per_gpu=512",
+         "actor_rollout_ref.rollout.sampling.n=2",
+         "actor_rollout_ref.rollout.response_length=16",
+         "actor_rollout_ref.rollout.min_stream_batch_size=4",
+         "actor_rollout_ref.rollout.num_rollout_ranks=1",
+         "actor_rollout_ref.rollout.rollout_port_base=31880",
+         "actor_rollout_ref.rollout.multi_turn.enable=true",
+         f"actor_rollout_ref.rollout.multi_turn.interaction_path={inter}",
+         "actor_rollout_ref.rollout.multi_turn.max_assistant_turns=2",
+         "actor_rollout_ref.rollout.multi_turn.per_turn_max_tokens=4",
+         "data.train_batch_size=8",
+         "data.max_prompt_length=16",
+         "data.synthetic_num_prompts=32",
+         f"trainer.default_local_dir={tmp_path}/ckpt",
+         "trainer.resume_mode=disable",
+         "reward=random",
+         "max_steps=1",
+         ],
+        capture_output=True, text=True, timeout=540, env=env)
+    assert r.returncode == 0, \
+        f"stdout:\n{r.stdout[-4000:]}\nstderr:\n{r.stderr[-4000:]}"
