@@ -90,7 +90,10 @@ class TcpWeightReceiver:
             self._threads.append(t)
 
     def _serve_one(self, ls: socket.socket):
-        conn, _ = ls.accept()
+        try:
+            conn, _ = ls.accept()
+        except OSError:
+            return                        # listener closed
         conn.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
         try:
             while True:
@@ -108,13 +111,15 @@ class TcpWeightReceiver:
                             f"span ({off}, {length}) exceeds buffer "
                             f"size {len(self._mv)}")
                     self._done.set()
-                    raise ConnectionError(self._failed)
+                    return                # error signaled via _failed
                 got = 0
                 while got < length:
                     n = conn.recv_into(self._mv[off + got: off + length],
                                        min(length - got, CHUNK))
                     if n == 0:
-                        raise ConnectionError("peer closed mid-span")
+                        return            # peer closed mid-span; the
+                                          # byte count will not complete
+                                          # and wait() times out cleanly
                     got += n
                 with self._lock:
                     self._received += length
