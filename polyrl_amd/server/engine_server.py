@@ -304,7 +304,7 @@ def main():
     import torch
     import uvicorn
 
-    from ..models.registry import get_config
+    from ..models.registry import get_model_config
 
     p = argparse.ArgumentParser()
     p.add_argument("--model", required=True)
@@ -329,7 +329,7 @@ def main():
                         "(reference: allowed_sender_ips)")
     args = p.parse_args()
 
-    cfg = get_config(args.model)
+    cfg = get_model_config(args.model)
     device = "cuda" if torch.cuda.is_available() else "cpu"
     dtype = getattr(torch, args.dtype) if device == "cuda" else torch.float32
     engine = Engine(cfg, device=device, dtype=dtype,
@@ -359,25 +359,48 @@ def main():
     app = create_app(engine, allowed_sender_cidrs=cidrs)
 
     if args.manager:
-        import requests
-        adv = args.advertise_addr
-        if adv is None and args.host not in ("0.0.0.0", "::", ""):
-            adv = args.host
-        if adv is None:
-            # the interface that routes to the manager (multi-node safe)
-            import socket
-            from urllib.parse import urlparse
-            u = urlparse(args.manager)
-            try:
-                sk = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
-                sk.connect((u.hostname, u.port or 80))
-                adv = sk.getsockname()[0]
-                sk.close()
-            except OSError:
-                adv = "127.0.0.1"
-        requests.post(f"{args.manager}/register_rollout_instance",
-                      json={"addr": f"http://{adv}:{args.port}"},
-                      timeout=10)
+        # register AFTER our own server answers /health (the manager
+        # health-gates joins, instance_manager.rs:5-37 capability) — from
+        # a background thread so uvicorn can start below
+        import threading
+
+        def _register():
+            import time as _t
+
+            import requests
+            adv = args.advertise_addr
+            if adv is None and args.host not in ("0.0.0.0", "::", ""):
+                adv = args.host
+            if adv is None:
+                # the interface that routes to the manager (multi-node)
+                import socket
+                from urllib.parse import urlparse
+                u = urlparse(args.manager)
+                try:
+                    sk = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
+                    sk.connect((u.hostname, u.port or 80))
+                    adv = sk.getsockname()[0]
+                    sk.close()
+                except OSError:
+                    adv = "127.0.0.1"
+            if ":" not in adv:
+                adv = f"{adv}:{args.port}"
+            deadline = _t.monotonic() + 120
+            while _t.monotonic() < deadline:
+                try:
+                    requests.get(f"http://{adv}/health", timeout=2)
+                    break
+                except Exception:
+                    _t.sleep(0.2)
+            for _ in range(30):
+                try:
+                    requests.post(
+                        f"{args.manager}/register_rollout_instance",
+                        json={"addr": f"http://{adv}"}, timeout=10)
+                    return
+                except Exception:
+                    _t.sleep(1.0)
+        threading.Thread(target=_register, daemon=True).start()
 
     uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
 

@@ -109,3 +109,70 @@ def test_manager_generate_and_batch_stream():
                 assert x["samples"][0]["output_ids"] == exp
         await sched.close()
     asyncio.run(go())
+
+
+def test_remote_instance_self_registration_e2e(tmp_path):
+    """The reference's elastic-join lifecycle (§3.4 / launch_sglang.sh
+    capability): a REAL `engine_server` subprocess started with --manager
+    self-registers against a live manager facade; the manager then serves
+    a /generate relayed through the joined instance."""
+    import asyncio
+    import subprocess
+    import sys
+    import time
+
+    import httpx
+
+    from polyrl_amd.scheduler.manager import RolloutScheduler, SchedulerConfig
+    from polyrl_amd.scheduler.http_api import serve_manager
+
+    loop = asyncio.new_event_loop()
+    import threading
+    threading.Thread(target=loop.run_forever, daemon=True).start()
+    sched = RolloutScheduler(SchedulerConfig(stats_interval_s=0.1,
+                                             health_check_timeout_s=60.0))
+    mgr_port = 31930
+    server = serve_manager(sched, host="127.0.0.1", port=mgr_port, loop=loop)
+    for _ in range(100):
+        try:
+            if httpx.get(f"http://127.0.0.1:{mgr_port}/health",
+                         timeout=1.0).status_code == 200:
+                break
+        except Exception:
+            time.sleep(0.05)
+
+    import os
+    env = dict(os.environ)
+    env["PYTHONPATH"] = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "polyrl_amd.server.engine_server",
+         "--model", "llama-debug-cpu", "--host", "127.0.0.1",
+         "--port", "31931", "--kv-gb", "0.02",
+         "--manager", f"http://127.0.0.1:{mgr_port}",
+         "--advertise-addr", "127.0.0.1:31931"],
+        env=env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT)
+    try:
+        # join is health-gated on the manager side
+        deadline = time.monotonic() + 60
+        while time.monotonic() < deadline:
+            if sched.num_active() >= 1:
+                break
+            time.sleep(0.2)
+        assert sched.num_active() >= 1, "instance never joined the pool"
+        # generate THROUGH the manager facade (relay route)
+        r = httpx.post(f"http://127.0.0.1:{mgr_port}/generate",
+                       json={"input_ids": [5, 6, 7],
+                             "sampling_params": {"n": 1,
+                                                 "max_new_tokens": 4},
+                             "return_logprob": True}, timeout=60.0)
+        assert r.status_code == 200, r.text
+        out = r.json()
+        sample = out["samples"][0]
+        assert len(sample["output_ids"]) == 4
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            proc.kill()
+        server.should_exit = True
