@@ -153,7 +153,7 @@ def test_remote_instance_self_registration_e2e(tmp_path):
         env=env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT)
     try:
         # join is health-gated on the manager side
-        deadline = time.monotonic() + 60
+        deadline = time.monotonic() + 180   # model init under suite load
         while time.monotonic() < deadline:
             if sched.num_active() >= 1:
                 break
