@@ -55,6 +55,17 @@ def test_tracking_jsonl_roundtrip(tmp_path):
     assert lines[1]["step"] == 4
 
 
+def test_tracking_tensorboard_degrades_without_package(tmp_path):
+    """tensorboard backend must be optional: on an image without the
+    tensorboard package Tracking falls back to _tb=None and log() still
+    works (core/metrics.py:159-165)."""
+    tr = Tracking("p", "e2", ["tensorboard", "jsonl"],
+                  default_local_dir=str(tmp_path))
+    tr.log({"a": 1.0}, step=1)          # must not raise either way
+    tr.close()
+    assert (tmp_path / "logs" / "e2" / "metrics.jsonl").exists()
+
+
 def test_marked_timer_accumulates():
     import time
     timing = {}
