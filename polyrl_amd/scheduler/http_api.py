@@ -44,7 +44,9 @@ def create_manager_app(scheduler, loop: Optional[asyncio.AbstractEventLoop]
         inst = HttpInstance(addr, weight_state_fn=remote_weight_state_fn)
         await _run(scheduler.register_instance(inst))
         return {"status": "registered", "instance_id": inst.instance_id,
-                "weight_version": scheduler.latest_weight_version}
+                "weight_version": scheduler.latest_weight_version,
+                "weight_sender_endpoint":
+                    scheduler.weight_sender_for(inst.instance_id)}
 
     @app.post("/register_local_rollout_instances")
     async def register_local_rollout_instances(request: Request):
@@ -162,6 +164,8 @@ def create_manager_app(scheduler, loop: Optional[asyncio.AbstractEventLoop]
                 "queued": st.num_queued,
                 "gen_throughput": st.gen_throughput,
                 "weight_version": getattr(inst, "weight_version", 0),
+                "weight_sender_endpoint":
+                    scheduler.weight_sender_for(inst.instance_id),
             })
         return {"instances": out,
                 "latest_weight_version": scheduler.latest_weight_version,
@@ -176,7 +180,19 @@ def create_manager_app(scheduler, loop: Optional[asyncio.AbstractEventLoop]
     @app.post("/get_receive_instances")
     async def get_receive_instances():
         insts = scheduler.get_receive_instances()
-        return {"instances": [i.instance_id for i in insts]}
+        return {"instances": [i.instance_id for i in insts],
+                "senders": {i.instance_id:
+                            scheduler.weight_sender_for(i.instance_id)
+                            for i in insts}}
+
+    @app.put("/update_weight_senders")
+    async def update_weight_senders(request: Request):
+        """Trainer registers its weight-sender endpoints; the scheduler
+        round-robins them over instances (handlers.rs PUT route +
+        state.rs:149-162)."""
+        body = await request.json()
+        scheduler.update_weight_senders(list(body.get("senders", [])))
+        return {"status": "ok", "num_senders": len(body.get("senders", []))}
 
     @app.post("/update_weights")
     async def update_weights(request: Request):

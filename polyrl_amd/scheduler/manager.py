@@ -59,6 +59,7 @@ class _InstState:
         self.stats = inst.get_stats()
         self.updating_weight = False
         self.weight_version = 0
+        self.weight_sender_endpoint: Optional[str] = None
         self.evicted = False
 
 
@@ -68,6 +69,8 @@ class RolloutScheduler:
         self._states: Dict[str, _InstState] = {}
         self._active: List[str] = []            # active pool (dispatchable)
         self._rr = 0                            # round-robin cursor
+        self._weight_senders: List[str] = []    # sender endpoints (PUT route)
+        self._sender_rr = 0                     # sender round-robin cursor
         self._cond: Optional[asyncio.Condition] = None
         self.latest_weight_version = 0
         self.balance = LoadBalanceState(
@@ -91,6 +94,7 @@ class RolloutScheduler:
                 await asyncio.sleep(self.cfg.health_check_interval_s)
         st = _InstState(inst)
         st.weight_version = getattr(inst, "weight_version", 0)
+        st.weight_sender_endpoint = self._assign_weight_sender()
         self._states[inst.instance_id] = st
         # local instances join the active pool immediately; remote instances
         # join once their weights reach the latest version (§3.4 lifecycle)
@@ -359,6 +363,29 @@ class RolloutScheduler:
                 st.weight_version = version
                 self._active.append(iid)
         await self._notify()
+
+    def _assign_weight_sender(self) -> Optional[str]:
+        """Round-robin the registered sender endpoints over instances
+        (reference state.rs:149-162 senders×groups rotation; our TCP
+        engine carries the per-sender stream fan-out internally)."""
+        if not self._weight_senders:
+            return None
+        ep = self._weight_senders[self._sender_rr % len(self._weight_senders)]
+        self._sender_rr += 1
+        return ep
+
+    def update_weight_senders(self, senders: List[str]):
+        """Replace the weight-sender endpoint registry and re-assign
+        endpoints to every known instance (PUT /update_weight_senders,
+        handlers.rs /update_weight_senders route)."""
+        self._weight_senders = list(senders)
+        self._sender_rr = 0
+        for st in self._states.values():
+            st.weight_sender_endpoint = self._assign_weight_sender()
+
+    def weight_sender_for(self, instance_id: str) -> Optional[str]:
+        st = self._states.get(instance_id)
+        return st.weight_sender_endpoint if st is not None else None
 
     def get_receive_instances(self) -> List[RolloutInstance]:
         """Instances needing the latest weights; CAS-marks them updating

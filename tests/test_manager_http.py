@@ -63,6 +63,51 @@ def test_manager_routes():
     asyncio.run(go())
 
 
+def test_weight_sender_registry():
+    """PUT /update_weight_senders + round-robin assignment over instances
+    (handlers.rs PUT route, state.rs:149-162): senders rotate across
+    registrations and re-assign on registry replacement."""
+    async def go():
+        sched = RolloutScheduler(SchedulerConfig(stats_interval_s=0.02))
+        app = create_manager_app(sched)
+        import httpx
+        async with httpx.AsyncClient(
+                transport=httpx.ASGITransport(app=app),
+                base_url="http://mgr") as c:
+            # before any senders: registration yields no endpoint
+            await sched.register_instance(FakeInstance("a", is_local=False),
+                                          skip_health_check=True)
+            assert sched.weight_sender_for("a") is None
+
+            r = await c.put("/update_weight_senders",
+                            json={"senders": ["10.0.0.1:7000",
+                                              "10.0.0.2:7000"]})
+            assert r.json()["num_senders"] == 2
+            # replacement re-assigns existing instances round-robin
+            assert sched.weight_sender_for("a") == "10.0.0.1:7000"
+
+            await sched.register_instance(FakeInstance("b", is_local=False),
+                                          skip_health_check=True)
+            await sched.register_instance(FakeInstance("d", is_local=False),
+                                          skip_health_check=True)
+            assert sched.weight_sender_for("b") == "10.0.0.2:7000"
+            assert sched.weight_sender_for("d") == "10.0.0.1:7000"
+
+            d = (await c.get("/get_instances_status")).json()
+            eps = {i["id"]: i["weight_sender_endpoint"]
+                   for i in d["instances"]}
+            assert eps == {"a": "10.0.0.1:7000", "b": "10.0.0.2:7000",
+                           "d": "10.0.0.1:7000"}
+
+            # receive-instances response carries the per-instance sender
+            await c.post("/update_weight_version", json={"version": 1})
+            r = (await c.post("/get_receive_instances")).json()
+            assert set(r["instances"]) == {"a", "b", "d"}
+            assert r["senders"]["b"] == "10.0.0.2:7000"
+        await sched.close()
+    asyncio.run(go())
+
+
 def test_manager_generate_and_batch_stream():
     """The manager's generation surface (handlers.rs /generate +
     /batch_generate_requests NDJSON contract): single group relay and the
