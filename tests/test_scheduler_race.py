@@ -231,3 +231,87 @@ def test_engine_update_weights_vs_generation_writer_lock():
         assert o.output_ids in refs, \
             f"torn generation (mixed weight versions): {o.output_ids}"
     assert runner.weight_version == 8
+
+
+@pytest.mark.timeout(180)
+def test_runner_admin_fuzz_no_deadlock_no_lost_futures():
+    """Seeded randomized interleaving of the runner's whole admin surface
+    (submit / abort(rid) / abort_all / update_weights with and without
+    in-flight aborts / stats) from several OS threads while the pump
+    thread generates.  Invariants under ANY schedule: every submitted
+    future resolves with a legal finish_reason and bounded length, nothing
+    deadlocks, and the runner still serves a fresh request afterwards.
+    (The round-2 lock-starvation bug lived exactly here: admin callers
+    were starved by the non-FIFO pump lock for whole generations.)"""
+    from polyrl_amd.models import create_model, get_model_config
+    from polyrl_amd.rollout.engine import Engine, SamplingParams
+    from polyrl_amd.rollout.runner import EngineRunner
+
+    cfg = get_model_config("llama-debug-cpu")
+    torch.manual_seed(1)
+    m = create_model(cfg, kind="actor", dtype="float32", device="cpu")
+    sd0 = {k: v.clone() for k, v in m.state_dict().items()}
+    sd1 = {k: (v + 0.03 * torch.randn_like(v)) for k, v in sd0.items()}
+
+    eng = Engine(cfg, device="cpu", dtype=torch.float32,
+                 kv_bytes_budget=16 << 20)
+    eng.model.load_state_dict(sd0)
+    runner = EngineRunner(eng)
+    runner.start()
+
+    N = 48
+    sp = SamplingParams(temperature=0.0, max_new_tokens=8)
+    stop_admins = threading.Event()
+
+    def admin(seed):
+        rng = random.Random(seed)
+        v = 100 * seed
+        while not stop_admins.is_set():
+            op = rng.random()
+            if op < 0.35:
+                runner.abort(rid=f"f{rng.randrange(N)}")
+            elif op < 0.45:
+                runner.abort(abort_all=True)
+            elif op < 0.70:
+                v += 1
+                runner.update_weights(sd1 if v % 2 else sd0, version=v,
+                                      abort_in_flight=bool(rng.random() < .5))
+            else:
+                runner.stats()
+            # tiny jitter so ops land at varied pump phases
+            if rng.random() < 0.5:
+                import time as _t
+                _t.sleep(rng.random() * 0.002)
+
+    async def drive():
+        futs = [runner.submit([3 + i % 7, 11, 4], sp, rid=f"f{i}")
+                for i in range(N)]
+        threads = [threading.Thread(target=admin, args=(s,), daemon=True)
+                   for s in (1, 2, 3)]
+        for t in threads:
+            t.start()
+        try:
+            outs = await asyncio.wait_for(asyncio.gather(*futs), timeout=90)
+        finally:
+            stop_admins.set()
+            for t in threads:
+                t.join(timeout=10)
+        assert not any(t.is_alive() for t in threads), "admin thread hung"
+        return outs
+
+    try:
+        outs = asyncio.run(drive())
+        for o in outs:
+            assert o.finish_reason in ("stop", "length", "abort"), \
+                o.finish_reason
+            assert len(o.output_ids) <= sp.max_new_tokens
+        # runner still serviceable after the storm
+        async def one_more():
+            return await asyncio.wait_for(
+                asyncio.gather(*[runner.submit([9, 9, 9], sp, rid="post")]),
+                timeout=30)
+        post = asyncio.run(one_more())[0]
+        assert post.finish_reason in ("stop", "length")
+        assert len(post.output_ids) >= 1
+    finally:
+        runner.stop()
