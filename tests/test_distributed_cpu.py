@@ -102,6 +102,17 @@ def test_bench_contract_world2():
     assert d["n_gpus"] == 2
     assert d["config"]["parallelism"] == "dp2"
     assert d["value"] > 0
+    # the driver's full field contract
+    for field in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                  "ms_per_step", "higher_is_better", "scaling",
+                  "vs_baseline", "dtype", "data", "config"):
+        assert field in d, f"missing {field}"
+    for field in ("model", "global_batch", "seq_len", "parallelism"):
+        assert field in d["config"], f"missing config.{field}"
+    assert d["steps"] == 1 and d["warmup"] == 1
+    assert d["scaling"] == "weak" and d["higher_is_better"] is True
+    assert d["ms_per_step"] > 0
+    assert d["data"].startswith("synthetic")
 
 
 @pytest.mark.timeout(600)
