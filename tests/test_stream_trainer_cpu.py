@@ -460,3 +460,40 @@ def test_main_stream_reward_config_dapo(tmp_path):
          "max_steps=1"],
         capture_output=True, text=True, timeout=420, env=env)
     assert r.returncode == 0, f"{r.stdout[-2500:]}\n{r.stderr[-2500:]}"
+
+
+def test_lora_end_to_end_freezes_base_and_publishes_merged(tmp_path):
+    """LoRA path through the FULL streamed loop (model.lora_rank>0,
+    reference capability: LoRA collect/summon in the FSDP machinery,
+    stream_fsdp_workers.py:221-233): base weights stay frozen, adapters
+    train, and the engine receives MERGED weights on publication."""
+    cfg = tiny_config(tmp_path, adv="grpo")
+    cfg.actor_rollout_ref.model.lora_rank = 4
+    cfg.actor_rollout_ref.model.lora_alpha = 8.0
+    trainer = StreamPPOTrainer(cfg, reward_fn=load_reward_manager("random"))
+
+    actor = trainer.actor.model
+    base_before = {k: v.detach().clone()
+                   for k, v in actor.state_dict().items()
+                   if ".base." in k or "embed" in k}
+    lora_before = {k: v.detach().clone()
+                   for k, v in actor.state_dict().items() if ".lora_" in k}
+    assert lora_before, "apply_lora did not wrap any linears"
+
+    trainer.fit(max_steps=2)
+
+    sd = actor.state_dict()
+    for k, v in base_before.items():
+        assert torch.equal(sd[k], v), f"frozen base weight changed: {k}"
+    assert any(not torch.equal(sd[k], v) for k, v in lora_before.items()), \
+        "no LoRA adapter weight changed after 2 steps"
+
+    # publication folded the deltas: engine weights == merged_state_dict
+    from polyrl_amd.models.lora import merged_state_dict
+    merged = merged_state_dict(actor)
+    eng_sd = dict(trainer.engine.model._name_map)   # fused-buffer views
+    common = [k for k in merged if k in eng_sd]
+    assert common, "no overlapping names between merged sd and engine"
+    for k in common:
+        assert torch.allclose(eng_sd[k].float(), merged[k].float(),
+                              atol=1e-5), f"engine got unmerged {k}"
