@@ -305,3 +305,61 @@ def test_disagg_multi_turn_world3(tmp_path):
         capture_output=True, text=True, timeout=540, env=env)
     assert r.returncode == 0, \
         f"stdout:\n{r.stdout[-4000:]}\nstderr:\n{r.stderr[-4000:]}"
+
+
+def test_multi_turn_survives_instance_failure_token_exact():
+    """Multi-turn + fault tolerance composed: an instance that dies
+    mid-assistant-turn is evicted and the turn CONTINUES token-exactly on
+    the survivor, so the final multi-turn batch (responses, presence mask,
+    assistant-only loss mask) is bitwise identical to a run with no
+    failure at all (handlers.rs continuation x interaction turns)."""
+    import asyncio
+
+    import numpy as np
+    import torch
+
+    from polyrl_amd.protocol import TensorBatch
+    from polyrl_amd.rollout.engine import SamplingParams
+    from polyrl_amd.scheduler.instances import FakeInstance
+    from polyrl_amd.trainer.disagg import DisaggCoordinator
+
+    mt = {"interaction":
+          lambda prompt, resp: (None, True) if len(resp) >= 10
+          else ([9, 9], False),
+          "max_assistant_turns": 3, "max_user_turns": 5,
+          "max_tool_response_length": 8, "per_turn_max_tokens": 4}
+
+    def run(insts):
+        coord = DisaggCoordinator(response_length=16, trainer_group=None,
+                                  rollout_urls=[], rank=0, n_trainer=1,
+                                  multi_turn=dict(mt))
+
+        async def _reg():
+            for i in insts:
+                await coord.scheduler.register_instance(
+                    i, skip_health_check=True)
+        asyncio.run_coroutine_threadsafe(_reg(), coord.loop).result(
+            timeout=30)
+        prompts = TensorBatch(
+            tensors={"input_ids": torch.tensor([[11, 12, 13]]),
+                     "attention_mask": torch.ones(1, 3, dtype=torch.long)},
+            non_tensors={"uid": np.array(["u0"], dtype=object)})
+        coord.submit(prompts,
+                     SamplingParams(temperature=1.0, max_new_tokens=4), n=1)
+        shards = coord._next_shards(1)
+        b = shards[0] if not isinstance(shards[0], list) else shards[0][0]
+        return b
+
+    ref = run([FakeInstance("healthy", is_local=True)])
+    # dies after 2 tokens of whichever turn it serves first -> evicted,
+    # the turn continues on the survivor
+    dying = FakeInstance("dies", is_local=True, fail_after_tokens=2)
+    survivor = FakeInstance("survivor", is_local=True)
+    got = run([dying, survivor])
+
+    assert dying.served_gids, \
+        "fault path not exercised: dying instance never picked"
+    assert survivor.served_gids, "survivor never picked up the continuation"
+    for key in ("responses", "attention_mask", "response_mask"):
+        assert torch.equal(ref[key], got[key]), \
+            f"{key} diverged under mid-turn failure:\n{ref[key]}\n{got[key]}"
