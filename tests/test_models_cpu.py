@@ -273,3 +273,37 @@ def test_pack_pad_to_guards(monkeypatch):
     monkeypatch.setenv("POLYRL_PACK_PAD_TO", "0")
     m(ids, attention_mask=am, position_ids=pos)
     assert sizes[-1] == 48, sizes
+
+def test_lora_composes_with_remove_padding():
+    """LoRA-wrapped linears through the packed varlen path == dense path
+    (the _lin router must leave wrapped modules their own forward)."""
+    import torch
+    from polyrl_amd.models import create_model, get_model_config
+    from polyrl_amd.models.lora import apply_lora
+    cfg = get_model_config("llama-debug-cpu")
+    torch.manual_seed(21)
+    m = create_model(cfg, kind="actor", dtype="float32", device="cpu")
+    apply_lora(m, r=4, alpha=8.0)
+    # non-zero B so the adapters actually contribute
+    for mod in m.modules():
+        if hasattr(mod, "lora_B"):
+            with torch.no_grad():
+                mod.lora_B.normal_(0, 0.05)
+    B, L = 2, 12
+    ids = torch.randint(0, cfg.vocab_size, (B, L))
+    am = torch.ones(B, L, dtype=torch.long)
+    am[1, :4] = 0
+    with torch.no_grad():
+        dense = m(ids, attention_mask=am)
+        m.model.use_remove_padding = True
+        packed = m(ids, attention_mask=am)
+        m.model.use_remove_padding = False
+    valid = am.bool()
+    err = (dense[valid] - packed[valid]).abs().max().item()
+    assert err < 1e-3, err
+    # and adapters really changed the output vs the un-adapted model
+    torch.manual_seed(21)   # identical base init to m
+    m2 = create_model(cfg, kind="actor", dtype="float32", device="cpu")
+    with torch.no_grad():
+        base = m2(ids, attention_mask=am)
+    assert not torch.allclose(base[valid], dense[valid], atol=1e-4)
