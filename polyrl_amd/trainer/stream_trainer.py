@@ -305,6 +305,19 @@ class StreamPPOTrainer:
         self.publisher = ElasticPublisher(
             self.actor.model, self.engine.model, self.coordinator,
             tie=model_cfg.tie_word_embeddings, trainer_group=self.pg)
+        if self.rank == 0:
+            # register this trainer as the weight-sender endpoint so the
+            # scheduler can assign it to joining instances (the reference's
+            # launcher.register_weight_senders PUT at startup)
+            from .disagg import advertise_addr
+            import asyncio as _aio
+            _sched = self.coordinator.scheduler
+            _ep = f"{advertise_addr()}:{ro.rollout_manager_port or 0}"
+
+            async def _reg_sender():
+                _sched.update_weight_senders([_ep])
+            _aio.run_coroutine_threadsafe(
+                _reg_sender(), self.coordinator.loop).result(timeout=30)
 
     def _kv_budget(self, ro) -> int:
         if self.device.startswith("cuda"):

@@ -154,6 +154,7 @@ def test_elastic_runtime_join(tmp_path):
                       if i["id"].endswith(":31885")]
                 if me:
                     joined["ok"] = True
+                    joined["sender"] = me[0].get("weight_sender_endpoint")
                     if me[0]["active"]:
                         joined["active"] = True
                         joined["version"] = me[0]["weight_version"]
@@ -167,6 +168,10 @@ def test_elastic_runtime_join(tmp_path):
         assert joined["ok"], "instance never appeared in the manager roster"
         assert joined["active"], "joined instance never became active"
         assert joined["version"] >= 1   # received a weight version via TCP
+        # the trainer registered itself as the weight sender and the
+        # scheduler assigned it to the joining instance (PUT
+        # /update_weight_senders -> round-robin at registration)
+        assert joined.get("sender"), "no weight_sender_endpoint assigned"
         # TCP push actually replaced our random init with trainer weights
         assert not torch.equal(eng.model.embed, embed_before)
     finally:
