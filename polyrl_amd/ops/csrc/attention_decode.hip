@@ -213,6 +213,22 @@ __global__ void paged_decode_merge_kernel(
     out[row * D + lane + j * 64] = f2bf(acc[j] * inv);
 }
 
+// EXPERIMENTAL round-3 candidate (default OFF, opt-in POLYRL_DECODE_MFMA=1
+// for on-box A/B): MFMA S/PV decode — moves the per-key dot products and
+// P·V off the VALU (PMC: decode is exactly VALU-throughput-bound at
+// B*Hk>=1024, profiles/PROFILES.md).  Textual include: -fno-gpu-rdc builds
+// cannot launch kernels defined in another TU.  Falls back to the
+// production VALU kernel for (D, G) combos the prototype doesn't cover.
+#include "experimental/attention_decode_mfma.hip"
+
+static bool use_mfma_decode() {
+  static const bool v = []() {
+    const char* e = getenv("POLYRL_DECODE_MFMA");
+    return e && e[0] == '1';
+  }();
+  return v;
+}
+
 template <int D>
 static void dispatch_g(torch::Tensor& out, const torch::Tensor& q,
                        const torch::Tensor& k_cache, const torch::Tensor& v_cache,
@@ -260,6 +276,17 @@ static void dispatch_g(torch::Tensor& out, const torch::Tensor& q,
       paged_decode_merge_kernel<D><<<dim3(B * Hq), dim3(64), 0, stream>>>(
           (bf16_t*)out.data_ptr(), ppart, nsplit);
   };
+  if (use_mfma_decode()) {
+    // prototype coverage: D=128 G in {1,4,8}; D=64 G=4 (llama/qwen shapes)
+    if constexpr (D == 128) {
+      if (G == 1) { args(paged_decode_attn_mfma_kernel<128, 1>); return; }
+      if (G == 4) { args(paged_decode_attn_mfma_kernel<128, 4>); return; }
+      if (G == 8) { args(paged_decode_attn_mfma_kernel<128, 8>); return; }
+    } else if constexpr (D == 64) {
+      if (G == 4) { args(paged_decode_attn_mfma_kernel<64, 4>); return; }
+    }
+    // uncovered combo: fall through to the production VALU kernel
+  }
   switch (G) {
     case 1: args(paged_decode_attn_kernel<D, 1>); break;
     case 2: args(paged_decode_attn_kernel<D, 2>); break;

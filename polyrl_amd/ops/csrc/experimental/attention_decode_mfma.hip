@@ -277,3 +277,9 @@ template __global__ void paged_decode_attn_mfma_kernel<128, 1>(
 template __global__ void paged_decode_attn_mfma_kernel<64, 4>(
     bf16_t*, float*, const bf16_t*, const bf16_t*, const bf16_t*,
     const int*, const int*, int, int, int, int, float, long, int);
+
+// this file is textually included into attention_decode.hip (the extension
+// builds -fno-gpu-rdc, so kernels cannot be launched across TUs) — keep the
+// local tile macros from leaking into the including TU
+#undef KT
+#undef NW
