@@ -4,6 +4,7 @@ All tests are @pytest.mark.gpu and run on a real MI355X via gpurun / the
 driver's round-end GPU tier.
 """
 import math
+import os
 
 import pytest
 import torch
@@ -213,6 +214,40 @@ def test_paged_attention_decode(Hq, Hk, D):
     q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device=DEV)
     scale = 1.0 / math.sqrt(D)
     out = ops.paged_attention_decode(q, kc, vc, pt.to(DEV), lens.to(DEV), scale)
+    expect = ref.paged_attention_decode(q.cpu(), kc.cpu(), vc.cpu(), pt,
+                                        lens, scale)
+    assert rel_err(out.cpu(), expect) < 3e-2
+
+
+@pytest.mark.skipif(os.environ.get("POLYRL_DECODE_MFMA") != "1",
+                    reason="experimental MFMA decode path: opt-in via "
+                           "POLYRL_DECODE_MFMA=1 (not yet validated)")
+@pytest.mark.parametrize("Hq,Hk,D", [(8, 8, 128), (32, 8, 128),
+                                     (8, 2, 128), (16, 4, 64)])
+def test_paged_attention_decode_mfma(Hq, Hk, D):
+    """Numerics gate for the EXPERIMENTAL MFMA decode prototype (covered
+    (D, G): 128/{1,4,8(2 falls back)}, 64/4).  Round-3 entry point:
+    POLYRL_DECODE_MFMA=1 pytest -k decode_mfma -m gpu."""
+    torch.manual_seed(10)
+    B, page_size = 9, 16
+    lens = torch.tensor([1, 5, 16, 17, 60, 64, 100, 255, 300],
+                        dtype=torch.int32)
+    max_pages = int((-(-lens.max().item() // page_size)))
+    num_pages = int(sum(-(-int(l) // page_size) for l in lens)) + 4
+    kc = torch.randn(num_pages, page_size, Hk, D, dtype=torch.bfloat16,
+                     device=DEV)
+    vc = torch.randn_like(kc)
+    perm = torch.randperm(num_pages)
+    pt = torch.zeros(B, max_pages, dtype=torch.int32)
+    pi = 0
+    for b in range(B):
+        for j in range(-(-int(lens[b]) // page_size)):
+            pt[b, j] = perm[pi]
+            pi += 1
+    q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device=DEV)
+    scale = 1.0 / math.sqrt(D)
+    out = ops.paged_attention_decode(q, kc, vc, pt.to(DEV), lens.to(DEV),
+                                     scale)
     expect = ref.paged_attention_decode(q.cpu(), kc.cpu(), vc.cpu(), pt,
                                         lens, scale)
     assert rel_err(out.cpu(), expect) < 3e-2
