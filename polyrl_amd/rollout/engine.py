@@ -37,6 +37,21 @@ def _mm_nt(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     return x @ w.t()
 
 
+# graphed-decode batch buckets (POLYRL_DECODE_BUCKETS=1): capture counts
+# chosen so padding waste stays <= 50% while EOS-staggered batches reuse
+# at most ~16 captures over the whole 1..256 running-batch range
+DECODE_BUCKETS = (1, 2, 3, 4, 6, 8, 12, 16, 24, 32, 48, 64, 96, 128,
+                  192, 256)
+
+
+def decode_bucket(b: int) -> int:
+    """Smallest bucket >= b (b beyond the table maps to itself)."""
+    for x in DECODE_BUCKETS:
+        if x >= b:
+            return x
+    return b
+
+
 def _tuned_decode_enabled() -> bool:
     import os
     return os.environ.get("POLYRL_TUNED_DECODE", "1") == "1"
@@ -397,6 +412,13 @@ class Engine:
         self._graphs: Dict[int, dict] = {}      # batch size -> capture state
         self._graph_pool = None
         self._graph_max_pages = (self.max_model_len + page_size - 1) // page_size
+        # EXPERIMENTAL (round-3 candidate, default OFF): pad the graphed
+        # decode batch up to a bucket size so EOS-staggered serving (B
+        # shrinking every few steps) replays existing captures instead of
+        # capturing a fresh graph per exact B.  Dummy rows write KV into a
+        # reserved scratch slot and their outputs are discarded.
+        self._decode_buckets = bool(int(__import__("os").environ.get(
+            "POLYRL_DECODE_BUCKETS", "0")))
 
     # ------------------------------------------------------------ public API
     def add_request(self, rid: str, input_ids: List[int],
@@ -879,6 +901,23 @@ class Engine:
         toks_h = out_tokens.t().cpu().tolist()
         self._finish_decode_chunk(reqs, C, toks_h, out_lps)
 
+    _SCRATCH_SEQ = -7777       # sentinel seq id owning the scratch page
+
+    def _decode_scratch_slot(self) -> int:
+        """Reserve (once per KV cache lifetime) one page as the dummy-row
+        KV sink for bucketed decode; zero it so dummy-row attention reads
+        finite bf16.  Re-reserves automatically after release/resume
+        (the KV cache is rebuilt there and seq maps reset)."""
+        if self.kv.seq_len(self._SCRATCH_SEQ) == 0:
+            if not self.kv.allocate(self._SCRATCH_SEQ, 1):
+                raise RuntimeError("no free page for decode-bucket scratch")
+            page = int(self.kv.slots_for(self._SCRATCH_SEQ, 0, 1)[0]) \
+                // self.kv.page_size
+            for li in range(len(self.kv.k_cache)):
+                self.kv.k_cache[li][page].zero_()
+                self.kv.v_cache[li][page].zero_()
+        return int(self.kv.slots_for(self._SCRATCH_SEQ, 0, 1)[0])
+
     def _decode_graphed(self, B, C, tokens, pos0, slots_all, page_table,
                         temp, tk, tp, seed0, out_tokens, out_lps):
         """hipGraph decode: capture one decode iteration per batch size and
@@ -887,22 +926,29 @@ class Engine:
         sampling seed advance INSIDE the graph, the slot column is a tiny
         D2D copy per replay."""
         dev = self.device
-        st = self._graphs.get(B)
+        # bucketed capture (EXPERIMENTAL, default off): pad to Bb rows; rows
+        # [B:Bb) are dummies writing KV into a reserved scratch slot
+        Bb = decode_bucket(B) if self._decode_buckets else B
+        scratch = self._decode_scratch_slot() if Bb > B else 0
+        st = self._graphs.get(Bb)
         if st is None:
             st = {
-                "tokens": torch.zeros(B, dtype=torch.int64, device=dev),
-                "pos": torch.zeros(B, dtype=torch.int32, device=dev),
-                "ctx": torch.zeros(B, dtype=torch.int32, device=dev),
-                "slots": torch.zeros(B, dtype=torch.int32, device=dev),
-                "ptab": torch.zeros(B, self._graph_max_pages,
+                "tokens": torch.zeros(Bb, dtype=torch.int64, device=dev),
+                "pos": torch.zeros(Bb, dtype=torch.int32, device=dev),
+                "ctx": torch.ones(Bb, dtype=torch.int32, device=dev),
+                "slots": torch.full((Bb,), scratch, dtype=torch.int32,
+                                    device=dev),
+                "ptab": torch.zeros(Bb, self._graph_max_pages,
                                     dtype=torch.int32, device=dev),
-                "temp": torch.ones(B, dtype=torch.float32, device=dev),
-                "tk": torch.full((B,), -1, dtype=torch.int32, device=dev),
-                "tp": torch.ones(B, dtype=torch.float32, device=dev),
+                "temp": torch.ones(Bb, dtype=torch.float32, device=dev),
+                "tk": torch.full((Bb,), -1, dtype=torch.int32, device=dev),
+                "tp": torch.ones(Bb, dtype=torch.float32, device=dev),
                 "seed": torch.zeros(1, dtype=torch.int64, device=dev),
-                "out_t": torch.zeros(B, dtype=torch.int64, device=dev),
-                "out_l": torch.zeros(B, dtype=torch.float32, device=dev),
+                "out_t": torch.zeros(Bb, dtype=torch.int64, device=dev),
+                "out_l": torch.zeros(Bb, dtype=torch.float32, device=dev),
             }
+            if Bb > B:
+                st["ptab"].fill_(scratch // self.kv.page_size)
 
             def body():
                 def attn_fn(li, q, k, v):
@@ -925,13 +971,13 @@ class Engine:
             # first real replay overwrites with identical data)
             side = torch.cuda.Stream()
             side.wait_stream(torch.cuda.current_stream())
-            st["tokens"].copy_(tokens)
-            st["pos"].copy_(pos0)
-            st["ctx"].copy_(pos0 + 1)
-            st["slots"].copy_(slots_all[0])
+            st["tokens"][:B].copy_(tokens)
+            st["pos"][:B].copy_(pos0)
+            st["ctx"][:B].copy_(pos0 + 1)
+            st["slots"][:B].copy_(slots_all[0])
             mp = min(page_table.shape[1], self._graph_max_pages)
-            st["ptab"].zero_()
-            st["ptab"][:, :mp].copy_(page_table[:, :mp])
+            st["ptab"][:B].zero_()
+            st["ptab"][:B, :mp].copy_(page_table[:, :mp])
             with torch.cuda.stream(side):
                 for _ in range(2):
                     body()
@@ -945,25 +991,35 @@ class Engine:
                 with torch.cuda.graph(g, pool=self._graph_pool):
                     body()
             st["graph"] = g
-            self._graphs[B] = st
+            self._graphs[Bb] = st
 
         # load chunk state into the static buffers
-        st["tokens"].copy_(tokens)
-        st["pos"].copy_(pos0)
-        st["ctx"].copy_(pos0 + 1)
+        st["tokens"][:B].copy_(tokens)
+        st["pos"][:B].copy_(pos0)
+        st["ctx"][:B].copy_(pos0 + 1)
         mp = min(page_table.shape[1], self._graph_max_pages)
-        st["ptab"].zero_()
-        st["ptab"][:, :mp].copy_(page_table[:, :mp])
-        st["temp"].copy_(temp)
-        st["tk"].copy_(tk)
-        st["tp"].copy_(tp)
+        st["ptab"][:B].zero_()
+        st["ptab"][:B, :mp].copy_(page_table[:, :mp])
+        st["temp"][:B].copy_(temp)
+        st["tk"][:B].copy_(tk)
+        st["tp"][:B].copy_(tp)
         st["seed"].fill_(seed0)
+        if Bb > B:
+            # reset dummy rows (a previous call on this bucket with a
+            # larger real B may have left real state here): KV writes go
+            # to the scratch slot, attention reads ctx=1 from the zeroed
+            # scratch page, outputs are discarded below
+            st["pos"][B:].zero_()
+            st["ctx"][B:].fill_(1)
+            st["slots"][B:].fill_(scratch)
+            st["ptab"][B:].fill_(scratch // self.kv.page_size)
+            st["tokens"][B:].zero_()
         g = st["graph"]
         for s in range(C):
-            st["slots"].copy_(slots_all[s])
+            st["slots"][:B].copy_(slots_all[s])
             g.replay()
-            out_tokens[s].copy_(st["out_t"])
-            out_lps[s].copy_(st["out_l"])
+            out_tokens[s].copy_(st["out_t"][:B])
+            out_lps[s].copy_(st["out_l"][:B])
 
     def _finish_decode_chunk(self, reqs, C, toks_h, out_lps):
         lps_h = out_lps.t().cpu().tolist()

@@ -483,3 +483,30 @@ def test_aborted_request_does_not_donate_radix(setup):
     pages2, matched2 = eng.radix.match(prompt + [0])
     assert matched2 > 0
     eng.kv.unref_pages(pages2)
+
+
+def test_decode_bucket_table():
+    """Bucketed graphed-decode helper (EXPERIMENTAL round-3 path): buckets
+    are monotone, cover 1..256 with <=50% padding waste, and B beyond the
+    table maps to itself."""
+    from polyrl_amd.rollout.engine import DECODE_BUCKETS, decode_bucket
+    assert list(DECODE_BUCKETS) == sorted(set(DECODE_BUCKETS))
+    for b in range(1, 257):
+        bb = decode_bucket(b)
+        assert bb >= b and bb in DECODE_BUCKETS
+        assert bb <= b * 3 / 2, (b, bb)   # bounded padding waste
+    assert decode_bucket(300) == 300          # beyond table: exact
+
+
+def test_decode_scratch_slot_reservation(setup):
+    """The dummy-row KV sink: reserved once, stable across calls, page
+    zeroed, re-reserved after the KV cache is rebuilt (release/resume)."""
+    _, _, eng = setup
+    s1 = eng._decode_scratch_slot()
+    s2 = eng._decode_scratch_slot()
+    assert s1 == s2                       # idempotent
+    page = s1 // eng.kv.page_size
+    assert eng.kv.k_cache[0][page].abs().sum().item() == 0.0
+    free_before = eng.kv.free_pages
+    eng._decode_scratch_slot()
+    assert eng.kv.free_pages == free_before   # no double reservation
