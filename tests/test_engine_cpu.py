@@ -510,3 +510,6 @@ def test_decode_scratch_slot_reservation(setup):
     free_before = eng.kv.free_pages
     eng._decode_scratch_slot()
     assert eng.kv.free_pages == free_before   # no double reservation
+    # release so later tests on this module-scoped engine see all pages
+    eng.kv.free_seq(eng._SCRATCH_SEQ)
+    assert eng.kv.free_pages == free_before + 1
