@@ -290,3 +290,38 @@ def test_load_reward_manager_from_config():
     cfg.custom_reward_function.path = path
     rm2 = load_reward_manager_from_config(cfg)
     assert isinstance(rm2, FunctionReward)
+
+
+def test_example_custom_reward_and_interaction_files_load():
+    """The shipped example files must keep matching the real loader
+    contracts (examples/reward/format_reward.py via FunctionReward;
+    examples/interactions/calc_tool.py via load_interaction)."""
+    import os
+
+    import torch
+
+    from polyrl_amd.protocol import TensorBatch
+    from polyrl_amd.reward import FunctionReward
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    mgr = FunctionReward(os.path.join(root, "examples/reward/format_reward.py"),
+                         length_target=4)
+    b = TensorBatch(tensors={
+        "input_ids": torch.ones(2, 6, dtype=torch.long),
+        "responses": torch.ones(2, 4, dtype=torch.long),
+        "response_mask": torch.tensor([[1, 1, 1, 0], [1, 0, 0, 0]]),
+        "attention_mask": torch.ones(2, 10, dtype=torch.long)})
+    r = mgr(b)
+    assert r.shape == (2, 4)
+    # in-band (3 tokens vs target 4) -> +1 on last valid token; 1 token -> -1
+    assert r[0].sum() == 1.0 and r[1].sum() == -1.0
+
+    from polyrl_amd.config import MultiTurnConfig
+    from polyrl_amd.trainer.rollout_coordinator import load_interaction
+    mt = load_interaction(MultiTurnConfig(
+        enable=True,
+        interaction_path=os.path.join(root,
+                                      "examples/interactions/calc_tool.py")))
+    user, done = mt["interaction"]([1, 2], [5, 7])
+    assert user == [12] and done is False
+    assert mt["interaction"]([1], [5])[1] is True     # too short: finalize
