@@ -67,3 +67,25 @@ def test_collective_plane_world2(tmp_path):
         capture_output=True, text=True, timeout=240, env=env)
     assert r.returncode == 0, f"{r.stdout[-3000:]}\n{r.stderr[-3000:]}"
     assert "plane OK" in r.stdout
+
+
+def test_tp_slice_qwen_biases_and_dims():
+    """tp_slice unit coverage: qwen2 QKV biases shard dim 0 alongside their
+    weights; consumers shard dim 1; norms/embeddings replicate; the engine
+    vocab-shards lm_head itself (engine.py receiver special-case)."""
+    import torch
+
+    from polyrl_amd.transfer.collective import tp_shard_dim, tp_slice
+
+    w = torch.arange(24, dtype=torch.float32).reshape(6, 4)
+    b = torch.arange(6, dtype=torch.float32)
+    assert tp_shard_dim("model.layers.0.self_attn.k_proj.bias") == 0
+    assert torch.equal(tp_slice("l.self_attn.q_proj.bias", b, 1, 2), b[3:])
+    assert torch.equal(tp_slice("l.self_attn.v_proj.weight", w, 0, 2), w[:3])
+    assert torch.equal(tp_slice("l.mlp.down_proj.weight", w, 1, 2), w[:, 2:])
+    # replicated tensors come back untouched for any rank
+    assert torch.equal(tp_slice("model.norm.weight", b, 1, 2), b)
+    assert torch.equal(tp_slice("model.embed_tokens.weight", w, 1, 2), w)
+    assert tp_shard_dim("lm_head.weight") is None   # engine special-case
+    # gpt2 tier replicates everything
+    assert tp_shard_dim("h.0.attn_qkv.weight", arch="gpt2") is None
