@@ -153,3 +153,13 @@ def test_profiling_helpers_cpu_noop():
 
     assert g(4) == 8
     log_gpu_memory("tag")     # must not raise without CUDA
+
+
+def test_get_tokenizer_offline_branches(tmp_path):
+    """hf_tokenizer capability is offline-friendly: registry model names
+    and directories without tokenizer files yield None (datasets carry
+    precomputed input_ids instead)."""
+    from polyrl_amd.utils.tokenizer import get_tokenizer
+    assert get_tokenizer("llama3-8b") is None          # registry name
+    assert get_tokenizer(str(tmp_path)) is None        # dir, no tok files
+    assert get_tokenizer(str(tmp_path / "nope")) is None
