@@ -94,6 +94,10 @@ def step_profiler(enabled: bool, trace_dir: str = "profiles/torch",
     acts = [torch.profiler.ProfilerActivity.CPU]
     if torch.cuda.is_available():
         acts.append(torch.profiler.ProfilerActivity.CUDA)
-    with torch.profiler.profile(activities=acts) as prof:
-        yield prof
+    import warnings
+    with warnings.catch_warnings():
+        # torch's informational "Profiler clears events each cycle" notice
+        warnings.filterwarnings("ignore", message=".*Profiler clears events.*")
+        with torch.profiler.profile(activities=acts) as prof:
+            yield prof
     prof.export_chrome_trace(os.path.join(trace_dir, f"step_{step}.json"))
