@@ -513,3 +513,21 @@ def test_decode_scratch_slot_reservation(setup):
     # release so later tests on this module-scoped engine see all pages
     eng.kv.free_seq(eng._SCRATCH_SEQ)
     assert eng.kv.free_pages == free_before + 1
+
+
+def test_decode_scratch_survives_release_resume(setup):
+    """release_memory/resume_memory rebuild the KV cache; the scratch
+    reservation must re-reserve lazily on the fresh cache (the seq maps
+    reset) instead of handing out a stale slot."""
+    _, _, eng = setup
+    s1 = eng._decode_scratch_slot()
+    assert eng.kv.seq_len(eng._SCRATCH_SEQ) == 1
+    eng.release_memory()
+    eng.resume_memory()
+    assert eng.kv.seq_len(eng._SCRATCH_SEQ) == 0    # fresh cache: no seq
+    s2 = eng._decode_scratch_slot()
+    assert eng.kv.seq_len(eng._SCRATCH_SEQ) == 1
+    page = s2 // eng.kv.page_size
+    assert 0 <= page < eng.kv.num_pages
+    assert eng.kv.k_cache[0][page].abs().sum().item() == 0.0
+    eng.kv.free_seq(eng._SCRATCH_SEQ)               # leave the engine clean
