@@ -210,19 +210,26 @@ def test_bench_contract_world8():
     8-GPU scaling tier, minus the GPUs."""
     import json
     repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-    cmd = [
-        sys.executable, "-m", "torch.distributed.run",
-        "--nnodes=1", "--nproc-per-node=8",
-        "--master-addr", "127.0.0.1", "--master-port", "29679",
-        os.path.join(repo, "bench.py"),
-        "--gpus", "8", "--steps", "1", "--warmup", "0",
-        "--model", "llama-debug-cpu", "--batch-per-gpu", "2",
-        "--n-samples", "2", "--prompt-len", "12", "--response-len", "6",
-    ]
     env = dict(os.environ)
     env["PYTHONPATH"] = repo
-    r = subprocess.run(cmd, capture_output=True, text=True, timeout=840,
-                       env=env)
+    for attempt in range(2):    # one retry: transient rendezvous/port races
+        import socket
+        with socket.socket() as s:       # fresh ephemeral rendezvous port
+            s.bind(("127.0.0.1", 0))
+            port = s.getsockname()[1]
+        cmd = [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node=8",
+            "--master-addr", "127.0.0.1", "--master-port", str(port),
+            os.path.join(repo, "bench.py"),
+            "--gpus", "8", "--steps", "1", "--warmup", "0",
+            "--model", "llama-debug-cpu", "--batch-per-gpu", "2",
+            "--n-samples", "2", "--prompt-len", "12", "--response-len", "6",
+        ]
+        r = subprocess.run(cmd, capture_output=True, text=True, timeout=840,
+                           env=env)
+        if r.returncode == 0:
+            break
     assert r.returncode == 0, f"{r.stdout[-2000:]}\n{r.stderr[-2000:]}"
     line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
     d = json.loads(line)
