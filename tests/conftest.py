@@ -14,3 +14,12 @@ def pytest_collection_modifyitems(config, items):
     for item in items:
         if "gpu" in item.keywords:
             item.add_marker(skip_gpu)
+
+
+def free_port() -> str:
+    """Fresh ephemeral port for torchrun rendezvous — fixed ports reused
+    across runs can race (TIME_WAIT / parallel soaks)."""
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return str(s.getsockname()[1])

@@ -7,12 +7,15 @@ import sys
 
 import pytest
 
+from conftest import free_port
+
 WORKER = r"""
 import torch, torch.distributed as dist
 dist.init_process_group("gloo")
 rank = dist.get_rank()
 
 from polyrl_amd.transfer.collective import CollectiveWeightPlane, tp_slice
+
 
 torch.manual_seed(0)
 named = {
@@ -63,7 +66,7 @@ def test_collective_plane_world2(tmp_path):
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node=2", "--master-addr", "127.0.0.1",
-         "--master-port", "29677", str(script)],
+         "--master-port", free_port(), str(script)],
         capture_output=True, text=True, timeout=240, env=env)
     assert r.returncode == 0, f"{r.stdout[-3000:]}\n{r.stderr[-3000:]}"
     assert "plane OK" in r.stdout

@@ -8,6 +8,8 @@ import sys
 
 import pytest
 
+from conftest import free_port
+
 
 @pytest.mark.timeout(600)
 def test_disagg_2train_1rollout(tmp_path):
@@ -16,7 +18,7 @@ def test_disagg_2train_1rollout(tmp_path):
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node=3", "--master-addr", "127.0.0.1",
-         "--master-port", "29661",
+         "--master-port", free_port(),
          "-m", "polyrl_amd.trainer.main_stream",
          "actor_rollout_ref.model.path=llama-debug-cpu",
          "actor_rollout_ref.model.dtype=float32",
@@ -51,7 +53,7 @@ def test_elastic_colocated_world2(tmp_path):
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node=2", "--master-addr", "127.0.0.1",
-         "--master-port", "29674",
+         "--master-port", free_port(),
          "-m", "polyrl_amd.trainer.main_stream",
          "actor_rollout_ref.model.path=llama-debug-cpu",
          "actor_rollout_ref.model.dtype=float32",
@@ -98,7 +100,7 @@ def test_elastic_runtime_join(tmp_path):
     proc = subprocess.Popen(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node=2", "--master-addr", "127.0.0.1",
-         "--master-port", "29676",
+         "--master-port", free_port(),
          "-m", "polyrl_amd.trainer.main_stream",
          "actor_rollout_ref.model.path=llama-debug-cpu",
          "actor_rollout_ref.model.dtype=float32",
@@ -190,7 +192,7 @@ def test_disagg_with_ulysses_sp2(tmp_path):
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node=3", "--master-addr", "127.0.0.1",
-         "--master-port", "29691",
+         "--master-port", free_port(),
          "-m", "polyrl_amd.trainer.main_stream",
          "actor_rollout_ref.model.path=llama-debug-cpu",
          "actor_rollout_ref.model.dtype=float32",
@@ -283,7 +285,7 @@ def test_disagg_multi_turn_world3(tmp_path):
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node=3", "--master-addr", "127.0.0.1",
-         "--master-port", "29676",
+         "--master-port", free_port(),
          "-m", "polyrl_amd.trainer.main_stream",
          "actor_rollout_ref.model.path=llama-debug-cpu",
          "actor_rollout_ref.model.dtype=float32",

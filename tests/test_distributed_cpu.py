@@ -9,12 +9,14 @@ import sys
 
 import pytest
 
+from conftest import free_port
+
 
 def _run_torchrun(args, nproc=2, timeout=600):
     cmd = [
         sys.executable, "-m", "torch.distributed.run",
         "--nnodes=1", f"--nproc-per-node={nproc}",
-        "--master-addr", "127.0.0.1", "--master-port", "29641",
+        "--master-addr", "127.0.0.1", "--master-port", free_port(),
         "-m", "polyrl_amd.trainer.main_stream",
     ] + args
     env = dict(os.environ)
@@ -86,7 +88,7 @@ def test_bench_contract_world2():
     cmd = [
         sys.executable, "-m", "torch.distributed.run",
         "--nnodes=1", "--nproc-per-node=2",
-        "--master-addr", "127.0.0.1", "--master-port", "29672",
+        "--master-addr", "127.0.0.1", "--master-port", free_port(),
         os.path.join(repo, "bench.py"),
         "--gpus", "2", "--steps", "1", "--warmup", "1",
         "--model", "llama-debug-cpu", "--batch-per-gpu", "4",

@@ -7,6 +7,8 @@ import sys
 
 import pytest
 
+from conftest import free_port
+
 WORKER = r"""
 import torch, torch.distributed as dist
 dist.init_process_group("gloo")
@@ -15,6 +17,7 @@ rank = dist.get_rank()
 from polyrl_amd.models import create_model, get_model_config
 from polyrl_amd.parallel.tp import TPContext
 from polyrl_amd.rollout.engine import Engine, SamplingParams
+
 
 cfg = get_model_config("llama-debug-cpu")
 torch.manual_seed(0)
@@ -66,7 +69,7 @@ def test_tp2_engine_matches_tp1(tmp_path):
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node=2", "--master-addr", "127.0.0.1",
-         "--master-port", "29658", str(script)],
+         "--master-port", free_port(), str(script)],
         capture_output=True, text=True, timeout=540, env=env)
     assert r.returncode == 0, f"stdout:\n{r.stdout[-3000:]}\nstderr:\n{r.stderr[-3000:]}"
     assert "TP OK" in r.stdout
@@ -83,6 +86,7 @@ dist.init_process_group("gloo")
 from polyrl_amd.models import create_model, get_model_config
 from polyrl_amd.parallel.tp import TPContext
 from polyrl_amd.rollout.engine import Engine, SamplingParams
+
 cfg = get_model_config("llama-debug-cpu")
 torch.manual_seed(0)
 model = create_model(cfg, kind="actor", dtype="float32", device="cpu")
@@ -113,7 +117,7 @@ print(f"rank {dist.get_rank()}: TP+group OK")
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node=2", "--master-addr", "127.0.0.1",
-         "--master-port", "29678", str(script)],
+         "--master-port", free_port(), str(script)],
         capture_output=True, text=True, timeout=540, env=env)
     assert r.returncode == 0, f"{r.stdout[-3000:]}\n{r.stderr[-3000:]}"
     assert "TP+group OK" in r.stdout

@@ -7,6 +7,8 @@ import sys
 
 import pytest
 
+from conftest import free_port
+
 WORKER = r"""
 import os, torch, torch.distributed as dist
 torch.manual_seed(0)
@@ -17,6 +19,7 @@ from polyrl_amd.config import ActorConfig, CriticConfig
 from polyrl_amd.models import create_model, get_model_config
 from polyrl_amd.protocol import TensorBatch
 from polyrl_amd.trainer.workers import ActorWorker, CriticWorker
+
 
 cfg = get_model_config("llama-debug-cpu")
 torch.manual_seed(42)
@@ -92,7 +95,7 @@ def test_ulysses_sp2_matches_full(tmp_path):
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node=2", "--master-addr", "127.0.0.1",
-         "--master-port", "29653", str(script)],
+         "--master-port", free_port(), str(script)],
         capture_output=True, text=True, timeout=540, env=env)
     assert r.returncode == 0, f"stdout:\n{r.stdout[-3000:]}\nstderr:\n{r.stderr[-3000:]}"
     assert "SP OK" in r.stdout
@@ -110,6 +113,7 @@ from polyrl_amd.models import create_model, get_model_config
 from polyrl_amd.protocol import TensorBatch
 from polyrl_amd.trainer.workers import ActorWorker, CriticWorker
 import polyrl_amd.core.algos as algos
+
 
 cfg = get_model_config("llama-debug-cpu")
 assert cfg.num_attention_heads % 2 == 0 and cfg.num_key_value_heads % 2 == 0
@@ -193,7 +197,7 @@ def test_ulysses_sp2_packed_matches_full(tmp_path):
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node=2", "--master-addr", "127.0.0.1",
-         "--master-port", "29657", str(script)],
+         "--master-port", free_port(), str(script)],
         capture_output=True, text=True, timeout=540, env=env)
     assert r.returncode == 0, f"stdout:\n{r.stdout[-3000:]}\nstderr:\n{r.stderr[-3000:]}"
     assert "PACKED-SP OK" in r.stdout
