@@ -23,3 +23,14 @@ def free_port() -> str:
     with socket.socket() as s:
         s.bind(("127.0.0.1", 0))
         return str(s.getsockname()[1])
+
+
+def retry_run(call, attempts=2):
+    """Re-invoke a subprocess launcher on transient nonzero exits —
+    torchrun agent/rendezvous spawns can race under box load; the caller
+    passes a lambda so each attempt draws a fresh free_port()."""
+    for _ in range(attempts):
+        r = call()
+        if r.returncode == 0:
+            break
+    return r

@@ -8,14 +8,14 @@ import sys
 
 import pytest
 
-from conftest import free_port
+from conftest import free_port, retry_run
 
 
 @pytest.mark.timeout(600)
 def test_disagg_2train_1rollout(tmp_path):
     env = dict(os.environ)
     env["PYTHONPATH"] = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-    r = subprocess.run(
+    r = retry_run(lambda: subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node=3", "--master-addr", "127.0.0.1",
          "--master-port", free_port(),
@@ -38,7 +38,7 @@ def test_disagg_2train_1rollout(tmp_path):
          "reward=random",
          "max_steps=2",
          ],
-        capture_output=True, text=True, timeout=540, env=env)
+        capture_output=True, text=True, timeout=540, env=env))
     assert r.returncode == 0, \
         f"stdout:\n{r.stdout[-4000:]}\nstderr:\n{r.stderr[-4000:]}"
 
@@ -50,7 +50,7 @@ def test_elastic_colocated_world2(tmp_path):
     serving on rank 0 (the reference's primary operating shape, §3.4)."""
     env = dict(os.environ)
     env["PYTHONPATH"] = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-    r = subprocess.run(
+    r = retry_run(lambda: subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node=2", "--master-addr", "127.0.0.1",
          "--master-port", free_port(),
@@ -74,7 +74,7 @@ def test_elastic_colocated_world2(tmp_path):
          "reward=random",
          "max_steps=2",
          ],
-        capture_output=True, text=True, timeout=540, env=env)
+        capture_output=True, text=True, timeout=540, env=env))
     assert r.returncode == 0, \
         f"stdout:\n{r.stdout[-4000:]}\nstderr:\n{r.stderr[-4000:]}"
 
@@ -189,7 +189,7 @@ def test_disagg_with_ulysses_sp2(tmp_path):
     (main_stream.py / workers.register_sp_groups)."""
     env = dict(os.environ)
     env["PYTHONPATH"] = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-    r = subprocess.run(
+    r = retry_run(lambda: subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node=3", "--master-addr", "127.0.0.1",
          "--master-port", free_port(),
@@ -214,7 +214,7 @@ def test_disagg_with_ulysses_sp2(tmp_path):
          "reward=random",
          "max_steps=2",
          ],
-        capture_output=True, text=True, timeout=540, env=env)
+        capture_output=True, text=True, timeout=540, env=env))
     assert r.returncode == 0, \
         f"stdout:\n{r.stdout[-4000:]}\nstderr:\n{r.stderr[-4000:]}"
     assert "fit done" in r.stdout
@@ -282,7 +282,7 @@ def test_disagg_multi_turn_world3(tmp_path):
         "    return [3, 4], False\n")
     env = dict(os.environ)
     env["PYTHONPATH"] = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-    r = subprocess.run(
+    r = retry_run(lambda: subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node=3", "--master-addr", "127.0.0.1",
          "--master-port", free_port(),
@@ -309,7 +309,7 @@ def test_disagg_multi_turn_world3(tmp_path):
          "reward=random",
          "max_steps=1",
          ],
-        capture_output=True, text=True, timeout=540, env=env)
+        capture_output=True, text=True, timeout=540, env=env))
     assert r.returncode == 0, \
         f"stdout:\n{r.stdout[-4000:]}\nstderr:\n{r.stderr[-4000:]}"
 

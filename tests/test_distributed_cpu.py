@@ -9,20 +9,23 @@ import sys
 
 import pytest
 
-from conftest import free_port
+from conftest import free_port, retry_run
 
 
 def _run_torchrun(args, nproc=2, timeout=600):
-    cmd = [
-        sys.executable, "-m", "torch.distributed.run",
-        "--nnodes=1", f"--nproc-per-node={nproc}",
-        "--master-addr", "127.0.0.1", "--master-port", free_port(),
-        "-m", "polyrl_amd.trainer.main_stream",
-    ] + args
     env = dict(os.environ)
     env["PYTHONPATH"] = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-    return subprocess.run(cmd, capture_output=True, text=True, timeout=timeout,
-                          env=env)
+
+    def attempt():
+        cmd = [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", f"--nproc-per-node={nproc}",
+            "--master-addr", "127.0.0.1", "--master-port", free_port(),
+            "-m", "polyrl_amd.trainer.main_stream",
+        ] + args
+        return subprocess.run(cmd, capture_output=True, text=True,
+                              timeout=timeout, env=env)
+    return retry_run(attempt)
 
 
 @pytest.mark.timeout(600)
@@ -85,7 +88,9 @@ def test_bench_contract_world2():
     --gpus 2 prints one whole-job JSON line from rank 0."""
     import json
     repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-    cmd = [
+    env = dict(os.environ)
+    env["PYTHONPATH"] = repo
+    r = retry_run(lambda: subprocess.run([
         sys.executable, "-m", "torch.distributed.run",
         "--nnodes=1", "--nproc-per-node=2",
         "--master-addr", "127.0.0.1", "--master-port", free_port(),
@@ -93,11 +98,7 @@ def test_bench_contract_world2():
         "--gpus", "2", "--steps", "1", "--warmup", "1",
         "--model", "llama-debug-cpu", "--batch-per-gpu", "4",
         "--n-samples", "2", "--prompt-len", "16", "--response-len", "8",
-    ]
-    env = dict(os.environ)
-    env["PYTHONPATH"] = repo
-    r = subprocess.run(cmd, capture_output=True, text=True, timeout=540,
-                      env=env)
+    ], capture_output=True, text=True, timeout=540, env=env))
     assert r.returncode == 0, f"{r.stdout[-2000:]}\n{r.stderr[-2000:]}"
     line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
     d = json.loads(line)

@@ -7,7 +7,7 @@ import sys
 
 import pytest
 
-from conftest import free_port
+from conftest import free_port, retry_run
 
 WORKER = r"""
 import torch, torch.distributed as dist
@@ -66,11 +66,11 @@ def test_tp2_engine_matches_tp1(tmp_path):
     script.write_text(WORKER)
     env = dict(os.environ)
     env["PYTHONPATH"] = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-    r = subprocess.run(
+    r = retry_run(lambda: subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node=2", "--master-addr", "127.0.0.1",
          "--master-port", free_port(), str(script)],
-        capture_output=True, text=True, timeout=540, env=env)
+        capture_output=True, text=True, timeout=540, env=env))
     assert r.returncode == 0, f"stdout:\n{r.stdout[-3000:]}\nstderr:\n{r.stderr[-3000:]}"
     assert "TP OK" in r.stdout
 
@@ -114,10 +114,10 @@ print(f"rank {dist.get_rank()}: TP+group OK")
 """)
     env = dict(os.environ)
     env["PYTHONPATH"] = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-    r = subprocess.run(
+    r = retry_run(lambda: subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node=2", "--master-addr", "127.0.0.1",
          "--master-port", free_port(), str(script)],
-        capture_output=True, text=True, timeout=540, env=env)
+        capture_output=True, text=True, timeout=540, env=env))
     assert r.returncode == 0, f"{r.stdout[-3000:]}\n{r.stderr[-3000:]}"
     assert "TP+group OK" in r.stdout

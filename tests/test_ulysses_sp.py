@@ -7,7 +7,7 @@ import sys
 
 import pytest
 
-from conftest import free_port
+from conftest import free_port, retry_run
 
 WORKER = r"""
 import os, torch, torch.distributed as dist
@@ -92,11 +92,11 @@ def test_ulysses_sp2_matches_full(tmp_path):
     script.write_text(WORKER)
     env = dict(os.environ)
     env["PYTHONPATH"] = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-    r = subprocess.run(
+    r = retry_run(lambda: subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node=2", "--master-addr", "127.0.0.1",
          "--master-port", free_port(), str(script)],
-        capture_output=True, text=True, timeout=540, env=env)
+        capture_output=True, text=True, timeout=540, env=env))
     assert r.returncode == 0, f"stdout:\n{r.stdout[-3000:]}\nstderr:\n{r.stderr[-3000:]}"
     assert "SP OK" in r.stdout
 
@@ -194,10 +194,10 @@ def test_ulysses_sp2_packed_matches_full(tmp_path):
     script.write_text(WORKER_PACKED)
     env = dict(os.environ)
     env["PYTHONPATH"] = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-    r = subprocess.run(
+    r = retry_run(lambda: subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node=2", "--master-addr", "127.0.0.1",
          "--master-port", free_port(), str(script)],
-        capture_output=True, text=True, timeout=540, env=env)
+        capture_output=True, text=True, timeout=540, env=env))
     assert r.returncode == 0, f"stdout:\n{r.stdout[-3000:]}\nstderr:\n{r.stderr[-3000:]}"
     assert "PACKED-SP OK" in r.stdout
